@@ -1,0 +1,205 @@
+"""Learner: GPU training process.
+
+Capability parity with the reference's agents/learner.py LearnerBase +
+variants: model on device + optimizer (67-70), weight PUB on learner_port+1
+broadcasting after every update (85-93, ppo/learning.py:108), metrics writer
+(77-79, 95-158), on-policy whole-buffer / off-policy random sampling from
+shared memory (168-233), latching replay-ready flag (385), checkpoint save
+every ``model_save_interval`` (ppo/learning.py:113-119).
+
+MI355X-first differences:
+* the algorithm lives in an Updater (learner_module/*) with a pure
+  ``step(batch)`` — the learner is a thin process shell;
+* multi-GPU data parallelism: each learner rank (one per GPU) drains its own
+  batch from the ring and all-reduces gradients over RCCL/xGMI via
+  GradReducer (reference has a single-GPU learner only);
+* batch staging goes through a pinned-host buffer + async H2D copy
+  (ops/staging.py) instead of per-field blocking copies.
+"""
+from __future__ import annotations
+
+import os
+import time
+from pathlib import Path
+
+import numpy as np
+import torch
+
+from pdrl_amd.agents.learner_module import is_on_policy, switch_module
+from pdrl_amd.buffers import SharedRolloutRing
+from pdrl_amd.transport import pub_bind
+from pdrl_amd.utils import ExecutionTimer, Protocol, SummaryWriter, encode
+
+
+class BatchStager:
+    """Host→device batch staging with pinned memory + a side copy stream
+    (the K15 staging design in SURVEY.md §2.4; reference path was
+    shm → np copy → to_torch → blocking .to(device), learner.py:197-233)."""
+
+    def __init__(self, device):
+        self.device = torch.device(device)
+        self.use_cuda = self.device.type == "cuda"
+        self._pinned: dict[str, torch.Tensor] = {}
+        self.copy_stream = torch.cuda.Stream(self.device) if self.use_cuda else None
+
+    def stage(self, batch_np: dict[str, np.ndarray]) -> dict[str, torch.Tensor]:
+        if not self.use_cuda:
+            return {k: torch.from_numpy(v) for k, v in batch_np.items()}
+        out = {}
+        with torch.cuda.stream(self.copy_stream):
+            for k, v in batch_np.items():
+                t = torch.from_numpy(v)
+                pin = self._pinned.get(k)
+                if pin is None or pin.shape != t.shape:
+                    pin = torch.empty_like(t, pin_memory=True)
+                    self._pinned[k] = pin
+                pin.copy_(t)
+                out[k] = pin.to(self.device, non_blocking=True)
+        torch.cuda.current_stream(self.device).wait_stream(self.copy_stream)
+        return out
+
+
+class Learner:
+    def __init__(
+        self,
+        ring: SharedRolloutRing,
+        learner_ip: str,
+        learner_port: int,
+        params,
+        device: str | None = None,
+        shared_stat=None,
+        stop_event=None,
+        heartbeat=None,
+        rank: int = 0,
+        world_size: int = 1,
+        grad_reducer=None,
+        resume_path: str | None = None,
+    ):
+        self.params = params
+        self.ring = ring
+        self.rank = rank
+        self.world_size = world_size
+        self.stop_event = stop_event
+        self.heartbeat = heartbeat
+        self.shared_stat = shared_stat
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        self.device = torch.device(device)
+
+        updater_cls, model_cls = switch_module(params.algo)
+        obs_dim, n_actions = params.obs_dim, params.n_actions
+        model = model_cls(obs_dim, n_actions, params.seq_len, params.hidden_size)
+        self.updater = updater_cls(model, params, self.device, grad_reducer=grad_reducer)
+        if resume_path:
+            self.updater.load(resume_path, map_location=self.device)
+
+        self.on_policy = is_on_policy(params.algo)
+        self.stager = BatchStager(self.device)
+        self._rng = np.random.default_rng(1234 + rank)
+        self.is_root = rank == 0
+
+        # weight plane: PUB bound at learner_port + 1 (rank 0 only)
+        self.pub = pub_bind(learner_ip, learner_port + 1) if self.is_root else None
+        self.writer = SummaryWriter(params.result_dir) if self.is_root else None
+        self.timer = ExecutionTimer(num_transition=params.seq_len * params.batch_size * world_size)
+        self._replay_ready = False  # latching (reference: learner.py:385)
+
+    # ------------------------------------------------------------------ #
+    def _stopped(self) -> bool:
+        return self.stop_event is not None and self.stop_event.is_set()
+
+    def sample_ready(self) -> bool:
+        b = self.params.batch_size
+        if self.on_policy:
+            return self.ring.available() >= b
+        if not self._replay_ready:
+            self._replay_ready = self.ring.ready(b)
+        return self._replay_ready
+
+    def next_batch(self, timeout: float = 30.0):
+        """Block until a batch is available (or timeout/stop). Stages to device."""
+        deadline = time.monotonic() + timeout
+        b = self.params.batch_size
+        while not self._stopped() and time.monotonic() < deadline:
+            if self.sample_ready():
+                batch_np = (
+                    self.ring.drain_batch(b)
+                    if self.on_policy
+                    else self.ring.sample_batch(b, self._rng)
+                )
+                if batch_np is not None:
+                    return self.stager.stage(batch_np)
+            time.sleep(0.001)
+        return None
+
+    def publish_model(self):
+        if self.pub is None:
+            return
+        header, payload = encode(Protocol.Model, self.updater.actor_state_dict())
+        self.pub.send(header, payload)
+
+    def log_stats(self, stats: dict):
+        if self.writer is None:
+            return
+        n = self.updater.update_count
+        if n % self.params.loss_log_interval == 0:
+            for k, v in stats.items():
+                self.writer.add_scalar(k, v, n)
+            for name, dq in self.timer.throughput_dict.items():
+                if dq:
+                    self.writer.add_scalar(f"{name}-tps", float(np.mean(dq)), n)
+            for name, dq in self.timer.timer_dict.items():
+                if dq:
+                    self.writer.add_scalar(f"{name}-sec", float(np.mean(dq)), n)
+            if self.shared_stat is not None and self.shared_stat[2] >= 1.0:
+                self.writer.add_scalar("game-count", self.shared_stat[0], n)
+                self.writer.add_scalar("50-game-mean-stat-of-epi-rew", self.shared_stat[1], n)
+                self.shared_stat[2] = 0.0
+
+    def save_checkpoint(self):
+        if not self.is_root:
+            return
+        n = self.updater.update_count
+        if n > 0 and n % self.params.model_save_interval == 0:
+            model_dir = Path(self.params.model_dir)
+            model_dir.mkdir(parents=True, exist_ok=True)
+            self.updater.save(model_dir / f"{self.params.algo}_{n}.pt")
+
+    # ------------------------------------------------------------------ #
+    def run(self, max_updates: int | None = None):
+        """The learning chain: batch → update → weight broadcast → log/save
+        (reference: learner.py:267-305 + each learning.py loop)."""
+        while not self._stopped():
+            with self.timer.timer("learner-batching-time"):
+                batch = self.next_batch()
+            if batch is None:
+                if max_updates is not None:
+                    break
+                continue
+            with self.timer.timer("learner-throughput", check_throughput=True):
+                stats = self.updater.step(batch)
+            self.publish_model()
+            self.log_stats(stats)
+            self.save_checkpoint()
+            if self.heartbeat is not None:
+                self.heartbeat.value = time.time()
+            if max_updates is not None and self.updater.update_count >= max_updates:
+                break
+
+    def close(self):
+        if self.pub is not None:
+            self.pub.close()
+        if self.writer is not None:
+            self.writer.close()
+
+
+def find_latest_checkpoint(model_dir: str, algo: str):
+    """Newest ``{algo}_{idx}.pt`` by trailing index (reference resume
+    behavior: main.py:128-146, utils.py:93-98)."""
+    from pdrl_amd.utils import extract_file_num
+
+    d = Path(model_dir)
+    if not d.is_dir():
+        return None
+    cands = sorted(d.glob(f"{algo}_*.pt"), key=extract_file_num)
+    return str(cands[-1]) if cands else None

@@ -1,0 +1,77 @@
+"""Shared updater machinery: gradient reduce + clip, initial recurrent state,
+checkpoint helpers.
+"""
+from __future__ import annotations
+
+import torch
+
+
+def batch_initial_state(batch: dict[str, torch.Tensor]):
+    """The batch's initial LSTM state: stored per-step hx/cx at t=0
+    (reference: each learning.py uses hx[:, 0], cx[:, 0] — e.g.
+    ppo/learning.py:37-41)."""
+    return batch["hx"][:, 0].contiguous(), batch["cx"][:, 0].contiguous()
+
+
+class BaseUpdater:
+    """Common skeleton: owns hyperparams, optional data-parallel gradient
+    reducer, grad-norm clipping, and an update counter."""
+
+    name = "base"
+
+    def __init__(self, params, device, grad_reducer=None):
+        self.params = params
+        self.device = torch.device(device)
+        self.grad_reducer = grad_reducer
+        self.update_count = 0
+
+    # -- to be provided by subclasses ----------------------------------- #
+    def trainable_modules(self) -> dict[str, torch.nn.Module]:
+        raise NotImplementedError
+
+    def optimizers(self) -> dict[str, torch.optim.Optimizer]:
+        raise NotImplementedError
+
+    def step(self, batch) -> dict:
+        raise NotImplementedError
+
+    # -- shared --------------------------------------------------------- #
+    def reduce_and_clip(self, parameters):
+        """All-reduce gradients across learner ranks (RCCL on GPU, gloo on
+        CPU), then global-norm clip. Called between backward() and step()."""
+        parameters = [p for p in parameters if p.grad is not None]
+        if self.grad_reducer is not None:
+            self.grad_reducer.all_reduce([p.grad for p in parameters])
+        torch.nn.utils.clip_grad_norm_(parameters, self.params.max_grad_norm)
+
+    def actor_state_dict(self):
+        """State broadcast to workers (the policy network only)."""
+        mods = self.trainable_modules()
+        m = mods.get("model") or next(iter(mods.values()))
+        actor = getattr(m, "actor", m)
+        return {k: v.cpu() for k, v in actor.state_dict().items()}
+
+    def save(self, path):
+        torch.save(
+            {
+                "algo": self.name,
+                "update_count": self.update_count,
+                "modules": {k: m.state_dict() for k, m in self.trainable_modules().items()},
+                "optimizers": {k: o.state_dict() for k, o in self.optimizers().items()},
+            },
+            path,
+        )
+
+    def load(self, path, map_location="cpu"):
+        ckpt = torch.load(path, map_location=map_location, weights_only=False)
+        for k, m in self.trainable_modules().items():
+            if k in ckpt.get("modules", {}):
+                m.load_state_dict(ckpt["modules"][k])
+        for k, o in self.optimizers().items():
+            if k in ckpt.get("optimizers", {}):
+                try:
+                    o.load_state_dict(ckpt["optimizers"][k])
+                except Exception:
+                    pass
+        self.update_count = int(ckpt.get("update_count", 0))
+        return ckpt

@@ -1,0 +1,128 @@
+"""SAC (discrete actions, expectation form) with auto-tuned entropy
+temperature and a genuinely deep-copied target critic.
+
+Capability parity with the reference's agents/learner_module/sac/learning.py
+(expectation-form actor loss: 36-62; alpha auto-tune vs target entropy:
+64-74; soft-Q TD target from the target critic: 76-103; twin-Q smooth-L1
+value loss gathered on the taken action: 105-120; Polyak soft update: 143)
+and learner.py:351-400 (three optimizers + log_alpha + latching replay-ready
+flag). The reference's target-critic aliasing bug (learner.py:357 — ``.to()``
+returns the same module, making soft_update a no-op) is fixed here via
+copy.deepcopy.
+"""
+from __future__ import annotations
+
+import copy
+
+import numpy as np
+import torch
+import torch.nn.functional as F
+
+from .compute_loss import soft_update
+from .common import BaseUpdater, batch_initial_state
+
+
+class SACUpdater(BaseUpdater):
+    name = "SAC"
+    TAU = 0.005
+
+    def __init__(self, model, params, device, grad_reducer=None):
+        super().__init__(params, device, grad_reducer)
+        self.model = model.to(device)
+        self.actor = self.model.actor
+        self.critic = self.model.critic
+        self.target_critic = copy.deepcopy(self.critic).to(device)
+        for p in self.target_critic.parameters():
+            p.requires_grad_(False)
+
+        n_actions = self.critic.n_outputs
+        # maximum-entropy target: 98% of uniform-policy entropy
+        self.target_entropy = 0.98 * float(-np.log(1.0 / n_actions))
+        self.log_alpha = torch.nn.Parameter(
+            torch.tensor(float(np.log(params.alpha)), device=self.device)
+        )
+
+        self.actor_optimizer = torch.optim.Adam(self.actor.parameters(), lr=params.lr)
+        self.critic_optimizer = torch.optim.Adam(self.critic.parameters(), lr=params.lr)
+        self.alpha_optimizer = torch.optim.Adam([self.log_alpha], lr=params.lr)
+
+    def trainable_modules(self):
+        return {"model": self.model, "target_critic": self.target_critic}
+
+    def optimizers(self):
+        return {
+            "actor_optimizer": self.actor_optimizer,
+            "critic_optimizer": self.critic_optimizer,
+            "alpha_optimizer": self.alpha_optimizer,
+        }
+
+    @property
+    def alpha(self):
+        return self.log_alpha.exp()
+
+    def step(self, batch: dict[str, torch.Tensor]) -> dict:
+        p = self.params
+        obs, act = batch["obs"], batch["act"]
+        rew = batch["rew"] * p.reward_scale
+        is_fir = batch["is_fir"]
+        hx0, cx0 = batch_initial_state(batch)
+        stats = {}
+
+        for _ in range(p.K_epoch):
+            probs, log_probs = self.actor(obs, (hx0, cx0))
+            q1, q2 = self.critic(obs, (hx0, cx0))
+
+            # -- actor: E_a~pi[ alpha*log pi - min Q ] ------------------- #
+            min_q = torch.min(q1, q2).detach()
+            actor_loss = (
+                (probs * (self.alpha.detach() * log_probs - min_q)).sum(-1).mean()
+            )
+            self.actor_optimizer.zero_grad(set_to_none=False)
+            actor_loss.backward()
+            self.reduce_and_clip(self.actor.parameters())
+            self.actor_optimizer.step()
+
+            # -- temperature auto-tune ----------------------------------- #
+            with torch.no_grad():
+                pi_entropy = -(probs * log_probs).sum(-1)
+            alpha_loss = (self.log_alpha * (pi_entropy - self.target_entropy).detach()).mean()
+            self.alpha_optimizer.zero_grad(set_to_none=False)
+            alpha_loss.backward()
+            if self.grad_reducer is not None:
+                self.grad_reducer.all_reduce([self.log_alpha.grad])
+            self.alpha_optimizer.step()
+
+            # -- critics: soft-Q TD target ------------------------------- #
+            with torch.no_grad():
+                next_probs, next_log_probs = self.actor(obs, (hx0, cx0))
+                tq1, tq2 = self.target_critic(obs, (hx0, cx0))
+                v_next = (
+                    next_probs[:, 1:]
+                    * (torch.min(tq1, tq2)[:, 1:] - self.alpha * next_log_probs[:, 1:])
+                ).sum(-1, keepdim=True)
+                mask = 1.0 - is_fir[:, 1:]
+                target_q = rew[:, :-1] + p.gamma * mask * v_next
+
+            q1c, q2c = self.critic(obs, (hx0, cx0))
+            a_idx = act[:, :-1].long()
+            q1_taken = q1c[:, :-1].gather(-1, a_idx)
+            q2_taken = q2c[:, :-1].gather(-1, a_idx)
+            value_loss = F.smooth_l1_loss(q1_taken, target_q) + F.smooth_l1_loss(
+                q2_taken, target_q
+            )
+            self.critic_optimizer.zero_grad(set_to_none=False)
+            value_loss.backward()
+            self.reduce_and_clip(self.critic.parameters())
+            self.critic_optimizer.step()
+
+            soft_update(self.critic, self.target_critic, self.TAU)
+
+            stats = {
+                "loss-actor": float(actor_loss.detach()),
+                "loss-value": float(value_loss.detach()),
+                "loss-alpha": float(alpha_loss.detach()),
+                "alpha": float(self.alpha.detach()),
+                "entropy": float(pi_entropy.mean()),
+            }
+        self.update_count += 1
+        return stats

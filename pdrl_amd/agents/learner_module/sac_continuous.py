@@ -1,0 +1,112 @@
+"""SAC (continuous actions) with reparameterized tanh-Gaussian actor.
+
+Capability parity with the reference's
+agents/learner_module/sac_continuous/learning.py (reparameterized actor loss
+alpha*log pi - min Q: 44-55; critic on (obs, act): 94-101; alpha auto-tune;
+Polyak soft update). Target critic is a real deep copy (the reference's
+aliasing bug is fixed — see sac.py docstring).
+"""
+from __future__ import annotations
+
+import copy
+
+import numpy as np
+import torch
+import torch.nn.functional as F
+
+from .compute_loss import soft_update
+from .common import BaseUpdater, batch_initial_state
+
+
+class SACContinuousUpdater(BaseUpdater):
+    name = "SAC-Continuous"
+    TAU = 0.005
+
+    def __init__(self, model, params, device, grad_reducer=None):
+        super().__init__(params, device, grad_reducer)
+        self.model = model.to(device)
+        self.actor = self.model.actor
+        self.critic = self.model.critic
+        self.target_critic = copy.deepcopy(self.critic).to(device)
+        for p in self.target_critic.parameters():
+            p.requires_grad_(False)
+
+        n_actions = self.actor.n_outputs
+        self.target_entropy = -float(n_actions)
+        self.log_alpha = torch.nn.Parameter(
+            torch.tensor(float(np.log(params.alpha)), device=self.device)
+        )
+
+        self.actor_optimizer = torch.optim.Adam(self.actor.parameters(), lr=params.lr)
+        self.critic_optimizer = torch.optim.Adam(self.critic.parameters(), lr=params.lr)
+        self.alpha_optimizer = torch.optim.Adam([self.log_alpha], lr=params.lr)
+
+    def trainable_modules(self):
+        return {"model": self.model, "target_critic": self.target_critic}
+
+    def optimizers(self):
+        return {
+            "actor_optimizer": self.actor_optimizer,
+            "critic_optimizer": self.critic_optimizer,
+            "alpha_optimizer": self.alpha_optimizer,
+        }
+
+    @property
+    def alpha(self):
+        return self.log_alpha.exp()
+
+    def step(self, batch: dict[str, torch.Tensor]) -> dict:
+        p = self.params
+        obs, act = batch["obs"], batch["act"]
+        rew = batch["rew"] * p.reward_scale
+        is_fir = batch["is_fir"]
+        hx0, cx0 = batch_initial_state(batch)
+        stats = {}
+
+        for _ in range(p.K_epoch):
+            # -- actor (reparameterized) --------------------------------- #
+            new_act, log_prob = self.actor(obs, (hx0, cx0))
+            q1_pi, q2_pi = self.critic(obs, new_act, (hx0, cx0))
+            actor_loss = (self.alpha.detach() * log_prob - torch.min(q1_pi, q2_pi)).mean()
+            self.actor_optimizer.zero_grad(set_to_none=False)
+            actor_loss.backward()
+            self.reduce_and_clip(self.actor.parameters())
+            self.actor_optimizer.step()
+
+            # -- temperature --------------------------------------------- #
+            alpha_loss = -(
+                self.log_alpha * (log_prob.detach() + self.target_entropy)
+            ).mean()
+            self.alpha_optimizer.zero_grad(set_to_none=False)
+            alpha_loss.backward()
+            if self.grad_reducer is not None:
+                self.grad_reducer.all_reduce([self.log_alpha.grad])
+            self.alpha_optimizer.step()
+
+            # -- critics -------------------------------------------------- #
+            with torch.no_grad():
+                next_act, next_log_prob = self.actor(obs, (hx0, cx0))
+                tq1, tq2 = self.target_critic(obs, next_act, (hx0, cx0))
+                v_next = torch.min(tq1, tq2)[:, 1:] - self.alpha * next_log_prob[:, 1:]
+                mask = 1.0 - is_fir[:, 1:]
+                target_q = rew[:, :-1] + p.gamma * mask * v_next
+
+            q1, q2 = self.critic(obs, act, (hx0, cx0))
+            value_loss = F.smooth_l1_loss(q1[:, :-1], target_q) + F.smooth_l1_loss(
+                q2[:, :-1], target_q
+            )
+            self.critic_optimizer.zero_grad(set_to_none=False)
+            value_loss.backward()
+            self.reduce_and_clip(self.critic.parameters())
+            self.critic_optimizer.step()
+
+            soft_update(self.critic, self.target_critic, self.TAU)
+
+            stats = {
+                "loss-actor": float(actor_loss.detach()),
+                "loss-value": float(value_loss.detach()),
+                "loss-alpha": float(alpha_loss.detach()),
+                "alpha": float(self.alpha.detach()),
+            }
+        self.update_count += 1
+        return stats

@@ -1,0 +1,83 @@
+"""LearnerStorage: learner-side ingest process.
+
+Capability parity with the reference's agents/learner_storage.py: SUB binds
+the learner data port (60-66); rollout messages feed the RolloutAssembler,
+stat messages update the shared stat array (77-90, 104-121); completed
+seq_len trajectories are written into shared memory for the learner
+(92-102, 123-159) — here into the explicitly-synchronized SharedRolloutRing
+instead of the reference's unlocked flat arrays.
+"""
+from __future__ import annotations
+
+import asyncio
+import time
+
+from pdrl_amd.buffers import RolloutAssembler, SharedRolloutRing
+from pdrl_amd.transport import sub_bind
+from pdrl_amd.utils import Protocol, decode
+
+
+class LearnerStorage:
+    def __init__(
+        self,
+        ring: SharedRolloutRing,
+        learner_ip: str,
+        learner_port: int,
+        params,
+        shared_stat=None,  # mp.Array('d', 3): [game_count, mean_rew, fresh-flag]
+        stop_event=None,
+        heartbeat=None,
+    ):
+        self.ring = ring
+        self.params = params
+        self.sub = sub_bind(learner_ip, learner_port)
+        self.shared_stat = shared_stat
+        self.stop_event = stop_event
+        self.heartbeat = heartbeat
+        self.assembler = RolloutAssembler(params.seq_len)
+        self.n_ingested = 0
+        self.n_stored = 0
+
+    def _stopped(self) -> bool:
+        return self.stop_event is not None and self.stop_event.is_set()
+
+    # ------------------------------------------------------------------ #
+    async def ingest_task(self):
+        """Data-plane SUB → assembler / stat array."""
+        while not self._stopped():
+            msg = self.sub.recv(timeout=0.0)
+            if msg is None:
+                await asyncio.sleep(0.001)
+                continue
+            protocol, data = decode(*msg)
+            if protocol is Protocol.Rollout:
+                await self.assembler.push(data)
+                self.n_ingested += 1
+            elif protocol is Protocol.Stat:
+                if self.shared_stat is not None:
+                    self.shared_stat[0] = float(data["game_count"])
+                    self.shared_stat[1] = float(data["mean_stat"])
+                    self.shared_stat[2] = 1.0  # fresh flag
+            else:
+                raise AssertionError(f"unexpected protocol at storage: {protocol}")
+            if self.heartbeat is not None:
+                self.heartbeat.value = time.time()
+
+    async def store_task(self):
+        """Assembler → shared ring."""
+        while not self._stopped():
+            try:
+                traj = await asyncio.wait_for(self.assembler.pop(), timeout=0.5)
+            except asyncio.TimeoutError:
+                continue
+            self.ring.put(traj)
+            self.n_stored += 1
+
+    async def chain(self):
+        await asyncio.gather(self.ingest_task(), self.store_task())
+
+    def run(self):
+        asyncio.run(self.chain())
+
+    def close(self):
+        self.sub.close()

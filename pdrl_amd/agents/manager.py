@@ -1,0 +1,67 @@
+"""Manager: per-worker-machine relay/aggregator between workers and the
+learner-side storage.
+
+Capability parity with the reference's agents/manager.py: SUB binds the
+manager data port collecting worker rollout/stat messages (manager.py:30-36),
+PUB connects to the learner port (38-40); rollouts are forwarded verbatim;
+episode-reward stats are aggregated into a mean every ``stat_interval`` (50)
+episodes before forwarding (51-83).
+"""
+from __future__ import annotations
+
+import time
+
+import numpy as np
+
+from pdrl_amd.transport import pub_connect, sub_bind
+from pdrl_amd.utils import Protocol, decode, encode
+
+
+class Manager:
+    STAT_INTERVAL = 50
+
+    def __init__(self, manager_ip, manager_port, learner_ip, learner_port, stop_event=None,
+                 heartbeat=None):
+        self.sub = sub_bind(manager_ip, manager_port)
+        self.pub = pub_connect(learner_ip, learner_port)
+        self.stop_event = stop_event
+        self.heartbeat = heartbeat
+        self.stat_q: list[float] = []
+        self.game_count = 0
+
+    def _stopped(self) -> bool:
+        return self.stop_event is not None and self.stop_event.is_set()
+
+    def relay_once(self, timeout: float = 0.5) -> bool:
+        """Handle one inbound message. Returns True if one was processed."""
+        msg = self.sub.recv(timeout=timeout)
+        if msg is None:
+            return False
+        header, payload = msg
+        protocol, data = decode(header, payload)
+        if protocol is Protocol.Rollout:
+            self.pub.send(header, payload)  # forward verbatim (no re-encode)
+        elif protocol is Protocol.Stat:
+            self.game_count += 1
+            self.stat_q.append(float(data["epi_rew"]))
+            if len(self.stat_q) >= self.STAT_INTERVAL:
+                mean_rew = float(np.mean(self.stat_q))
+                out = encode(
+                    Protocol.Stat,
+                    {"game_count": self.game_count, "mean_stat": mean_rew},
+                )
+                self.pub.send(*out)
+                self.stat_q.clear()
+        else:
+            raise AssertionError(f"unexpected protocol at manager: {protocol}")
+        return True
+
+    def run(self):
+        while not self._stopped():
+            self.relay_once(timeout=0.5)
+            if self.heartbeat is not None:
+                self.heartbeat.value = time.time()
+
+    def close(self):
+        self.sub.close()
+        self.pub.close()

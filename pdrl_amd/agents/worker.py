@@ -1,0 +1,126 @@
+"""Worker: CPU rollout collector.
+
+Capability parity with the reference's agents/worker.py: owns an env + the
+actor module on CPU in eval mode (worker.py:32); steps the env with
+``model.act`` building a per-step record of 10 fields + episode uuid
+(110-123); publishes each step over the data plane (58-60); hot-reloads
+actor weights from the learner's weight plane (62-72); publishes episode
+reward stats at episode end (74-78); stamps a heartbeat each step (133-134).
+
+Differences by design:
+* one loop interleaving collection and weight polling (the threaded transport
+  makes the reference's two asyncio tasks unnecessary);
+* the reference's hard-coded ``asyncio.sleep(0.05)`` per-step throttle
+  (worker.py:131 — capping every worker at 20 steps/s) is a config knob
+  ``worker_step_sleep`` defaulting to 0.
+"""
+from __future__ import annotations
+
+import time
+import uuid
+
+import torch
+
+from pdrl_amd.agents.env_maker import EnvBase
+from pdrl_amd.transport import pub_connect, sub_connect
+from pdrl_amd.utils import Protocol, encode, decode
+
+
+class Worker:
+    def __init__(
+        self,
+        model,
+        worker_idx: int,
+        manager_ip: str,
+        manager_port: int,
+        learner_ip: str,
+        learner_port: int,
+        params,
+        heartbeat=None,
+        stop_event=None,
+        seed: int | None = None,
+    ):
+        self.params = params
+        self.worker_idx = worker_idx
+        self.model = model.cpu().eval()
+        self.env = EnvBase(params.env, seed=seed)
+        self.heartbeat = heartbeat
+        self.stop_event = stop_event
+        # data plane: PUB → manager; weight plane: SUB ← learner (port+1)
+        self.pub = pub_connect(manager_ip, manager_port)
+        self.sub = sub_connect(learner_ip, learner_port + 1)
+        self.step_sleep = float(getattr(params, "worker_step_sleep", 0.0))
+
+    # ------------------------------------------------------------------ #
+    def pub_rollout(self, step_data: dict):
+        header, payload = encode(Protocol.Rollout, step_data)
+        self.pub.send(header, payload)
+
+    def pub_stat(self, epi_rew: float):
+        header, payload = encode(Protocol.Stat, {"epi_rew": float(epi_rew)})
+        self.pub.send(header, payload)
+
+    def poll_model(self):
+        msg = self.sub.recv(timeout=0.0)
+        # drain to the newest weight broadcast
+        newest = None
+        while msg is not None:
+            newest = msg
+            msg = self.sub.recv(timeout=0.0)
+        if newest is not None:
+            protocol, state_dict = decode(*newest)
+            if protocol is Protocol.Model:
+                actor = getattr(self.model, "actor", self.model)
+                actor.load_state_dict(state_dict)
+
+    def _stopped(self) -> bool:
+        return self.stop_event is not None and self.stop_event.is_set()
+
+    # ------------------------------------------------------------------ #
+    def collect(self, max_episodes: int | None = None):
+        """Roll out episodes forever (or for max_episodes, for tests)."""
+        H = self.params.hidden_size
+        episodes = 0
+        while not self._stopped():
+            obs = self.env.reset()
+            hx = torch.zeros(1, H)
+            cx = torch.zeros(1, H)
+            epi_rew = 0.0
+            epi_id = uuid.uuid4().hex
+            is_fir = 1.0
+            for _ in range(self.params.time_horizon):
+                if self._stopped():
+                    break
+                self.poll_model()
+                action, logits, log_prob, (next_hx, next_cx) = self.model.act(obs, (hx, cx))
+                next_obs, rew, done, _ = self.env.step(action)
+                epi_rew += rew
+                step_data = {
+                    "obs": obs.squeeze(0).numpy(),
+                    "act": action.reshape(-1).float().numpy(),
+                    "rew": rew,
+                    "logits": logits.squeeze(0).numpy(),
+                    "log_prob": log_prob.reshape(-1).numpy(),
+                    "is_fir": is_fir,
+                    "done": float(done),
+                    "hx": hx.squeeze(0).numpy(),
+                    "cx": cx.squeeze(0).numpy(),
+                    "id": epi_id,
+                }
+                self.pub_rollout(step_data)
+                obs, hx, cx = next_obs, next_hx, next_cx
+                is_fir = 0.0
+                if self.heartbeat is not None:
+                    self.heartbeat.value = time.time()
+                if self.step_sleep > 0:
+                    time.sleep(self.step_sleep)
+                if done:
+                    break
+            self.pub_stat(epi_rew)
+            episodes += 1
+            if max_episodes is not None and episodes >= max_episodes:
+                break
+
+    def close(self):
+        self.pub.close()
+        self.sub.close()

@@ -1,0 +1,95 @@
+"""RolloutAssembler: reassembles per-step worker messages into fixed-length
+(seq_len) trajectories keyed by episode uuid.
+
+Behavioral parity with the reference's buffers/rollout_assembler.py:25-83,
+including its two quirky-but-load-bearing semantics:
+
+* **Staleness eviction** — a partial (not-done) trajectory older than
+  ``stale_s`` (0.5 s) is dropped, bounding policy lag
+  (reference: rollout_assembler.py:51-56).
+* **Done-splice** — an episode that finishes before reaching seq_len is
+  parked; the next new episode's steps are appended onto the SMALLEST parked
+  finished trajectory, with ``is_fir`` (is-first-step flag) forced to 1.0 at
+  the splice point so the learner resets recurrent state there
+  (reference: rollout_assembler.py:61-73).
+
+Completed trajectories are stacked into ``{key: tensor(seq, feat)}`` and
+pushed to an asyncio queue.
+"""
+from __future__ import annotations
+
+import asyncio
+
+import torch
+
+from .trajectory import Trajectory
+
+REQUIRED_KEYS = {
+    "obs",
+    "act",
+    "rew",
+    "logits",
+    "log_prob",
+    "is_fir",
+    "done",
+    "hx",
+    "cx",
+    "id",
+}
+
+
+def stack_trajectory(steps: list[dict]) -> dict[str, torch.Tensor]:
+    """[{field: tensor(feat)} x seq] → {field: tensor(seq, feat)}."""
+    out = {}
+    for key in steps[0]:
+        if key == "id":
+            continue
+        vals = [torch.as_tensor(s[key], dtype=torch.float32).reshape(-1) for s in steps]
+        out[key] = torch.stack(vals, dim=0)
+    return out
+
+
+class RolloutAssembler:
+    def __init__(self, seq_len: int, out_queue: asyncio.Queue | None = None, stale_s: float = 0.5):
+        self.seq_len = seq_len
+        self.stale_s = stale_s
+        self.active: dict[str, Trajectory] = {}  # partial, still-running episodes
+        self.parked_done: dict[str, Trajectory] = {}  # finished but short episodes
+        self.out_queue = out_queue if out_queue is not None else asyncio.Queue(1024)
+
+    async def push(self, step: dict):
+        missing = REQUIRED_KEYS - set(step)
+        assert not missing, f"rollout step missing fields: {missing}"
+        eid = step["id"]
+
+        # staleness eviction of partial trajectories
+        for k in [k for k, tr in self.active.items() if tr.age > self.stale_s]:
+            del self.active[k]
+
+        if eid in self.active:
+            traj = self.active[eid]
+        elif self.parked_done:
+            # splice a fresh episode onto the smallest parked finished one
+            smallest = min(self.parked_done, key=lambda k: len(self.parked_done[k]))
+            traj = self.parked_done.pop(smallest)
+            step = dict(step)
+            step["is_fir"] = 1.0
+            self.active[eid] = traj
+        else:
+            traj = Trajectory(self.seq_len)
+            self.active[eid] = traj
+
+        traj.append(step)
+
+        if len(traj) == self.seq_len:
+            del self.active[eid]
+            await self.out_queue.put(stack_trajectory(traj.steps))
+        elif step.get("done", False):
+            self.active.pop(eid, None)
+            self.parked_done[eid] = traj
+
+    async def pop(self) -> dict[str, torch.Tensor]:
+        return await self.out_queue.get()
+
+    def qsize(self) -> int:
+        return self.out_queue.qsize()

@@ -1,0 +1,263 @@
+"""Core utilities: config namespaces, wire protocol, serialization, timers,
+process management.
+
+Capability parity with the reference framework's ``utils/utils.py``
+(reference: utils/utils.py:24-44 config loaders, 229-249 Protocol/encode/decode,
+167-189 ExecutionTimer, 106-164 helpers) — re-designed, not translated:
+
+* ``encode``/``decode`` use pickle + zlib (the container has no blosc2; zlib
+  level 1 gives the same "cheap wire compression" semantics).
+* Config is reloadable (``load_params``/``load_machines``) instead of frozen at
+  import, but module-level ``Params``/``Machines`` namespaces are kept so call
+  sites read identically.
+* GPU selection targets ROCm (torch.cuda is HIP on ROCm builds).
+"""
+from __future__ import annotations
+
+import json
+import os
+import pickle
+import signal
+import time
+import zlib
+from collections import deque
+from contextlib import contextmanager
+from enum import Enum
+from pathlib import Path
+from types import SimpleNamespace
+
+import numpy as np
+import torch
+
+UTILS_DIR = Path(__file__).resolve().parent
+REPO_ROOT = UTILS_DIR.parent.parent
+
+
+# --------------------------------------------------------------------------- #
+# Config
+# --------------------------------------------------------------------------- #
+def _ns(d: dict) -> SimpleNamespace:
+    return SimpleNamespace(**d)
+
+
+def load_params(path: str | os.PathLike | None = None) -> SimpleNamespace:
+    """Load hyperparameter JSON (same key schema as the reference's
+    utils/parameters.json) into a namespace."""
+    p = Path(path) if path is not None else UTILS_DIR / "parameters.json"
+    with open(p) as f:
+        params = _ns(json.load(f))
+    # Derived paths (reference: utils/utils.py:79-81): results/<ts>/models
+    if getattr(params, "result_dir", None) is None:
+        ts = time.strftime("%Y-%m-%d_%H-%M-%S")
+        params.result_dir = str(REPO_ROOT / "results" / ts)
+    if getattr(params, "model_dir", None) is None:
+        params.model_dir = str(Path(params.result_dir) / "models")
+    return params
+
+
+def load_machines(path: str | os.PathLike | None = None) -> SimpleNamespace:
+    """Load cluster topology JSON (same schema as the reference's
+    utils/machines.json)."""
+    p = Path(path) if path is not None else UTILS_DIR / "machines.json"
+    with open(p) as f:
+        m = json.load(f)
+    machines = SimpleNamespace()
+    machines.learner = _ns(m["learner"])
+    machines.workers = [_ns(w) for w in m["workers"]]
+    return machines
+
+
+Params = load_params(os.environ.get("PDRL_PARAMS"))
+Machines = load_machines(os.environ.get("PDRL_MACHINES"))
+
+
+def refresh_result_dirs(params=None):
+    """Stamp a fresh results/<ts> tree for a new training run."""
+    params = params or Params
+    ts = time.strftime("%Y-%m-%d_%H-%M-%S")
+    params.result_dir = str(REPO_ROOT / "results" / ts)
+    params.model_dir = str(Path(params.result_dir) / "models")
+    return params
+
+
+# --------------------------------------------------------------------------- #
+# Wire protocol + serialization
+# --------------------------------------------------------------------------- #
+class Protocol(Enum):
+    """Message classes on the actor plane (reference: utils/utils.py:229-232)."""
+
+    Model = "model"
+    Rollout = "rollout"
+    Stat = "stat"
+
+
+def encode(protocol: Protocol, data) -> tuple[bytes, bytes]:
+    """Serialize a (protocol, payload) message: pickle + zlib level 1."""
+    return (
+        pickle.dumps(protocol, protocol=pickle.HIGHEST_PROTOCOL),
+        zlib.compress(pickle.dumps(data, protocol=pickle.HIGHEST_PROTOCOL), 1),
+    )
+
+
+def decode(header: bytes, payload: bytes):
+    return pickle.loads(header), pickle.loads(zlib.decompress(payload))
+
+
+# --------------------------------------------------------------------------- #
+# Timers / throughput accounting
+# --------------------------------------------------------------------------- #
+class ExecutionTimer:
+    """Named wall-clock timer with throughput tracking.
+
+    ``with timer.timer("learner-throughput", check_throughput=True): ...``
+    records elapsed seconds and, when requested, transitions/sec computed from
+    ``num_transition`` (reference: utils/utils.py:167-189, learner.py:34-36).
+    """
+
+    def __init__(self, num_transition: int = 0, maxlen: int = 100):
+        self.num_transition = num_transition
+        self.timer_dict: dict[str, deque] = {}
+        self.throughput_dict: dict[str, deque] = {}
+        self.maxlen = maxlen
+
+    @contextmanager
+    def timer(self, name: str, check_throughput: bool = False):
+        start = time.perf_counter()
+        yield self
+        elapsed = time.perf_counter() - start
+        self.timer_dict.setdefault(name, deque(maxlen=self.maxlen)).append(elapsed)
+        if check_throughput and elapsed > 0:
+            self.throughput_dict.setdefault(name, deque(maxlen=self.maxlen)).append(
+                self.num_transition / elapsed
+            )
+
+
+# --------------------------------------------------------------------------- #
+# Tensor helpers
+# --------------------------------------------------------------------------- #
+def to_torch(x: np.ndarray) -> torch.Tensor:
+    return torch.from_numpy(np.ascontiguousarray(x)).float()
+
+
+def mul(shape) -> int:
+    out = 1
+    for s in shape:
+        out *= int(s)
+    return out
+
+
+def make_gpu_batch(*args, device):
+    return tuple(t.to(device, non_blocking=True) for t in args)
+
+
+def obs_preprocess(obs, need_conv: bool = False) -> torch.Tensor:
+    """Flatten raw observation to a (1, feat) float tensor.
+
+    Conv/pixel path is not supported (same restriction the reference asserts,
+    main.py:95); vector observations only.
+    """
+    assert not need_conv, "conv observation path not supported"
+    arr = np.asarray(obs, dtype=np.float32).reshape(1, -1)
+    return torch.from_numpy(arr)
+
+
+def extract_file_num(path) -> int:
+    """Trailing integer of a checkpoint filename ``{algo}_{idx}.pt`` (-1 if none)."""
+    stem = Path(path).stem
+    tail = stem.split("_")[-1]
+    try:
+        return int(tail)
+    except ValueError:
+        return -1
+
+
+# --------------------------------------------------------------------------- #
+# Process management
+# --------------------------------------------------------------------------- #
+ChildProcesses: list = []  # registry of mp.Process handles for cleanup
+
+
+def register_child(p):
+    ChildProcesses.append(p)
+    return p
+
+
+def terminate_children(timeout: float = 5.0):
+    for p in ChildProcesses:
+        if p.is_alive():
+            p.terminate()
+    deadline = time.time() + timeout
+    for p in ChildProcesses:
+        p.join(max(0.0, deadline - time.time()))
+    for p in ChildProcesses:
+        if p.is_alive():
+            p.kill()
+    ChildProcesses.clear()
+
+
+def kill_process_tree(pid: int | None = None):
+    """Kill a process and its children (psutil if available, else best effort)."""
+    pid = pid or os.getpid()
+    try:
+        import psutil
+
+        parent = psutil.Process(pid)
+        for child in parent.children(recursive=True):
+            try:
+                child.send_signal(signal.SIGTERM)
+            except psutil.NoSuchProcess:
+                pass
+    except Exception:
+        pass
+
+
+def select_least_used_gpu() -> int:
+    """Pick the CUDA/HIP device with the least reserved memory
+    (reference behavior: utils/utils.py:106-117)."""
+    if not torch.cuda.is_available():
+        return 0
+    n = torch.cuda.device_count()
+    if n <= 1:
+        return 0
+    reserved = [torch.cuda.memory_reserved(i) for i in range(n)]
+    return int(np.argmin(reserved))
+
+
+def save_error_log(role: str, text: str, result_dir: str | None = None):
+    """Append a crash report under logs/<role>/ (reference: utils.py:192-198)."""
+    log_dir = REPO_ROOT / "logs" / role
+    log_dir.mkdir(parents=True, exist_ok=True)
+    fname = log_dir / f"error_log_{time.strftime('%Y%m%d_%H%M%S')}_{os.getpid()}.txt"
+    with open(fname, "a") as f:
+        f.write(text + "\n")
+    return str(fname)
+
+
+class counted:
+    """Decorator counting invocations (reference: utils/utils.py:120-127)."""
+
+    def __init__(self, fn):
+        self.fn = fn
+        self.calls = 0
+
+    def __get__(self, obj, objtype=None):
+        import functools
+
+        return functools.partial(self.__call__, obj)
+
+    def __call__(self, *args, **kwargs):
+        self.calls += 1
+        return self.fn(*args, **kwargs)
+
+
+# Field names of a trajectory batch, in canonical storage order.
+DataFrameKeyword = [
+    "obs_batch",
+    "act_batch",
+    "rew_batch",
+    "logits_batch",
+    "log_prob_batch",
+    "is_fir_batch",
+    "hx_batch",
+    "cx_batch",
+]

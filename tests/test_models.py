@@ -1,0 +1,137 @@
+"""Model zoo tests: SeqLSTMCore parity vs torch.nn.LSTMCell, shapes of all
+13 model classes, act/forward contracts."""
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from pdrl_amd.networks import (
+    MlpLSTMActor,
+    MlpLSTMActorContinuous,
+    MlpLSTMBase,
+    MlpLSTMContinuous,
+    MlpLSTMCritic,
+    MlpLSTMCriticContinuous,
+    MlpLSTMDoubleCritic,
+    MlpLSTMDoubleCriticContinuous,
+    MlpLSTMSeperate,
+    MlpLSTMSeperateContinuous,
+    MlpLSTMSingle,
+    MlpLSTMSingleContinuous,
+    SeqLSTMCore,
+    categorical_stats,
+)
+
+B, S, F_DIM, H, A = 6, 5, 4, 64, 2
+
+
+def test_core_matches_nn_lstmcell():
+    """The explicit gate math must equal torch's LSTMCell given the same
+    weights (this is the oracle the HIP kernel is later tested against)."""
+    torch.manual_seed(0)
+    core = SeqLSTMCore(F_DIM, H, {"out": 3})
+    cell = nn.LSTMCell(H, H)
+    with torch.no_grad():
+        cell.weight_ih.copy_(core.w_ih.t())
+        cell.weight_hh.copy_(core.w_hh.t())
+        cell.bias_ih.copy_(core.b_g)
+        cell.bias_hh.zero_()
+
+    x = torch.randn(B, S, F_DIM)
+    hx, cx = torch.randn(B, H) * 0.1, torch.randn(B, H) * 0.1
+    outs, h_end, c_end = core(x, hx, cx)
+
+    xb = F.relu(x.reshape(B * S, F_DIM) @ core.body_w + core.body_b).view(B, S, H)
+    h, c = hx, cx
+    for t in range(S):
+        h, c = cell(xb[:, t], (h, c))
+    torch.testing.assert_close(h_end, h, rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(c_end, c, rtol=1e-5, atol=1e-5)
+    w, b = core.head_params("out")
+    torch.testing.assert_close(outs["out"][:, -1], h @ w + b, rtol=1e-5, atol=1e-5)
+
+
+def test_core_backward_flows():
+    core = SeqLSTMCore(F_DIM, H, {"v": 1})
+    x = torch.randn(B, S, F_DIM, requires_grad=True)
+    outs, _, _ = core(x, torch.zeros(B, H), torch.zeros(B, H))
+    outs["v"].sum().backward()
+    assert x.grad is not None and torch.isfinite(x.grad).all()
+    for p in core.parameters():
+        assert p.grad is not None and torch.isfinite(p.grad).all()
+
+
+def test_base_act_and_forward_shapes():
+    m = MlpLSTMBase(F_DIM, A, S, H)
+    obs = torch.randn(1, F_DIM)
+    action, logits, log_prob, (hx, cx) = m.act(obs, (torch.zeros(1, H), torch.zeros(1, H)))
+    assert action.shape == (1, 1) and logits.shape == (1, A)
+    assert log_prob.shape == (1, 1) and hx.shape == (1, H)
+    assert action.item() in (0, 1)
+
+    obs_b = torch.randn(B, S, F_DIM)
+    acts = torch.randint(0, A, (B, S, 1)).float()
+    lg, lp, ent, val = m(obs_b, (torch.zeros(B, H), torch.zeros(B, H)), acts)
+    assert lg.shape == (B, S, A) and lp.shape == (B, S, 1)
+    assert ent.shape == (B, S, 1) and val.shape == (B, S, 1)
+    assert (ent >= 0).all()
+
+
+def test_categorical_stats_match_distributions():
+    torch.manual_seed(1)
+    logits = torch.randn(B, S, A)
+    acts = torch.randint(0, A, (B, S, 1))
+    lp, ent = categorical_stats(logits, acts)
+    d = torch.distributions.Categorical(logits=logits)
+    torch.testing.assert_close(lp.squeeze(-1), d.log_prob(acts.squeeze(-1)), rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(ent.squeeze(-1), d.entropy(), rtol=1e-5, atol=1e-6)
+
+
+def test_continuous_models():
+    m = MlpLSTMContinuous(2, 1, S, H)
+    a, logits, lp, (hx, cx) = m.act(torch.randn(1, 2), (torch.zeros(1, H), torch.zeros(1, H)))
+    assert a.shape == (1, 1) and logits.shape == (1, 2)
+    lg, lp2, ent, val = m(torch.randn(B, S, 2), (torch.zeros(B, H), torch.zeros(B, H)),
+                          torch.randn(B, S, 1))
+    assert lp2.shape == (B, S, 1) and val.shape == (B, S, 1)
+
+
+def test_sac_discrete_stack():
+    m = MlpLSTMSeperate(F_DIM, A, S, H)
+    probs, log_probs = m.actor(torch.randn(B, S, F_DIM), (torch.zeros(B, H), torch.zeros(B, H)))
+    assert probs.shape == (B, S, A)
+    torch.testing.assert_close(probs.sum(-1), torch.ones(B, S), rtol=1e-5, atol=1e-5)
+    q1, q2 = m.critic(torch.randn(B, S, F_DIM), (torch.zeros(B, H), torch.zeros(B, H)))
+    assert q1.shape == (B, S, A)
+    # twin critics are independent networks
+    assert not torch.allclose(q1, q2)
+
+
+def test_sac_continuous_stack():
+    m = MlpLSTMSeperateContinuous(2, 1, S, H)
+    act, lp = m.actor(torch.randn(B, S, 2), (torch.zeros(B, H), torch.zeros(B, H)))
+    assert act.shape == (B, S, 1) and (act.abs() <= 1).all()
+    assert lp.shape == (B, S, 1)
+    q1, q2 = m.critic(torch.randn(B, S, 2), act, (torch.zeros(B, H), torch.zeros(B, H)))
+    assert q1.shape == (B, S, 1)
+    # reparameterized action carries gradient
+    act.sum().backward()
+    assert any(p.grad is not None for p in m.actor.parameters())
+
+
+def test_single_wrapper_aliases_critic():
+    m = MlpLSTMSingle(F_DIM, A, S, H)
+    assert m.critic is m.actor  # single-network algos share the torso
+
+
+def test_all_13_classes_instantiate():
+    classes = [
+        MlpLSTMBase, MlpLSTMContinuous, MlpLSTMActor, MlpLSTMActorContinuous,
+        MlpLSTMCritic, MlpLSTMCriticContinuous, MlpLSTMDoubleCritic,
+        MlpLSTMDoubleCriticContinuous, MlpLSTMSingle, MlpLSTMSingleContinuous,
+        MlpLSTMSeperate, MlpLSTMSeperateContinuous, SeqLSTMCore,
+    ]
+    for cls in classes:
+        if cls is SeqLSTMCore:
+            cls(F_DIM, H, {"o": 1})
+        else:
+            cls(F_DIM, A, S, H)
