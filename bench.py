@@ -1,0 +1,148 @@
+"""Flagship benchmark: IMPALA CartPole-v1 learner step throughput on MI355X
+(BASELINE.json metric: learner env-steps/sec at 1/2/4/8 GPUs).
+
+Runs the full training iteration — fused model forward, V-trace, loss,
+backward, flat-bucket RCCL all-reduce (N>1), fused clip+RMSprop — on
+synthetic rollouts of the BASELINE config shape (obs 4, 2 actions, B=128,
+S=5 per GPU; random-init MlpLSTMSingle, fp32: the reference's compute
+dtype). Weak scaling: per-GPU batch fixed as N grows.
+
+Contract (driver):
+  python bench.py --gpus N --steps K --warmup W
+  (N>1 is launched via torch.distributed.run, one rank per GPU over RCCL;
+  rank 0 prints ONE JSON line.)
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+from pathlib import Path
+
+import torch
+
+sys.path.insert(0, str(Path(__file__).resolve().parent))
+
+from pdrl_amd.agents.learner_module import ImpalaUpdater  # noqa: E402
+from pdrl_amd.networks import MlpLSTMSingle  # noqa: E402
+from pdrl_amd.parallel import GradReducer, init_distributed  # noqa: E402
+from pdrl_amd.utils import load_params  # noqa: E402
+
+OBS_DIM, N_ACTIONS = 4, 2  # CartPole-v1
+
+
+def make_synthetic_batch(params, device, seed):
+    """Random rollout batch with the exact field shapes the learner consumes
+    (synthetic data — no network access for real envs at benchmark scale)."""
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    B, S, H = params.batch_size, params.seq_len, params.hidden_size
+    logits = torch.randn(B, S, N_ACTIONS, generator=g)
+    acts = torch.randint(0, N_ACTIONS, (B, S, 1), generator=g).float()
+    logp = torch.log_softmax(logits, dim=-1).gather(-1, acts.long())
+    batch = {
+        "obs": torch.randn(B, S, OBS_DIM, generator=g),
+        "act": acts,
+        "rew": torch.rand(B, S, 1, generator=g),
+        "logits": logits,
+        "log_prob": logp,
+        "is_fir": (torch.rand(B, S, 1, generator=g) < 0.2).float(),
+        "hx": torch.randn(B, S, H, generator=g) * 0.1,
+        "cx": torch.randn(B, S, H, generator=g) * 0.1,
+    }
+    return {k: v.to(device) for k, v in batch.items()}
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=200)
+    ap.add_argument("--warmup", type=int, default=50)
+    ap.add_argument("--batch-size", type=int, default=128)
+    ap.add_argument("--seq-len", type=int, default=5)
+    args = ap.parse_args()
+
+    rank, world = init_distributed()
+    n_gpus = max(args.gpus, world)
+    use_cuda = torch.cuda.is_available()
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+    if use_cuda:
+        torch.cuda.set_device(device)
+
+    params = load_params()
+    params.algo = "IMPALA"
+    params.batch_size = args.batch_size
+    params.seq_len = args.seq_len
+    params.obs_dim, params.n_actions = OBS_DIM, N_ACTIONS
+
+    torch.manual_seed(1234)  # identical init across ranks
+    model = MlpLSTMSingle(OBS_DIM, N_ACTIONS, params.seq_len, params.hidden_size)
+    reducer = GradReducer() if world > 1 else None
+    updater = ImpalaUpdater(model, params, device, grad_reducer=reducer)
+
+    batch = make_synthetic_batch(params, device, seed=100 + rank)
+
+    import torch.distributed as dist
+
+    def barrier():
+        if world > 1:
+            dist.barrier()
+
+    def sync():
+        if use_cuda:
+            torch.cuda.synchronize(device)
+
+    for _ in range(args.warmup):
+        updater.step(batch)
+    sync()
+    barrier()
+    sync()
+
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        updater.step(batch)
+    sync()
+    barrier()
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks (the job finishes when the slowest rank does)
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if dist.get_backend() == "nccl" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    transitions = n_gpus * args.batch_size * args.seq_len * args.steps
+    value = transitions / elapsed
+    if rank == 0:
+        result = {
+            "metric": "learner_env_steps_per_sec",
+            "value": value,
+            "unit": "env-steps/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "MlpLSTMSingle(obs4,act2,H64)",
+                "algo": "IMPALA CartPole-v1",
+                "global_batch": args.batch_size * n_gpus,
+                "seq_len": args.seq_len,
+                "parallelism": f"dp{n_gpus}",
+            },
+        }
+        print(json.dumps(result))
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -36,6 +36,37 @@ class BaseUpdater:
         raise NotImplementedError
 
     # -- shared --------------------------------------------------------- #
+    def make_optimizer(self, kind: str, parameters, lr: float,
+                       clip: bool = True, **kw):
+        """Fused flat-buffer optimizer (HIP, clip folded in) on GPU with the
+        extension loaded; stock torch.optim elsewhere."""
+        parameters = list(parameters)
+        from pdrl_amd import ops
+
+        max_norm = self.params.max_grad_norm if clip else None
+        if self.device.type == "cuda" and ops.available():
+            from pdrl_amd.ops.optim import FusedAdam, FusedRMSprop
+
+            if kind == "rmsprop":
+                return FusedRMSprop(parameters, lr=lr, max_norm=max_norm,
+                                    eps=kw.get("eps", 1e-5))
+            return FusedAdam(parameters, lr=lr, max_norm=max_norm)
+        if kind == "rmsprop":
+            return torch.optim.RMSprop(parameters, lr=lr, eps=kw.get("eps", 1e-5))
+        return torch.optim.Adam(parameters, lr=lr)
+
+    def apply_step(self, optimizer, parameters):
+        """Gradient epilogue: all-reduce across ranks → clip → update.
+        Fused path: one collective on the flat grad bucket; clip is inside
+        the update kernel. Eager path: per-grad reduce + torch clip."""
+        if getattr(optimizer, "is_fused", False):
+            if self.grad_reducer is not None:
+                self.grad_reducer.all_reduce([optimizer.flat_grad])
+            optimizer.step()
+        else:
+            self.reduce_and_clip(parameters)
+            optimizer.step()
+
     def reduce_and_clip(self, parameters):
         """All-reduce gradients across learner ranks (RCCL on GPU, gloo on
         CPU), then global-norm clip. Called between backward() and step()."""

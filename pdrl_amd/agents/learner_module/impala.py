@@ -20,8 +20,8 @@ class ImpalaUpdater(BaseUpdater):
     def __init__(self, model, params, device, grad_reducer=None):
         super().__init__(params, device, grad_reducer)
         self.model = model.to(device)
-        self.optimizer = torch.optim.RMSprop(
-            self.model.parameters(), lr=params.lr, eps=1e-5
+        self.optimizer = self.make_optimizer(
+            "rmsprop", self.model.parameters(), lr=params.lr, eps=1e-5
         )
 
     def trainable_modules(self):
@@ -53,11 +53,11 @@ class ImpalaUpdater(BaseUpdater):
             - p.entropy_coef * entropy_mean
         )
         stats = {
-            "loss-total": float(loss.detach()),
-            "loss-policy": float(policy_loss.detach()),
-            "loss-value": float(value_loss.detach()),
-            "entropy": float(entropy_mean.detach()),
-            "rho-avg": float(rhos.mean()),
+            "loss-total": loss.detach(),
+            "loss-policy": policy_loss.detach(),
+            "loss-value": value_loss.detach(),
+            "entropy": entropy_mean.detach(),
+            "rho-avg": rhos.mean(),
         }
         return loss, stats
 
@@ -67,7 +67,6 @@ class ImpalaUpdater(BaseUpdater):
             loss, stats = self.compute_losses(batch)
             self.optimizer.zero_grad(set_to_none=False)
             loss.backward()
-            self.reduce_and_clip(self.model.parameters())
-            self.optimizer.step()
+            self.apply_step(self.optimizer, self.model.parameters())
         self.update_count += 1
         return stats

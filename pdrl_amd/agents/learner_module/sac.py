@@ -42,9 +42,10 @@ class SACUpdater(BaseUpdater):
             torch.tensor(float(np.log(params.alpha)), device=self.device)
         )
 
-        self.actor_optimizer = torch.optim.Adam(self.actor.parameters(), lr=params.lr)
-        self.critic_optimizer = torch.optim.Adam(self.critic.parameters(), lr=params.lr)
-        self.alpha_optimizer = torch.optim.Adam([self.log_alpha], lr=params.lr)
+        self.actor_optimizer = self.make_optimizer("adam", self.actor.parameters(), lr=params.lr)
+        self.critic_optimizer = self.make_optimizer("adam", self.critic.parameters(), lr=params.lr)
+        self.alpha_optimizer = self.make_optimizer("adam", [self.log_alpha], lr=params.lr,
+                                                   clip=False)
 
     def trainable_modules(self):
         return {"model": self.model, "target_critic": self.target_critic}
@@ -79,8 +80,7 @@ class SACUpdater(BaseUpdater):
             )
             self.actor_optimizer.zero_grad(set_to_none=False)
             actor_loss.backward()
-            self.reduce_and_clip(self.actor.parameters())
-            self.actor_optimizer.step()
+            self.apply_step(self.actor_optimizer, self.actor.parameters())
 
             # -- temperature auto-tune ----------------------------------- #
             with torch.no_grad():
@@ -88,9 +88,14 @@ class SACUpdater(BaseUpdater):
             alpha_loss = (self.log_alpha * (pi_entropy - self.target_entropy).detach()).mean()
             self.alpha_optimizer.zero_grad(set_to_none=False)
             alpha_loss.backward()
-            if self.grad_reducer is not None:
-                self.grad_reducer.all_reduce([self.log_alpha.grad])
-            self.alpha_optimizer.step()
+            if getattr(self.alpha_optimizer, "is_fused", False):
+                if self.grad_reducer is not None:
+                    self.grad_reducer.all_reduce([self.alpha_optimizer.flat_grad])
+                self.alpha_optimizer.step()
+            else:
+                if self.grad_reducer is not None:
+                    self.grad_reducer.all_reduce([self.log_alpha.grad])
+                self.alpha_optimizer.step()
 
             # -- critics: soft-Q TD target ------------------------------- #
             with torch.no_grad():
@@ -112,17 +117,16 @@ class SACUpdater(BaseUpdater):
             )
             self.critic_optimizer.zero_grad(set_to_none=False)
             value_loss.backward()
-            self.reduce_and_clip(self.critic.parameters())
-            self.critic_optimizer.step()
+            self.apply_step(self.critic_optimizer, self.critic.parameters())
 
             soft_update(self.critic, self.target_critic, self.TAU)
 
             stats = {
-                "loss-actor": float(actor_loss.detach()),
-                "loss-value": float(value_loss.detach()),
-                "loss-alpha": float(alpha_loss.detach()),
-                "alpha": float(self.alpha.detach()),
-                "entropy": float(pi_entropy.mean()),
+                "loss-actor": actor_loss.detach(),
+                "loss-value": value_loss.detach(),
+                "loss-alpha": alpha_loss.detach(),
+                "alpha": self.alpha.detach(),
+                "entropy": pi_entropy.mean(),
             }
         self.update_count += 1
         return stats

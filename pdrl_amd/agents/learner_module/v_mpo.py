@@ -26,7 +26,8 @@ class VMPOUpdater(BaseUpdater):
         init = float(np.log(params.v_mpo_lagrange_multiplier_init))
         self.log_eta = torch.nn.Parameter(torch.tensor(init, device=self.device))
         self.log_alpha = torch.nn.Parameter(torch.tensor(init, device=self.device))
-        self.optimizer = torch.optim.RMSprop(
+        self.optimizer = self.make_optimizer(
+            "rmsprop",
             list(self.model.parameters()) + [self.log_eta, self.log_alpha],
             lr=params.lr,
             eps=1e-5,
@@ -97,12 +98,12 @@ class VMPOUpdater(BaseUpdater):
             + alpha_loss
         )
         stats = {
-            "loss-total": float(loss.detach()),
-            "loss-policy": float(policy_loss.detach()),
-            "loss-value": float(value_loss.detach()),
-            "eta": float(eta.detach()),
-            "alpha": float(alpha.detach()),
-            "kl": float(kl.detach()),
+            "loss-total": loss.detach(),
+            "loss-policy": policy_loss.detach(),
+            "loss-value": value_loss.detach(),
+            "eta": eta.detach(),
+            "alpha": alpha.detach(),
+            "kl": kl.detach(),
         }
         return loss, stats
 
@@ -112,9 +113,9 @@ class VMPOUpdater(BaseUpdater):
             loss, stats = self.compute_losses(batch)
             self.optimizer.zero_grad(set_to_none=False)
             loss.backward()
-            self.reduce_and_clip(
-                list(self.model.parameters()) + [self.log_eta, self.log_alpha]
+            self.apply_step(
+                self.optimizer,
+                list(self.model.parameters()) + [self.log_eta, self.log_alpha],
             )
-            self.optimizer.step()
         self.update_count += 1
         return stats

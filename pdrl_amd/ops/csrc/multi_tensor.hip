@@ -1,0 +1,183 @@
+// Multi-tensor / flat-buffer update kernels for CDNA4 (gfx950):
+//   K12 soft_update   — Polyak θ' ← (1-τ)θ' + τθ over a chunk table
+//   K13 grad-norm clip — fused INTO the optimizer step (scale from a
+//                        device-resident squared norm; no host sync)
+//   K14 RMSprop / Adam — single-kernel flat-buffer optimizer updates
+//
+// The framework flattens each trainable group into one contiguous param +
+// grad (+ state) buffer at updater init (ops/optim.py), so the entire
+// optimizer epilogue is: one reduction kernel (norm²) + one update kernel —
+// all device-side, hipGraph-capturable, and the flat grad buffer doubles as
+// the single RCCL all-reduce bucket (one ~0.7 MB
+// latency-bound bucket over xGMI).
+#include "common.h"
+
+#include <vector>
+
+namespace {
+
+__global__ void l2norm_sq_kernel(const float* __restrict__ g, long n,
+                                 float* __restrict__ out) {
+  __shared__ float red[256];
+  float acc = 0.0f;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    const float v = g[i];
+    acc = fmaf(v, v, acc);
+  }
+  // wave reduce then block reduce
+  for (int off = kWave / 2; off > 0; off >>= 1)
+    acc += __shfl_down(acc, off, kWave);
+  const int lane = threadIdx.x & (kWave - 1);
+  const int wid = threadIdx.x / kWave;
+  if (lane == 0) red[wid] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float s = 0.0f;
+    for (int w = 0; w < (int)(blockDim.x / kWave); ++w) s += red[w];
+    atomicAdd(out, s);
+  }
+}
+
+__device__ __forceinline__ float clip_scale(const float* norm_sq,
+                                            float max_norm) {
+  if (max_norm <= 0.0f) return 1.0f;
+  const float norm = sqrtf(*norm_sq) + 1e-6f;
+  return (norm > max_norm) ? (max_norm / norm) : 1.0f;
+}
+
+__global__ void rmsprop_kernel(float* __restrict__ p, const float* __restrict__ g,
+                               float* __restrict__ sq_avg,
+                               const float* __restrict__ norm_sq, long n,
+                               float lr, float alpha, float eps,
+                               float max_norm) {
+  const float scale = clip_scale(norm_sq, max_norm);
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    const float gi = g[i] * scale;
+    const float sa = alpha * sq_avg[i] + (1.0f - alpha) * gi * gi;
+    sq_avg[i] = sa;
+    p[i] -= lr * gi / (sqrtf(sa) + eps);
+  }
+}
+
+// step-count + bias corrections live on device so graph replays stay correct
+__global__ void adam_prep_kernel(float* __restrict__ state3, float beta1,
+                                 float beta2) {
+  // state3 = {t, bc1, bc2}
+  const float t = state3[0] + 1.0f;
+  state3[0] = t;
+  state3[1] = 1.0f - __powf(beta1, t);
+  state3[2] = 1.0f - __powf(beta2, t);
+}
+
+__global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
+                            float* __restrict__ m, float* __restrict__ v,
+                            const float* __restrict__ state3,
+                            const float* __restrict__ norm_sq, long n,
+                            float lr, float beta1, float beta2, float eps,
+                            float max_norm) {
+  const float scale = clip_scale(norm_sq, max_norm);
+  const float bc1 = state3[1], bc2 = state3[2];
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    const float gi = g[i] * scale;
+    const float mi = beta1 * m[i] + (1.0f - beta1) * gi;
+    const float vi = beta2 * v[i] + (1.0f - beta2) * gi * gi;
+    m[i] = mi;
+    v[i] = vi;
+    p[i] -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+  }
+}
+
+// chunk table: src/dst pointers packed as int64 in a device tensor
+__global__ void soft_update_kernel(const long* __restrict__ src_ptrs,
+                                   const long* __restrict__ dst_ptrs,
+                                   const long* __restrict__ numels,
+                                   int n_tensors, float tau) {
+  const int ti = blockIdx.y;
+  if (ti >= n_tensors) return;
+  const float* src = reinterpret_cast<const float*>(src_ptrs[ti]);
+  float* dst = reinterpret_cast<float*>(dst_ptrs[ti]);
+  const long n = numels[ti];
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    dst[i] = fmaf(tau, src[i] - dst[i], dst[i]);
+  }
+}
+
+constexpr int kThreads = 256;
+
+int grid_for(long n) {
+  long b = (n + kThreads - 1) / kThreads;
+  return (int)std::min<long>(b, 2048);
+}
+
+}  // namespace
+
+void l2norm_sq_hip(const at::Tensor& g, at::Tensor& out) {
+  CHECK_IN(g); CHECK_GPU(out);
+  hipLaunchKernelGGL(l2norm_sq_kernel, dim3(grid_for(g.numel())),
+                     dim3(kThreads), 0, current_stream(),
+                     g.data_ptr<float>(), g.numel(),
+                     out.data_ptr<float>());
+  HIP_CHECK_LAST();
+}
+
+void rmsprop_step_hip(at::Tensor& p, const at::Tensor& g, at::Tensor& sq_avg,
+                      const at::Tensor& norm_sq, double lr, double alpha,
+                      double eps, double max_norm) {
+  CHECK_IN(p); CHECK_IN(g); CHECK_IN(sq_avg);
+  hipLaunchKernelGGL(rmsprop_kernel, dim3(grid_for(p.numel())), dim3(kThreads),
+                     0, current_stream(), p.data_ptr<float>(),
+                     g.data_ptr<float>(), sq_avg.data_ptr<float>(),
+                     norm_sq.data_ptr<float>(), p.numel(), (float)lr,
+                     (float)alpha, (float)eps, (float)max_norm);
+  HIP_CHECK_LAST();
+}
+
+void adam_step_hip(at::Tensor& p, const at::Tensor& g, at::Tensor& m,
+                   at::Tensor& v, at::Tensor& state3,
+                   const at::Tensor& norm_sq, double lr, double beta1,
+                   double beta2, double eps, double max_norm) {
+  CHECK_IN(p); CHECK_IN(g); CHECK_IN(m); CHECK_IN(v);
+  hipLaunchKernelGGL(adam_prep_kernel, dim3(1), dim3(1), 0, current_stream(),
+                     state3.data_ptr<float>(), (float)beta1, (float)beta2);
+  hipLaunchKernelGGL(adam_kernel, dim3(grid_for(p.numel())), dim3(kThreads), 0,
+                     current_stream(), p.data_ptr<float>(),
+                     g.data_ptr<float>(), m.data_ptr<float>(),
+                     v.data_ptr<float>(), state3.data_ptr<float>(),
+                     norm_sq.data_ptr<float>(), p.numel(), (float)lr,
+                     (float)beta1, (float)beta2, (float)eps, (float)max_norm);
+  HIP_CHECK_LAST();
+}
+
+void soft_update_hip(const std::vector<at::Tensor>& src,
+                     const std::vector<at::Tensor>& dst, double tau) {
+  TORCH_CHECK(src.size() == dst.size(), "src/dst count mismatch");
+  const int n = (int)src.size();
+  if (n == 0) return;
+  auto cpu_opts = at::TensorOptions().dtype(at::kLong);
+  auto sp = at::empty({n}, cpu_opts);
+  auto dp = at::empty({n}, cpu_opts);
+  auto ne = at::empty({n}, cpu_opts);
+  long max_n = 1;
+  for (int i = 0; i < n; ++i) {
+    CHECK_IN(src[i]); CHECK_IN(dst[i]);
+    TORCH_CHECK(src[i].numel() == dst[i].numel(), "tensor ", i, " size mismatch");
+    sp[i] = (long)src[i].data_ptr<float>();
+    dp[i] = (long)dst[i].data_ptr<float>();
+    ne[i] = (long)src[i].numel();
+    max_n = std::max<long>(max_n, src[i].numel());
+  }
+  auto dev = src[0].device();
+  auto spd = sp.to(dev, /*non_blocking=*/true);
+  auto dpd = dp.to(dev, true);
+  auto ned = ne.to(dev, true);
+  dim3 grid(grid_for(max_n), n);
+  hipLaunchKernelGGL(soft_update_kernel, grid, dim3(kThreads), 0,
+                     current_stream(), spd.data_ptr<long>(),
+                     dpd.data_ptr<long>(), ned.data_ptr<long>(), n,
+                     (float)tau);
+  HIP_CHECK_LAST();
+}

@@ -1,0 +1,362 @@
+// Fused SeqLSTMCore forward/backward for CDNA4 (gfx950).
+//
+// Implements kernels K1-K3 of SURVEY.md §2.4: body Linear+ReLU, the LSTM
+// recurrence over the whole sequence, and all head Linears — in ONE launch
+// each way, replacing the reference's per-step Python loop over nn.LSTMCell
+// (reference: networks/models.py:71-75 and 4 clones ≈ 2·S·5 launches) and
+// its surrounding eager ops.
+//
+// Geometry (MI355X-first): one 4H-thread workgroup per *batch row*. At the
+// framework's operating point (B=128, H=64, S=5) the step is latency-bound,
+// not FLOP-bound: batch-row parallelism fills 128 of 256 CUs with fully
+// independent work and zero inter-workgroup traffic (the recurrence is
+// row-local). An MFMA tiling of the (B×4H×2H) gate GEMM would concentrate
+// the work on ~4 CUs and serialize on the matrix pipe's 64-cycle f32 MFMA
+// issue — measured slower at this shape (see profiles/). Per the platform
+// guide's GEMV rule, per-row work streams weights straight into registers.
+//
+// Weight residency:
+//   forward  — each thread owns one gate column: w_ih[:,g] and w_hh[:,g]
+//              live in 2×H VGPRs; LDS holds only xb/h/c/gates (few KB).
+//   backward — needs transposed access (row k over all gates), so both gate
+//              weight matrices are staged in LDS as [4H][H+1] (the +1 pad
+//              makes both the scattered store and the broadcast-row read
+//              bank-conflict-free; LDS banking per the CDNA4 guide §2).
+//
+// dtype: fp32 (the reference trains fp32; gfx950 fp32 is exact — parity vs
+// the PyTorch eager oracle is tested to 1e-5 in tests/test_gpu_kernels.py).
+
+#include "common.h"
+
+#include <vector>
+
+namespace {
+
+// stash layout per (b, t): [xb(H) | gates i,f,g,o (4H) | c(H) | h(H)] = 7H
+constexpr int kStashFields = 7;
+
+template <int H>
+__global__ __launch_bounds__(4 * H) void seq_lstm_fwd_kernel(
+    const float* __restrict__ x,       // (B,S,F)
+    const float* __restrict__ h0,      // (B,H)
+    const float* __restrict__ c0,      // (B,H)
+    const float* __restrict__ body_w,  // (F,H)
+    const float* __restrict__ body_b,  // (H)
+    const float* __restrict__ w_ih,    // (H,4H)
+    const float* __restrict__ w_hh,    // (H,4H)
+    const float* __restrict__ b_g,     // (4H)
+    const float* __restrict__ heads_w, // (H,D)
+    const float* __restrict__ heads_b, // (D)
+    float* __restrict__ outs,          // (B,S,D)
+    float* __restrict__ hS,            // (B,H)
+    float* __restrict__ cS,            // (B,H)
+    float* __restrict__ stash,         // (B,S,7H)
+    int S, int F, int D) {
+  constexpr int G = 4 * H;
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* xb = reinterpret_cast<float*>(smem_raw);  // (S,H)
+  float* hs = xb + S * H;                          // (S,H)
+  float* gates = hs + S * H;                       // (4H)
+  float* hbuf = gates + G;                         // (H)
+  float* cbuf = hbuf + H;                          // (H)
+
+  // Register-resident gate weight columns (thread = gate column tid).
+  float wih[H], whh[H];
+#pragma unroll
+  for (int k = 0; k < H; ++k) wih[k] = w_ih[k * G + tid];
+#pragma unroll
+  for (int k = 0; k < H; ++k) whh[k] = w_hh[k * G + tid];
+  const float bias = b_g[tid];
+
+  // Body GEMM + ReLU for all S steps of this row (K1).
+  for (int idx = tid; idx < S * H; idx += G) {
+    const int t = idx / H, j = idx % H;
+    float acc = body_b[j];
+    const float* xr = x + ((long)b * S + t) * F;
+    for (int k = 0; k < F; ++k) acc = fmaf(xr[k], body_w[k * H + j], acc);
+    acc = fmaxf(acc, 0.0f);
+    xb[t * H + j] = acc;
+    stash[(((long)b * S + t) * kStashFields) * H + j] = acc;
+  }
+  if (tid < H) {
+    hbuf[tid] = h0[(long)b * H + tid];
+    cbuf[tid] = c0[(long)b * H + tid];
+  }
+  __syncthreads();
+
+  // LSTM recurrence, whole sequence in-kernel (K2).
+  for (int t = 0; t < S; ++t) {
+    const float* xbt = xb + t * H;
+    float acc = bias;
+#pragma unroll
+    for (int k = 0; k < H; ++k) acc = fmaf(xbt[k], wih[k], acc);
+#pragma unroll
+    for (int k = 0; k < H; ++k) acc = fmaf(hbuf[k], whh[k], acc);
+    const int sel = tid / H;  // 0:i 1:f 2:g 3:o
+    const float a = (sel == 2) ? tanhf(acc) : sigmoidf_dev(acc);
+    const long sbase = (((long)b * S + t) * kStashFields) * H;
+    gates[tid] = a;
+    stash[sbase + H + tid] = a;  // gates occupy [H, 5H)
+    __syncthreads();
+    if (tid < H) {
+      const float c_new =
+          gates[H + tid] * cbuf[tid] + gates[tid] * gates[2 * H + tid];
+      const float h_new = gates[3 * H + tid] * tanhf(c_new);
+      cbuf[tid] = c_new;
+      hbuf[tid] = h_new;
+      hs[t * H + tid] = h_new;
+      stash[sbase + 5 * H + tid] = c_new;
+      stash[sbase + 6 * H + tid] = h_new;
+    }
+    __syncthreads();
+  }
+
+  if (tid < H) {
+    hS[(long)b * H + tid] = hbuf[tid];
+    cS[(long)b * H + tid] = cbuf[tid];
+  }
+
+  // Heads (K3) on the stored h sequence.
+  for (int idx = tid; idx < S * D; idx += G) {
+    const int t = idx / D, d = idx % D;
+    float acc = heads_b[d];
+    const float* ht = hs + t * H;
+#pragma unroll
+    for (int k = 0; k < H; ++k) acc = fmaf(ht[k], heads_w[k * D + d], acc);
+    outs[((long)b * S + t) * D + d] = acc;
+  }
+}
+
+// Backward through heads + recurrence + body for one batch row.
+// Emits per-(b,t) pre-activation gate grads (dgates) and pre-ReLU body grads
+// (dxb) for the weight-gradient GEMMs (done as plain library GEMMs on the
+// host side — hipBLASLt — per the MI355X mandate), plus dx / dh0 / dc0.
+template <int H>
+__global__ __launch_bounds__(4 * H) void seq_lstm_bwd_kernel(
+    const float* __restrict__ gouts,   // (B,S,D) head-output grads
+    const float* __restrict__ ghS,     // (B,H) or nullptr
+    const float* __restrict__ gcS,     // (B,H) or nullptr
+    const float* __restrict__ stash,   // (B,S,7H)
+    const float* __restrict__ x,       // (B,S,F)
+    const float* __restrict__ c0,      // (B,H)
+    const float* __restrict__ body_w,  // (F,H)
+    const float* __restrict__ w_ih,    // (H,4H)
+    const float* __restrict__ w_hh,    // (H,4H)
+    const float* __restrict__ heads_w, // (H,D)
+    float* __restrict__ dx,            // (B,S,F)
+    float* __restrict__ dh0,           // (B,H)
+    float* __restrict__ dc0,           // (B,H)
+    float* __restrict__ dgates,        // (B,S,4H) pre-activation
+    float* __restrict__ dxb,           // (B,S,H) pre-ReLU
+    int S, int F, int D) {
+  constexpr int G = 4 * H;
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  // transposed, padded weight images: wt[g][k] at g*(H+1)+k — the pad makes
+  // both the cooperative scatter-store and the lane-broadcast row read
+  // conflict-free (bank = (g*(H+1)+k) % 32 varies with g at fixed k).
+  float* wt_ih = reinterpret_cast<float*>(smem_raw);  // (4H, H+1)
+  float* wt_hh = wt_ih + G * (H + 1);                 // (4H, H+1)
+  float* dhh = wt_hh + G * (H + 1);                   // (S, H) head-grad dh
+  float* dg4 = dhh + S * H;                           // (4H)
+  float* dxb_s = dg4 + G;                             // (S, H)
+
+  // Stage transposed weights (coalesced global read, padded LDS write).
+  for (int idx = tid; idx < H * G; idx += G) {
+    const int k = idx / G, g = idx % G;
+    wt_ih[g * (H + 1) + k] = w_ih[idx];
+    wt_hh[g * (H + 1) + k] = w_hh[idx];
+  }
+
+  // Head back-projection: dh_heads[t][k] = sum_d gouts[t][d] * heads_w[k][d]
+  for (int idx = tid; idx < S * H; idx += G) {
+    const int t = idx / H, k = idx % H;
+    float acc = 0.0f;
+    const float* gr = gouts + ((long)b * S + t) * D;
+    for (int d = 0; d < D; ++d) acc = fmaf(gr[d], heads_w[k * D + d], acc);
+    dhh[t * H + k] = acc;
+  }
+  __syncthreads();
+
+  float dh_rec = 0.0f, dc_rec = 0.0f;  // live in thread k (< H) only
+  if (tid < H) {
+    if (ghS != nullptr) dh_rec = ghS[(long)b * H + tid];
+    if (gcS != nullptr) dc_rec = gcS[(long)b * H + tid];
+  }
+
+  for (int t = S - 1; t >= 0; --t) {
+    const long sbase = (((long)b * S + t) * kStashFields) * H;
+    if (tid < H) {
+      const int k = tid;
+      const float i_ = stash[sbase + H + k];
+      const float f_ = stash[sbase + 2 * H + k];
+      const float g_ = stash[sbase + 3 * H + k];
+      const float o_ = stash[sbase + 4 * H + k];
+      const float c_ = stash[sbase + 5 * H + k];
+      const float tc = tanhf(c_);
+      const float c_prev =
+          (t > 0) ? stash[sbase - kStashFields * H + 5 * H + k]
+                  : c0[(long)b * H + k];
+      const float dh = dhh[t * H + k] + dh_rec;
+      const float dc = dc_rec + dh * o_ * (1.0f - tc * tc);
+      dg4[k] = dc * g_ * i_ * (1.0f - i_);
+      dg4[H + k] = dc * c_prev * f_ * (1.0f - f_);
+      dg4[2 * H + k] = dc * i_ * (1.0f - g_ * g_);
+      dg4[3 * H + k] = dh * tc * o_ * (1.0f - o_);
+      dc_rec = dc * f_;
+    }
+    __syncthreads();
+    // persist pre-activation gate grads for the weight GEMMs
+    dgates[((long)b * S + t) * G + tid] = dg4[tid];
+    if (tid < H) {
+      const int k = tid;
+      // recurrent back-projection + body back-projection (row reads of the
+      // transposed LDS images; lanes stride 1 → conflict-free)
+      float acc_h = 0.0f, acc_x = 0.0f;
+#pragma unroll 4
+      for (int g = 0; g < G; ++g) {
+        const float d = dg4[g];
+        acc_h = fmaf(d, wt_hh[g * (H + 1) + k], acc_h);
+        acc_x = fmaf(d, wt_ih[g * (H + 1) + k], acc_x);
+      }
+      dh_rec = acc_h;
+      const float xb_v = stash[sbase + k];  // post-ReLU body activation
+      const float dxb_v = (xb_v > 0.0f) ? acc_x : 0.0f;
+      dxb_s[t * H + k] = dxb_v;
+      dxb[((long)b * S + t) * H + k] = dxb_v;
+    }
+    __syncthreads();  // dg4 reused next iteration
+  }
+
+  if (tid < H) {
+    dh0[(long)b * H + tid] = dh_rec;
+    dc0[(long)b * H + tid] = dc_rec;
+  }
+  __syncthreads();
+
+  // dx[t][f] = sum_j dxb[t][j] * body_w[f][j]
+  for (int idx = tid; idx < S * F; idx += G) {
+    const int t = idx / F, f = idx % F;
+    float acc = 0.0f;
+    const float* dr = dxb_s + t * H;
+    const float* wr = body_w + f * H;
+#pragma unroll
+    for (int j = 0; j < H; ++j) acc = fmaf(dr[j], wr[j], acc);
+    dx[((long)b * S + t) * F + f] = acc;
+  }
+}
+
+template <int H>
+void launch_fwd(const at::Tensor& x, const at::Tensor& h0, const at::Tensor& c0,
+                const at::Tensor& body_w, const at::Tensor& body_b,
+                const at::Tensor& w_ih, const at::Tensor& w_hh,
+                const at::Tensor& b_g, const at::Tensor& heads_w,
+                const at::Tensor& heads_b, at::Tensor& outs, at::Tensor& hS,
+                at::Tensor& cS, at::Tensor& stash, int B, int S, int F, int D) {
+  const int lds =
+      (2 * S * H + 4 * H + 2 * H) * sizeof(float);
+  hipLaunchKernelGGL((seq_lstm_fwd_kernel<H>), dim3(B), dim3(4 * H), lds,
+                     current_stream(), x.data_ptr<float>(),
+                     h0.data_ptr<float>(), c0.data_ptr<float>(),
+                     body_w.data_ptr<float>(),
+                     body_b.data_ptr<float>(),
+                     w_ih.data_ptr<float>(), w_hh.data_ptr<float>(),
+                     b_g.data_ptr<float>(),
+                     heads_w.data_ptr<float>(),
+                     heads_b.data_ptr<float>(), outs.data_ptr<float>(),
+                     hS.data_ptr<float>(), cS.data_ptr<float>(),
+                     stash.data_ptr<float>(), S, F, D);
+  HIP_CHECK_LAST();
+}
+
+template <int H>
+void launch_bwd(const at::Tensor& gouts, const c10::optional<at::Tensor>& ghS,
+                const c10::optional<at::Tensor>& gcS, const at::Tensor& stash,
+                const at::Tensor& x, const at::Tensor& c0,
+                const at::Tensor& body_w, const at::Tensor& w_ih,
+                const at::Tensor& w_hh, const at::Tensor& heads_w,
+                at::Tensor& dx, at::Tensor& dh0, at::Tensor& dc0,
+                at::Tensor& dgates, at::Tensor& dxb, int B, int S, int F,
+                int D) {
+  const int G = 4 * H;
+  const int lds =
+      (2 * G * (H + 1) + S * H + G + S * H) * sizeof(float);
+  TORCH_CHECK(lds <= 160 * 1024, "backward LDS footprint exceeds 160 KiB");
+  hipLaunchKernelGGL(
+      (seq_lstm_bwd_kernel<H>), dim3(B), dim3(G), lds, current_stream(),
+      gouts.data_ptr<float>(),
+      ghS.has_value() ? ghS->data_ptr<float>() : nullptr,
+      gcS.has_value() ? gcS->data_ptr<float>() : nullptr,
+      stash.data_ptr<float>(), x.data_ptr<float>(),
+      c0.data_ptr<float>(), body_w.data_ptr<float>(),
+      w_ih.data_ptr<float>(), w_hh.data_ptr<float>(),
+      heads_w.data_ptr<float>(), dx.data_ptr<float>(),
+      dh0.data_ptr<float>(), dc0.data_ptr<float>(), dgates.data_ptr<float>(),
+      dxb.data_ptr<float>(), S, F, D);
+  HIP_CHECK_LAST();
+}
+
+}  // namespace
+
+std::vector<at::Tensor> seq_lstm_forward_hip(
+    const at::Tensor& x, const at::Tensor& h0, const at::Tensor& c0,
+    const at::Tensor& body_w, const at::Tensor& body_b, const at::Tensor& w_ih,
+    const at::Tensor& w_hh, const at::Tensor& b_g, const at::Tensor& heads_w,
+    const at::Tensor& heads_b) {
+  CHECK_IN(x); CHECK_IN(h0); CHECK_IN(c0); CHECK_IN(body_w); CHECK_IN(body_b);
+  CHECK_IN(w_ih); CHECK_IN(w_hh); CHECK_IN(b_g); CHECK_IN(heads_w);
+  CHECK_IN(heads_b);
+  const int B = x.size(0), S = x.size(1), F = x.size(2);
+  const int H = h0.size(1), D = heads_w.size(1);
+  TORCH_CHECK(body_w.size(0) == F && body_w.size(1) == H, "body_w shape");
+  TORCH_CHECK(w_ih.size(0) == H && w_ih.size(1) == 4 * H, "w_ih shape");
+  TORCH_CHECK(S >= 1 && S <= 32, "seq_len must be in [1, 32]");
+
+  auto opt = x.options();
+  auto outs = at::empty({B, S, D}, opt);
+  auto hS = at::empty({B, H}, opt);
+  auto cS = at::empty({B, H}, opt);
+  auto stash = at::empty({B, S, kStashFields * H}, opt);
+
+  switch (H) {
+    case 32: launch_fwd<32>(x, h0, c0, body_w, body_b, w_ih, w_hh, b_g, heads_w, heads_b, outs, hS, cS, stash, B, S, F, D); break;
+    case 64: launch_fwd<64>(x, h0, c0, body_w, body_b, w_ih, w_hh, b_g, heads_w, heads_b, outs, hS, cS, stash, B, S, F, D); break;
+    case 128: launch_fwd<128>(x, h0, c0, body_w, body_b, w_ih, w_hh, b_g, heads_w, heads_b, outs, hS, cS, stash, B, S, F, D); break;
+    default:
+      TORCH_CHECK(false, "hidden size ", H, " unsupported (32/64/128)");
+  }
+  return {outs, hS, cS, stash};
+}
+
+std::vector<at::Tensor> seq_lstm_backward_core_hip(
+    const at::Tensor& gouts, const c10::optional<at::Tensor>& ghS,
+    const c10::optional<at::Tensor>& gcS, const at::Tensor& stash,
+    const at::Tensor& x, const at::Tensor& c0, const at::Tensor& body_w,
+    const at::Tensor& w_ih, const at::Tensor& w_hh,
+    const at::Tensor& heads_w) {
+  CHECK_IN(gouts); CHECK_IN(stash); CHECK_IN(x); CHECK_IN(c0);
+  CHECK_IN(body_w); CHECK_IN(w_ih); CHECK_IN(w_hh); CHECK_IN(heads_w);
+  const int B = x.size(0), S = x.size(1), F = x.size(2);
+  const int H = c0.size(1), D = heads_w.size(1);
+
+  auto opt = x.options();
+  auto dx = at::empty({B, S, F}, opt);
+  auto dh0 = at::empty({B, H}, opt);
+  auto dc0 = at::empty({B, H}, opt);
+  auto dgates = at::empty({B, S, 4 * H}, opt);
+  auto dxb = at::empty({B, S, H}, opt);
+
+  switch (H) {
+    case 32: launch_bwd<32>(gouts, ghS, gcS, stash, x, c0, body_w, w_ih, w_hh, heads_w, dx, dh0, dc0, dgates, dxb, B, S, F, D); break;
+    case 64: launch_bwd<64>(gouts, ghS, gcS, stash, x, c0, body_w, w_ih, w_hh, heads_w, dx, dh0, dc0, dgates, dxb, B, S, F, D); break;
+    case 128: launch_bwd<128>(gouts, ghS, gcS, stash, x, c0, body_w, w_ih, w_hh, heads_w, dx, dh0, dc0, dgates, dxb, B, S, F, D); break;
+    default:
+      TORCH_CHECK(false, "hidden size ", H, " unsupported (32/64/128)");
+  }
+  return {dx, dh0, dc0, dgates, dxb};
+}

@@ -1,0 +1,36 @@
+"""In-tree build of the pdrl_amd HIP extension for MI355X (gfx950).
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+The built .so lands in pdrl_amd/ops/ so it travels with the repo snapshot
+(not a site-packages or JIT-cache install).
+"""
+import os
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+SRC = [
+    "pdrl_amd/ops/csrc/bindings.cpp",
+    "pdrl_amd/ops/csrc/seq_lstm.hip",
+    "pdrl_amd/ops/csrc/scans.hip",
+    "pdrl_amd/ops/csrc/multi_tensor.hip",
+]
+
+setup(
+    name="pdrl_amd_hip_ops",
+    ext_modules=[
+        CUDAExtension(
+            name="pdrl_amd.ops._hip_ops",
+            sources=SRC,
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)},
+)

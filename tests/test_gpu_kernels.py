@@ -1,0 +1,228 @@
+"""GPU kernel parity tests: every HIP kernel vs the PyTorch fp32 eager
+oracle (SURVEY.md §4's test mandate). All tests here require an MI355X."""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+B, S, F_DIM, H, A = 16, 5, 4, 64, 2
+
+
+def _ops():
+    from pdrl_amd import ops
+
+    assert ops.available(), "HIP extension must be loaded on a GPU box"
+    return ops
+
+
+def make_core(heads, seed=0):
+    from pdrl_amd.networks import SeqLSTMCore
+
+    torch.manual_seed(seed)
+    return SeqLSTMCore(F_DIM, H, heads).to(DEV)
+
+
+def test_extension_loads_on_gpu():
+    _ops()
+
+
+def test_seq_lstm_forward_parity():
+    _ops()
+    core = make_core({"logits": A, "value": 1})
+    torch.manual_seed(1)
+    x = torch.randn(B, S, F_DIM, device=DEV)
+    hx = torch.randn(B, H, device=DEV) * 0.3
+    cx = torch.randn(B, H, device=DEV) * 0.3
+
+    outs_f, h_f, c_f = core._forward_fused(x, hx, cx)
+    outs_e, h_e, c_e = core._forward_eager(x, hx, cx)
+
+    torch.testing.assert_close(h_f, h_e, rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(c_f, c_e, rtol=1e-5, atol=1e-5)
+    for k in outs_e:
+        torch.testing.assert_close(outs_f[k], outs_e[k], rtol=1e-5, atol=1e-5)
+
+
+def test_seq_lstm_backward_parity():
+    _ops()
+    for seed in (0, 7):
+        core_f = make_core({"logits": A, "value": 1}, seed=seed)
+        core_e = make_core({"logits": A, "value": 1}, seed=seed)
+        torch.manual_seed(seed + 100)
+        x = torch.randn(B, S, F_DIM, device=DEV)
+        hx = torch.randn(B, H, device=DEV, requires_grad=True) * 0.3
+        cx = torch.randn(B, H, device=DEV) * 0.3
+        xf = x.clone().requires_grad_(True)
+        xe = x.clone().requires_grad_(True)
+        hf = hx.detach().clone().requires_grad_(True)
+        he = hx.detach().clone().requires_grad_(True)
+
+        outs_f, hSf, _ = core_f._forward_fused(xf, hf, cx)
+        outs_e, hSe, _ = core_e._forward_eager(xe, he, cx)
+        # mixed loss touching both heads + the final state
+        loss_f = (outs_f["logits"].square().mean() + outs_f["value"].abs().mean()
+                  + hSf.mean())
+        loss_e = (outs_e["logits"].square().mean() + outs_e["value"].abs().mean()
+                  + hSe.mean())
+        loss_f.backward()
+        loss_e.backward()
+
+        torch.testing.assert_close(xf.grad, xe.grad, rtol=1e-4, atol=1e-5)
+        torch.testing.assert_close(hf.grad, he.grad, rtol=1e-4, atol=1e-5)
+        for (n1, p1), (n2, p2) in zip(core_f.named_parameters(),
+                                      core_e.named_parameters()):
+            assert n1 == n2
+            torch.testing.assert_close(p1.grad, p2.grad, rtol=1e-4, atol=1e-5,
+                                       msg=lambda m: f"{n1}: {m}")
+
+
+def test_gae_parity():
+    _ops()
+    from pdrl_amd.agents.learner_module.compute_loss import compute_gae
+
+    torch.manual_seed(2)
+    deltas = torch.randn(B, S - 1, 1, device=DEV)
+    dones = (torch.rand(B, S - 1, 1, device=DEV) < 0.3).float()
+    got = compute_gae(deltas, 0.99, 0.95, dones)  # dispatches to HIP on GPU
+    want = compute_gae(deltas.cpu(), 0.99, 0.95, dones.cpu())
+    torch.testing.assert_close(got.cpu(), want, rtol=1e-5, atol=1e-6)
+
+
+def test_vtrace_parity():
+    _ops()
+    from pdrl_amd.agents.learner_module.compute_loss import compute_v_trace
+
+    torch.manual_seed(3)
+    behav = -torch.rand(B, S, 1, device=DEV)
+    target = behav + 0.3 * torch.randn(B, S, 1, device=DEV)
+    is_fir = (torch.rand(B, S, 1, device=DEV) < 0.2).float()
+    rew = torch.randn(B, S, 1, device=DEV)
+    val = torch.randn(B, S, 1, device=DEV)
+    rhos_g, adv_g, vs_g = compute_v_trace(behav, target, is_fir, rew, val, 0.99)
+    rhos_c, adv_c, vs_c = compute_v_trace(
+        behav.cpu(), target.cpu(), is_fir.cpu(), rew.cpu(), val.cpu(), 0.99
+    )
+    torch.testing.assert_close(rhos_g.cpu(), rhos_c, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(vs_g.cpu(), vs_c, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(adv_g.cpu(), adv_c, rtol=1e-4, atol=1e-5)
+
+
+def test_soft_update_parity():
+    _ops()
+    from pdrl_amd.agents.learner_module.compute_loss import soft_update
+    from pdrl_amd.networks import MlpLSTMCritic
+
+    torch.manual_seed(4)
+    net = MlpLSTMCritic(F_DIM, A, S, H).to(DEV)
+    tgt_g = MlpLSTMCritic(F_DIM, A, S, H).to(DEV)
+    tgt_c = MlpLSTMCritic(F_DIM, A, S, H)
+    tgt_c.load_state_dict({k: v.cpu() for k, v in tgt_g.state_dict().items()})
+    net_c = MlpLSTMCritic(F_DIM, A, S, H)
+    net_c.load_state_dict({k: v.cpu() for k, v in net.state_dict().items()})
+
+    soft_update(net, tgt_g, 0.005)  # HIP multi-tensor path
+    soft_update(net_c, tgt_c, 0.005)  # eager path
+    for pg, pc in zip(tgt_g.parameters(), tgt_c.parameters()):
+        torch.testing.assert_close(pg.cpu(), pc, rtol=1e-6, atol=1e-7)
+
+
+def _clone_params(n=3, seed=5):
+    torch.manual_seed(seed)
+    shapes = [(64, 256), (37,), (4, 64)]
+    base = [torch.randn(*s) * 0.1 for s in shapes]
+    grads = [torch.randn(*s) for s in shapes]
+    return base, grads
+
+
+@pytest.mark.parametrize("kind", ["rmsprop", "adam"])
+def test_fused_optimizer_parity(kind):
+    _ops()
+    from pdrl_amd.ops.optim import FusedAdam, FusedRMSprop
+
+    base, grads = _clone_params()
+    max_norm = 1.5
+
+    # fused (GPU)
+    ps_f = [torch.nn.Parameter(b.clone().to(DEV)) for b in base]
+    opt_f = (FusedRMSprop(ps_f, lr=1e-2, eps=1e-5, max_norm=max_norm)
+             if kind == "rmsprop" else
+             FusedAdam(ps_f, lr=1e-2, max_norm=max_norm))
+
+    # reference (torch, CPU)
+    ps_r = [torch.nn.Parameter(b.clone()) for b in base]
+    opt_r = (torch.optim.RMSprop(ps_r, lr=1e-2, eps=1e-5)
+             if kind == "rmsprop" else torch.optim.Adam(ps_r, lr=1e-2))
+
+    for step in range(5):
+        gs = [g * (0.5 + step) for g in grads]
+        opt_f.zero_grad()
+        for p, g in zip(ps_f, gs):
+            p.grad.copy_(g.to(DEV))
+        opt_f.step()
+
+        for p, g in zip(ps_r, gs):
+            p.grad = g.clone()
+        torch.nn.utils.clip_grad_norm_(ps_r, max_norm)
+        opt_r.step()
+
+    for pf, pr in zip(ps_f, ps_r):
+        torch.testing.assert_close(pf.detach().cpu(), pr.detach(),
+                                   rtol=1e-4, atol=1e-6)
+
+
+@pytest.mark.parametrize("algo,continuous", [
+    ("PPO", False), ("IMPALA", False), ("V-MPO", False),
+    ("SAC", False), ("SAC-Continuous", True), ("PPO-Continuous", True),
+])
+def test_updater_step_on_gpu(algo, continuous):
+    _ops()
+    from pdrl_amd.agents.learner_module import switch_module
+    from pdrl_amd.utils import load_params
+    from tests.conftest import make_batch
+
+    torch.manual_seed(0)
+    p = load_params()
+    p.algo = algo
+    p.batch_size, p.seq_len = 16, 5
+    p.obs_dim = 2 if continuous else 4
+    p.n_actions = 1 if continuous else 2
+    upd_cls, model_cls = switch_module(algo)
+    model = model_cls(p.obs_dim, p.n_actions, p.seq_len, p.hidden_size)
+    upd = upd_cls(model, p, DEV)
+    batch = make_batch(p, n_actions=p.n_actions, continuous=continuous, device=DEV)
+    batch["obs"] = torch.randn(p.batch_size, p.seq_len, p.obs_dim, device=DEV)
+    for _ in range(3):
+        stats = upd.step(batch)
+    assert all(np.isfinite(v) for v in stats.values()), stats
+
+
+def test_gpu_vs_cpu_updater_trajectories_match():
+    """Full IMPALA update on GPU (fused kernels) vs CPU (eager) from identical
+    init: parameters must agree to fp32 tolerance after 3 steps."""
+    _ops()
+    from pdrl_amd.agents.learner_module import ImpalaUpdater
+    from pdrl_amd.networks import MlpLSTMSingle
+    from pdrl_amd.utils import load_params
+    from tests.conftest import make_batch
+
+    p = load_params()
+    p.batch_size, p.seq_len, p.obs_dim, p.n_actions = 16, 5, 4, 2
+
+    torch.manual_seed(42)
+    model_g = MlpLSTMSingle(4, 2, p.seq_len, p.hidden_size)
+    torch.manual_seed(42)
+    model_c = MlpLSTMSingle(4, 2, p.seq_len, p.hidden_size)
+
+    upd_g = ImpalaUpdater(model_g, p, DEV)
+    upd_c = ImpalaUpdater(model_c, p, "cpu")
+    batch_c = make_batch(p, seed=9)
+    batch_g = {k: v.to(DEV) for k, v in batch_c.items()}
+    for _ in range(3):
+        sg = upd_g.step(batch_g)
+        sc = upd_c.step(batch_c)
+    for (n, pg), pc in zip(model_g.named_parameters(), model_c.parameters()):
+        torch.testing.assert_close(pg.detach().cpu(), pc.detach(), rtol=2e-3,
+                                   atol=2e-5, msg=lambda m: f"{n}: {m}")
+    assert abs(sg["loss-total"] - sc["loss-total"]) < 1e-2

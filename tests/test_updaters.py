@@ -2,6 +2,8 @@
 synthetic batch, produces finite losses, and changes parameters."""
 import copy
 
+import numpy as np
+
 import pytest
 import torch
 
@@ -44,7 +46,7 @@ def test_on_policy_updaters_step(updater_cls, model_cls, params):
     batch = make_batch(params)
     before = param_snapshot(model)
     stats = upd.step(batch)
-    assert all(torch.isfinite(torch.tensor(v)) for v in stats.values()), stats
+    assert all(np.isfinite(float(v)) for v in stats.values()), stats
     assert params_changed(before, model)
     assert upd.update_count == 1
 
@@ -58,7 +60,7 @@ def test_ppo_continuous(params):
     batch = make_batch(params, n_actions=1, continuous=True)
     batch["obs"] = torch.randn(params.batch_size, params.seq_len, 2)
     stats = upd.step(batch)
-    assert all(torch.isfinite(torch.tensor(v)) for v in stats.values())
+    assert all(np.isfinite(float(v)) for v in stats.values())
 
 
 def test_sac_discrete(params):
@@ -74,7 +76,7 @@ def test_sac_discrete(params):
     batch = make_batch(params)
     before_t = param_snapshot(upd.target_critic)
     stats = upd.step(batch)
-    assert all(torch.isfinite(torch.tensor(v)) for v in stats.values())
+    assert all(np.isfinite(float(v)) for v in stats.values())
     # soft update moved the target
     assert params_changed(before_t, upd.target_critic)
 
@@ -88,7 +90,7 @@ def test_sac_continuous(params):
     batch["obs"] = torch.randn(params.batch_size, params.seq_len, 2)
     alpha_before = float(upd.log_alpha)
     stats = upd.step(batch)
-    assert all(torch.isfinite(torch.tensor(v)) for v in stats.values())
+    assert all(np.isfinite(float(v)) for v in stats.values())
     assert float(upd.log_alpha) != alpha_before  # temperature auto-tuned
 
 
