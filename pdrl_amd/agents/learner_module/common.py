@@ -55,6 +55,26 @@ class BaseUpdater:
             return torch.optim.RMSprop(parameters, lr=lr, eps=kw.get("eps", 1e-5))
         return torch.optim.Adam(parameters, lr=lr)
 
+    def make_fused_step(self, algo: str, model, optimizer):
+        """Whole-step fused HIP DAG (IMPALA/PPO, discrete policies) when the
+        extension is loaded on GPU; None → the eager-autograd path runs."""
+        from pdrl_amd import ops
+
+        if not (self.device.type == "cuda" and ops.available()):
+            return None
+        if not getattr(optimizer, "is_fused", False):
+            return None
+        actor = getattr(model, "actor", model)
+        core = getattr(actor, "core", None)
+        if core is None or core.head_names != ["logits", "value"]:
+            return None  # continuous-policy fused step not implemented yet
+        from pdrl_amd.ops.fused_step import FusedOnPolicyStep
+
+        use_graph = bool(int((__import__("os").environ.get("PDRL_USE_GRAPH", "1"))))
+        return FusedOnPolicyStep(algo, core, self.params, optimizer,
+                                 grad_reducer=self.grad_reducer,
+                                 use_graph=use_graph)
+
     def apply_step(self, optimizer, parameters):
         """Gradient epilogue: all-reduce across ranks → clip → update.
         Fused path: one collective on the flat grad bucket; clip is inside

@@ -32,8 +32,8 @@ def compute_gae(
     if _use_hip(deltas):
         from pdrl_amd import ops
 
-        return ops.ext().gae(deltas, float(gamma), float(lmbda),
-                             dones if dones is not None else torch.zeros_like(deltas))
+        d = dones.contiguous() if dones is not None else torch.zeros_like(deltas)
+        return ops.ext().gae(deltas.contiguous(), float(gamma), float(lmbda), d)
     B, T, _ = deltas.shape
     adv = torch.zeros_like(deltas)
     running = torch.zeros(B, 1, dtype=deltas.dtype, device=deltas.device)
@@ -67,7 +67,8 @@ def compute_v_trace(
         from pdrl_amd import ops
 
         return ops.ext().vtrace(
-            behav_log_probs, target_log_probs, is_fir, rewards, values,
+            behav_log_probs.contiguous(), target_log_probs.contiguous(),
+            is_fir.contiguous(), rewards.contiguous(), values.contiguous(),
             float(gamma), float(rho_bar), float(rho_min), float(c_bar),
         )
     log_rhos = (target_log_probs - behav_log_probs)[:, :-1]

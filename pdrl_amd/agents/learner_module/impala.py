@@ -23,6 +23,7 @@ class ImpalaUpdater(BaseUpdater):
         self.optimizer = self.make_optimizer(
             "rmsprop", self.model.parameters(), lr=params.lr, eps=1e-5
         )
+        self.fused_step = self.make_fused_step("IMPALA", self.model, self.optimizer)
 
     def trainable_modules(self):
         return {"model": self.model}
@@ -62,6 +63,10 @@ class ImpalaUpdater(BaseUpdater):
         return loss, stats
 
     def step(self, batch: dict[str, torch.Tensor]) -> dict:
+        if self.fused_step is not None:
+            stats = self.fused_step.run(batch)
+            self.update_count += 1
+            return stats
         stats = {}
         for _ in range(self.params.K_epoch):
             loss, stats = self.compute_losses(batch)

@@ -24,6 +24,7 @@ class PPOUpdater(BaseUpdater):
         self.optimizer = self.make_optimizer(
             "rmsprop", self.model.parameters(), lr=params.lr, eps=1e-5
         )
+        self.fused_step = self.make_fused_step("PPO", self.model, self.optimizer)
 
     def trainable_modules(self):
         return {"model": self.model}
@@ -70,6 +71,10 @@ class PPOUpdater(BaseUpdater):
         return loss, stats
 
     def step(self, batch: dict[str, torch.Tensor]) -> dict:
+        if self.fused_step is not None:
+            stats = self.fused_step.run(batch)
+            self.update_count += 1
+            return stats
         stats = {}
         for _ in range(self.params.K_epoch):
             loss, stats = self.compute_losses(batch)

@@ -53,6 +53,7 @@ class SeqLSTMCore(nn.Module):
         self.input_dim = input_dim
         self.hidden = hidden
         self.head_names = list(heads.keys())
+        self.head_dims = dict(heads)
         H = hidden
         self.body_w = nn.Parameter(torch.empty(input_dim, H))
         self.body_b = nn.Parameter(torch.empty(H))
@@ -62,15 +63,28 @@ class SeqLSTMCore(nn.Module):
         _init_linear_t(self.body_w, self.body_b, input_dim)
         _init_linear_t(self.w_ih, None, H)
         _init_linear_t(self.w_hh, self.b_g, H)
+        # heads live CONCATENATED along the output dim: one (H, Dtot) GEMM in
+        # both the eager and the fused path, and the parameter layout matches
+        # the HIP weight-grad kernel exactly (no cats/splits anywhere).
+        self.out_dim = sum(heads.values())
+        self.heads_w = nn.Parameter(torch.empty(H, self.out_dim))
+        self.heads_b = nn.Parameter(torch.empty(self.out_dim))
+        _init_linear_t(self.heads_w, self.heads_b, H)
+        self._head_slices = {}
+        off = 0
         for name, dim in heads.items():
-            w = nn.Parameter(torch.empty(H, dim))
-            b = nn.Parameter(torch.empty(dim))
-            _init_linear_t(w, b, H)
-            self.register_parameter(f"head_{name}_w", w)
-            self.register_parameter(f"head_{name}_b", b)
+            self._head_slices[name] = (off, off + dim)
+            off += dim
 
     def head_params(self, name: str):
-        return getattr(self, f"head_{name}_w"), getattr(self, f"head_{name}_b")
+        """(weight, bias) views of the concatenated head parameters."""
+        lo, hi = self._head_slices[name]
+        return self.heads_w[:, lo:hi], self.heads_b[lo:hi]
+
+    def split_heads(self, outs_cat: torch.Tensor) -> dict[str, torch.Tensor]:
+        return {
+            name: outs_cat[..., lo:hi] for name, (lo, hi) in self._head_slices.items()
+        }
 
     # ------------------------------------------------------------------ #
     def forward(
@@ -99,12 +113,10 @@ class SeqLSTMCore(nn.Module):
             h = go * torch.tanh(c)
             hs.append(h)
         hseq = torch.stack(hs, dim=1)  # (B, S, H)
-        outs = {}
-        for name in self.head_names:
-            w, b = self.head_params(name)
-            outs[name] = hseq.reshape(B * S, H) @ w + b
-            outs[name] = outs[name].view(B, S, -1)
-        return outs, h, c
+        outs_cat = (hseq.reshape(B * S, H) @ self.heads_w + self.heads_b).view(
+            B, S, self.out_dim
+        )
+        return self.split_heads(outs_cat), h, c
 
     def _forward_fused(self, x, hx, cx):
         from pdrl_amd import ops

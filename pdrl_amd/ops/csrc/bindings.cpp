@@ -12,11 +12,41 @@ std::vector<at::Tensor> seq_lstm_backward_core_hip(
     const c10::optional<at::Tensor>&, const at::Tensor&, const at::Tensor&,
     const at::Tensor&, const at::Tensor&, const at::Tensor&, const at::Tensor&,
     const at::Tensor&);
+std::vector<at::Tensor> seq_lstm_wgrad_hip(
+    const at::Tensor&, const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&);
+void seq_lstm_wgrad_out_hip(const at::Tensor&, const at::Tensor&,
+                            const at::Tensor&, const at::Tensor&,
+                            const at::Tensor&, const at::Tensor&, at::Tensor&,
+                            at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
+                            at::Tensor&, at::Tensor&);
 at::Tensor gae_hip(const at::Tensor&, double, double, const at::Tensor&);
 std::vector<at::Tensor> vtrace_hip(const at::Tensor&, const at::Tensor&,
                                    const at::Tensor&, const at::Tensor&,
                                    const at::Tensor&, double, double, double,
-                                   double);
+                                   double, long, long, double);
+std::vector<at::Tensor> cat_stats_hip(const at::Tensor&, const at::Tensor&,
+                                      long);
+std::vector<at::Tensor> ppo_td_gae_hip(const at::Tensor&, const at::Tensor&,
+                                       const at::Tensor&, long, double, double,
+                                       double);
+void impala_loss_reduce_hip(const at::Tensor&, const at::Tensor&,
+                            const at::Tensor&, long, const at::Tensor&,
+                            const at::Tensor&, const at::Tensor&, at::Tensor&,
+                            double, double, double);
+void ppo_loss_reduce_hip(const at::Tensor&, const at::Tensor&,
+                         const at::Tensor&, const at::Tensor&, long,
+                         const at::Tensor&, const at::Tensor&, at::Tensor&,
+                         double, double, double, double);
+at::Tensor impala_loss_bwd_hip(const at::Tensor&, long, const at::Tensor&,
+                               const at::Tensor&, const at::Tensor&,
+                               const at::Tensor&, const at::Tensor&, double,
+                               double, double);
+at::Tensor ppo_loss_bwd_hip(const at::Tensor&, long, const at::Tensor&,
+                            const at::Tensor&, const at::Tensor&,
+                            const at::Tensor&, const at::Tensor&,
+                            const at::Tensor&, const at::Tensor&, double,
+                            double, double, double);
 void l2norm_sq_hip(const at::Tensor&, at::Tensor&);
 void rmsprop_step_hip(at::Tensor&, const at::Tensor&, at::Tensor&,
                       const at::Tensor&, double, double, double, double);
@@ -31,8 +61,24 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused body+LSTM+heads forward (gfx950)");
   m.def("seq_lstm_backward_core", &seq_lstm_backward_core_hip,
         "fused BPTT backward core (gfx950)");
+  m.def("seq_lstm_wgrad", &seq_lstm_wgrad_hip,
+        "MFMA weight-gradient GEMMs + wave-per-element small grads");
+  m.def("seq_lstm_wgrad_out", &seq_lstm_wgrad_out_hip,
+        "wgrad writing into caller buffers (flat grad views)");
   m.def("gae", &gae_hip, "GAE reverse scan");
-  m.def("vtrace", &vtrace_hip, "fused V-trace scan");
+  m.def("vtrace", &vtrace_hip, "fused V-trace scan",
+        pybind11::arg("behav_lp"), pybind11::arg("target_lp"),
+        pybind11::arg("is_fir"), pybind11::arg("rew"), pybind11::arg("val"),
+        pybind11::arg("gamma"), pybind11::arg("rho_bar"),
+        pybind11::arg("rho_min"), pybind11::arg("c_bar"),
+        pybind11::arg("vD") = 1, pybind11::arg("val_off") = 0,
+        pybind11::arg("rew_scale") = 1.0);
+  m.def("cat_stats", &cat_stats_hip, "categorical log-softmax stats");
+  m.def("ppo_td_gae", &ppo_td_gae_hip, "fused TD target + GAE scan");
+  m.def("impala_loss_reduce", &impala_loss_reduce_hip, "IMPALA loss stats");
+  m.def("ppo_loss_reduce", &ppo_loss_reduce_hip, "PPO loss stats");
+  m.def("impala_loss_bwd", &impala_loss_bwd_hip, "analytic IMPALA loss grad");
+  m.def("ppo_loss_bwd", &ppo_loss_bwd_hip, "analytic PPO loss grad");
   m.def("l2norm_sq", &l2norm_sq_hip, "squared L2 norm into a device scalar");
   m.def("rmsprop_step", &rmsprop_step_hip, "fused clip+RMSprop on flat buffers");
   m.def("adam_step", &adam_step_hip, "fused clip+Adam on flat buffers");

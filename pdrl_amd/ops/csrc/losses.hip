@@ -1,0 +1,372 @@
+// Fused RL loss kernels for CDNA4 (gfx950) — K4 + K8 + K9 of SURVEY.md §2.4.
+//
+// Replaces the learner's eager loss chain (~40 elementwise/reduce launches
+// per step in PyTorch eager: softmax, gather, exp, clamp, min, smooth-L1,
+// means — reference: ppo/learning.py:43-106, impala/learning.py:48-94) with:
+//   cat_stats        — log-softmax stats per (b,t): log pi(a), entropy, lse
+//   ppo_td_gae       — TD target + GAE reverse scan, one thread per row
+//   {impala,ppo}_loss_reduce — single-block reduction to a device stats
+//                      vector {loss parts + monitoring stats}
+//   {impala,ppo}_loss_bwd    — ANALYTIC gradient straight into the packed
+//                      head-grad buffer gouts = [dlogits | dvalue] (B,S,D)
+//
+// All kernels consume the fused forward's PACKED head output
+// model_out (B,S,D) with D = A+1: logits in cols [0,A), value in col A.
+// Everything stays on device (no host syncs); the whole training step is a
+// fixed ~10-kernel DAG (ops/fused_step.py), hipGraph-capturable.
+#include "common.h"
+
+#include <vector>
+
+namespace {
+
+__device__ __forceinline__ float huber(float d) {  // smooth_l1, beta = 1
+  const float a = fabsf(d);
+  return (a < 1.0f) ? 0.5f * d * d : a - 0.5f;
+}
+__device__ __forceinline__ float huber_grad(float d) {
+  return fminf(fmaxf(d, -1.0f), 1.0f);
+}
+
+// ---- categorical stats ---------------------------------------------------
+// one thread per (b,t) over the packed model_out rows (stride D)
+__global__ void cat_stats_kernel(const float* __restrict__ mo,   // (N,D)
+                                 const float* __restrict__ act,  // (N)
+                                 float* __restrict__ logp,       // (N)
+                                 float* __restrict__ ent,        // (N)
+                                 float* __restrict__ lse,        // (N)
+                                 long N, int A, int D) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= N) return;
+  const float* z = mo + i * D;
+  float m = z[0];
+  for (int j = 1; j < A; ++j) m = fmaxf(m, z[j]);
+  float s = 0.0f;
+  for (int j = 0; j < A; ++j) s += __expf(z[j] - m);
+  const float l = m + __logf(s);
+  float h = 0.0f;
+  for (int j = 0; j < A; ++j) {
+    const float lp = z[j] - l;
+    h -= __expf(lp) * lp;
+  }
+  const int a = (int)act[i];
+  logp[i] = z[a] - l;
+  ent[i] = h;
+  lse[i] = l;
+}
+
+// ---- PPO targets: td + delta + GAE scan (one thread per batch row) -------
+// value = column A of model_out (element stride D)
+__global__ void ppo_td_gae_kernel(const float* __restrict__ rew,     // (B,S)
+                                  const float* __restrict__ is_fir,  // (B,S)
+                                  const float* __restrict__ val,     // strided
+                                  float* __restrict__ td,            // (B,T)
+                                  float* __restrict__ adv,           // (B,T)
+                                  int B, int S, int D, float gamma,
+                                  float lmbda, float rew_scale) {
+  const int b = blockIdx.x * blockDim.x + threadIdx.x;
+  if (b >= B) return;
+  const int T = S - 1;
+  const long sb = (long)b * S, tb = (long)b * T;
+  float run = 0.0f;
+  for (int t = T - 1; t >= 0; --t) {
+    const float mask = 1.0f - is_fir[sb + t + 1];
+    const float tdv =
+        rew[sb + t] * rew_scale + gamma * mask * val[(sb + t + 1) * D];
+    const float delta = tdv - val[(sb + t) * D];
+    run = fmaf(gamma * lmbda * mask, run, delta);
+    td[tb + t] = tdv;
+    adv[tb + t] = run;
+  }
+}
+
+// ---- single-block loss reductions ---------------------------------------
+// stats_impala = {total, policy, value, entropy, rho_avg}
+__global__ void impala_loss_reduce_kernel(
+    const float* __restrict__ logp,  // (B,S)
+    const float* __restrict__ ent,   // (B,S)
+    const float* __restrict__ val,   // strided by D
+    const float* __restrict__ adv,   // (B,T)
+    const float* __restrict__ vs,    // (B,T)
+    const float* __restrict__ rhos,  // (B,T)
+    float* __restrict__ stats,       // (5)
+    int B, int S, int D, float cp, float cv, float ce) {
+  const int T = S - 1;
+  const int N = B * T;
+  float pl = 0, vl = 0, es = 0, rs = 0;
+  for (int i = threadIdx.x; i < N; i += blockDim.x) {
+    const int b = i / T, t = i % T;
+    const long si = (long)b * S + t;
+    pl -= logp[si] * adv[i];
+    vl += huber(val[si * D] - vs[i]);
+    es += ent[si];
+    rs += rhos[i];
+  }
+  __shared__ float red[4][256];
+  red[0][threadIdx.x] = pl; red[1][threadIdx.x] = vl;
+  red[2][threadIdx.x] = es; red[3][threadIdx.x] = rs;
+  __syncthreads();
+  for (int off = blockDim.x / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      for (int r = 0; r < 4; ++r)
+        red[r][threadIdx.x] += red[r][threadIdx.x + off];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    const float inv = 1.0f / N;
+    const float p = red[0][0] * inv, v = red[1][0] * inv, e = red[2][0] * inv;
+    stats[0] = cp * p + cv * v - ce * e;
+    stats[1] = p; stats[2] = v; stats[3] = e; stats[4] = red[3][0] * inv;
+  }
+}
+
+// stats_ppo = {total, policy, value, entropy, ratio_avg, ratio_min, ratio_max}
+__global__ void ppo_loss_reduce_kernel(
+    const float* __restrict__ logp,       // (B,S) target
+    const float* __restrict__ behav_lp,   // (B,S)
+    const float* __restrict__ ent,        // (B,S)
+    const float* __restrict__ val,        // strided by D
+    const float* __restrict__ adv,        // (B,T)
+    const float* __restrict__ td,         // (B,T)
+    float* __restrict__ stats,            // (7)
+    int B, int S, int D, float cp, float cv, float ce, float eps_clip) {
+  const int T = S - 1;
+  const int N = B * T;
+  float pl = 0, vl = 0, es = 0, ravg = 0;
+  float rmin = 1e30f, rmax = -1e30f;
+  for (int i = threadIdx.x; i < N; i += blockDim.x) {
+    const int b = i / T, t = i % T;
+    const long si = (long)b * S + t;
+    const float r = __expf(logp[si] - behav_lp[si]);
+    const float a = adv[i];
+    const float s1 = r * a;
+    const float s2 = fminf(fmaxf(r, 1.0f - eps_clip), 1.0f + eps_clip) * a;
+    pl -= fminf(s1, s2);
+    vl += huber(val[si * D] - td[i]);
+    es += ent[si];
+    ravg += r;
+    rmin = fminf(rmin, r);
+    rmax = fmaxf(rmax, r);
+  }
+  __shared__ float red[4][256];
+  __shared__ float rmn[256], rmx[256];
+  red[0][threadIdx.x] = pl; red[1][threadIdx.x] = vl;
+  red[2][threadIdx.x] = es; red[3][threadIdx.x] = ravg;
+  rmn[threadIdx.x] = rmin; rmx[threadIdx.x] = rmax;
+  __syncthreads();
+  for (int off = blockDim.x / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      for (int r = 0; r < 4; ++r)
+        red[r][threadIdx.x] += red[r][threadIdx.x + off];
+      rmn[threadIdx.x] = fminf(rmn[threadIdx.x], rmn[threadIdx.x + off]);
+      rmx[threadIdx.x] = fmaxf(rmx[threadIdx.x], rmx[threadIdx.x + off]);
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    const float inv = 1.0f / N;
+    const float p = red[0][0] * inv, v = red[1][0] * inv, e = red[2][0] * inv;
+    stats[0] = cp * p + cv * v - ce * e;
+    stats[1] = p; stats[2] = v; stats[3] = e;
+    stats[4] = red[3][0] * inv; stats[5] = rmn[0]; stats[6] = rmx[0];
+  }
+}
+
+// ---- analytic loss backward → packed head grads [dlogits | dvalue] -------
+__global__ void impala_loss_bwd_kernel(
+    const float* __restrict__ mo,    // (N,D) packed model out
+    const float* __restrict__ act,   // (N)
+    const float* __restrict__ lse,   // (N)
+    const float* __restrict__ ent,   // (N)
+    const float* __restrict__ adv,   // (B,T)
+    const float* __restrict__ vs,    // (B,T)
+    float* __restrict__ gouts,       // (B,S,D)
+    int B, int S, int A, float cp, float cv, float ce) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long N = (long)B * S;
+  if (i >= N) return;
+  const int T = S - 1;
+  const int b = (int)(i / S), t = (int)(i % S);
+  const int D = A + 1;
+  float* g = gouts + i * D;
+  if (t >= T) {
+    for (int j = 0; j < D; ++j) g[j] = 0.0f;
+    return;
+  }
+  const float invN = 1.0f / (B * T);
+  const long ti = (long)b * T + t;
+  const float dlogp = -cp * adv[ti] * invN;
+  const float dH = -ce * invN;
+  const float H = ent[i];
+  const float* z = mo + i * D;
+  const int a = (int)act[i];
+  for (int j = 0; j < A; ++j) {
+    const float lp = z[j] - lse[i];
+    const float p = __expf(lp);
+    g[j] = dlogp * ((j == a ? 1.0f : 0.0f) - p) + dH * (-p * (lp + H));
+  }
+  g[A] = cv * huber_grad(z[A] - vs[ti]) * invN;
+}
+
+__global__ void ppo_loss_bwd_kernel(
+    const float* __restrict__ mo,        // (N,D)
+    const float* __restrict__ act,       // (N)
+    const float* __restrict__ lse,       // (N)
+    const float* __restrict__ ent,       // (N)
+    const float* __restrict__ logp,      // (N) target log pi(a)
+    const float* __restrict__ behav_lp,  // (N)
+    const float* __restrict__ adv,       // (B,T)
+    const float* __restrict__ td,        // (B,T)
+    float* __restrict__ gouts,           // (B,S,D)
+    int B, int S, int A, float cp, float cv, float ce, float eps_clip) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long N = (long)B * S;
+  if (i >= N) return;
+  const int T = S - 1;
+  const int b = (int)(i / S), t = (int)(i % S);
+  const int D = A + 1;
+  float* g = gouts + i * D;
+  if (t >= T) {
+    for (int j = 0; j < D; ++j) g[j] = 0.0f;
+    return;
+  }
+  const float invN = 1.0f / (B * T);
+  const long ti = (long)b * T + t;
+  const float a_v = adv[ti];
+  const float r = __expf(logp[i] - behav_lp[i]);
+  const bool inside = (r > 1.0f - eps_clip) && (r < 1.0f + eps_clip);
+  const float s1 = r * a_v;
+  const float s2 = fminf(fmaxf(r, 1.0f - eps_clip), 1.0f + eps_clip) * a_v;
+  // d min(s1, s2)/d logp: the clipped branch has zero grad outside the band
+  const float gr = (inside || s1 < s2) ? a_v * r : 0.0f;
+  const float dlogp = -cp * gr * invN;
+  const float dH = -ce * invN;
+  const float H = ent[i];
+  const float* z = mo + i * D;
+  const int a = (int)act[i];
+  for (int j = 0; j < A; ++j) {
+    const float lp = z[j] - lse[i];
+    const float p = __expf(lp);
+    g[j] = dlogp * ((j == a ? 1.0f : 0.0f) - p) + dH * (-p * (lp + H));
+  }
+  g[A] = cv * huber_grad(z[A] - td[ti]) * invN;
+}
+
+}  // namespace
+
+std::vector<at::Tensor> cat_stats_hip(const at::Tensor& model_out,
+                                      const at::Tensor& act, long A) {
+  CHECK_IN(model_out); CHECK_IN(act);
+  const int D = model_out.size(-1);
+  const long N = model_out.numel() / D;
+  auto opt = model_out.options();
+  auto logp = at::empty({N}, opt);
+  auto ent = at::empty({N}, opt);
+  auto lse = at::empty({N}, opt);
+  const int threads = 256;
+  const long blocks = (N + threads - 1) / threads;
+  hipLaunchKernelGGL(cat_stats_kernel, dim3(blocks), dim3(threads), 0,
+                     current_stream(), model_out.data_ptr<float>(),
+                     act.data_ptr<float>(), logp.data_ptr<float>(),
+                     ent.data_ptr<float>(), lse.data_ptr<float>(), N, (int)A,
+                     D);
+  HIP_CHECK_LAST();
+  return {logp, ent, lse};
+}
+
+std::vector<at::Tensor> ppo_td_gae_hip(const at::Tensor& rew,
+                                       const at::Tensor& is_fir,
+                                       const at::Tensor& model_out, long A,
+                                       double gamma, double lmbda,
+                                       double rew_scale) {
+  CHECK_IN(rew); CHECK_IN(is_fir); CHECK_IN(model_out);
+  const int B = model_out.size(0), S = model_out.size(1);
+  const int D = model_out.size(2);
+  auto opt = model_out.options();
+  auto td = at::empty({B, S - 1}, opt);
+  auto adv = at::empty({B, S - 1}, opt);
+  const int threads = 256;
+  hipLaunchKernelGGL(ppo_td_gae_kernel, dim3((B + threads - 1) / threads),
+                     dim3(threads), 0, current_stream(),
+                     rew.data_ptr<float>(), is_fir.data_ptr<float>(),
+                     model_out.data_ptr<float>() + A, td.data_ptr<float>(),
+                     adv.data_ptr<float>(), B, S, D, (float)gamma,
+                     (float)lmbda, (float)rew_scale);
+  HIP_CHECK_LAST();
+  return {td, adv};
+}
+
+void impala_loss_reduce_hip(const at::Tensor& logp, const at::Tensor& ent,
+                            const at::Tensor& model_out, long A,
+                            const at::Tensor& adv, const at::Tensor& vs,
+                            const at::Tensor& rhos, at::Tensor& stats,
+                            double cp, double cv, double ce) {
+  const int B = model_out.size(0), S = model_out.size(1);
+  const int D = model_out.size(2);
+  hipLaunchKernelGGL(impala_loss_reduce_kernel, dim3(1), dim3(256), 0,
+                     current_stream(), logp.data_ptr<float>(),
+                     ent.data_ptr<float>(), model_out.data_ptr<float>() + A,
+                     adv.data_ptr<float>(), vs.data_ptr<float>(),
+                     rhos.data_ptr<float>(), stats.data_ptr<float>(), B, S, D,
+                     (float)cp, (float)cv, (float)ce);
+  HIP_CHECK_LAST();
+}
+
+void ppo_loss_reduce_hip(const at::Tensor& logp, const at::Tensor& behav_lp,
+                         const at::Tensor& ent, const at::Tensor& model_out,
+                         long A, const at::Tensor& adv, const at::Tensor& td,
+                         at::Tensor& stats, double cp, double cv, double ce,
+                         double eps_clip) {
+  const int B = model_out.size(0), S = model_out.size(1);
+  const int D = model_out.size(2);
+  hipLaunchKernelGGL(ppo_loss_reduce_kernel, dim3(1), dim3(256), 0,
+                     current_stream(), logp.data_ptr<float>(),
+                     behav_lp.data_ptr<float>(), ent.data_ptr<float>(),
+                     model_out.data_ptr<float>() + A, adv.data_ptr<float>(),
+                     td.data_ptr<float>(), stats.data_ptr<float>(), B, S, D,
+                     (float)cp, (float)cv, (float)ce, (float)eps_clip);
+  HIP_CHECK_LAST();
+}
+
+at::Tensor impala_loss_bwd_hip(const at::Tensor& model_out, long A,
+                               const at::Tensor& act, const at::Tensor& lse,
+                               const at::Tensor& ent, const at::Tensor& adv,
+                               const at::Tensor& vs, double cp, double cv,
+                               double ce) {
+  const int B = model_out.size(0), S = model_out.size(1);
+  auto gouts = at::empty_like(model_out);
+  const long N = (long)B * S;
+  const int threads = 256;
+  hipLaunchKernelGGL(impala_loss_bwd_kernel,
+                     dim3((N + threads - 1) / threads), dim3(threads), 0,
+                     current_stream(), model_out.data_ptr<float>(),
+                     act.data_ptr<float>(), lse.data_ptr<float>(),
+                     ent.data_ptr<float>(), adv.data_ptr<float>(),
+                     vs.data_ptr<float>(), gouts.data_ptr<float>(), B, S,
+                     (int)A, (float)cp, (float)cv, (float)ce);
+  HIP_CHECK_LAST();
+  return gouts;
+}
+
+at::Tensor ppo_loss_bwd_hip(const at::Tensor& model_out, long A,
+                            const at::Tensor& act, const at::Tensor& lse,
+                            const at::Tensor& ent, const at::Tensor& logp,
+                            const at::Tensor& behav_lp, const at::Tensor& adv,
+                            const at::Tensor& td, double cp, double cv,
+                            double ce, double eps_clip) {
+  const int B = model_out.size(0), S = model_out.size(1);
+  auto gouts = at::empty_like(model_out);
+  const long N = (long)B * S;
+  const int threads = 256;
+  hipLaunchKernelGGL(ppo_loss_bwd_kernel, dim3((N + threads - 1) / threads),
+                     dim3(threads), 0, current_stream(),
+                     model_out.data_ptr<float>(), act.data_ptr<float>(),
+                     lse.data_ptr<float>(), ent.data_ptr<float>(),
+                     logp.data_ptr<float>(), behav_lp.data_ptr<float>(),
+                     adv.data_ptr<float>(), td.data_ptr<float>(),
+                     gouts.data_ptr<float>(), B, S, (int)A, (float)cp,
+                     (float)cv, (float)ce, (float)eps_clip);
+  HIP_CHECK_LAST();
+  return gouts;
+}

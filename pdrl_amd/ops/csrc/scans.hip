@@ -33,11 +33,12 @@ __global__ void vtrace_kernel(
     const float* __restrict__ target_lp,  // (B,S)
     const float* __restrict__ is_fir,     // (B,S)
     const float* __restrict__ rew,        // (B,S)
-    const float* __restrict__ val,        // (B,S)
+    const float* __restrict__ val,        // (B,S) element stride vD
     float* __restrict__ rhos,             // (B,T), T = S-1
     float* __restrict__ adv,              // (B,T)
     float* __restrict__ vs,               // (B,T)
-    int B, int S, float gamma, float rho_bar, float rho_min, float c_bar) {
+    int B, int S, int vD, float gamma, float rho_bar, float rho_min,
+    float c_bar, float rew_scale) {
   const int b = blockIdx.x * blockDim.x + threadIdx.x;
   if (b >= B) return;
   const int T = S - 1;
@@ -53,18 +54,19 @@ __global__ void vtrace_kernel(
     const float rho = fminf(fmaxf(ratio, rho_min), rho_bar);
     const float c = fminf(ratio, c_bar);
     const float mask = 1.0f - is_fir[sb + t + 1];
-    const float delta =
-        rho * (rew[sb + t] + gamma * mask * val[sb + t + 1] - val[sb + t]);
+    const float delta = rho * (rew[sb + t] * rew_scale +
+                               gamma * mask * val[(sb + t + 1) * vD] -
+                               val[(sb + t) * vD]);
     acc = fmaf(gamma * mask * c, acc, delta);
     rhos[tb + t] = rho;
-    vs[tb + t] = val[sb + t] + acc;
+    vs[tb + t] = val[(sb + t) * vD] + acc;
   }
   // forward pass: advantages against vs_{t+1} (bootstrap from val_S-1 tail)
   for (int t = 0; t < T; ++t) {
     const float mask = 1.0f - is_fir[sb + t + 1];
-    const float vnext = (t + 1 < T) ? vs[tb + t + 1] : val[sb + T];
-    adv[tb + t] =
-        rhos[tb + t] * (rew[sb + t] + gamma * mask * vnext - val[sb + t]);
+    const float vnext = (t + 1 < T) ? vs[tb + t + 1] : val[(long)(sb + T) * vD];
+    adv[tb + t] = rhos[tb + t] * (rew[sb + t] * rew_scale +
+                                  gamma * mask * vnext - val[(sb + t) * vD]);
   }
 }
 
@@ -90,7 +92,8 @@ std::vector<at::Tensor> vtrace_hip(const at::Tensor& behav_lp,
                                    const at::Tensor& is_fir,
                                    const at::Tensor& rew, const at::Tensor& val,
                                    double gamma, double rho_bar, double rho_min,
-                                   double c_bar) {
+                                   double c_bar, long vD, long val_off,
+                                   double rew_scale) {
   CHECK_IN(behav_lp); CHECK_IN(target_lp); CHECK_IN(is_fir);
   CHECK_IN(rew); CHECK_IN(val);
   const int B = val.size(0), S = val.size(1);
@@ -105,10 +108,11 @@ std::vector<at::Tensor> vtrace_hip(const at::Tensor& behav_lp,
                      current_stream(), behav_lp.data_ptr<float>(),
                      target_lp.data_ptr<float>(),
                      is_fir.data_ptr<float>(),
-                     rew.data_ptr<float>(), val.data_ptr<float>(),
+                     rew.data_ptr<float>(), val.data_ptr<float>() + val_off,
                      rhos.data_ptr<float>(), adv.data_ptr<float>(),
-                     vs.data_ptr<float>(), B, S, (float)gamma, (float)rho_bar,
-                     (float)rho_min, (float)c_bar);
+                     vs.data_ptr<float>(), B, S, (int)vD, (float)gamma,
+                     (float)rho_bar, (float)rho_min, (float)c_bar,
+                     (float)rew_scale);
   HIP_CHECK_LAST();
   return {rhos, adv, vs};
 }

@@ -165,6 +165,8 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_kernel(
   float* dhh = wt_hh + G * (H + 1);                   // (S, H) head-grad dh
   float* dg4 = dhh + S * H;                           // (4H)
   float* dxb_s = dg4 + G;                             // (S, H)
+  float* part_h = dxb_s + S * H;                      // (4, H) partial sums
+  float* part_x = part_h + G;                         // (4, H)
 
   // Stage transposed weights (coalesced global read, padded LDS write).
   for (int idx = tid; idx < H * G; idx += G) {
@@ -213,18 +215,29 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_kernel(
     __syncthreads();
     // persist pre-activation gate grads for the weight GEMMs
     dgates[((long)b * S + t) * G + tid] = dg4[tid];
-    if (tid < H) {
-      const int k = tid;
-      // recurrent back-projection + body back-projection (row reads of the
+    {
+      // recurrent + body back-projection, split over all 4H threads:
+      // thread (part, k) sums its quarter of the g range (row reads of the
       // transposed LDS images; lanes stride 1 → conflict-free)
+      const int part = tid / H, k = tid % H;
+      const int gbeg = part * H;
       float acc_h = 0.0f, acc_x = 0.0f;
 #pragma unroll 4
-      for (int g = 0; g < G; ++g) {
+      for (int gg = 0; gg < H; ++gg) {
+        const int g = gbeg + gg;
         const float d = dg4[g];
         acc_h = fmaf(d, wt_hh[g * (H + 1) + k], acc_h);
         acc_x = fmaf(d, wt_ih[g * (H + 1) + k], acc_x);
       }
-      dh_rec = acc_h;
+      part_h[part * H + k] = acc_h;
+      part_x[part * H + k] = acc_x;
+    }
+    __syncthreads();
+    if (tid < H) {
+      const int k = tid;
+      dh_rec = part_h[k] + part_h[H + k] + part_h[2 * H + k] + part_h[3 * H + k];
+      const float acc_x =
+          part_x[k] + part_x[H + k] + part_x[2 * H + k] + part_x[3 * H + k];
       const float xb_v = stash[sbase + k];  // post-ReLU body activation
       const float dxb_v = (xb_v > 0.0f) ? acc_x : 0.0f;
       dxb_s[t * H + k] = dxb_v;
@@ -285,7 +298,7 @@ void launch_bwd(const at::Tensor& gouts, const c10::optional<at::Tensor>& ghS,
                 int D) {
   const int G = 4 * H;
   const int lds =
-      (2 * G * (H + 1) + S * H + G + S * H) * sizeof(float);
+      (2 * G * (H + 1) + S * H + G + S * H + 2 * G) * sizeof(float);
   TORCH_CHECK(lds <= 160 * 1024, "backward LDS footprint exceeds 160 KiB");
   hipLaunchKernelGGL(
       (seq_lstm_bwd_kernel<H>), dim3(B), dim3(G), lds, current_stream(),

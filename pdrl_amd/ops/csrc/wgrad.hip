@@ -1,0 +1,212 @@
+// Weight-gradient kernels for the fused SeqLSTMCore backward (gfx950).
+//
+// The gate-weight gradients ARE the GEMM-shaped hot op of this workload:
+//   dW_ih[m][g] = Σ_n xb[n][m]   · dgates[n][g]     (M=H, N=4H, K=B·S)
+//   dW_hh[m][g] = Σ_n hprev[n][m] · dgates[n][g]
+// i.e. two (64 × 256 × 640) Aᵀ·B GEMMs. hipBLASLt schedules a 256×64
+// macro-tile for this shape (measured 40 µs per GEMM on MI355X — see
+// profiles/), so these run on hand-written MFMA tiles instead:
+// v_mfma_f32_16x16x4_f32 (exact fp32 at the f32 vector rate — CDNA4 guide
+// §3), one 16×16 C-tile per wave, K swept 4 rows per instruction. The hprev
+// operand (h shifted one step, h0 at t=0) is materialized by the load
+// addressing — no torch.cat, no extra memory.
+//
+// The small gradients (body/head weights + all biases) are K=B·S reductions
+// with one WAVE per output element (lane-strided loads + shuffle reduce).
+#include "common.h"
+
+#include <vector>
+
+namespace {
+
+constexpr int kStashFields = 7;  // must match seq_lstm.hip
+
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+// A-operand row n of the gate GEMMs: xb (sel 0) or hprev (sel 1).
+template <int H>
+__device__ __forceinline__ const float* gate_a_row(const float* stash,
+                                                   const float* h0, int n,
+                                                   int S, int sel) {
+  if (sel == 0) {
+    return stash + (long)n * kStashFields * H;  // xb field
+  }
+  const int t = n % S;
+  if (t == 0) {
+    return h0 + (long)(n / S) * H;
+  }
+  return stash + (long)(n - 1) * kStashFields * H + 6 * H;  // h field
+}
+
+// One wave computes one 16x16 tile of C = A^T · B.
+// blockIdx.x → (m0, g_base); 4 waves fan out over g; blockIdx.y → which GEMM.
+template <int H>
+__global__ __launch_bounds__(256) void wgrad_gates_mfma_kernel(
+    const float* __restrict__ stash,   // (B,S,7H)
+    const float* __restrict__ h0,      // (B,H)
+    const float* __restrict__ dgates,  // (N,4H)
+    float* __restrict__ dw_ih,         // (H,4H)
+    float* __restrict__ dw_hh,         // (H,4H)
+    int N, int S) {
+  constexpr int G = 4 * H;
+  const int wave = threadIdx.x / kWave;
+  const int lane = threadIdx.x % kWave;
+  const int g_blocks = G / kWave;  // 64-wide g blocks
+  const int m0 = (blockIdx.x / g_blocks) * 16;
+  const int g0 = (blockIdx.x % g_blocks) * kWave + wave * 16;
+  const int sel = blockIdx.y;
+  float* out = (sel == 0) ? dw_ih : dw_hh;
+
+  const int i = lane & 15;   // row within A frag / col within B frag
+  const int k = lane >> 4;   // inner (n) offset 0..3
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  int n0 = 0;
+  for (; n0 + 4 <= N; n0 += 4) {
+    const float a = gate_a_row<H>(stash, h0, n0 + k, S, sel)[m0 + i];
+    const float b = dgates[(long)(n0 + k) * G + g0 + i];
+    acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+  }
+  // ragged tail (N not divisible by 4): zero-pad the missing rows
+  if (n0 < N) {
+    const bool live = (n0 + k) < N;
+    const float a =
+        live ? gate_a_row<H>(stash, h0, n0 + k, S, sel)[m0 + i] : 0.f;
+    const float b = live ? dgates[(long)(n0 + k) * G + g0 + i] : 0.f;
+    acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+  }
+
+  // C map: col = lane&15, row = (lane>>4)*4 + reg
+  const int c_col = g0 + (lane & 15);
+  const int c_row0 = m0 + (lane >> 4) * 4;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    out[(long)(c_row0 + r) * G + c_col] = acc[r];
+  }
+}
+
+// One wave per output element; lanes stride the K=N reduction.
+// Segments: dbody_w (F*H) | dbody_b (H) | db_g (4H) | dheads_w (H*D) |
+//           dheads_b (D)
+template <int H>
+__global__ void wgrad_small_kernel(
+    const float* __restrict__ x,       // (N,F)
+    const float* __restrict__ dxb,     // (N,H)
+    const float* __restrict__ stash,   // (N,7H)
+    const float* __restrict__ dgates,  // (N,4H)
+    const float* __restrict__ gouts,   // (N,D)
+    float* __restrict__ dbody_w, float* __restrict__ dbody_b,
+    float* __restrict__ db_g, float* __restrict__ dheads_w,
+    float* __restrict__ dheads_b, int N, int F, int D) {
+  constexpr int G = 4 * H;
+  const int wave_id = (blockIdx.x * blockDim.x + threadIdx.x) / kWave;
+  const int lane = threadIdx.x % kWave;
+  const int n_fw = F * H, n_hw = H * D;
+  const int total = n_fw + H + G + n_hw + D;
+  if (wave_id >= total) return;
+
+  // resolve segment
+  const float *pa = nullptr, *pb = nullptr;
+  long stride_a = 0, stride_b = 0;
+  float* out = nullptr;
+  int oi = 0;
+  int e = wave_id;
+  if (e < n_fw) {  // dbody_w[f][j] = sum x[n][f]*dxb[n][j]
+    const int f = e / H, j = e % H;
+    pa = x + f; stride_a = F;
+    pb = dxb + j; stride_b = H;
+    out = dbody_w; oi = e;
+  } else if ((e -= n_fw) < H) {  // dbody_b[j] = sum dxb[n][j]
+    pb = dxb + e; stride_b = H;
+    out = dbody_b; oi = e;
+  } else if ((e -= H) < G) {  // db_g[g] = sum dgates[n][g]
+    pb = dgates + e; stride_b = G;
+    out = db_g; oi = e;
+  } else if ((e -= G) < n_hw) {  // dheads_w[k][d] = sum h[n][k]*gouts[n][d]
+    const int k = e / D, d = e % D;
+    pa = stash + 6 * H + k; stride_a = kStashFields * H;
+    pb = gouts + d; stride_b = D;
+    out = dheads_w; oi = e;
+  } else {  // dheads_b[d] = sum gouts[n][d]
+    e -= n_hw;
+    pb = gouts + e; stride_b = D;
+    out = dheads_b; oi = e;
+  }
+
+  float acc = 0.f;
+  for (int n = lane; n < N; n += kWave) {
+    const float bv = pb[(long)n * stride_b];
+    acc = (pa != nullptr) ? fmaf(pa[(long)n * stride_a], bv, acc) : acc + bv;
+  }
+#pragma unroll
+  for (int off = kWave / 2; off > 0; off >>= 1) acc += __shfl_down(acc, off, kWave);
+  if (lane == 0) out[oi] = acc;
+}
+
+template <int H>
+void launch_wgrad(const at::Tensor& x, const at::Tensor& h0,
+                  const at::Tensor& stash, const at::Tensor& dgates,
+                  const at::Tensor& dxb, const at::Tensor& gouts,
+                  at::Tensor& dw_ih, at::Tensor& dw_hh, at::Tensor& dbody_w,
+                  at::Tensor& dbody_b, at::Tensor& db_g, at::Tensor& dheads_w,
+                  at::Tensor& dheads_b, int N, int S, int F, int D) {
+  constexpr int G = 4 * H;
+  // gate GEMMs: (H/16) m-tiles × (G/64) g-blocks blocks, 2 GEMMs on y
+  dim3 grid((H / 16) * (G / kWave), 2);
+  hipLaunchKernelGGL((wgrad_gates_mfma_kernel<H>), grid, dim3(256), 0,
+                     current_stream(), stash.data_ptr<float>(),
+                     h0.data_ptr<float>(), dgates.data_ptr<float>(),
+                     dw_ih.data_ptr<float>(), dw_hh.data_ptr<float>(), N, S);
+  HIP_CHECK_LAST();
+
+  const int total_waves = F * H + H + G + H * D + D;
+  const int threads = 256;
+  const int blocks = (total_waves * kWave + threads - 1) / threads;
+  hipLaunchKernelGGL((wgrad_small_kernel<H>), dim3(blocks), dim3(threads), 0,
+                     current_stream(), x.data_ptr<float>(),
+                     dxb.data_ptr<float>(), stash.data_ptr<float>(),
+                     dgates.data_ptr<float>(), gouts.data_ptr<float>(),
+                     dbody_w.data_ptr<float>(), dbody_b.data_ptr<float>(),
+                     db_g.data_ptr<float>(), dheads_w.data_ptr<float>(),
+                     dheads_b.data_ptr<float>(), N, F, D);
+  HIP_CHECK_LAST();
+}
+
+}  // namespace
+
+void seq_lstm_wgrad_out_hip(const at::Tensor& x, const at::Tensor& h0,
+                            const at::Tensor& stash, const at::Tensor& dgates,
+                            const at::Tensor& dxb, const at::Tensor& gouts,
+                            at::Tensor& dw_ih, at::Tensor& dw_hh,
+                            at::Tensor& dbody_w, at::Tensor& dbody_b,
+                            at::Tensor& db_g, at::Tensor& dheads_w,
+                            at::Tensor& dheads_b) {
+  CHECK_IN(x); CHECK_IN(h0); CHECK_IN(stash); CHECK_IN(dgates);
+  CHECK_IN(dxb); CHECK_IN(gouts);
+  const int B = x.size(0), S = x.size(1), F = x.size(2);
+  const int H = h0.size(1), D = gouts.size(2);
+  const int N = B * S;
+  switch (H) {
+    case 32: launch_wgrad<32>(x, h0, stash, dgates, dxb, gouts, dw_ih, dw_hh, dbody_w, dbody_b, db_g, dheads_w, dheads_b, N, S, F, D); break;
+    case 64: launch_wgrad<64>(x, h0, stash, dgates, dxb, gouts, dw_ih, dw_hh, dbody_w, dbody_b, db_g, dheads_w, dheads_b, N, S, F, D); break;
+    case 128: launch_wgrad<128>(x, h0, stash, dgates, dxb, gouts, dw_ih, dw_hh, dbody_w, dbody_b, db_g, dheads_w, dheads_b, N, S, F, D); break;
+    default: TORCH_CHECK(false, "hidden size ", H, " unsupported");
+  }
+}
+
+std::vector<at::Tensor> seq_lstm_wgrad_hip(
+    const at::Tensor& x, const at::Tensor& h0, const at::Tensor& stash,
+    const at::Tensor& dgates, const at::Tensor& dxb, const at::Tensor& gouts) {
+  const int F = x.size(2), H = h0.size(1), D = gouts.size(2);
+  auto opt = x.options();
+  auto dw_ih = at::empty({H, 4 * H}, opt);
+  auto dw_hh = at::empty({H, 4 * H}, opt);
+  auto dbody_w = at::empty({F, H}, opt);
+  auto dbody_b = at::empty({H}, opt);
+  auto db_g = at::empty({4 * H}, opt);
+  auto dheads_w = at::empty({H, D}, opt);
+  auto dheads_b = at::empty({D}, opt);
+  seq_lstm_wgrad_out_hip(x, h0, stash, dgates, dxb, gouts, dw_ih, dw_hh,
+                         dbody_w, dbody_b, db_g, dheads_w, dheads_b);
+  return {dw_ih, dw_hh, dbody_w, dbody_b, db_g, dheads_w, dheads_b};
+}

@@ -16,6 +16,8 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 SRC = [
     "pdrl_amd/ops/csrc/bindings.cpp",
     "pdrl_amd/ops/csrc/seq_lstm.hip",
+    "pdrl_amd/ops/csrc/wgrad.hip",
+    "pdrl_amd/ops/csrc/losses.hip",
     "pdrl_amd/ops/csrc/scans.hip",
     "pdrl_amd/ops/csrc/multi_tensor.hip",
 ]

@@ -195,7 +195,7 @@ def test_updater_step_on_gpu(algo, continuous):
     batch["obs"] = torch.randn(p.batch_size, p.seq_len, p.obs_dim, device=DEV)
     for _ in range(3):
         stats = upd.step(batch)
-    assert all(np.isfinite(v) for v in stats.values()), stats
+    assert all(np.isfinite(float(v)) for v in stats.values()), stats
 
 
 def test_gpu_vs_cpu_updater_trajectories_match():
@@ -226,3 +226,36 @@ def test_gpu_vs_cpu_updater_trajectories_match():
         torch.testing.assert_close(pg.detach().cpu(), pc.detach(), rtol=2e-3,
                                    atol=2e-5, msg=lambda m: f"{n}: {m}")
     assert abs(sg["loss-total"] - sc["loss-total"]) < 1e-2
+
+
+def test_fused_step_engaged_and_ppo_parity():
+    """PPO on GPU runs the fused whole-step DAG; parameters track the CPU
+    eager oracle across 3 updates."""
+    _ops()
+    from pdrl_amd.agents.learner_module import PPOUpdater
+    from pdrl_amd.networks import MlpLSTMSingle
+    from pdrl_amd.utils import load_params
+    from tests.conftest import make_batch
+
+    p = load_params()
+    p.batch_size, p.seq_len, p.obs_dim, p.n_actions = 16, 5, 4, 2
+
+    torch.manual_seed(7)
+    model_g = MlpLSTMSingle(4, 2, p.seq_len, p.hidden_size)
+    torch.manual_seed(7)
+    model_c = MlpLSTMSingle(4, 2, p.seq_len, p.hidden_size)
+
+    upd_g = PPOUpdater(model_g, p, DEV)
+    assert upd_g.fused_step is not None, "fused step must engage on GPU"
+    upd_c = PPOUpdater(model_c, p, "cpu")
+    batch_c = make_batch(p, seed=11)
+    batch_g = {k: v.to(DEV) for k, v in batch_c.items()}
+    for _ in range(3):
+        sg = upd_g.step(batch_g)
+        sc = upd_c.step(batch_c)
+    for (n, pg), pc in zip(model_g.named_parameters(), model_c.parameters()):
+        torch.testing.assert_close(pg.detach().cpu(), pc.detach(), rtol=2e-3,
+                                   atol=2e-5, msg=lambda m: f"{n}: {m}")
+    for k in ("loss-total", "loss-policy", "loss-value", "entropy",
+              "ratio-avg", "ratio-min", "ratio-max"):
+        assert abs(float(sg[k]) - float(sc[k])) < 5e-3, (k, float(sg[k]), float(sc[k]))
