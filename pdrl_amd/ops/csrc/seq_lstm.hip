@@ -51,7 +51,7 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_fwd_kernel(
     float* __restrict__ hS,            // (B,H)
     float* __restrict__ cS,            // (B,H)
     float* __restrict__ stash,         // (B,S,7H)
-    int S, int F, int D) {
+    int S, int F, int D, long h0s) {   // h0s: row stride of h0/c0
   constexpr int G = 4 * H;
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
@@ -82,8 +82,8 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_fwd_kernel(
     stash[(((long)b * S + t) * kStashFields) * H + j] = acc;
   }
   if (tid < H) {
-    hbuf[tid] = h0[(long)b * H + tid];
-    cbuf[tid] = c0[(long)b * H + tid];
+    hbuf[tid] = h0[(long)b * h0s + tid];
+    cbuf[tid] = c0[(long)b * h0s + tid];
   }
   __syncthreads();
 
@@ -151,7 +151,7 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_kernel(
     float* __restrict__ dc0,           // (B,H)
     float* __restrict__ dgates,        // (B,S,4H) pre-activation
     float* __restrict__ dxb,           // (B,S,H) pre-ReLU
-    int S, int F, int D) {
+    int S, int F, int D, long h0s) {
   constexpr int G = 4 * H;
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
@@ -203,7 +203,7 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_kernel(
       const float tc = tanhf(c_);
       const float c_prev =
           (t > 0) ? stash[sbase - kStashFields * H + 5 * H + k]
-                  : c0[(long)b * H + k];
+                  : c0[(long)b * h0s + k];
       const float dh = dhh[t * H + k] + dh_rec;
       const float dc = dc_rec + dh * o_ * (1.0f - tc * tc);
       dg4[k] = dc * g_ * i_ * (1.0f - i_);
@@ -283,7 +283,7 @@ void launch_fwd(const at::Tensor& x, const at::Tensor& h0, const at::Tensor& c0,
                      heads_w.data_ptr<float>(),
                      heads_b.data_ptr<float>(), outs.data_ptr<float>(),
                      hS.data_ptr<float>(), cS.data_ptr<float>(),
-                     stash.data_ptr<float>(), S, F, D);
+                     stash.data_ptr<float>(), S, F, D, (long)h0.stride(0));
   HIP_CHECK_LAST();
 }
 
@@ -310,7 +310,7 @@ void launch_bwd(const at::Tensor& gouts, const c10::optional<at::Tensor>& ghS,
       w_ih.data_ptr<float>(), w_hh.data_ptr<float>(),
       heads_w.data_ptr<float>(), dx.data_ptr<float>(),
       dh0.data_ptr<float>(), dc0.data_ptr<float>(), dgates.data_ptr<float>(),
-      dxb.data_ptr<float>(), S, F, D);
+      dxb.data_ptr<float>(), S, F, D, (long)c0.stride(0));
   HIP_CHECK_LAST();
 }
 
@@ -321,9 +321,12 @@ std::vector<at::Tensor> seq_lstm_forward_hip(
     const at::Tensor& body_w, const at::Tensor& body_b, const at::Tensor& w_ih,
     const at::Tensor& w_hh, const at::Tensor& b_g, const at::Tensor& heads_w,
     const at::Tensor& heads_b) {
-  CHECK_IN(x); CHECK_IN(h0); CHECK_IN(c0); CHECK_IN(body_w); CHECK_IN(body_b);
+  CHECK_IN(x); CHECK_IN(body_w); CHECK_IN(body_b);
   CHECK_IN(w_ih); CHECK_IN(w_hh); CHECK_IN(b_g); CHECK_IN(heads_w);
   CHECK_IN(heads_b);
+  CHECK_GPU(h0); CHECK_F32(h0); CHECK_GPU(c0); CHECK_F32(c0);
+  TORCH_CHECK(h0.stride(1) == 1 && c0.stride(1) == 1, "h0/c0 inner stride");
+  TORCH_CHECK(h0.stride(0) == c0.stride(0), "h0/c0 stride mismatch");
   const int B = x.size(0), S = x.size(1), F = x.size(2);
   const int H = h0.size(1), D = heads_w.size(1);
   TORCH_CHECK(body_w.size(0) == F && body_w.size(1) == H, "body_w shape");
@@ -352,8 +355,10 @@ std::vector<at::Tensor> seq_lstm_backward_core_hip(
     const at::Tensor& x, const at::Tensor& c0, const at::Tensor& body_w,
     const at::Tensor& w_ih, const at::Tensor& w_hh,
     const at::Tensor& heads_w) {
-  CHECK_IN(gouts); CHECK_IN(stash); CHECK_IN(x); CHECK_IN(c0);
+  CHECK_IN(gouts); CHECK_IN(stash); CHECK_IN(x);
   CHECK_IN(body_w); CHECK_IN(w_ih); CHECK_IN(w_hh); CHECK_IN(heads_w);
+  CHECK_GPU(c0); CHECK_F32(c0);
+  TORCH_CHECK(c0.stride(1) == 1, "c0 inner stride must be 1");
   const int B = x.size(0), S = x.size(1), F = x.size(2);
   const int H = c0.size(1), D = heads_w.size(1);
 

@@ -27,13 +27,13 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 template <int H>
 __device__ __forceinline__ const float* gate_a_row(const float* stash,
                                                    const float* h0, int n,
-                                                   int S, int sel) {
+                                                   int S, int sel, long h0s) {
   if (sel == 0) {
     return stash + (long)n * kStashFields * H;  // xb field
   }
   const int t = n % S;
   if (t == 0) {
-    return h0 + (long)(n / S) * H;
+    return h0 + (long)(n / S) * h0s;
   }
   return stash + (long)(n - 1) * kStashFields * H + 6 * H;  // h field
 }
@@ -47,7 +47,7 @@ __global__ __launch_bounds__(256) void wgrad_gates_mfma_kernel(
     const float* __restrict__ dgates,  // (N,4H)
     float* __restrict__ dw_ih,         // (H,4H)
     float* __restrict__ dw_hh,         // (H,4H)
-    int N, int S) {
+    int N, int S, long h0s) {
   constexpr int G = 4 * H;
   const int wave = threadIdx.x / kWave;
   const int lane = threadIdx.x % kWave;
@@ -60,18 +60,30 @@ __global__ __launch_bounds__(256) void wgrad_gates_mfma_kernel(
   const int i = lane & 15;   // row within A frag / col within B frag
   const int k = lane >> 4;   // inner (n) offset 0..3
 
+  // Software-pipelined K sweep: issue a chunk of 2×U independent loads, then
+  // U MFMAs — keeps ~16 loads in flight so the 16x16x4 f32 MFMA chain never
+  // stalls on L2 latency (the naive load→mfma→load loop measured 61 µs;
+  // pipelined ~an order less).
+  constexpr int U = 8;  // MFMAs per chunk, 32 n-rows
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
   int n0 = 0;
-  for (; n0 + 4 <= N; n0 += 4) {
-    const float a = gate_a_row<H>(stash, h0, n0 + k, S, sel)[m0 + i];
-    const float b = dgates[(long)(n0 + k) * G + g0 + i];
-    acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+  for (; n0 + 4 * U <= N; n0 += 4 * U) {
+    float a[U], b[U];
+#pragma unroll
+    for (int u = 0; u < U; ++u) {
+      const int n = n0 + 4 * u + k;
+      a[u] = gate_a_row<H>(stash, h0, n, S, sel, h0s)[m0 + i];
+      b[u] = dgates[(long)n * G + g0 + i];
+    }
+#pragma unroll
+    for (int u = 0; u < U; ++u) {
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a[u], b[u], acc, 0, 0, 0);
+    }
   }
-  // ragged tail (N not divisible by 4): zero-pad the missing rows
-  if (n0 < N) {
+  for (; n0 < N; n0 += 4) {  // ragged tail, zero-padded
     const bool live = (n0 + k) < N;
     const float a =
-        live ? gate_a_row<H>(stash, h0, n0 + k, S, sel)[m0 + i] : 0.f;
+        live ? gate_a_row<H>(stash, h0, n0 + k, S, sel, h0s)[m0 + i] : 0.f;
     const float b = live ? dgates[(long)(n0 + k) * G + g0 + i] : 0.f;
     acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
   }
@@ -156,7 +168,8 @@ void launch_wgrad(const at::Tensor& x, const at::Tensor& h0,
   hipLaunchKernelGGL((wgrad_gates_mfma_kernel<H>), grid, dim3(256), 0,
                      current_stream(), stash.data_ptr<float>(),
                      h0.data_ptr<float>(), dgates.data_ptr<float>(),
-                     dw_ih.data_ptr<float>(), dw_hh.data_ptr<float>(), N, S);
+                     dw_ih.data_ptr<float>(), dw_hh.data_ptr<float>(), N, S,
+                     (long)h0.stride(0));
   HIP_CHECK_LAST();
 
   const int total_waves = F * H + H + G + H * D + D;
@@ -181,8 +194,10 @@ void seq_lstm_wgrad_out_hip(const at::Tensor& x, const at::Tensor& h0,
                             at::Tensor& dbody_w, at::Tensor& dbody_b,
                             at::Tensor& db_g, at::Tensor& dheads_w,
                             at::Tensor& dheads_b) {
-  CHECK_IN(x); CHECK_IN(h0); CHECK_IN(stash); CHECK_IN(dgates);
+  CHECK_IN(x); CHECK_IN(stash); CHECK_IN(dgates);
   CHECK_IN(dxb); CHECK_IN(gouts);
+  CHECK_GPU(h0); CHECK_F32(h0);
+  TORCH_CHECK(h0.stride(1) == 1, "h0 inner stride must be 1");
   const int B = x.size(0), S = x.size(1), F = x.size(2);
   const int H = h0.size(1), D = gouts.size(2);
   const int N = B * S;

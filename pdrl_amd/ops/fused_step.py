@@ -63,13 +63,20 @@ class FusedOnPolicyStep:
         # wgrad_out writes (dw_ih, dw_hh, dbody_w, dbody_b, db_g, dheads_w, dheads_b)
         return [gs[2], gs[3], gs[0], gs[1], gs[4], gs[5], gs[6]]
 
-    def _body(self, batch):
+    def compute_grads_only(self, batch):
+        """Kernels 1-7 only (for gradient parity tests): fills the flat grad
+        buffer without touching the optimizer."""
+        self._body(batch, update=False)
+
+    def _body(self, batch, update: bool = True):
         c, p, A = self.core, self.params, self.A
         e = ext()
         x = batch["obs"]
         B, S, _ = x.shape
-        hx0 = batch["hx"][:, 0].contiguous()
-        cx0 = batch["cx"][:, 0].contiguous()
+        # strided row views (stride S*H) — the kernels take h0/c0 strides,
+        # so no .contiguous() copies here
+        hx0 = batch["hx"][:, 0]
+        cx0 = batch["cx"][:, 0]
         act = batch["act"].reshape(-1)
         rew = batch["rew"].reshape(B, S)
         behav = batch["log_prob"].reshape(B, S)
@@ -115,6 +122,8 @@ class FusedOnPolicyStep:
         )
         e.seq_lstm_wgrad_out(x, hx0, stash, dgates, dxb, gouts,
                              *self._grad_views())
+        if not update:
+            return
         if self.grad_reducer is not None:
             self.grad_reducer.all_reduce([self.optimizer.flat_grad])
         self.optimizer.step()
@@ -125,9 +134,10 @@ class FusedOnPolicyStep:
 
     # ------------------------------------------------------------------ #
     def _try_capture(self, batch):
-        self._static = {
-            k: batch[k].detach().clone().contiguous() for k in BATCH_FIELDS
-        }
+        # capture on the CALLER's tensors: when later batches reuse the same
+        # storage (bench; packed-staging learner) replay needs ZERO copies
+        self._static = {k: batch[k] for k in BATCH_FIELDS}
+        self._static_ptrs = {k: batch[k].data_ptr() for k in BATCH_FIELDS}
         try:
             side = torch.cuda.Stream()
             side.wait_stream(torch.cuda.current_stream())
@@ -151,7 +161,8 @@ class FusedOnPolicyStep:
             self._try_capture(batch)
         if self._graph is not None:
             for k in BATCH_FIELDS:
-                self._static[k].copy_(batch[k], non_blocking=True)
+                if batch[k].data_ptr() != self._static_ptrs[k]:
+                    self._static[k].copy_(batch[k], non_blocking=True)
             self._graph.replay()
         else:
             self._full(batch)
