@@ -47,6 +47,15 @@ at::Tensor ppo_loss_bwd_hip(const at::Tensor&, long, const at::Tensor&,
                             const at::Tensor&, const at::Tensor&,
                             const at::Tensor&, const at::Tensor&, double,
                             double, double, double);
+bool impala_loss_mega_hip(const at::Tensor&, const at::Tensor&,
+                          const at::Tensor&, const at::Tensor&,
+                          const at::Tensor&, at::Tensor&, at::Tensor&, long,
+                          double, double, double, double, double, double,
+                          double, double);
+bool ppo_loss_mega_hip(const at::Tensor&, const at::Tensor&, const at::Tensor&,
+                       const at::Tensor&, const at::Tensor&, at::Tensor&,
+                       at::Tensor&, long, double, double, double, double,
+                       double, double, double);
 void l2norm_sq_hip(const at::Tensor&, at::Tensor&);
 void rmsprop_step_hip(at::Tensor&, const at::Tensor&, at::Tensor&,
                       const at::Tensor&, double, double, double, double);
@@ -79,6 +88,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ppo_loss_reduce", &ppo_loss_reduce_hip, "PPO loss stats");
   m.def("impala_loss_bwd", &impala_loss_bwd_hip, "analytic IMPALA loss grad");
   m.def("ppo_loss_bwd", &ppo_loss_bwd_hip, "analytic PPO loss grad");
+  m.def("impala_loss_mega", &impala_loss_mega_hip,
+        "single-launch IMPALA loss: stats+vtrace+reduce+bwd");
+  m.def("ppo_loss_mega", &ppo_loss_mega_hip,
+        "single-launch PPO loss: stats+gae+reduce+bwd");
   m.def("l2norm_sq", &l2norm_sq_hip, "squared L2 norm into a device scalar");
   m.def("rmsprop_step", &rmsprop_step_hip, "fused clip+RMSprop on flat buffers");
   m.def("adam_step", &adam_step_hip, "fused clip+Adam on flat buffers");

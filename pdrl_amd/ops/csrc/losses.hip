@@ -370,3 +370,301 @@ at::Tensor ppo_loss_bwd_hip(const at::Tensor& model_out, long A,
   HIP_CHECK_LAST();
   return gouts;
 }
+
+// ------------------------------------------------------------------------
+// Single-block MEGA loss kernels: cat_stats + scan + reduce + analytic bwd
+// in ONE launch (the 4-kernel sequence above costs ~4 dependent-kernel
+// boundaries ≈ 18 µs at this size; the whole loss fits one CU's LDS for
+// B·S ≤ ~2500). Falls back to the 4-kernel path for larger shapes.
+namespace {
+
+constexpr int kMegaThreads = 256;
+
+// LDS layout helpers (floats): logp[N] | ent[N] | lse[N] | w0[BT] | w1[BT]
+// | w2[BT] — w* = {rhos, adv, vs} for IMPALA, {td, adv, -} for PPO.
+
+__global__ __launch_bounds__(kMegaThreads) void impala_loss_mega_kernel(
+    const float* __restrict__ mo,     // (N,D)
+    const float* __restrict__ act,    // (N)
+    const float* __restrict__ behav,  // (B,S)
+    const float* __restrict__ rew,    // (B,S)
+    const float* __restrict__ fir,    // (B,S)
+    float* __restrict__ gouts,        // (B,S,D)
+    float* __restrict__ stats,        // (5)
+    int B, int S, int A, float gamma, float rho_bar, float rho_min,
+    float c_bar, float rew_scale, float cp, float cv, float ce) {
+  const int D = A + 1;
+  const int T = S - 1;
+  const int N = B * S;
+  const int BT = B * T;
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* s_logp = reinterpret_cast<float*>(smem_raw);
+  float* s_ent = s_logp + N;
+  float* s_lse = s_ent + N;
+  float* s_rho = s_lse + N;
+  float* s_adv = s_rho + BT;
+  float* s_vs = s_adv + BT;
+  const int tid = threadIdx.x;
+
+  // phase A: categorical stats
+  for (int i = tid; i < N; i += kMegaThreads) {
+    const float* z = mo + (long)i * D;
+    float m = z[0];
+    for (int j = 1; j < A; ++j) m = fmaxf(m, z[j]);
+    float s = 0.f;
+    for (int j = 0; j < A; ++j) s += __expf(z[j] - m);
+    const float l = m + __logf(s);
+    float h = 0.f;
+    for (int j = 0; j < A; ++j) {
+      const float lp = z[j] - l;
+      h -= __expf(lp) * lp;
+    }
+    s_logp[i] = z[(int)act[i]] - l;
+    s_ent[i] = h;
+    s_lse[i] = l;
+  }
+  __syncthreads();
+
+  // phase B: V-trace scan, one thread per batch row
+  for (int b = tid; b < B; b += kMegaThreads) {
+    const long sb = (long)b * S, tb = (long)b * T;
+    float acc = 0.f;
+    for (int t = T - 1; t >= 0; --t) {
+      const float ratio = __expf(s_logp[sb + t] - behav[sb + t]);
+      const float rho = fminf(fmaxf(ratio, rho_min), rho_bar);
+      const float c = fminf(ratio, c_bar);
+      const float mask = 1.f - fir[sb + t + 1];
+      const float vt = mo[(sb + t) * D + A], vn = mo[(sb + t + 1) * D + A];
+      const float delta = rho * (rew[sb + t] * rew_scale + gamma * mask * vn - vt);
+      acc = fmaf(gamma * mask * c, acc, delta);
+      s_rho[tb + t] = rho;
+      s_vs[tb + t] = vt + acc;
+    }
+    for (int t = 0; t < T; ++t) {
+      const float mask = 1.f - fir[sb + t + 1];
+      const float vnext = (t + 1 < T) ? s_vs[tb + t + 1] : mo[(sb + T) * D + A];
+      s_adv[tb + t] = s_rho[tb + t] * (rew[sb + t] * rew_scale +
+                                       gamma * mask * vnext - mo[(sb + t) * D + A]);
+    }
+  }
+  __syncthreads();
+
+  // phase C: loss reduction
+  {
+    float pl = 0, vl = 0, es = 0, rs = 0;
+    for (int i = tid; i < BT; i += kMegaThreads) {
+      const int b = i / T, t = i % T;
+      const long si = (long)b * S + t;
+      pl -= s_logp[si] * s_adv[i];
+      vl += huber(mo[si * D + A] - s_vs[i]);
+      es += s_ent[si];
+      rs += s_rho[i];
+    }
+    __shared__ float red[4][kMegaThreads];
+    red[0][tid] = pl; red[1][tid] = vl; red[2][tid] = es; red[3][tid] = rs;
+    __syncthreads();
+    for (int off = kMegaThreads / 2; off > 0; off >>= 1) {
+      if (tid < off)
+        for (int r = 0; r < 4; ++r) red[r][tid] += red[r][tid + off];
+      __syncthreads();
+    }
+    if (tid == 0) {
+      const float inv = 1.0f / BT;
+      const float p = red[0][0] * inv, v = red[1][0] * inv, e = red[2][0] * inv;
+      stats[0] = cp * p + cv * v - ce * e;
+      stats[1] = p; stats[2] = v; stats[3] = e; stats[4] = red[3][0] * inv;
+    }
+  }
+
+  // phase D: analytic backward into packed gouts
+  const float invN = 1.0f / BT;
+  for (int i = tid; i < N; i += kMegaThreads) {
+    const int t = i % S, b = i / S;
+    float* g = gouts + (long)i * D;
+    if (t >= T) {
+      for (int j = 0; j < D; ++j) g[j] = 0.f;
+      continue;
+    }
+    const long ti = (long)b * T + t;
+    const float dlogp = -cp * s_adv[ti] * invN;
+    const float dH = -ce * invN;
+    const float H = s_ent[i];
+    const float* z = mo + (long)i * D;
+    const int a = (int)act[i];
+    for (int j = 0; j < A; ++j) {
+      const float lp = z[j] - s_lse[i];
+      const float pj = __expf(lp);
+      g[j] = dlogp * ((j == a ? 1.f : 0.f) - pj) + dH * (-pj * (lp + H));
+    }
+    g[A] = cv * huber_grad(z[A] - s_vs[ti]) * invN;
+  }
+}
+
+__global__ __launch_bounds__(kMegaThreads) void ppo_loss_mega_kernel(
+    const float* __restrict__ mo,     // (N,D)
+    const float* __restrict__ act,    // (N)
+    const float* __restrict__ behav,  // (B,S)
+    const float* __restrict__ rew,    // (B,S)
+    const float* __restrict__ fir,    // (B,S)
+    float* __restrict__ gouts,        // (B,S,D)
+    float* __restrict__ stats,        // (7)
+    int B, int S, int A, float gamma, float lmbda, float rew_scale, float cp,
+    float cv, float ce, float eps_clip) {
+  const int D = A + 1;
+  const int T = S - 1;
+  const int N = B * S;
+  const int BT = B * T;
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* s_logp = reinterpret_cast<float*>(smem_raw);
+  float* s_ent = s_logp + N;
+  float* s_lse = s_ent + N;
+  float* s_td = s_lse + N;
+  float* s_adv = s_td + BT;
+  const int tid = threadIdx.x;
+
+  for (int i = tid; i < N; i += kMegaThreads) {
+    const float* z = mo + (long)i * D;
+    float m = z[0];
+    for (int j = 1; j < A; ++j) m = fmaxf(m, z[j]);
+    float s = 0.f;
+    for (int j = 0; j < A; ++j) s += __expf(z[j] - m);
+    const float l = m + __logf(s);
+    float h = 0.f;
+    for (int j = 0; j < A; ++j) {
+      const float lp = z[j] - l;
+      h -= __expf(lp) * lp;
+    }
+    s_logp[i] = z[(int)act[i]] - l;
+    s_ent[i] = h;
+    s_lse[i] = l;
+  }
+  __syncthreads();
+
+  for (int b = tid; b < B; b += kMegaThreads) {
+    const long sb = (long)b * S, tb = (long)b * T;
+    float run = 0.f;
+    for (int t = T - 1; t >= 0; --t) {
+      const float mask = 1.f - fir[sb + t + 1];
+      const float tdv = rew[sb + t] * rew_scale +
+                        gamma * mask * mo[(sb + t + 1) * D + A];
+      const float delta = tdv - mo[(sb + t) * D + A];
+      run = fmaf(gamma * lmbda * mask, run, delta);
+      s_td[tb + t] = tdv;
+      s_adv[tb + t] = run;
+    }
+  }
+  __syncthreads();
+
+  {
+    float pl = 0, vl = 0, es = 0, ravg = 0;
+    float rmin = 1e30f, rmax = -1e30f;
+    for (int i = tid; i < BT; i += kMegaThreads) {
+      const int b = i / T, t = i % T;
+      const long si = (long)b * S + t;
+      const float r = __expf(s_logp[si] - behav[si]);
+      const float a = s_adv[i];
+      const float s1 = r * a;
+      const float s2 = fminf(fmaxf(r, 1.f - eps_clip), 1.f + eps_clip) * a;
+      pl -= fminf(s1, s2);
+      vl += huber(mo[si * D + A] - s_td[i]);
+      es += s_ent[si];
+      ravg += r;
+      rmin = fminf(rmin, r);
+      rmax = fmaxf(rmax, r);
+    }
+    __shared__ float red[4][kMegaThreads];
+    __shared__ float rmn[kMegaThreads], rmx[kMegaThreads];
+    red[0][tid] = pl; red[1][tid] = vl; red[2][tid] = es; red[3][tid] = ravg;
+    rmn[tid] = rmin; rmx[tid] = rmax;
+    __syncthreads();
+    for (int off = kMegaThreads / 2; off > 0; off >>= 1) {
+      if (tid < off) {
+        for (int r = 0; r < 4; ++r) red[r][tid] += red[r][tid + off];
+        rmn[tid] = fminf(rmn[tid], rmn[tid + off]);
+        rmx[tid] = fmaxf(rmx[tid], rmx[tid + off]);
+      }
+      __syncthreads();
+    }
+    if (tid == 0) {
+      const float inv = 1.0f / BT;
+      const float p = red[0][0] * inv, v = red[1][0] * inv, e = red[2][0] * inv;
+      stats[0] = cp * p + cv * v - ce * e;
+      stats[1] = p; stats[2] = v; stats[3] = e;
+      stats[4] = red[3][0] * inv; stats[5] = rmn[0]; stats[6] = rmx[0];
+    }
+  }
+
+  const float invN = 1.0f / BT;
+  for (int i = tid; i < N; i += kMegaThreads) {
+    const int t = i % S, b = i / S;
+    float* g = gouts + (long)i * D;
+    if (t >= T) {
+      for (int j = 0; j < D; ++j) g[j] = 0.f;
+      continue;
+    }
+    const long ti = (long)b * T + t;
+    const float a_v = s_adv[ti];
+    const float r = __expf(s_logp[i] - behav[i]);
+    const bool inside = (r > 1.f - eps_clip) && (r < 1.f + eps_clip);
+    const float s1 = r * a_v;
+    const float s2 = fminf(fmaxf(r, 1.f - eps_clip), 1.f + eps_clip) * a_v;
+    const float gr = (inside || s1 < s2) ? a_v * r : 0.f;
+    const float dlogp = -cp * gr * invN;
+    const float dH = -ce * invN;
+    const float H = s_ent[i];
+    const float* z = mo + (long)i * D;
+    const int a = (int)act[i];
+    for (int j = 0; j < A; ++j) {
+      const float lp = z[j] - s_lse[i];
+      const float pj = __expf(lp);
+      g[j] = dlogp * ((j == a ? 1.f : 0.f) - pj) + dH * (-pj * (lp + H));
+    }
+    g[A] = cv * huber_grad(z[A] - s_td[ti]) * invN;
+  }
+}
+
+}  // namespace
+
+// returns true when the mega (single-launch) path handled this shape
+bool impala_loss_mega_hip(const at::Tensor& mo, const at::Tensor& act,
+                          const at::Tensor& behav, const at::Tensor& rew,
+                          const at::Tensor& fir, at::Tensor& gouts,
+                          at::Tensor& stats, long A, double gamma,
+                          double rho_bar, double rho_min, double c_bar,
+                          double rew_scale, double cp, double cv, double ce) {
+  const int B = mo.size(0), S = mo.size(1);
+  const int N = B * S, BT = B * (S - 1);
+  const long lds = (3L * N + 3L * BT) * sizeof(float);
+  if (lds > 64 * 1024) return false;
+  hipLaunchKernelGGL(impala_loss_mega_kernel, dim3(1), dim3(256), lds,
+                     current_stream(), mo.data_ptr<float>(),
+                     act.data_ptr<float>(), behav.data_ptr<float>(),
+                     rew.data_ptr<float>(), fir.data_ptr<float>(),
+                     gouts.data_ptr<float>(), stats.data_ptr<float>(), B, S,
+                     (int)A, (float)gamma, (float)rho_bar, (float)rho_min,
+                     (float)c_bar, (float)rew_scale, (float)cp, (float)cv,
+                     (float)ce);
+  HIP_CHECK_LAST();
+  return true;
+}
+
+bool ppo_loss_mega_hip(const at::Tensor& mo, const at::Tensor& act,
+                       const at::Tensor& behav, const at::Tensor& rew,
+                       const at::Tensor& fir, at::Tensor& gouts,
+                       at::Tensor& stats, long A, double gamma, double lmbda,
+                       double rew_scale, double cp, double cv, double ce,
+                       double eps_clip) {
+  const int B = mo.size(0), S = mo.size(1);
+  const int N = B * S, BT = B * (S - 1);
+  const long lds = (3L * N + 2L * BT) * sizeof(float);
+  if (lds > 64 * 1024) return false;
+  hipLaunchKernelGGL(ppo_loss_mega_kernel, dim3(1), dim3(256), lds,
+                     current_stream(), mo.data_ptr<float>(),
+                     act.data_ptr<float>(), behav.data_ptr<float>(),
+                     rew.data_ptr<float>(), fir.data_ptr<float>(),
+                     gouts.data_ptr<float>(), stats.data_ptr<float>(), B, S,
+                     (int)A, (float)gamma, (float)lmbda, (float)rew_scale,
+                     (float)cp, (float)cv, (float)ce, (float)eps_clip);
+  HIP_CHECK_LAST();
+  return true;
+}

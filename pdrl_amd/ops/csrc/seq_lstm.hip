@@ -157,22 +157,22 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_kernel(
   const int tid = threadIdx.x;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  // transposed, padded weight images: wt[g][k] at g*(H+1)+k — the pad makes
-  // both the cooperative scatter-store and the lane-broadcast row read
-  // conflict-free (bank = (g*(H+1)+k) % 32 varies with g at fixed k).
-  float* wt_ih = reinterpret_cast<float*>(smem_raw);  // (4H, H+1)
-  float* wt_hh = wt_ih + G * (H + 1);                 // (4H, H+1)
-  float* dhh = wt_hh + G * (H + 1);                   // (S, H) head-grad dh
+  float* dhh = reinterpret_cast<float*>(smem_raw);    // (S, H) head-grad dh
   float* dg4 = dhh + S * H;                           // (4H)
   float* dxb_s = dg4 + G;                             // (S, H)
   float* part_h = dxb_s + S * H;                      // (4, H) partial sums
   float* part_x = part_h + G;                         // (4, H)
 
-  // Stage transposed weights (coalesced global read, padded LDS write).
-  for (int idx = tid; idx < H * G; idx += G) {
-    const int k = idx / G, g = idx % G;
-    wt_ih[g * (H + 1) + k] = w_ih[idx];
-    wt_hh[g * (H + 1) + k] = w_hh[idx];
+  // Register-resident weight rows for the back-projections: thread
+  // (part, k) = (tid/H, tid%H) owns its quarter of rows k of w_ih / w_hh —
+  // the reductions then run entirely on registers + dg4 LDS broadcasts
+  // (no 131 KiB LDS staging, no staging barrier).
+  const int part = tid / H, kk = tid % H;
+  float wih_row[H], whh_row[H];
+#pragma unroll
+  for (int gg = 0; gg < H; ++gg) {
+    wih_row[gg] = w_ih[(long)kk * G + part * H + gg];
+    whh_row[gg] = w_hh[(long)kk * G + part * H + gg];
   }
 
   // Head back-projection: dh_heads[t][k] = sum_d gouts[t][d] * heads_w[k][d]
@@ -216,21 +216,18 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_kernel(
     // persist pre-activation gate grads for the weight GEMMs
     dgates[((long)b * S + t) * G + tid] = dg4[tid];
     {
-      // recurrent + body back-projection, split over all 4H threads:
-      // thread (part, k) sums its quarter of the g range (row reads of the
-      // transposed LDS images; lanes stride 1 → conflict-free)
-      const int part = tid / H, k = tid % H;
+      // recurrent + body back-projection, split over all 4H threads on
+      // register-resident weight rows (dg4 reads broadcast from LDS)
       const int gbeg = part * H;
       float acc_h = 0.0f, acc_x = 0.0f;
 #pragma unroll 4
       for (int gg = 0; gg < H; ++gg) {
-        const int g = gbeg + gg;
-        const float d = dg4[g];
-        acc_h = fmaf(d, wt_hh[g * (H + 1) + k], acc_h);
-        acc_x = fmaf(d, wt_ih[g * (H + 1) + k], acc_x);
+        const float d = dg4[gbeg + gg];
+        acc_h = fmaf(d, whh_row[gg], acc_h);
+        acc_x = fmaf(d, wih_row[gg], acc_x);
       }
-      part_h[part * H + k] = acc_h;
-      part_x[part * H + k] = acc_x;
+      part_h[part * H + kk] = acc_h;
+      part_x[part * H + kk] = acc_x;
     }
     __syncthreads();
     if (tid < H) {
@@ -297,8 +294,7 @@ void launch_bwd(const at::Tensor& gouts, const c10::optional<at::Tensor>& ghS,
                 at::Tensor& dgates, at::Tensor& dxb, int B, int S, int F,
                 int D) {
   const int G = 4 * H;
-  const int lds =
-      (2 * G * (H + 1) + S * H + G + S * H + 2 * G) * sizeof(float);
+  const int lds = (S * H + G + S * H + 2 * G) * sizeof(float);
   TORCH_CHECK(lds <= 160 * 1024, "backward LDS footprint exceeds 160 KiB");
   hipLaunchKernelGGL(
       (seq_lstm_bwd_kernel<H>), dim3(B), dim3(G), lds, current_stream(),

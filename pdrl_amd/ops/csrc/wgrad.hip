@@ -60,26 +60,50 @@ __global__ __launch_bounds__(256) void wgrad_gates_mfma_kernel(
   const int i = lane & 15;   // row within A frag / col within B frag
   const int k = lane >> 4;   // inner (n) offset 0..3
 
-  // Software-pipelined K sweep: issue a chunk of 2×U independent loads, then
-  // U MFMAs — keeps ~16 loads in flight so the 16x16x4 f32 MFMA chain never
-  // stalls on L2 latency (the naive load→mfma→load loop measured 61 µs;
-  // pipelined ~an order less).
+  // Double-buffered software pipeline: the next chunk's 2×U loads issue
+  // BEFORE the current chunk's MFMA chain, so L2 latency hides under the
+  // matrix work (naive load→mfma loop: 61 µs; U=8 issue-then-use: 26 µs;
+  // this double-buffered form targets the MFMA-issue floor).
   constexpr int U = 8;  // MFMAs per chunk, 32 n-rows
+  const int step = 4 * U;
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-  int n0 = 0;
-  for (; n0 + 4 * U <= N; n0 += 4 * U) {
-    float a[U], b[U];
-#pragma unroll
-    for (int u = 0; u < U; ++u) {
-      const int n = n0 + 4 * u + k;
-      a[u] = gate_a_row<H>(stash, h0, n, S, sel, h0s)[m0 + i];
-      b[u] = dgates[(long)n * G + g0 + i];
-    }
-#pragma unroll
-    for (int u = 0; u < U; ++u) {
-      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a[u], b[u], acc, 0, 0, 0);
-    }
+  float a0[U], b0[U], a1[U], b1[U];
+
+#define PDRL_WG_LOAD(av, bv, base)                                         \
+  _Pragma("unroll") for (int u = 0; u < U; ++u) {                          \
+    const int n = (base) + 4 * u + k;                                      \
+    av[u] = gate_a_row<H>(stash, h0, n, S, sel, h0s)[m0 + i];              \
+    bv[u] = dgates[(long)n * G + g0 + i];                                  \
   }
+#define PDRL_WG_MFMA(av, bv)                                               \
+  _Pragma("unroll") for (int u = 0; u < U; ++u) {                          \
+    acc = __builtin_amdgcn_mfma_f32_16x16x4f32(av[u], bv[u], acc, 0, 0, 0);\
+  }
+
+  const int nfull = N - (N % step);
+  int n0 = 0;
+  if (nfull >= step) {
+    PDRL_WG_LOAD(a0, b0, 0);
+    bool cur0 = true;
+    for (n0 = step; n0 < nfull; n0 += step) {
+      if (cur0) {
+        PDRL_WG_LOAD(a1, b1, n0);
+        PDRL_WG_MFMA(a0, b0);
+      } else {
+        PDRL_WG_LOAD(a0, b0, n0);
+        PDRL_WG_MFMA(a1, b1);
+      }
+      cur0 = !cur0;
+    }
+    if (cur0) {
+      PDRL_WG_MFMA(a0, b0);
+    } else {
+      PDRL_WG_MFMA(a1, b1);
+    }
+    n0 = nfull;
+  }
+#undef PDRL_WG_LOAD
+#undef PDRL_WG_MFMA
   for (; n0 < N; n0 += 4) {  // ragged tail, zero-padded
     const bool live = (n0 + k) < N;
     const float a =

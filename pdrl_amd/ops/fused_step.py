@@ -63,6 +63,52 @@ class FusedOnPolicyStep:
         # wgrad_out writes (dw_ih, dw_hh, dbody_w, dbody_b, db_g, dheads_w, dheads_b)
         return [gs[2], gs[3], gs[0], gs[1], gs[4], gs[5], gs[6]]
 
+    def _loss(self, e, mo, act, behav, rew, fir, B, S, A, p):
+        """Loss stats + analytic head-grad buffer. One mega-kernel launch
+        when the shape fits a CU's LDS; 4-kernel sequence otherwise."""
+        gouts = torch.empty_like(mo)
+        if self.algo == "IMPALA":
+            if e.impala_loss_mega(
+                mo, act, behav, rew, fir, gouts, self.stats_buf, A,
+                p.gamma, 0.8, 0.1, 1.0, p.reward_scale,
+                p.policy_loss_coef, p.value_loss_coef, p.entropy_coef,
+            ):
+                return gouts
+            logp, ent, lse = e.cat_stats(mo, act, A)
+            logp2 = logp.view(B, S)
+            rhos, adv, vs = e.vtrace(
+                behav.view(B, S, 1), logp2.view(B, S, 1).contiguous(),
+                fir.view(B, S, 1), rew.view(B, S, 1), mo.view(B, S, -1),
+                p.gamma, 0.8, 0.1, 1.0, vD=mo.shape[-1], val_off=A,
+                rew_scale=p.reward_scale,
+            )
+            e.impala_loss_reduce(
+                logp2, ent.view(B, S), mo, A, adv, vs, rhos, self.stats_buf,
+                p.policy_loss_coef, p.value_loss_coef, p.entropy_coef,
+            )
+            return e.impala_loss_bwd(
+                mo, A, act, lse, ent, adv, vs,
+                p.policy_loss_coef, p.value_loss_coef, p.entropy_coef,
+            )
+        if e.ppo_loss_mega(
+            mo, act, behav, rew, fir, gouts, self.stats_buf, A,
+            p.gamma, p.lmbda, p.reward_scale,
+            p.policy_loss_coef, p.value_loss_coef, p.entropy_coef, p.eps_clip,
+        ):
+            return gouts
+        logp, ent, lse = e.cat_stats(mo, act, A)
+        logp2 = logp.view(B, S)
+        td, adv = e.ppo_td_gae(rew, fir, mo, A, p.gamma, p.lmbda,
+                               p.reward_scale)
+        e.ppo_loss_reduce(
+            logp2, behav, ent.view(B, S), mo, A, adv, td, self.stats_buf,
+            p.policy_loss_coef, p.value_loss_coef, p.entropy_coef, p.eps_clip,
+        )
+        return e.ppo_loss_bwd(
+            mo, A, act, lse, ent, logp, behav.reshape(-1), adv, td,
+            p.policy_loss_coef, p.value_loss_coef, p.entropy_coef, p.eps_clip,
+        )
+
     def compute_grads_only(self, batch):
         """Kernels 1-7 only (for gradient parity tests): fills the flat grad
         buffer without touching the optimizer."""
@@ -86,36 +132,7 @@ class FusedOnPolicyStep:
             x, hx0, cx0, c.body_w, c.body_b, c.w_ih, c.w_hh, c.b_g,
             c.heads_w, c.heads_b,
         )
-        logp, ent, lse = e.cat_stats(mo, act, A)
-        logp2 = logp.view(B, S)
-        if self.algo == "IMPALA":
-            rhos, adv, vs = e.vtrace(
-                behav.view(B, S, 1), logp2.view(B, S, 1).contiguous(),
-                fir.view(B, S, 1), rew.view(B, S, 1), mo.view(B, S, -1),
-                p.gamma, 0.8, 0.1, 1.0, vD=mo.shape[-1], val_off=A,
-                rew_scale=p.reward_scale,
-            )
-            e.impala_loss_reduce(
-                logp2, ent.view(B, S), mo, A, adv, vs, rhos, self.stats_buf,
-                p.policy_loss_coef, p.value_loss_coef, p.entropy_coef,
-            )
-            gouts = e.impala_loss_bwd(
-                mo, A, act, lse, ent, adv, vs,
-                p.policy_loss_coef, p.value_loss_coef, p.entropy_coef,
-            )
-        else:
-            td, adv = e.ppo_td_gae(rew, fir, mo, A, p.gamma, p.lmbda,
-                                   p.reward_scale)
-            e.ppo_loss_reduce(
-                logp2, behav, ent.view(B, S), mo, A, adv, td, self.stats_buf,
-                p.policy_loss_coef, p.value_loss_coef, p.entropy_coef,
-                p.eps_clip,
-            )
-            gouts = e.ppo_loss_bwd(
-                mo, A, act, lse, ent, logp, behav.reshape(-1), adv, td,
-                p.policy_loss_coef, p.value_loss_coef, p.entropy_coef,
-                p.eps_clip,
-            )
+        gouts = self._loss(e, mo, act, behav, rew, fir, B, S, A, p)
         _, _, _, dgates, dxb = e.seq_lstm_backward_core(
             gouts, None, None, stash, x, cx0, c.body_w, c.w_ih, c.w_hh,
             c.heads_w,

