@@ -220,7 +220,9 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_kernel(
       // register-resident weight rows (dg4 reads broadcast from LDS)
       const int gbeg = part * H;
       float acc_h = 0.0f, acc_x = 0.0f;
-#pragma unroll 4
+      // FULL unroll: static indices keep wih_row/whh_row in registers
+      // (partial unroll → runtime indices → scratch, rule 20)
+#pragma unroll
       for (int gg = 0; gg < H; ++gg) {
         const float d = dg4[gbeg + gg];
         acc_h = fmaf(d, whh_row[gg], acc_h);

@@ -88,9 +88,11 @@ __global__ __launch_bounds__(256) void wgrad_gates_mfma_kernel(
     for (n0 = step; n0 < nfull; n0 += step) {
       if (cur0) {
         PDRL_WG_LOAD(a1, b1, n0);
+        __builtin_amdgcn_sched_barrier(0);  // keep loads issued ahead
         PDRL_WG_MFMA(a0, b0);
       } else {
         PDRL_WG_LOAD(a0, b0, n0);
+        __builtin_amdgcn_sched_barrier(0);
         PDRL_WG_MFMA(a1, b1);
       }
       cur0 = !cur0;
