@@ -19,7 +19,8 @@ void seq_lstm_wgrad_out_hip(const at::Tensor&, const at::Tensor&,
                             const at::Tensor&, const at::Tensor&,
                             const at::Tensor&, const at::Tensor&, at::Tensor&,
                             at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
-                            at::Tensor&, at::Tensor&);
+                            at::Tensor&, at::Tensor&,
+                            const c10::optional<at::Tensor>&);
 at::Tensor gae_hip(const at::Tensor&, double, double, const at::Tensor&);
 std::vector<at::Tensor> vtrace_hip(const at::Tensor&, const at::Tensor&,
                                    const at::Tensor&, const at::Tensor&,
@@ -49,13 +50,15 @@ at::Tensor ppo_loss_bwd_hip(const at::Tensor&, long, const at::Tensor&,
                             double, double, double);
 bool impala_loss_mega_hip(const at::Tensor&, const at::Tensor&,
                           const at::Tensor&, const at::Tensor&,
-                          const at::Tensor&, at::Tensor&, at::Tensor&, long,
+                          const at::Tensor&, at::Tensor&, at::Tensor&,
+                          const c10::optional<at::Tensor>&, long, double,
                           double, double, double, double, double, double,
-                          double, double);
+                          double);
 bool ppo_loss_mega_hip(const at::Tensor&, const at::Tensor&, const at::Tensor&,
                        const at::Tensor&, const at::Tensor&, at::Tensor&,
-                       at::Tensor&, long, double, double, double, double,
-                       double, double, double);
+                       at::Tensor&, const c10::optional<at::Tensor>&, long,
+                       double, double, double, double, double, double,
+                       double);
 void l2norm_sq_hip(const at::Tensor&, at::Tensor&);
 void rmsprop_step_hip(at::Tensor&, const at::Tensor&, at::Tensor&,
                       const at::Tensor&, double, double, double, double);

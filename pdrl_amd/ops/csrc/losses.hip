@@ -391,6 +391,7 @@ __global__ __launch_bounds__(kMegaThreads) void impala_loss_mega_kernel(
     const float* __restrict__ fir,    // (B,S)
     float* __restrict__ gouts,        // (B,S,D)
     float* __restrict__ stats,        // (5)
+    float* __restrict__ norm_sq,      // optional: zeroed here for the wgrads
     int B, int S, int A, float gamma, float rho_bar, float rho_min,
     float c_bar, float rew_scale, float cp, float cv, float ce) {
   const int D = A + 1;
@@ -405,6 +406,7 @@ __global__ __launch_bounds__(kMegaThreads) void impala_loss_mega_kernel(
   float* s_adv = s_rho + BT;
   float* s_vs = s_adv + BT;
   const int tid = threadIdx.x;
+  if (tid == 0 && norm_sq != nullptr) *norm_sq = 0.f;
 
   // phase A: categorical stats
   for (int i = tid; i < N; i += kMegaThreads) {
@@ -508,6 +510,7 @@ __global__ __launch_bounds__(kMegaThreads) void ppo_loss_mega_kernel(
     const float* __restrict__ fir,    // (B,S)
     float* __restrict__ gouts,        // (B,S,D)
     float* __restrict__ stats,        // (7)
+    float* __restrict__ norm_sq,      // optional: zeroed here for the wgrads
     int B, int S, int A, float gamma, float lmbda, float rew_scale, float cp,
     float cv, float ce, float eps_clip) {
   const int D = A + 1;
@@ -521,6 +524,7 @@ __global__ __launch_bounds__(kMegaThreads) void ppo_loss_mega_kernel(
   float* s_td = s_lse + N;
   float* s_adv = s_td + BT;
   const int tid = threadIdx.x;
+  if (tid == 0 && norm_sq != nullptr) *norm_sq = 0.f;
 
   for (int i = tid; i < N; i += kMegaThreads) {
     const float* z = mo + (long)i * D;
@@ -629,9 +633,11 @@ __global__ __launch_bounds__(kMegaThreads) void ppo_loss_mega_kernel(
 bool impala_loss_mega_hip(const at::Tensor& mo, const at::Tensor& act,
                           const at::Tensor& behav, const at::Tensor& rew,
                           const at::Tensor& fir, at::Tensor& gouts,
-                          at::Tensor& stats, long A, double gamma,
-                          double rho_bar, double rho_min, double c_bar,
-                          double rew_scale, double cp, double cv, double ce) {
+                          at::Tensor& stats,
+                          const c10::optional<at::Tensor>& norm_sq, long A,
+                          double gamma, double rho_bar, double rho_min,
+                          double c_bar, double rew_scale, double cp, double cv,
+                          double ce) {
   const int B = mo.size(0), S = mo.size(1);
   const int N = B * S, BT = B * (S - 1);
   const long lds = (3L * N + 3L * BT) * sizeof(float);
@@ -640,7 +646,9 @@ bool impala_loss_mega_hip(const at::Tensor& mo, const at::Tensor& act,
                      current_stream(), mo.data_ptr<float>(),
                      act.data_ptr<float>(), behav.data_ptr<float>(),
                      rew.data_ptr<float>(), fir.data_ptr<float>(),
-                     gouts.data_ptr<float>(), stats.data_ptr<float>(), B, S,
+                     gouts.data_ptr<float>(), stats.data_ptr<float>(),
+                     norm_sq.has_value() ? norm_sq->data_ptr<float>() : nullptr,
+                     B, S,
                      (int)A, (float)gamma, (float)rho_bar, (float)rho_min,
                      (float)c_bar, (float)rew_scale, (float)cp, (float)cv,
                      (float)ce);
@@ -651,9 +659,10 @@ bool impala_loss_mega_hip(const at::Tensor& mo, const at::Tensor& act,
 bool ppo_loss_mega_hip(const at::Tensor& mo, const at::Tensor& act,
                        const at::Tensor& behav, const at::Tensor& rew,
                        const at::Tensor& fir, at::Tensor& gouts,
-                       at::Tensor& stats, long A, double gamma, double lmbda,
-                       double rew_scale, double cp, double cv, double ce,
-                       double eps_clip) {
+                       at::Tensor& stats,
+                       const c10::optional<at::Tensor>& norm_sq, long A,
+                       double gamma, double lmbda, double rew_scale, double cp,
+                       double cv, double ce, double eps_clip) {
   const int B = mo.size(0), S = mo.size(1);
   const int N = B * S, BT = B * (S - 1);
   const long lds = (3L * N + 2L * BT) * sizeof(float);
@@ -662,7 +671,9 @@ bool ppo_loss_mega_hip(const at::Tensor& mo, const at::Tensor& act,
                      current_stream(), mo.data_ptr<float>(),
                      act.data_ptr<float>(), behav.data_ptr<float>(),
                      rew.data_ptr<float>(), fir.data_ptr<float>(),
-                     gouts.data_ptr<float>(), stats.data_ptr<float>(), B, S,
+                     gouts.data_ptr<float>(), stats.data_ptr<float>(),
+                     norm_sq.has_value() ? norm_sq->data_ptr<float>() : nullptr,
+                     B, S,
                      (int)A, (float)gamma, (float)lmbda, (float)rew_scale,
                      (float)cp, (float)cv, (float)ce, (float)eps_clip);
   HIP_CHECK_LAST();

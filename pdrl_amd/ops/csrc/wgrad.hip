@@ -47,6 +47,7 @@ __global__ __launch_bounds__(256) void wgrad_gates_mfma_kernel(
     const float* __restrict__ dgates,  // (N,4H)
     float* __restrict__ dw_ih,         // (H,4H)
     float* __restrict__ dw_hh,         // (H,4H)
+    float* __restrict__ norm_sq,       // optional ||grad||² accumulator
     int N, int S, long h0s) {
   constexpr int G = 4 * H;
   const int wave = threadIdx.x / kWave;
@@ -60,10 +61,19 @@ __global__ __launch_bounds__(256) void wgrad_gates_mfma_kernel(
   const int i = lane & 15;   // row within A frag / col within B frag
   const int k = lane >> 4;   // inner (n) offset 0..3
 
+  // A-row pointer table, built ONCE per block: the per-row n%S / n/S integer
+  // divisions in the load loop (640 of them per lane) serialized the K sweep
+  // — with the table each load is ptr[n] (LDS broadcast) + one global load.
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  const float** tab = reinterpret_cast<const float**>(smem_raw);
+  for (int n = threadIdx.x; n < N; n += 256) {
+    tab[n] = gate_a_row<H>(stash, h0, n, S, sel, h0s);
+  }
+  __syncthreads();
+
   // Double-buffered software pipeline: the next chunk's 2×U loads issue
   // BEFORE the current chunk's MFMA chain, so L2 latency hides under the
-  // matrix work (naive load→mfma loop: 61 µs; U=8 issue-then-use: 26 µs;
-  // this double-buffered form targets the MFMA-issue floor).
+  // matrix work (naive load→mfma loop: 61 µs on this shape).
   constexpr int U = 8;  // MFMAs per chunk, 32 n-rows
   const int step = 4 * U;
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
@@ -72,7 +82,7 @@ __global__ __launch_bounds__(256) void wgrad_gates_mfma_kernel(
 #define PDRL_WG_LOAD(av, bv, base)                                         \
   _Pragma("unroll") for (int u = 0; u < U; ++u) {                          \
     const int n = (base) + 4 * u + k;                                      \
-    av[u] = gate_a_row<H>(stash, h0, n, S, sel, h0s)[m0 + i];              \
+    av[u] = tab[n][m0 + i];                                                \
     bv[u] = dgates[(long)n * G + g0 + i];                                  \
   }
 #define PDRL_WG_MFMA(av, bv)                                               \
@@ -108,8 +118,7 @@ __global__ __launch_bounds__(256) void wgrad_gates_mfma_kernel(
 #undef PDRL_WG_MFMA
   for (; n0 < N; n0 += 4) {  // ragged tail, zero-padded
     const bool live = (n0 + k) < N;
-    const float a =
-        live ? gate_a_row<H>(stash, h0, n0 + k, S, sel, h0s)[m0 + i] : 0.f;
+    const float a = live ? tab[n0 + k][m0 + i] : 0.f;
     const float b = live ? dgates[(long)(n0 + k) * G + g0 + i] : 0.f;
     acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
   }
@@ -117,9 +126,21 @@ __global__ __launch_bounds__(256) void wgrad_gates_mfma_kernel(
   // C map: col = lane&15, row = (lane>>4)*4 + reg
   const int c_col = g0 + (lane & 15);
   const int c_row0 = m0 + (lane >> 4) * 4;
+  float nrm = 0.f;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     out[(long)(c_row0 + r) * G + c_col] = acc[r];
+    nrm = fmaf(acc[r], acc[r], nrm);
+  }
+  if (norm_sq != nullptr) {  // block-reduce ||grad||² → one atomic per block
+    __shared__ float red[4];
+#pragma unroll
+    for (int off = kWave / 2; off > 0; off >>= 1)
+      nrm += __shfl_down(nrm, off, kWave);
+    if (lane == 0) red[wave] = nrm;
+    __syncthreads();
+    if (threadIdx.x == 0)
+      atomicAdd(norm_sq, red[0] + red[1] + red[2] + red[3]);
   }
 }
 
@@ -135,21 +156,24 @@ __global__ void wgrad_small_kernel(
     const float* __restrict__ gouts,   // (N,D)
     float* __restrict__ dbody_w, float* __restrict__ dbody_b,
     float* __restrict__ db_g, float* __restrict__ dheads_w,
-    float* __restrict__ dheads_b, int N, int F, int D) {
+    float* __restrict__ dheads_b, float* __restrict__ norm_sq,
+    int N, int F, int D) {
   constexpr int G = 4 * H;
   const int wave_id = (blockIdx.x * blockDim.x + threadIdx.x) / kWave;
   const int lane = threadIdx.x % kWave;
   const int n_fw = F * H, n_hw = H * D;
   const int total = n_fw + H + G + n_hw + D;
-  if (wave_id >= total) return;
+  const bool live_wave = wave_id < total;
 
   // resolve segment
   const float *pa = nullptr, *pb = nullptr;
   long stride_a = 0, stride_b = 0;
   float* out = nullptr;
   int oi = 0;
-  int e = wave_id;
-  if (e < n_fw) {  // dbody_w[f][j] = sum x[n][f]*dxb[n][j]
+  int e = live_wave ? wave_id : 0;
+  if (!live_wave) {
+    pb = dgates; stride_b = G; out = nullptr;
+  } else if (e < n_fw) {  // dbody_w[f][j] = sum x[n][f]*dxb[n][j]
     const int f = e / H, j = e % H;
     pa = x + f; stride_a = F;
     pb = dxb + j; stride_b = H;
@@ -172,13 +196,23 @@ __global__ void wgrad_small_kernel(
   }
 
   float acc = 0.f;
-  for (int n = lane; n < N; n += kWave) {
-    const float bv = pb[(long)n * stride_b];
-    acc = (pa != nullptr) ? fmaf(pa[(long)n * stride_a], bv, acc) : acc + bv;
-  }
+  if (live_wave) {
+    for (int n = lane; n < N; n += kWave) {
+      const float bv = pb[(long)n * stride_b];
+      acc = (pa != nullptr) ? fmaf(pa[(long)n * stride_a], bv, acc) : acc + bv;
+    }
 #pragma unroll
-  for (int off = kWave / 2; off > 0; off >>= 1) acc += __shfl_down(acc, off, kWave);
-  if (lane == 0) out[oi] = acc;
+    for (int off = kWave / 2; off > 0; off >>= 1)
+      acc += __shfl_down(acc, off, kWave);
+    if (lane == 0) out[oi] = acc;
+  }
+  if (norm_sq != nullptr) {  // one atomic per block
+    __shared__ float red[4];
+    if (lane == 0) red[threadIdx.x / kWave] = live_wave ? acc * acc : 0.f;
+    __syncthreads();
+    if (threadIdx.x == 0)
+      atomicAdd(norm_sq, red[0] + red[1] + red[2] + red[3]);
+  }
 }
 
 template <int H>
@@ -187,15 +221,19 @@ void launch_wgrad(const at::Tensor& x, const at::Tensor& h0,
                   const at::Tensor& dxb, const at::Tensor& gouts,
                   at::Tensor& dw_ih, at::Tensor& dw_hh, at::Tensor& dbody_w,
                   at::Tensor& dbody_b, at::Tensor& db_g, at::Tensor& dheads_w,
-                  at::Tensor& dheads_b, int N, int S, int F, int D) {
+                  at::Tensor& dheads_b, const c10::optional<at::Tensor>& norm_sq,
+                  int N, int S, int F, int D) {
   constexpr int G = 4 * H;
   // gate GEMMs: (H/16) m-tiles × (G/64) g-blocks blocks, 2 GEMMs on y
   dim3 grid((H / 16) * (G / kWave), 2);
-  hipLaunchKernelGGL((wgrad_gates_mfma_kernel<H>), grid, dim3(256), 0,
+  const int tab_lds = N * sizeof(const float*);
+  TORCH_CHECK(tab_lds <= 64 * 1024, "wgrad row table exceeds LDS (B*S too big)");
+  float* nrm = norm_sq.has_value() ? norm_sq->data_ptr<float>() : nullptr;
+  hipLaunchKernelGGL((wgrad_gates_mfma_kernel<H>), grid, dim3(256), tab_lds,
                      current_stream(), stash.data_ptr<float>(),
                      h0.data_ptr<float>(), dgates.data_ptr<float>(),
-                     dw_ih.data_ptr<float>(), dw_hh.data_ptr<float>(), N, S,
-                     (long)h0.stride(0));
+                     dw_ih.data_ptr<float>(), dw_hh.data_ptr<float>(), nrm,
+                     N, S, (long)h0.stride(0));
   HIP_CHECK_LAST();
 
   const int total_waves = F * H + H + G + H * D + D;
@@ -207,7 +245,7 @@ void launch_wgrad(const at::Tensor& x, const at::Tensor& h0,
                      dgates.data_ptr<float>(), gouts.data_ptr<float>(),
                      dbody_w.data_ptr<float>(), dbody_b.data_ptr<float>(),
                      db_g.data_ptr<float>(), dheads_w.data_ptr<float>(),
-                     dheads_b.data_ptr<float>(), N, F, D);
+                     dheads_b.data_ptr<float>(), nrm, N, F, D);
   HIP_CHECK_LAST();
 }
 
@@ -219,7 +257,8 @@ void seq_lstm_wgrad_out_hip(const at::Tensor& x, const at::Tensor& h0,
                             at::Tensor& dw_ih, at::Tensor& dw_hh,
                             at::Tensor& dbody_w, at::Tensor& dbody_b,
                             at::Tensor& db_g, at::Tensor& dheads_w,
-                            at::Tensor& dheads_b) {
+                            at::Tensor& dheads_b,
+                            const c10::optional<at::Tensor>& norm_sq) {
   CHECK_IN(x); CHECK_IN(stash); CHECK_IN(dgates);
   CHECK_IN(dxb); CHECK_IN(gouts);
   CHECK_GPU(h0); CHECK_F32(h0);
@@ -228,9 +267,9 @@ void seq_lstm_wgrad_out_hip(const at::Tensor& x, const at::Tensor& h0,
   const int H = h0.size(1), D = gouts.size(2);
   const int N = B * S;
   switch (H) {
-    case 32: launch_wgrad<32>(x, h0, stash, dgates, dxb, gouts, dw_ih, dw_hh, dbody_w, dbody_b, db_g, dheads_w, dheads_b, N, S, F, D); break;
-    case 64: launch_wgrad<64>(x, h0, stash, dgates, dxb, gouts, dw_ih, dw_hh, dbody_w, dbody_b, db_g, dheads_w, dheads_b, N, S, F, D); break;
-    case 128: launch_wgrad<128>(x, h0, stash, dgates, dxb, gouts, dw_ih, dw_hh, dbody_w, dbody_b, db_g, dheads_w, dheads_b, N, S, F, D); break;
+    case 32: launch_wgrad<32>(x, h0, stash, dgates, dxb, gouts, dw_ih, dw_hh, dbody_w, dbody_b, db_g, dheads_w, dheads_b, norm_sq, N, S, F, D); break;
+    case 64: launch_wgrad<64>(x, h0, stash, dgates, dxb, gouts, dw_ih, dw_hh, dbody_w, dbody_b, db_g, dheads_w, dheads_b, norm_sq, N, S, F, D); break;
+    case 128: launch_wgrad<128>(x, h0, stash, dgates, dxb, gouts, dw_ih, dw_hh, dbody_w, dbody_b, db_g, dheads_w, dheads_b, norm_sq, N, S, F, D); break;
     default: TORCH_CHECK(false, "hidden size ", H, " unsupported");
   }
 }
@@ -248,6 +287,7 @@ std::vector<at::Tensor> seq_lstm_wgrad_hip(
   auto dheads_w = at::empty({H, D}, opt);
   auto dheads_b = at::empty({D}, opt);
   seq_lstm_wgrad_out_hip(x, h0, stash, dgates, dxb, gouts, dw_ih, dw_hh,
-                         dbody_w, dbody_b, db_g, dheads_w, dheads_b);
+                         dbody_w, dbody_b, db_g, dheads_w, dheads_b,
+                         c10::nullopt);
   return {dw_ih, dw_hh, dbody_w, dbody_b, db_g, dheads_w, dheads_b};
 }

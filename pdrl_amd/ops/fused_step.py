@@ -55,6 +55,13 @@ class FusedOnPolicyStep:
         self._graph_failed = False
 
     # ------------------------------------------------------------------ #
+    def _norm_buf(self):
+        """The optimizer's device-resident ||grad||² scalar (filled by the
+        loss-mega zero + wgrad accumulation in single-rank mode)."""
+        if self.grad_reducer is not None:
+            return None  # post-allreduce norm is computed by optimizer.step()
+        return self.optimizer.norm_sq
+
     def _grad_views(self):
         c = self.core
         ps = [c.body_w, c.body_b, c.w_ih, c.w_hh, c.b_g, c.heads_w, c.heads_b]
@@ -67,13 +74,16 @@ class FusedOnPolicyStep:
         """Loss stats + analytic head-grad buffer. One mega-kernel launch
         when the shape fits a CU's LDS; 4-kernel sequence otherwise."""
         gouts = torch.empty_like(mo)
+        norm = self._norm_buf()
         if self.algo == "IMPALA":
             if e.impala_loss_mega(
-                mo, act, behav, rew, fir, gouts, self.stats_buf, A,
+                mo, act, behav, rew, fir, gouts, self.stats_buf, norm, A,
                 p.gamma, 0.8, 0.1, 1.0, p.reward_scale,
                 p.policy_loss_coef, p.value_loss_coef, p.entropy_coef,
             ):
                 return gouts
+            if norm is not None:
+                norm.zero_()  # fallback path: mega kernel didn't zero it
             logp, ent, lse = e.cat_stats(mo, act, A)
             logp2 = logp.view(B, S)
             rhos, adv, vs = e.vtrace(
@@ -91,11 +101,13 @@ class FusedOnPolicyStep:
                 p.policy_loss_coef, p.value_loss_coef, p.entropy_coef,
             )
         if e.ppo_loss_mega(
-            mo, act, behav, rew, fir, gouts, self.stats_buf, A,
+            mo, act, behav, rew, fir, gouts, self.stats_buf, norm, A,
             p.gamma, p.lmbda, p.reward_scale,
             p.policy_loss_coef, p.value_loss_coef, p.entropy_coef, p.eps_clip,
         ):
             return gouts
+        if norm is not None:
+            norm.zero_()  # fallback path: mega kernel didn't zero it
         logp, ent, lse = e.cat_stats(mo, act, A)
         logp2 = logp.view(B, S)
         td, adv = e.ppo_td_gae(rew, fir, mo, A, p.gamma, p.lmbda,
@@ -138,12 +150,17 @@ class FusedOnPolicyStep:
             c.heads_w,
         )
         e.seq_lstm_wgrad_out(x, hx0, stash, dgates, dxb, gouts,
-                             *self._grad_views())
+                             *self._grad_views(), self._norm_buf())
         if not update:
             return
         if self.grad_reducer is not None:
+            # multi-rank: norm must be of the AVERAGED grads → recompute
             self.grad_reducer.all_reduce([self.optimizer.flat_grad])
-        self.optimizer.step()
+            self.optimizer.step()
+        else:
+            # single rank: ||grad||² was accumulated by the wgrad kernels
+            # (zeroed in the mega loss kernel) — skip fill + l2norm launches
+            self.optimizer._update()
 
     def _full(self, batch):
         for _ in range(self.params.K_epoch):
