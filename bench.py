@@ -11,6 +11,10 @@ Contract (driver):
   python bench.py --gpus N --steps K --warmup W
   (N>1 is launched via torch.distributed.run, one rank per GPU over RCCL;
   rank 0 prints ONE JSON line.)
+
+Secondary configs (BASELINE.json configs[2..5]) via --algo:
+  --algo PPO | V-MPO | SAC | PPO-Continuous | SAC-Continuous
+(continuous algos switch to the MountainCarContinuous shape: obs 2, 1 act).
 """
 from __future__ import annotations
 
@@ -25,12 +29,18 @@ import torch
 
 sys.path.insert(0, str(Path(__file__).resolve().parent))
 
-from pdrl_amd.agents.learner_module import ImpalaUpdater  # noqa: E402
-from pdrl_amd.networks import MlpLSTMSingle  # noqa: E402
+from pdrl_amd.agents.learner_module import switch_module  # noqa: E402
 from pdrl_amd.parallel import GradReducer, init_distributed  # noqa: E402
 from pdrl_amd.utils import load_params  # noqa: E402
 
 OBS_DIM, N_ACTIONS = 4, 2  # CartPole-v1
+
+
+def env_shape(algo: str):
+    """(obs_dim, n_actions, continuous) of the BASELINE env for the algo."""
+    if algo.endswith("Continuous"):
+        return 2, 1, True  # MountainCarContinuous-v0
+    return OBS_DIM, N_ACTIONS, False  # CartPole-v1
 
 
 def make_synthetic_batch(params, device, seed):
@@ -38,11 +48,17 @@ def make_synthetic_batch(params, device, seed):
     (synthetic data — no network access for real envs at benchmark scale)."""
     g = torch.Generator(device="cpu").manual_seed(seed)
     B, S, H = params.batch_size, params.seq_len, params.hidden_size
-    logits = torch.randn(B, S, N_ACTIONS, generator=g)
-    acts = torch.randint(0, N_ACTIONS, (B, S, 1), generator=g).float()
-    logp = torch.log_softmax(logits, dim=-1).gather(-1, acts.long())
+    obs_dim, n_act, continuous = env_shape(params.algo)
+    if continuous:
+        acts = torch.tanh(torch.randn(B, S, n_act, generator=g))
+        logits = torch.randn(B, S, 2 * n_act, generator=g)
+        logp = -torch.rand(B, S, 1, generator=g)
+    else:
+        logits = torch.randn(B, S, n_act, generator=g)
+        acts = torch.randint(0, n_act, (B, S, 1), generator=g).float()
+        logp = torch.log_softmax(logits, dim=-1).gather(-1, acts.long())
     batch = {
-        "obs": torch.randn(B, S, OBS_DIM, generator=g),
+        "obs": torch.randn(B, S, obs_dim, generator=g),
         "act": acts,
         "rew": torch.rand(B, S, 1, generator=g),
         "logits": logits,
@@ -61,6 +77,9 @@ def main():
     ap.add_argument("--warmup", type=int, default=50)
     ap.add_argument("--batch-size", type=int, default=128)
     ap.add_argument("--seq-len", type=int, default=5)
+    ap.add_argument("--algo", default="IMPALA",
+                    choices=["IMPALA", "PPO", "V-MPO", "SAC",
+                             "PPO-Continuous", "SAC-Continuous"])
     args = ap.parse_args()
 
     rank, world = init_distributed()
@@ -72,15 +91,17 @@ def main():
         torch.cuda.set_device(device)
 
     params = load_params()
-    params.algo = "IMPALA"
+    params.algo = args.algo
     params.batch_size = args.batch_size
     params.seq_len = args.seq_len
-    params.obs_dim, params.n_actions = OBS_DIM, N_ACTIONS
+    obs_dim, n_act, continuous = env_shape(args.algo)
+    params.obs_dim, params.n_actions = obs_dim, n_act
 
     torch.manual_seed(1234)  # identical init across ranks
-    model = MlpLSTMSingle(OBS_DIM, N_ACTIONS, params.seq_len, params.hidden_size)
+    updater_cls, model_cls = switch_module(args.algo)
+    model = model_cls(obs_dim, n_act, params.seq_len, params.hidden_size)
     reducer = GradReducer() if world > 1 else None
-    updater = ImpalaUpdater(model, params, device, grad_reducer=reducer)
+    updater = updater_cls(model, params, device, grad_reducer=reducer)
 
     batch = make_synthetic_batch(params, device, seed=100 + rank)
 
@@ -132,8 +153,9 @@ def main():
             "dtype": "fp32",
             "data": "synthetic",
             "config": {
-                "model": "MlpLSTMSingle(obs4,act2,H64)",
-                "algo": "IMPALA CartPole-v1",
+                "model": f"{type(model).__name__}(obs{obs_dim},act{n_act},H{params.hidden_size})",
+                "algo": f"{args.algo} "
+                        + ("MountainCarContinuous-v0" if continuous else "CartPole-v1"),
                 "global_batch": args.batch_size * n_gpus,
                 "seq_len": args.seq_len,
                 "parallelism": f"dp{n_gpus}",

@@ -32,31 +32,58 @@ from pdrl_amd.utils import ExecutionTimer, Protocol, SummaryWriter, encode
 
 
 class BatchStager:
-    """Host→device batch staging with pinned memory + a side copy stream
-    (the K15 staging design in SURVEY.md §2.4; reference path was
-    shm → np copy → to_torch → blocking .to(device), learner.py:197-233)."""
+    """Host→device batch staging: ONE packed pinned buffer → ONE async H2D
+    copy → STABLE per-field device views (the K15 staging design in
+    SURVEY.md §2.4; reference path was shm → per-field np copy → to_torch →
+    blocking .to(device), learner.py:197-233).
+
+    Because the device views are the same tensors every call, the fused-step
+    hipGraph captures directly on them and replays with zero extra copies.
+    """
 
     def __init__(self, device):
         self.device = torch.device(device)
         self.use_cuda = self.device.type == "cuda"
-        self._pinned: dict[str, torch.Tensor] = {}
         self.copy_stream = torch.cuda.Stream(self.device) if self.use_cuda else None
+        self._layout: list[tuple[str, tuple, int, int]] | None = None  # (k, shape, off, n)
+        self._pinned: torch.Tensor | None = None
+        self._dev: torch.Tensor | None = None
+        self._views: dict[str, torch.Tensor] = {}
+        self._host_views: dict[str, np.ndarray] = {}
+
+    def _build(self, batch_np: dict[str, np.ndarray]):
+        layout = []
+        off = 0
+        for k in sorted(batch_np):
+            v = batch_np[k]
+            n = int(v.size)
+            layout.append((k, tuple(v.shape), off, n))
+            off += n
+        self._layout = layout
+        self._pinned = torch.empty(off, dtype=torch.float32, pin_memory=True)
+        self._dev = torch.empty(off, dtype=torch.float32, device=self.device)
+        flat_host = self._pinned.numpy()
+        for k, shape, o, n in layout:
+            self._views[k] = self._dev[o : o + n].view(shape)
+            self._host_views[k] = flat_host[o : o + n].reshape(shape)
 
     def stage(self, batch_np: dict[str, np.ndarray]) -> dict[str, torch.Tensor]:
         if not self.use_cuda:
             return {k: torch.from_numpy(v) for k, v in batch_np.items()}
-        out = {}
+        if self._layout is None:
+            self._build(batch_np)
+            self._ev = torch.cuda.Event()
+            self._ev.record()
+        self._ev.synchronize()  # previous H2D must be done reading pinned mem
+        for k, shape, o, n in self._layout:
+            self._host_views[k][...] = batch_np[k]
+        # the copy must not overtake the previous step still reading _dev
+        self.copy_stream.wait_stream(torch.cuda.current_stream(self.device))
         with torch.cuda.stream(self.copy_stream):
-            for k, v in batch_np.items():
-                t = torch.from_numpy(v)
-                pin = self._pinned.get(k)
-                if pin is None or pin.shape != t.shape:
-                    pin = torch.empty_like(t, pin_memory=True)
-                    self._pinned[k] = pin
-                pin.copy_(t)
-                out[k] = pin.to(self.device, non_blocking=True)
+            self._dev.copy_(self._pinned, non_blocking=True)
+            self._ev.record(self.copy_stream)
         torch.cuda.current_stream(self.device).wait_stream(self.copy_stream)
-        return out
+        return dict(self._views)
 
 
 class Learner:

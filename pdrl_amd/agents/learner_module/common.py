@@ -70,7 +70,13 @@ class BaseUpdater:
             return None  # continuous-policy fused step not implemented yet
         from pdrl_amd.ops.fused_step import FusedOnPolicyStep
 
-        use_graph = bool(int((__import__("os").environ.get("PDRL_USE_GRAPH", "1"))))
+        # graph replay is single-rank only for now: capturing the RCCL
+        # all-reduce is unverified on this stack (stream-ordered fused path
+        # runs for multi-rank; correctness identical)
+        use_graph = (
+            bool(int(__import__("os").environ.get("PDRL_USE_GRAPH", "1")))
+            and self.grad_reducer is None
+        )
         return FusedOnPolicyStep(algo, core, self.params, optimizer,
                                  grad_reducer=self.grad_reducer,
                                  use_graph=use_graph)
