@@ -89,12 +89,27 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_fwd_kernel(
 
   // LSTM recurrence, whole sequence in-kernel (K2).
   for (int t = 0; t < S; ++t) {
-    const float* xbt = xb + t * H;
+    // vectorized LDS broadcast reads: ds_read_b128 moves 4 floats per
+    // 4 cycles vs 4× ds_read_b32 at 2 cycles each (§LDS table)
+    const float4* xbt4 = reinterpret_cast<const float4*>(xb + t * H);
+    const float4* h4 = reinterpret_cast<const float4*>(hbuf);
     float acc = bias;
 #pragma unroll
-    for (int k = 0; k < H; ++k) acc = fmaf(xbt[k], wih[k], acc);
+    for (int k = 0; k < H / 4; ++k) {
+      const float4 xv = xbt4[k];
+      acc = fmaf(xv.x, wih[4 * k], acc);
+      acc = fmaf(xv.y, wih[4 * k + 1], acc);
+      acc = fmaf(xv.z, wih[4 * k + 2], acc);
+      acc = fmaf(xv.w, wih[4 * k + 3], acc);
+    }
 #pragma unroll
-    for (int k = 0; k < H; ++k) acc = fmaf(hbuf[k], whh[k], acc);
+    for (int k = 0; k < H / 4; ++k) {
+      const float4 hv = h4[k];
+      acc = fmaf(hv.x, whh[4 * k], acc);
+      acc = fmaf(hv.y, whh[4 * k + 1], acc);
+      acc = fmaf(hv.z, whh[4 * k + 2], acc);
+      acc = fmaf(hv.w, whh[4 * k + 3], acc);
+    }
     const int sel = tid / H;  // 0:i 1:f 2:g 3:o
     const float a = (sel == 2) ? tanhf(acc) : sigmoidf_dev(acc);
     const long sbase = (((long)b * S + t) * kStashFields) * H;
@@ -119,21 +134,27 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_fwd_kernel(
     cS[(long)b * H + tid] = cbuf[tid];
   }
 
-  // Heads (K3) on the stored h sequence.
+  // Heads (K3) on the stored h sequence (LDS reads vectorized).
   for (int idx = tid; idx < S * D; idx += G) {
     const int t = idx / D, d = idx % D;
     float acc = heads_b[d];
-    const float* ht = hs + t * H;
+    const float4* ht4 = reinterpret_cast<const float4*>(hs + t * H);
 #pragma unroll
-    for (int k = 0; k < H; ++k) acc = fmaf(ht[k], heads_w[k * D + d], acc);
+    for (int k = 0; k < H / 4; ++k) {
+      const float4 hv = ht4[k];
+      acc = fmaf(hv.x, heads_w[(4 * k) * D + d], acc);
+      acc = fmaf(hv.y, heads_w[(4 * k + 1) * D + d], acc);
+      acc = fmaf(hv.z, heads_w[(4 * k + 2) * D + d], acc);
+      acc = fmaf(hv.w, heads_w[(4 * k + 3) * D + d], acc);
+    }
     outs[((long)b * S + t) * D + d] = acc;
   }
 }
 
 // Backward through heads + recurrence + body for one batch row.
 // Emits per-(b,t) pre-activation gate grads (dgates) and pre-ReLU body grads
-// (dxb) for the weight-gradient GEMMs (done as plain library GEMMs on the
-// host side — hipBLASLt — per the MI355X mandate), plus dx / dh0 / dc0.
+// (dxb) consumed by the MFMA weight-gradient kernels (wgrad.hip), plus
+// dx / dh0 / dc0.
 template <int H>
 __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_kernel(
     const float* __restrict__ gouts,   // (B,S,D) head-output grads
@@ -218,15 +239,21 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_kernel(
     {
       // recurrent + body back-projection, split over all 4H threads on
       // register-resident weight rows (dg4 reads broadcast from LDS)
-      const int gbeg = part * H;
       float acc_h = 0.0f, acc_x = 0.0f;
+      const float4* dg44 = reinterpret_cast<const float4*>(dg4 + part * H);
       // FULL unroll: static indices keep wih_row/whh_row in registers
-      // (partial unroll → runtime indices → scratch, rule 20)
+      // (partial unroll → runtime indices → scratch, rule 20); b128 reads
 #pragma unroll
-      for (int gg = 0; gg < H; ++gg) {
-        const float d = dg4[gbeg + gg];
-        acc_h = fmaf(d, whh_row[gg], acc_h);
-        acc_x = fmaf(d, wih_row[gg], acc_x);
+      for (int gg = 0; gg < H / 4; ++gg) {
+        const float4 dv = dg44[gg];
+        acc_h = fmaf(dv.x, whh_row[4 * gg], acc_h);
+        acc_x = fmaf(dv.x, wih_row[4 * gg], acc_x);
+        acc_h = fmaf(dv.y, whh_row[4 * gg + 1], acc_h);
+        acc_x = fmaf(dv.y, wih_row[4 * gg + 1], acc_x);
+        acc_h = fmaf(dv.z, whh_row[4 * gg + 2], acc_h);
+        acc_x = fmaf(dv.z, wih_row[4 * gg + 2], acc_x);
+        acc_h = fmaf(dv.w, whh_row[4 * gg + 3], acc_h);
+        acc_x = fmaf(dv.w, wih_row[4 * gg + 3], acc_x);
       }
       part_h[part * H + kk] = acc_h;
       part_x[part * H + kk] = acc_x;
