@@ -24,6 +24,16 @@ __device__ __forceinline__ float sigmoidf_dev(float x) {
 // wave-width on CDNA4 is 64 lanes; hard-coded per the platform guide.
 constexpr int kWave = 64;
 
+// Force an array to stay materialized in VGPRs: without this the compiler
+// happily sinks the loads back into every use (re-reading global memory per
+// loop iteration) when keeping N registers live across barriers looks
+// expensive to it. The empty asm makes each element opaque at this point
+// (guide §5.7 item 3).
+#define PDRL_PIN_REGS(arr, n)                                   \
+  _Pragma("unroll") for (int _pi = 0; _pi < (n); ++_pi) {       \
+    asm volatile("" : "+v"((arr)[_pi]));                        \
+  }
+
 static inline hipStream_t current_stream() {
   return at::cuda::getCurrentCUDAStream().stream();
 }

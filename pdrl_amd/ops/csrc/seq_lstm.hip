@@ -69,6 +69,8 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_fwd_kernel(
   for (int k = 0; k < H; ++k) wih[k] = w_ih[k * G + tid];
 #pragma unroll
   for (int k = 0; k < H; ++k) whh[k] = w_hh[k * G + tid];
+  PDRL_PIN_REGS(wih, H);
+  PDRL_PIN_REGS(whh, H);
   const float bias = b_g[tid];
 
   // Body GEMM + ReLU for all S steps of this row (K1).
@@ -195,6 +197,8 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_kernel(
     wih_row[gg] = w_ih[(long)kk * G + part * H + gg];
     whh_row[gg] = w_hh[(long)kk * G + part * H + gg];
   }
+  PDRL_PIN_REGS(wih_row, H);
+  PDRL_PIN_REGS(whh_row, H);
 
   // Head back-projection: dh_heads[t][k] = sum_d gouts[t][d] * heads_w[k][d]
   for (int idx = tid; idx < S * H; idx += G) {

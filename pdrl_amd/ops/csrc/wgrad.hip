@@ -41,6 +41,15 @@ __device__ __forceinline__ const float* gate_a_row(const float* stash,
 // One wave computes one 16x16 tile of C = A^T · B.
 // blockIdx.x → (m0, g_base); 4 waves fan out over g; blockIdx.y → which GEMM.
 template <int H>
+__device__ void wgrad_small_body(const float*, const float*, const float*,
+                                 const float*, const float*, float*, float*,
+                                 float*, float*, float*, float*, int, int,
+                                 int, int);
+
+// One launch covers BOTH gate-GEMMs (MFMA tiles) and the small grads:
+// blocks [0, gemm_blocks) × y∈{0,1} run the two GEMMs; blocks beyond that on
+// y==0 run the wave-per-element small-grad reductions concurrently.
+template <int H>
 __global__ __launch_bounds__(256) void wgrad_gates_mfma_kernel(
     const float* __restrict__ stash,   // (B,S,7H)
     const float* __restrict__ h0,      // (B,H)
@@ -48,11 +57,25 @@ __global__ __launch_bounds__(256) void wgrad_gates_mfma_kernel(
     float* __restrict__ dw_ih,         // (H,4H)
     float* __restrict__ dw_hh,         // (H,4H)
     float* __restrict__ norm_sq,       // optional ||grad||² accumulator
+    const float* __restrict__ x,       // (N,F) small-grad inputs …
+    const float* __restrict__ dxb,     // (N,H)
+    const float* __restrict__ gouts,   // (N,D)
+    float* __restrict__ dbody_w, float* __restrict__ dbody_b,
+    float* __restrict__ db_g, float* __restrict__ dheads_w,
+    float* __restrict__ dheads_b, int F, int D,
     int N, int S, long h0s) {
   constexpr int G = 4 * H;
+  const int g_blocks = G / kWave;  // 64-wide g blocks
+  const int gemm_blocks = (H / 16) * g_blocks;
+  if ((int)blockIdx.x >= gemm_blocks) {
+    if (blockIdx.y != 0) return;
+    wgrad_small_body<H>(x, dxb, stash, dgates, gouts, dbody_w, dbody_b, db_g,
+                        dheads_w, dheads_b, norm_sq, N, F, D,
+                        ((int)blockIdx.x - gemm_blocks) * (256 / kWave));
+    return;
+  }
   const int wave = threadIdx.x / kWave;
   const int lane = threadIdx.x % kWave;
-  const int g_blocks = G / kWave;  // 64-wide g blocks
   const int m0 = (blockIdx.x / g_blocks) * 16;
   const int g0 = (blockIdx.x % g_blocks) * kWave + wave * 16;
   const int sel = blockIdx.y;
@@ -148,7 +171,7 @@ __global__ __launch_bounds__(256) void wgrad_gates_mfma_kernel(
 // Segments: dbody_w (F*H) | dbody_b (H) | db_g (4H) | dheads_w (H*D) |
 //           dheads_b (D)
 template <int H>
-__global__ void wgrad_small_kernel(
+__device__ void wgrad_small_body(
     const float* __restrict__ x,       // (N,F)
     const float* __restrict__ dxb,     // (N,H)
     const float* __restrict__ stash,   // (N,7H)
@@ -157,9 +180,9 @@ __global__ void wgrad_small_kernel(
     float* __restrict__ dbody_w, float* __restrict__ dbody_b,
     float* __restrict__ db_g, float* __restrict__ dheads_w,
     float* __restrict__ dheads_b, float* __restrict__ norm_sq,
-    int N, int F, int D) {
+    int N, int F, int D, int wave_base) {
   constexpr int G = 4 * H;
-  const int wave_id = (blockIdx.x * blockDim.x + threadIdx.x) / kWave;
+  const int wave_id = wave_base + (int)threadIdx.x / kWave;
   const int lane = threadIdx.x % kWave;
   const int n_fw = F * H, n_hw = H * D;
   const int total = n_fw + H + G + n_hw + D;
@@ -207,11 +230,11 @@ __global__ void wgrad_small_kernel(
     if (lane == 0) out[oi] = acc;
   }
   if (norm_sq != nullptr) {  // one atomic per block
-    __shared__ float red[4];
-    if (lane == 0) red[threadIdx.x / kWave] = live_wave ? acc * acc : 0.f;
+    __shared__ float red_s[4];
+    if (lane == 0) red_s[threadIdx.x / kWave] = live_wave ? acc * acc : 0.f;
     __syncthreads();
     if (threadIdx.x == 0)
-      atomicAdd(norm_sq, red[0] + red[1] + red[2] + red[3]);
+      atomicAdd(norm_sq, red_s[0] + red_s[1] + red_s[2] + red_s[3]);
   }
 }
 
@@ -224,8 +247,11 @@ void launch_wgrad(const at::Tensor& x, const at::Tensor& h0,
                   at::Tensor& dheads_b, const c10::optional<at::Tensor>& norm_sq,
                   int N, int S, int F, int D) {
   constexpr int G = 4 * H;
-  // gate GEMMs: (H/16) m-tiles × (G/64) g-blocks blocks, 2 GEMMs on y
-  dim3 grid((H / 16) * (G / kWave), 2);
+  // one fused launch: gate GEMM tiles + small-grad blocks side by side
+  const int gemm_blocks = (H / 16) * (G / kWave);
+  const int total_waves = F * H + H + G + H * D + D;
+  const int small_blocks = (total_waves * kWave + 255) / 256;
+  dim3 grid(gemm_blocks + small_blocks, 2);
   const int tab_lds = N * sizeof(const float*);
   TORCH_CHECK(tab_lds <= 64 * 1024, "wgrad row table exceeds LDS (B*S too big)");
   float* nrm = norm_sq.has_value() ? norm_sq->data_ptr<float>() : nullptr;
@@ -233,19 +259,11 @@ void launch_wgrad(const at::Tensor& x, const at::Tensor& h0,
                      current_stream(), stash.data_ptr<float>(),
                      h0.data_ptr<float>(), dgates.data_ptr<float>(),
                      dw_ih.data_ptr<float>(), dw_hh.data_ptr<float>(), nrm,
-                     N, S, (long)h0.stride(0));
-  HIP_CHECK_LAST();
-
-  const int total_waves = F * H + H + G + H * D + D;
-  const int threads = 256;
-  const int blocks = (total_waves * kWave + threads - 1) / threads;
-  hipLaunchKernelGGL((wgrad_small_kernel<H>), dim3(blocks), dim3(threads), 0,
-                     current_stream(), x.data_ptr<float>(),
-                     dxb.data_ptr<float>(), stash.data_ptr<float>(),
-                     dgates.data_ptr<float>(), gouts.data_ptr<float>(),
-                     dbody_w.data_ptr<float>(), dbody_b.data_ptr<float>(),
-                     db_g.data_ptr<float>(), dheads_w.data_ptr<float>(),
-                     dheads_b.data_ptr<float>(), nrm, N, F, D);
+                     x.data_ptr<float>(), dxb.data_ptr<float>(),
+                     gouts.data_ptr<float>(), dbody_w.data_ptr<float>(),
+                     dbody_b.data_ptr<float>(), db_g.data_ptr<float>(),
+                     dheads_w.data_ptr<float>(), dheads_b.data_ptr<float>(),
+                     F, D, N, S, (long)h0.stride(0));
   HIP_CHECK_LAST();
 }
 
