@@ -125,6 +125,17 @@ class Learner:
         self._rng = np.random.default_rng(1234 + rank)
         self.is_root = rank == 0
 
+        # Off-policy on GPU: the replay lives in HBM (BASELINE configs[5]) —
+        # the host ring becomes an ingest conveyor, sampling is device-side.
+        self.device_replay = None
+        if not self.on_policy and self.device.type == "cuda":
+            from pdrl_amd.buffers.device_replay import DeviceReplay
+
+            self.device_replay = DeviceReplay(
+                ring.field_dims, ring.seq_len, params.buffer_size, self.device,
+                seed=1234 + rank,
+            )
+
         # weight plane: PUB bound at learner_port + 1 (rank 0 only)
         self.pub = pub_bind(learner_ip, learner_port + 1) if self.is_root else None
         self.writer = SummaryWriter(params.result_dir) if self.is_root else None
@@ -139,6 +150,13 @@ class Learner:
         b = self.params.batch_size
         if self.on_policy:
             return self.ring.available() >= b
+        if self.device_replay is not None:
+            new = self.ring.drain_new()
+            if new is not None:
+                self.device_replay.append_batch(new)
+            if not self._replay_ready:
+                self._replay_ready = self.device_replay.size >= b
+            return self._replay_ready
         if not self._replay_ready:
             self._replay_ready = self.ring.ready(b)
         return self._replay_ready
@@ -149,13 +167,18 @@ class Learner:
         b = self.params.batch_size
         while not self._stopped() and time.monotonic() < deadline:
             if self.sample_ready():
-                batch_np = (
-                    self.ring.drain_batch(b)
-                    if self.on_policy
-                    else self.ring.sample_batch(b, self._rng)
-                )
-                if batch_np is not None:
-                    return self.stager.stage(batch_np)
+                if self.device_replay is not None:
+                    batch = self.device_replay.sample(b)
+                    if batch is not None:
+                        return batch
+                else:
+                    batch_np = (
+                        self.ring.drain_batch(b)
+                        if self.on_policy
+                        else self.ring.sample_batch(b, self._rng)
+                    )
+                    if batch_np is not None:
+                        return self.stager.stage(batch_np)
             time.sleep(0.001)
         return None
 

@@ -95,6 +95,24 @@ class SharedRolloutRing:
             self._consumed.value = base + batch
             return out
 
+    def drain_new(self, max_n: int | None = None) -> dict[str, np.ndarray] | None:
+        """Off-policy ingest conveyor: copy out every slot written since the
+        last drain (oldest lost on overwrite), advancing the consumed cursor.
+        Used to feed the DEVICE-resident replay (buffers/device_replay.py)."""
+        with self._lock:
+            head, base = self._head.value, self._consumed.value
+            if head - base > self.capacity:  # producer lapped us
+                base = head - self.capacity
+            n = head - base
+            if max_n is not None:
+                n = min(n, max_n)
+            if n == 0:
+                return None
+            idx = np.arange(base, base + n) % self.capacity
+            out = {name: self._view(name)[idx].copy() for name in self.field_dims}
+            self._consumed.value = base + n
+            return out
+
     def sample_batch(self, batch: int, rng: np.random.Generator) -> dict[str, np.ndarray] | None:
         """Off-policy: random sample of ``batch`` filled slots."""
         with self._lock:

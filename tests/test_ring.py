@@ -79,3 +79,17 @@ def test_cross_process_producer_consumer():
     p.join(10)
     assert not p.is_alive()
     assert seen == [float(v) for v in range(24)]
+
+
+def test_drain_new_off_policy():
+    ring = SharedRolloutRing(FIELDS, seq_len=5, capacity=8, on_policy=False)
+    assert ring.drain_new() is None
+    for v in range(5):
+        ring.put(make_traj(v))
+    out = ring.drain_new()
+    np.testing.assert_array_equal(out["rew"][:, 0, 0], [0, 1, 2, 3, 4])
+    assert ring.drain_new() is None  # nothing new
+    for v in range(5, 20):  # overwrites: only last 8 retrievable
+        ring.put(make_traj(v))
+    out = ring.drain_new()
+    np.testing.assert_array_equal(out["rew"][:, 0, 0], list(range(12, 20)))
