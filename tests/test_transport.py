@@ -7,7 +7,7 @@ from pdrl_amd.transport import Endpoint
 from pdrl_amd.utils import Protocol, decode, encode
 
 
-def _wait(cond, timeout=5.0):
+def _wait(cond, timeout=15.0):
     deadline = time.monotonic() + timeout
     while time.monotonic() < deadline:
         if cond():
