@@ -1,0 +1,147 @@
+"""End-to-end ingest bench: N workers → manager → storage → shared ring,
+no learner. Reports steps ingested and trajectories stored per second, plus
+per-stage counters to localize bottlenecks.
+
+    python scripts/ingest_bench.py --workers 8 --seconds 15
+"""
+from __future__ import annotations
+
+import argparse
+import socket
+import sys
+import time
+from pathlib import Path
+
+REPO = Path(__file__).resolve().parent.parent
+sys.path.insert(0, str(REPO))
+
+
+def free_port() -> int:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def _worker(model, idx, mport, seconds):
+    import torch
+
+    torch.set_num_threads(1)
+    sys.path.insert(0, str(REPO))
+    from pdrl_amd.agents import Worker
+    from pdrl_amd.utils import load_params
+    import main as main_mod
+
+    p = load_params()
+    p.algo, p.env = "IMPALA", "CartPole-v1"
+    main_mod.probe_env_spaces(p)
+    if model is None:
+        model = main_mod.build_model(p).cpu().eval()
+    w = Worker(model, idx, "127.0.0.1", mport, "127.0.0.1", 1, p, seed=idx)
+    deadline = time.monotonic() + seconds
+    while time.monotonic() < deadline:
+        w.collect(max_episodes=1)
+
+
+def _manager(mport, lport, seconds, relayed):
+    sys.path.insert(0, str(REPO))
+    import torch
+
+    torch.set_num_threads(1)
+    from pdrl_amd.agents import Manager
+
+    m = Manager("127.0.0.1", mport, "127.0.0.1", lport)
+    deadline = time.monotonic() + seconds
+    n = 0
+    while time.monotonic() < deadline:
+        if m.relay_once(timeout=0.2):
+            n += 1
+            relayed.value = n
+
+
+def _storage(ring, lport, seconds, ingested, stored):
+    sys.path.insert(0, str(REPO))
+    import asyncio
+
+    import torch
+
+    torch.set_num_threads(1)
+    from pdrl_amd.agents import LearnerStorage
+    from pdrl_amd.utils import load_params
+    import main as main_mod
+
+    p = load_params()
+    p.algo, p.env = "IMPALA", "CartPole-v1"
+    main_mod.probe_env_spaces(p)
+    s = LearnerStorage(ring, "127.0.0.1", lport, p)
+
+    async def run():
+        t = asyncio.get_event_loop().time
+        deadline = t() + seconds
+        tasks = [asyncio.create_task(s.ingest_task()),
+                 asyncio.create_task(s.store_task())]
+        while t() < deadline:
+            await asyncio.sleep(0.5)
+            ingested.value = s.n_ingested
+            stored.value = s.n_stored
+        for task in tasks:
+            task.cancel()
+
+    asyncio.run(run())
+
+
+def main():
+    import torch.multiprocessing as mp
+
+    from pdrl_amd.buffers import SharedRolloutRing, rollout_fields
+    from pdrl_amd.utils import load_params
+    import main as main_mod
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--workers", type=int, default=8)
+    ap.add_argument("--seconds", type=float, default=15.0)
+    ap.add_argument("--shared-model", action="store_true")
+    args = ap.parse_args()
+
+    p = load_params()
+    p.algo, p.env = "IMPALA", "CartPole-v1"
+    main_mod.probe_env_spaces(p)
+    ctx = mp.get_context("spawn")
+    mport, lport = free_port(), free_port()
+    fields = rollout_fields(p.obs_dim, p.n_actions, p.hidden_size, False)
+    # big off-policy ring so storage never blocks on a full ring
+    ring = SharedRolloutRing(fields, p.seq_len, 65536, on_policy=False)
+
+    model = None
+    if args.shared_model:
+        model = main_mod.build_model(p).cpu().eval()
+        model.share_memory()
+
+    relayed = ctx.Value("q", 0)
+    ingested = ctx.Value("q", 0)
+    stored = ctx.Value("q", 0)
+    run_s = args.seconds + 10  # children outlive the measure window
+    procs = [ctx.Process(target=_manager, args=(mport, lport, run_s, relayed)),
+             ctx.Process(target=_storage, args=(ring, lport, run_s, ingested, stored))]
+    procs += [ctx.Process(target=_worker, args=(model, i, mport, run_s))
+              for i in range(args.workers)]
+    for pr in procs:
+        pr.start()
+    time.sleep(5)  # spin-up
+    r0, i0, s0 = relayed.value, ingested.value, stored.value
+    t0 = time.monotonic()
+    time.sleep(args.seconds)
+    dt = time.monotonic() - t0
+    r1, i1, s1 = relayed.value, ingested.value, stored.value
+    print(f"workers={args.workers} relayed={(r1-r0)/dt:8.0f}/s "
+          f"ingested={(i1-i0)/dt:8.0f}/s stored={(s1-s0)/dt:8.0f} traj/s "
+          f"(~{(s1-s0)*p.seq_len/dt:.0f} steps/s)")
+    for pr in procs:
+        pr.terminate()
+    for pr in procs:
+        pr.join(5)
+
+
+if __name__ == "__main__":
+    main()
