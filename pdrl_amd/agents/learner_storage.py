@@ -43,23 +43,27 @@ class LearnerStorage:
 
     # ------------------------------------------------------------------ #
     async def ingest_task(self):
-        """Data-plane SUB → assembler / stat array."""
+        """Data-plane SUB → assembler / stat array. Messages are drained in
+        batches; a Rollout payload is a CHUNK (list of steps) or one step."""
         while not self._stopped():
-            msg = self.sub.recv(timeout=0.0)
-            if msg is None:
+            msgs = self.sub.recv_many(256)
+            if not msgs:
                 await asyncio.sleep(0.001)
                 continue
-            protocol, data = decode(*msg)
-            if protocol is Protocol.Rollout:
-                await self.assembler.push(data)
-                self.n_ingested += 1
-            elif protocol is Protocol.Stat:
-                if self.shared_stat is not None:
-                    self.shared_stat[0] = float(data["game_count"])
-                    self.shared_stat[1] = float(data["mean_stat"])
-                    self.shared_stat[2] = 1.0  # fresh flag
-            else:
-                raise AssertionError(f"unexpected protocol at storage: {protocol}")
+            for msg in msgs:
+                protocol, data = decode(*msg)
+                if protocol is Protocol.Rollout:
+                    steps = data if isinstance(data, list) else [data]
+                    for step in steps:
+                        await self.assembler.push(step)
+                    self.n_ingested += len(steps)
+                elif protocol is Protocol.Stat:
+                    if self.shared_stat is not None:
+                        self.shared_stat[0] = float(data["game_count"])
+                        self.shared_stat[1] = float(data["mean_stat"])
+                        self.shared_stat[2] = 1.0  # fresh flag
+                else:
+                    raise AssertionError(f"unexpected protocol at storage: {protocol}")
             if self.heartbeat is not None:
                 self.heartbeat.value = time.time()
 

@@ -38,10 +38,13 @@ class Manager:
         if msg is None:
             return False
         header, payload = msg
-        protocol, data = decode(header, payload)
-        if protocol is Protocol.Rollout:
-            self.pub.send(header, payload)  # forward verbatim (no re-encode)
+        import pickle
+
+        protocol = pickle.loads(header)  # header-only inspect: rollouts are
+        if protocol is Protocol.Rollout:  # forwarded without decompressing
+            self.pub.send(header, payload)
         elif protocol is Protocol.Stat:
+            _, data = decode(header, payload)
             self.game_count += 1
             self.stat_q.append(float(data["epi_rew"]))
             if len(self.stat_q) >= self.STAT_INTERVAL:

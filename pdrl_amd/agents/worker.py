@@ -50,11 +50,22 @@ class Worker:
         self.pub = pub_connect(manager_ip, manager_port)
         self.sub = sub_connect(learner_ip, learner_port + 1)
         self.step_sleep = float(getattr(params, "worker_step_sleep", 0.0))
+        # steps per wire message: one decode per chunk instead of per step
+        # (a per-step wire protocol capped the single storage process at
+        # ~10K steps/s; seq_len-sized chunks amortize pickle+zlib 5×)
+        self.batch_steps = int(getattr(params, "worker_batch_steps", 0)) or params.seq_len
+        self._step_buf: list[dict] = []
 
     # ------------------------------------------------------------------ #
-    def pub_rollout(self, step_data: dict):
-        header, payload = encode(Protocol.Rollout, step_data)
-        self.pub.send(header, payload)
+    def pub_rollout(self, step_data: dict, flush: bool = False):
+        """Buffer the step; publish a chunk every ``batch_steps`` steps and at
+        episode end (the storage-side assembler consumes chunks step-wise, so
+        assembly semantics are unchanged)."""
+        self._step_buf.append(step_data)
+        if len(self._step_buf) >= self.batch_steps or flush:
+            header, payload = encode(Protocol.Rollout, self._step_buf)
+            self.pub.send(header, payload)
+            self._step_buf = []
 
     def pub_stat(self, epi_rew: float):
         header, payload = encode(Protocol.Stat, {"epi_rew": float(epi_rew)})
@@ -107,7 +118,7 @@ class Worker:
                     "cx": cx.squeeze(0).numpy(),
                     "id": epi_id,
                 }
-                self.pub_rollout(step_data)
+                self.pub_rollout(step_data, flush=done)
                 obs, hx, cx = next_obs, next_hx, next_cx
                 is_fir = 0.0
                 if self.heartbeat is not None:

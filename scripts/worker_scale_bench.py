@@ -84,10 +84,16 @@ def main():
         pr.join(args.seconds + 60)
     dt = time.monotonic() - t0
 
-    # drain sink to count actual step messages
+    # drain sink and count steps (messages carry step CHUNKS)
+    from pdrl_amd.utils import decode
+
     n = 0
-    while sink.recv(timeout=0.5) is not None:
-        n += 1
+    while True:
+        msg = sink.recv(timeout=0.5)
+        if msg is None:
+            break
+        _, data = decode(*msg)
+        n += len(data) if isinstance(data, list) else 1
     eps = sum(c.value for c in counters)
     mode = "private" if args.private else "shared"
     print(f"{args.workers} workers ({mode}): {n} steps, {eps} episodes in "
