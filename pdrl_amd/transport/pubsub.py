@@ -14,6 +14,13 @@ the system relies on:
   automatic reconnect) — matching the reference topology where workers
   connect-PUB to a bound manager SUB, and workers connect-SUB to the
   learner's bound weight PUB.
+* Flow control: when an endpoint's RX queue fills, it PAUSES reading its
+  sockets; TCP backpressure then stalls upstream senders, whose bounded send
+  queues drop THEIR oldest — so overload sheds load at the producer (the
+  cheapest place, and what flows through is the freshest data). Without
+  this, an overloaded consumer burns its CPU parsing messages it will drop
+  (measured: a 64-worker firehose halved the storage process's useful
+  ingest rate).
 * A message is a (header, payload) byte pair — the ZMQ multipart shape the
   Protocol enum + encode/decode produce.
 
@@ -73,6 +80,11 @@ class Endpoint:
         self._rx_cv = threading.Condition()
         self._send_hwm = send_hwm
         self._closed = False
+        # read-pause flow control thresholds
+        self._rx_high = max(2, int(recv_hwm * 0.9))
+        self._rx_low = recv_hwm // 2
+        self._paused = False
+        self._resume_req = False
         self._listener: socket.socket | None = None
         self._connect_addr = connect
         self._connecting: socket.socket | None = None
@@ -110,7 +122,13 @@ class Endpoint:
         except OSError:
             pass
         peer = _Peer(sock)
-        self._sel.register(sock, selectors.EVENT_READ, peer)
+        events = 0 if self._paused else selectors.EVENT_READ
+        self._sel.register(sock, events or selectors.EVENT_READ, peer)
+        if self._paused:
+            try:  # registered with READ then immediately muted (API needs >0)
+                self._sel.modify(sock, selectors.EVENT_WRITE, peer)
+            except (KeyError, ValueError):
+                pass
         with self._peers_lock:
             self._peers[sock.fileno()] = peer
 
@@ -128,13 +146,23 @@ class Endpoint:
             pass
 
     def _update_interest(self, peer: _Peer):
-        want = selectors.EVENT_READ
+        want = 0 if self._paused else selectors.EVENT_READ
         if peer.tx:
             want |= selectors.EVENT_WRITE
         try:
-            self._sel.modify(peer.sock, want, peer)
+            self._sel.modify(peer.sock, want or selectors.EVENT_WRITE, peer)
         except (KeyError, ValueError):
             pass
+
+    def _set_paused(self, paused: bool):
+        if paused == self._paused:
+            return
+        self._paused = paused
+        with self._peers_lock:
+            peers = list(self._peers.values())
+        for peer in peers:
+            if peer.alive:
+                self._update_interest(peer)
 
     def _io_loop(self):
         while not self._closed:
@@ -165,6 +193,9 @@ class Endpoint:
                         self._write_ready(peer)
             # distribute frames queued by send() to peer tx queues
             self._flush_pending()
+            if self._resume_req:
+                self._resume_req = False
+                self._set_paused(False)
         # teardown
         for key in list(self._sel.get_map().values()):
             if isinstance(key.data, _Peer):
@@ -248,6 +279,9 @@ class Endpoint:
             with self._rx_cv:
                 self._rx.extend(msgs)  # deque(maxlen) drops oldest
                 self._rx_cv.notify()
+                depth = len(self._rx)
+            if depth >= self._rx_high:
+                self._set_paused(True)  # io thread: safe to touch interests
 
     def _write_ready(self, peer: _Peer):
         try:
@@ -309,7 +343,12 @@ class Endpoint:
                 if remaining is not None and remaining <= 0:
                     return None
                 self._rx_cv.wait(timeout=remaining if remaining is not None else 0.5)
-            return self._rx.popleft()
+            out = self._rx.popleft()
+            depth = len(self._rx)
+        if self._paused and depth <= self._rx_low:
+            self._resume_req = True
+            self._wake()
+        return out
 
     def recv_many(self, max_n: int = 1024) -> list:
         """Drain up to max_n queued messages without blocking."""
@@ -317,6 +356,10 @@ class Endpoint:
         with self._rx_cv:
             while self._rx and len(out) < max_n:
                 out.append(self._rx.popleft())
+            depth = len(self._rx)
+        if self._paused and depth <= self._rx_low:
+            self._resume_req = True
+            self._wake()
         return out
 
     def n_peers(self) -> int:
