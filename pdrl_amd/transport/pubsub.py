@@ -47,7 +47,7 @@ _CHUNK = 1 << 18
 
 
 class _Peer:
-    __slots__ = ("sock", "rx", "tx", "tx_offset", "alive", "want_write")
+    __slots__ = ("sock", "rx", "tx", "tx_offset", "alive", "registered")
 
     def __init__(self, sock: socket.socket):
         self.sock = sock
@@ -55,7 +55,7 @@ class _Peer:
         self.tx: deque = deque()  # framed bytes objects
         self.tx_offset = 0
         self.alive = True
-        self.want_write = False
+        self.registered = False
 
 
 class Endpoint:
@@ -122,22 +122,18 @@ class Endpoint:
         except OSError:
             pass
         peer = _Peer(sock)
-        events = 0 if self._paused else selectors.EVENT_READ
-        self._sel.register(sock, events or selectors.EVENT_READ, peer)
-        if self._paused:
-            try:  # registered with READ then immediately muted (API needs >0)
-                self._sel.modify(sock, selectors.EVENT_WRITE, peer)
-            except (KeyError, ValueError):
-                pass
         with self._peers_lock:
             self._peers[sock.fileno()] = peer
+        self._update_interest(peer)
 
     def _drop_peer(self, peer: _Peer):
         peer.alive = False
-        try:
-            self._sel.unregister(peer.sock)
-        except (KeyError, ValueError):
-            pass
+        if peer.registered:
+            try:
+                self._sel.unregister(peer.sock)
+            except (KeyError, ValueError):
+                pass
+            peer.registered = False
         with self._peers_lock:
             self._peers.pop(peer.sock.fileno(), -1)
         try:
@@ -146,12 +142,23 @@ class Endpoint:
             pass
 
     def _update_interest(self, peer: _Peer):
+        # selectors require events != 0: a fully-muted peer (paused, nothing
+        # to send) is UNREGISTERED — registering it write-only would make
+        # select() always-ready and spin the reactor at 100% CPU
         want = 0 if self._paused else selectors.EVENT_READ
         if peer.tx:
             want |= selectors.EVENT_WRITE
         try:
-            self._sel.modify(peer.sock, want or selectors.EVENT_WRITE, peer)
-        except (KeyError, ValueError):
+            if want == 0:
+                if peer.registered:
+                    self._sel.unregister(peer.sock)
+                    peer.registered = False
+            elif peer.registered:
+                self._sel.modify(peer.sock, want, peer)
+            else:
+                self._sel.register(peer.sock, want, peer)
+                peer.registered = True
+        except (KeyError, ValueError, OSError):
             pass
 
     def _set_paused(self, paused: bool):
