@@ -47,13 +47,15 @@ _CHUNK = 1 << 18
 
 
 class _Peer:
-    __slots__ = ("sock", "rx", "tx", "tx_offset", "alive", "registered")
+    __slots__ = ("sock", "rx", "tx", "inflight", "inflight_off", "alive",
+                 "registered")
 
     def __init__(self, sock: socket.socket):
         self.sock = sock
         self.rx = bytearray()
-        self.tx: deque = deque()  # framed bytes objects
-        self.tx_offset = 0
+        self.tx: deque = deque()  # droppable framed bytes objects
+        self.inflight: bytes | None = None  # partially-sent frame — NEVER
+        self.inflight_off = 0               # dropped (would desync framing)
         self.alive = True
         self.registered = False
 
@@ -146,7 +148,7 @@ class Endpoint:
         # to send) is UNREGISTERED — registering it write-only would make
         # select() always-ready and spin the reactor at 100% CPU
         want = 0 if self._paused else selectors.EVENT_READ
-        if peer.tx:
+        if peer.tx or peer.inflight is not None:
             want |= selectors.EVENT_WRITE
         try:
             if want == 0:
@@ -292,16 +294,19 @@ class Endpoint:
 
     def _write_ready(self, peer: _Peer):
         try:
-            while peer.tx:
-                frame = peer.tx[0]
-                sent = peer.sock.send(frame[peer.tx_offset:] if peer.tx_offset
-                                      else frame)
-                peer.tx_offset += sent
-                if peer.tx_offset >= len(frame):
-                    peer.tx.popleft()
-                    peer.tx_offset = 0
-                elif sent == 0:
+            while True:
+                if peer.inflight is None:
+                    if not peer.tx:
+                        break
+                    peer.inflight = peer.tx.popleft()
+                    peer.inflight_off = 0
+                sent = peer.sock.send(peer.inflight[peer.inflight_off:])
+                if sent == 0:
                     break
+                peer.inflight_off += sent
+                if peer.inflight_off >= len(peer.inflight):
+                    peer.inflight = None
+                    peer.inflight_off = 0
         except (BlockingIOError, InterruptedError):
             pass
         except OSError:
@@ -322,15 +327,10 @@ class Endpoint:
                 continue
             for f in frames:
                 if len(peer.tx) >= self._send_hwm:
-                    peer.tx.popleft()  # drop oldest (PUB HWM semantics)
-                    if peer.tx_offset:
-                        # dropped a partially-sent frame: resync is impossible
-                        # mid-frame, so drop the peer instead
-                        self._drop_peer(peer)
-                        break
+                    peer.tx.popleft()  # drop oldest (PUB HWM semantics);
+                    # the in-flight frame lives outside tx, so framing is safe
                 peer.tx.append(f)
-            if peer.alive:
-                self._write_ready(peer)
+            self._write_ready(peer)
 
     # -- API ---------------------------------------------------------------- #
     def send(self, header: bytes, payload: bytes):
