@@ -74,6 +74,14 @@ def main():
     base = json.loads((REPO / "pdrl_amd/utils/parameters.json").read_text())
     base["algo"] = args.algo
     base["env"] = args.env
+    if args.lr is not None:
+        base["lr"] = args.lr
+    if args.entropy is not None:
+        base["entropy_coef"] = args.entropy
+    if args.clip is not None:
+        base["max_grad_norm"] = args.clip
+    if args.batch_size is not None:
+        base["batch_size"] = args.batch_size
     tmp = tempfile.NamedTemporaryFile("w", suffix=".json", delete=False)
     json.dump(base, tmp)
     tmp.close()
