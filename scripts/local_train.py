@@ -65,6 +65,10 @@ def main():
     ap.add_argument("--workers", type=int, default=12)
     ap.add_argument("--minutes", type=float, default=5.0)
     ap.add_argument("--target", type=float, default=475.0)
+    ap.add_argument("--lr", type=float, default=None)
+    ap.add_argument("--entropy", type=float, default=None)
+    ap.add_argument("--clip", type=float, default=None, help="max_grad_norm")
+    ap.add_argument("--batch-size", type=int, default=None)
     args = ap.parse_args()
 
     from pdrl_amd.utils import load_params
