@@ -38,7 +38,7 @@ class ImpalaUpdater(BaseUpdater):
         behav_log_prob, is_fir = batch["log_prob"], batch["is_fir"]
         hx0, cx0 = batch_initial_state(batch)
 
-        _, log_probs, entropy, value = self.model.actor(obs, (hx0, cx0), act)
+        logits, log_probs, entropy, value = self.model.actor(obs, (hx0, cx0), act)
 
         rhos, advantages, values_target = compute_v_trace(
             behav_log_prob, log_probs, is_fir, rew, value, p.gamma
@@ -48,10 +48,16 @@ class ImpalaUpdater(BaseUpdater):
         value_loss = F.smooth_l1_loss(value[:, :-1], values_target)
         entropy_mean = entropy[:, :-1].mean()
 
+        # logit L2 keeps the policy out of exact one-hot saturation (an fp32
+        # one-hot has ZERO policy/entropy gradients — an absorbing collapse
+        # state observed at high update rates)
+        logit_reg = float(getattr(p, "logit_reg", 0.0))
+        reg_loss = logits[:, :-1].pow(2).mean() if logit_reg > 0 else 0.0
         loss = (
             p.policy_loss_coef * policy_loss
             + p.value_loss_coef * value_loss
             - p.entropy_coef * entropy_mean
+            + logit_reg * reg_loss
         )
         stats = {
             "loss-total": loss.detach(),

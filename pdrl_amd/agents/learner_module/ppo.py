@@ -54,10 +54,16 @@ class PPOUpdater(BaseUpdater):
         value_loss = F.smooth_l1_loss(value[:, :-1], td_target)
         entropy_mean = entropy[:, :-1].mean()
 
+        # logit L2 keeps the policy out of exact one-hot saturation (an fp32
+        # one-hot has ZERO policy/entropy gradients — an absorbing collapse
+        # state observed at high update rates)
+        logit_reg = float(getattr(p, "logit_reg", 0.0))
+        reg_loss = logits[:, :-1].pow(2).mean() if logit_reg > 0 else 0.0
         loss = (
             p.policy_loss_coef * policy_loss
             + p.value_loss_coef * value_loss
             - p.entropy_coef * entropy_mean
+            + logit_reg * reg_loss
         )
         stats = {
             "loss-total": loss.detach(),

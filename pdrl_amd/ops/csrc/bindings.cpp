@@ -34,31 +34,31 @@ std::vector<at::Tensor> ppo_td_gae_hip(const at::Tensor&, const at::Tensor&,
 void impala_loss_reduce_hip(const at::Tensor&, const at::Tensor&,
                             const at::Tensor&, long, const at::Tensor&,
                             const at::Tensor&, const at::Tensor&, at::Tensor&,
-                            double, double, double);
+                            double, double, double, double);
 void ppo_loss_reduce_hip(const at::Tensor&, const at::Tensor&,
                          const at::Tensor&, const at::Tensor&, long,
                          const at::Tensor&, const at::Tensor&, at::Tensor&,
-                         double, double, double, double);
+                         double, double, double, double, double);
 at::Tensor impala_loss_bwd_hip(const at::Tensor&, long, const at::Tensor&,
                                const at::Tensor&, const at::Tensor&,
                                const at::Tensor&, const at::Tensor&, double,
-                               double, double);
+                               double, double, double);
 at::Tensor ppo_loss_bwd_hip(const at::Tensor&, long, const at::Tensor&,
                             const at::Tensor&, const at::Tensor&,
                             const at::Tensor&, const at::Tensor&,
                             const at::Tensor&, const at::Tensor&, double,
-                            double, double, double);
+                            double, double, double, double);
 bool impala_loss_mega_hip(const at::Tensor&, const at::Tensor&,
                           const at::Tensor&, const at::Tensor&,
                           const at::Tensor&, at::Tensor&, at::Tensor&,
                           const c10::optional<at::Tensor>&, long, double,
                           double, double, double, double, double, double,
-                          double);
+                          double, double);
 bool ppo_loss_mega_hip(const at::Tensor&, const at::Tensor&, const at::Tensor&,
                        const at::Tensor&, const at::Tensor&, at::Tensor&,
                        at::Tensor&, const c10::optional<at::Tensor>&, long,
                        double, double, double, double, double, double,
-                       double);
+                       double, double);
 void l2norm_sq_hip(const at::Tensor&, at::Tensor&);
 void rmsprop_step_hip(at::Tensor&, const at::Tensor&, at::Tensor&,
                       const at::Tensor&, double, double, double, double);

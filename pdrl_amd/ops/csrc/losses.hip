@@ -90,10 +90,11 @@ __global__ void impala_loss_reduce_kernel(
     const float* __restrict__ vs,    // (B,T)
     const float* __restrict__ rhos,  // (B,T)
     float* __restrict__ stats,       // (5)
-    int B, int S, int D, float cp, float cv, float ce) {
+    int B, int S, int D, float cp, float cv, float ce, float creg) {
   const int T = S - 1;
   const int N = B * T;
-  float pl = 0, vl = 0, es = 0, rs = 0;
+  const int A = D - 1;
+  float pl = 0, vl = 0, es = 0, rs = 0, rg = 0;
   for (int i = threadIdx.x; i < N; i += blockDim.x) {
     const int b = i / T, t = i % T;
     const long si = (long)b * S + t;
@@ -101,6 +102,10 @@ __global__ void impala_loss_reduce_kernel(
     vl += huber(val[si * D] - vs[i]);
     es += ent[si];
     rs += rhos[i];
+    for (int j = 0; j < A; ++j) {
+      const float z = val[si * D + j - A];  // logits live at cols [0, A)
+      rg = fmaf(z, z, rg);
+    }
   }
   __shared__ float red[4][256];
   red[0][threadIdx.x] = pl; red[1][threadIdx.x] = vl;
@@ -113,10 +118,17 @@ __global__ void impala_loss_reduce_kernel(
     }
     __syncthreads();
   }
+  __shared__ float rgred[256];
+  rgred[threadIdx.x] = rg;
+  __syncthreads();
+  for (int off = blockDim.x / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) rgred[threadIdx.x] += rgred[threadIdx.x + off];
+    __syncthreads();
+  }
   if (threadIdx.x == 0) {
     const float inv = 1.0f / N;
     const float p = red[0][0] * inv, v = red[1][0] * inv, e = red[2][0] * inv;
-    stats[0] = cp * p + cv * v - ce * e;
+    stats[0] = cp * p + cv * v - ce * e + creg * rgred[0] / (N * A);
     stats[1] = p; stats[2] = v; stats[3] = e; stats[4] = red[3][0] * inv;
   }
 }
@@ -130,10 +142,12 @@ __global__ void ppo_loss_reduce_kernel(
     const float* __restrict__ adv,        // (B,T)
     const float* __restrict__ td,         // (B,T)
     float* __restrict__ stats,            // (7)
-    int B, int S, int D, float cp, float cv, float ce, float eps_clip) {
+    int B, int S, int D, float cp, float cv, float ce, float eps_clip,
+    float creg) {
   const int T = S - 1;
   const int N = B * T;
-  float pl = 0, vl = 0, es = 0, ravg = 0;
+  const int A = D - 1;
+  float pl = 0, vl = 0, es = 0, ravg = 0, rg = 0;
   float rmin = 1e30f, rmax = -1e30f;
   for (int i = threadIdx.x; i < N; i += blockDim.x) {
     const int b = i / T, t = i % T;
@@ -148,6 +162,10 @@ __global__ void ppo_loss_reduce_kernel(
     ravg += r;
     rmin = fminf(rmin, r);
     rmax = fmaxf(rmax, r);
+    for (int j = 0; j < A; ++j) {
+      const float z = val[si * D + j - A];
+      rg = fmaf(z, z, rg);
+    }
   }
   __shared__ float red[4][256];
   __shared__ float rmn[256], rmx[256];
@@ -164,10 +182,17 @@ __global__ void ppo_loss_reduce_kernel(
     }
     __syncthreads();
   }
+  __shared__ float rgred[256];
+  rgred[threadIdx.x] = rg;
+  __syncthreads();
+  for (int off = blockDim.x / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) rgred[threadIdx.x] += rgred[threadIdx.x + off];
+    __syncthreads();
+  }
   if (threadIdx.x == 0) {
     const float inv = 1.0f / N;
     const float p = red[0][0] * inv, v = red[1][0] * inv, e = red[2][0] * inv;
-    stats[0] = cp * p + cv * v - ce * e;
+    stats[0] = cp * p + cv * v - ce * e + creg * rgred[0] / (N * A);
     stats[1] = p; stats[2] = v; stats[3] = e;
     stats[4] = red[3][0] * inv; stats[5] = rmn[0]; stats[6] = rmx[0];
   }
@@ -182,7 +207,7 @@ __global__ void impala_loss_bwd_kernel(
     const float* __restrict__ adv,   // (B,T)
     const float* __restrict__ vs,    // (B,T)
     float* __restrict__ gouts,       // (B,S,D)
-    int B, int S, int A, float cp, float cv, float ce) {
+    int B, int S, int A, float cp, float cv, float ce, float creg) {
   const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long N = (long)B * S;
   if (i >= N) return;
@@ -195,6 +220,7 @@ __global__ void impala_loss_bwd_kernel(
     return;
   }
   const float invN = 1.0f / (B * T);
+  const float dreg = 2.0f * creg * invN / A;  // d/dz of creg*mean(z^2)
   const long ti = (long)b * T + t;
   const float dlogp = -cp * adv[ti] * invN;
   const float dH = -ce * invN;
@@ -204,7 +230,8 @@ __global__ void impala_loss_bwd_kernel(
   for (int j = 0; j < A; ++j) {
     const float lp = z[j] - lse[i];
     const float p = __expf(lp);
-    g[j] = dlogp * ((j == a ? 1.0f : 0.0f) - p) + dH * (-p * (lp + H));
+    g[j] = dlogp * ((j == a ? 1.0f : 0.0f) - p) + dH * (-p * (lp + H)) +
+           dreg * z[j];
   }
   g[A] = cv * huber_grad(z[A] - vs[ti]) * invN;
 }
@@ -219,7 +246,8 @@ __global__ void ppo_loss_bwd_kernel(
     const float* __restrict__ adv,       // (B,T)
     const float* __restrict__ td,        // (B,T)
     float* __restrict__ gouts,           // (B,S,D)
-    int B, int S, int A, float cp, float cv, float ce, float eps_clip) {
+    int B, int S, int A, float cp, float cv, float ce, float eps_clip,
+    float creg) {
   const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long N = (long)B * S;
   if (i >= N) return;
@@ -240,6 +268,7 @@ __global__ void ppo_loss_bwd_kernel(
   const float s2 = fminf(fmaxf(r, 1.0f - eps_clip), 1.0f + eps_clip) * a_v;
   // d min(s1, s2)/d logp: the clipped branch has zero grad outside the band
   const float gr = (inside || s1 < s2) ? a_v * r : 0.0f;
+  const float dreg = 2.0f * creg * invN / A;
   const float dlogp = -cp * gr * invN;
   const float dH = -ce * invN;
   const float H = ent[i];
@@ -248,7 +277,8 @@ __global__ void ppo_loss_bwd_kernel(
   for (int j = 0; j < A; ++j) {
     const float lp = z[j] - lse[i];
     const float p = __expf(lp);
-    g[j] = dlogp * ((j == a ? 1.0f : 0.0f) - p) + dH * (-p * (lp + H));
+    g[j] = dlogp * ((j == a ? 1.0f : 0.0f) - p) + dH * (-p * (lp + H)) +
+           dreg * z[j];
   }
   g[A] = cv * huber_grad(z[A] - td[ti]) * invN;
 }
@@ -301,7 +331,7 @@ void impala_loss_reduce_hip(const at::Tensor& logp, const at::Tensor& ent,
                             const at::Tensor& model_out, long A,
                             const at::Tensor& adv, const at::Tensor& vs,
                             const at::Tensor& rhos, at::Tensor& stats,
-                            double cp, double cv, double ce) {
+                            double cp, double cv, double ce, double creg) {
   const int B = model_out.size(0), S = model_out.size(1);
   const int D = model_out.size(2);
   hipLaunchKernelGGL(impala_loss_reduce_kernel, dim3(1), dim3(256), 0,
@@ -309,7 +339,7 @@ void impala_loss_reduce_hip(const at::Tensor& logp, const at::Tensor& ent,
                      ent.data_ptr<float>(), model_out.data_ptr<float>() + A,
                      adv.data_ptr<float>(), vs.data_ptr<float>(),
                      rhos.data_ptr<float>(), stats.data_ptr<float>(), B, S, D,
-                     (float)cp, (float)cv, (float)ce);
+                     (float)cp, (float)cv, (float)ce, (float)creg);
   HIP_CHECK_LAST();
 }
 
@@ -317,7 +347,7 @@ void ppo_loss_reduce_hip(const at::Tensor& logp, const at::Tensor& behav_lp,
                          const at::Tensor& ent, const at::Tensor& model_out,
                          long A, const at::Tensor& adv, const at::Tensor& td,
                          at::Tensor& stats, double cp, double cv, double ce,
-                         double eps_clip) {
+                         double eps_clip, double creg) {
   const int B = model_out.size(0), S = model_out.size(1);
   const int D = model_out.size(2);
   hipLaunchKernelGGL(ppo_loss_reduce_kernel, dim3(1), dim3(256), 0,
@@ -325,7 +355,8 @@ void ppo_loss_reduce_hip(const at::Tensor& logp, const at::Tensor& behav_lp,
                      behav_lp.data_ptr<float>(), ent.data_ptr<float>(),
                      model_out.data_ptr<float>() + A, adv.data_ptr<float>(),
                      td.data_ptr<float>(), stats.data_ptr<float>(), B, S, D,
-                     (float)cp, (float)cv, (float)ce, (float)eps_clip);
+                     (float)cp, (float)cv, (float)ce, (float)eps_clip,
+                     (float)creg);
   HIP_CHECK_LAST();
 }
 
@@ -333,7 +364,7 @@ at::Tensor impala_loss_bwd_hip(const at::Tensor& model_out, long A,
                                const at::Tensor& act, const at::Tensor& lse,
                                const at::Tensor& ent, const at::Tensor& adv,
                                const at::Tensor& vs, double cp, double cv,
-                               double ce) {
+                               double ce, double creg) {
   const int B = model_out.size(0), S = model_out.size(1);
   auto gouts = at::empty_like(model_out);
   const long N = (long)B * S;
@@ -344,7 +375,7 @@ at::Tensor impala_loss_bwd_hip(const at::Tensor& model_out, long A,
                      act.data_ptr<float>(), lse.data_ptr<float>(),
                      ent.data_ptr<float>(), adv.data_ptr<float>(),
                      vs.data_ptr<float>(), gouts.data_ptr<float>(), B, S,
-                     (int)A, (float)cp, (float)cv, (float)ce);
+                     (int)A, (float)cp, (float)cv, (float)ce, (float)creg);
   HIP_CHECK_LAST();
   return gouts;
 }
@@ -354,7 +385,7 @@ at::Tensor ppo_loss_bwd_hip(const at::Tensor& model_out, long A,
                             const at::Tensor& ent, const at::Tensor& logp,
                             const at::Tensor& behav_lp, const at::Tensor& adv,
                             const at::Tensor& td, double cp, double cv,
-                            double ce, double eps_clip) {
+                            double ce, double eps_clip, double creg) {
   const int B = model_out.size(0), S = model_out.size(1);
   auto gouts = at::empty_like(model_out);
   const long N = (long)B * S;
@@ -366,7 +397,7 @@ at::Tensor ppo_loss_bwd_hip(const at::Tensor& model_out, long A,
                      logp.data_ptr<float>(), behav_lp.data_ptr<float>(),
                      adv.data_ptr<float>(), td.data_ptr<float>(),
                      gouts.data_ptr<float>(), B, S, (int)A, (float)cp,
-                     (float)cv, (float)ce, (float)eps_clip);
+                     (float)cv, (float)ce, (float)eps_clip, (float)creg);
   HIP_CHECK_LAST();
   return gouts;
 }
@@ -393,7 +424,7 @@ __global__ __launch_bounds__(kMegaThreads) void impala_loss_mega_kernel(
     float* __restrict__ stats,        // (5)
     float* __restrict__ norm_sq,      // optional: zeroed here for the wgrads
     int B, int S, int A, float gamma, float rho_bar, float rho_min,
-    float c_bar, float rew_scale, float cp, float cv, float ce) {
+    float c_bar, float rew_scale, float cp, float cv, float ce, float creg) {
   const int D = A + 1;
   const int T = S - 1;
   const int N = B * S;
@@ -453,7 +484,7 @@ __global__ __launch_bounds__(kMegaThreads) void impala_loss_mega_kernel(
 
   // phase C: loss reduction
   {
-    float pl = 0, vl = 0, es = 0, rs = 0;
+    float pl = 0, vl = 0, es = 0, rs = 0, rg = 0;
     for (int i = tid; i < BT; i += kMegaThreads) {
       const int b = i / T, t = i % T;
       const long si = (long)b * S + t;
@@ -461,25 +492,31 @@ __global__ __launch_bounds__(kMegaThreads) void impala_loss_mega_kernel(
       vl += huber(mo[si * D + A] - s_vs[i]);
       es += s_ent[si];
       rs += s_rho[i];
+      for (int j = 0; j < A; ++j) {
+        const float z = mo[si * D + j];
+        rg = fmaf(z, z, rg);
+      }
     }
-    __shared__ float red[4][kMegaThreads];
+    __shared__ float red[5][kMegaThreads];
     red[0][tid] = pl; red[1][tid] = vl; red[2][tid] = es; red[3][tid] = rs;
+    red[4][tid] = rg;
     __syncthreads();
     for (int off = kMegaThreads / 2; off > 0; off >>= 1) {
       if (tid < off)
-        for (int r = 0; r < 4; ++r) red[r][tid] += red[r][tid + off];
+        for (int r = 0; r < 5; ++r) red[r][tid] += red[r][tid + off];
       __syncthreads();
     }
     if (tid == 0) {
       const float inv = 1.0f / BT;
       const float p = red[0][0] * inv, v = red[1][0] * inv, e = red[2][0] * inv;
-      stats[0] = cp * p + cv * v - ce * e;
+      stats[0] = cp * p + cv * v - ce * e + creg * red[4][0] * inv / A;
       stats[1] = p; stats[2] = v; stats[3] = e; stats[4] = red[3][0] * inv;
     }
   }
 
   // phase D: analytic backward into packed gouts
   const float invN = 1.0f / BT;
+  const float dreg = 2.0f * creg * invN / A;
   for (int i = tid; i < N; i += kMegaThreads) {
     const int t = i % S, b = i / S;
     float* g = gouts + (long)i * D;
@@ -496,7 +533,8 @@ __global__ __launch_bounds__(kMegaThreads) void impala_loss_mega_kernel(
     for (int j = 0; j < A; ++j) {
       const float lp = z[j] - s_lse[i];
       const float pj = __expf(lp);
-      g[j] = dlogp * ((j == a ? 1.f : 0.f) - pj) + dH * (-pj * (lp + H));
+      g[j] = dlogp * ((j == a ? 1.f : 0.f) - pj) + dH * (-pj * (lp + H)) +
+             dreg * z[j];
     }
     g[A] = cv * huber_grad(z[A] - s_vs[ti]) * invN;
   }
@@ -512,7 +550,7 @@ __global__ __launch_bounds__(kMegaThreads) void ppo_loss_mega_kernel(
     float* __restrict__ stats,        // (7)
     float* __restrict__ norm_sq,      // optional: zeroed here for the wgrads
     int B, int S, int A, float gamma, float lmbda, float rew_scale, float cp,
-    float cv, float ce, float eps_clip) {
+    float cv, float ce, float eps_clip, float creg) {
   const int D = A + 1;
   const int T = S - 1;
   const int N = B * S;
@@ -560,7 +598,7 @@ __global__ __launch_bounds__(kMegaThreads) void ppo_loss_mega_kernel(
   __syncthreads();
 
   {
-    float pl = 0, vl = 0, es = 0, ravg = 0;
+    float pl = 0, vl = 0, es = 0, ravg = 0, rg = 0;
     float rmin = 1e30f, rmax = -1e30f;
     for (int i = tid; i < BT; i += kMegaThreads) {
       const int b = i / T, t = i % T;
@@ -575,15 +613,20 @@ __global__ __launch_bounds__(kMegaThreads) void ppo_loss_mega_kernel(
       ravg += r;
       rmin = fminf(rmin, r);
       rmax = fmaxf(rmax, r);
+      for (int j = 0; j < A; ++j) {
+        const float z = mo[si * D + j];
+        rg = fmaf(z, z, rg);
+      }
     }
-    __shared__ float red[4][kMegaThreads];
+    __shared__ float red[5][kMegaThreads];
     __shared__ float rmn[kMegaThreads], rmx[kMegaThreads];
     red[0][tid] = pl; red[1][tid] = vl; red[2][tid] = es; red[3][tid] = ravg;
+    red[4][tid] = rg;
     rmn[tid] = rmin; rmx[tid] = rmax;
     __syncthreads();
     for (int off = kMegaThreads / 2; off > 0; off >>= 1) {
       if (tid < off) {
-        for (int r = 0; r < 4; ++r) red[r][tid] += red[r][tid + off];
+        for (int r = 0; r < 5; ++r) red[r][tid] += red[r][tid + off];
         rmn[tid] = fminf(rmn[tid], rmn[tid + off]);
         rmx[tid] = fmaxf(rmx[tid], rmx[tid + off]);
       }
@@ -592,13 +635,14 @@ __global__ __launch_bounds__(kMegaThreads) void ppo_loss_mega_kernel(
     if (tid == 0) {
       const float inv = 1.0f / BT;
       const float p = red[0][0] * inv, v = red[1][0] * inv, e = red[2][0] * inv;
-      stats[0] = cp * p + cv * v - ce * e;
+      stats[0] = cp * p + cv * v - ce * e + creg * red[4][0] * inv / A;
       stats[1] = p; stats[2] = v; stats[3] = e;
       stats[4] = red[3][0] * inv; stats[5] = rmn[0]; stats[6] = rmx[0];
     }
   }
 
   const float invN = 1.0f / BT;
+  const float dreg = 2.0f * creg * invN / A;
   for (int i = tid; i < N; i += kMegaThreads) {
     const int t = i % S, b = i / S;
     float* g = gouts + (long)i * D;
@@ -621,7 +665,8 @@ __global__ __launch_bounds__(kMegaThreads) void ppo_loss_mega_kernel(
     for (int j = 0; j < A; ++j) {
       const float lp = z[j] - s_lse[i];
       const float pj = __expf(lp);
-      g[j] = dlogp * ((j == a ? 1.f : 0.f) - pj) + dH * (-pj * (lp + H));
+      g[j] = dlogp * ((j == a ? 1.f : 0.f) - pj) + dH * (-pj * (lp + H)) +
+             dreg * z[j];
     }
     g[A] = cv * huber_grad(z[A] - s_td[ti]) * invN;
   }
@@ -637,7 +682,7 @@ bool impala_loss_mega_hip(const at::Tensor& mo, const at::Tensor& act,
                           const c10::optional<at::Tensor>& norm_sq, long A,
                           double gamma, double rho_bar, double rho_min,
                           double c_bar, double rew_scale, double cp, double cv,
-                          double ce) {
+                          double ce, double creg) {
   const int B = mo.size(0), S = mo.size(1);
   const int N = B * S, BT = B * (S - 1);
   const long lds = (3L * N + 3L * BT) * sizeof(float);
@@ -651,7 +696,7 @@ bool impala_loss_mega_hip(const at::Tensor& mo, const at::Tensor& act,
                      B, S,
                      (int)A, (float)gamma, (float)rho_bar, (float)rho_min,
                      (float)c_bar, (float)rew_scale, (float)cp, (float)cv,
-                     (float)ce);
+                     (float)ce, (float)creg);
   HIP_CHECK_LAST();
   return true;
 }
@@ -662,7 +707,7 @@ bool ppo_loss_mega_hip(const at::Tensor& mo, const at::Tensor& act,
                        at::Tensor& stats,
                        const c10::optional<at::Tensor>& norm_sq, long A,
                        double gamma, double lmbda, double rew_scale, double cp,
-                       double cv, double ce, double eps_clip) {
+                       double cv, double ce, double eps_clip, double creg) {
   const int B = mo.size(0), S = mo.size(1);
   const int N = B * S, BT = B * (S - 1);
   const long lds = (3L * N + 2L * BT) * sizeof(float);
@@ -675,7 +720,8 @@ bool ppo_loss_mega_hip(const at::Tensor& mo, const at::Tensor& act,
                      norm_sq.has_value() ? norm_sq->data_ptr<float>() : nullptr,
                      B, S,
                      (int)A, (float)gamma, (float)lmbda, (float)rew_scale,
-                     (float)cp, (float)cv, (float)ce, (float)eps_clip);
+                     (float)cp, (float)cv, (float)ce, (float)eps_clip,
+                     (float)creg);
   HIP_CHECK_LAST();
   return true;
 }

@@ -75,11 +75,12 @@ class FusedOnPolicyStep:
         when the shape fits a CU's LDS; 4-kernel sequence otherwise."""
         gouts = torch.empty_like(mo)
         norm = self._norm_buf()
+        creg = float(getattr(p, "logit_reg", 0.0))
         if self.algo == "IMPALA":
             if e.impala_loss_mega(
                 mo, act, behav, rew, fir, gouts, self.stats_buf, norm, A,
                 p.gamma, 0.8, 0.1, 1.0, p.reward_scale,
-                p.policy_loss_coef, p.value_loss_coef, p.entropy_coef,
+                p.policy_loss_coef, p.value_loss_coef, p.entropy_coef, creg,
             ):
                 return gouts
             if norm is not None:
@@ -94,16 +95,17 @@ class FusedOnPolicyStep:
             )
             e.impala_loss_reduce(
                 logp2, ent.view(B, S), mo, A, adv, vs, rhos, self.stats_buf,
-                p.policy_loss_coef, p.value_loss_coef, p.entropy_coef,
+                p.policy_loss_coef, p.value_loss_coef, p.entropy_coef, creg,
             )
             return e.impala_loss_bwd(
                 mo, A, act, lse, ent, adv, vs,
-                p.policy_loss_coef, p.value_loss_coef, p.entropy_coef,
+                p.policy_loss_coef, p.value_loss_coef, p.entropy_coef, creg,
             )
         if e.ppo_loss_mega(
             mo, act, behav, rew, fir, gouts, self.stats_buf, norm, A,
             p.gamma, p.lmbda, p.reward_scale,
             p.policy_loss_coef, p.value_loss_coef, p.entropy_coef, p.eps_clip,
+            creg,
         ):
             return gouts
         if norm is not None:
@@ -115,10 +117,12 @@ class FusedOnPolicyStep:
         e.ppo_loss_reduce(
             logp2, behav, ent.view(B, S), mo, A, adv, td, self.stats_buf,
             p.policy_loss_coef, p.value_loss_coef, p.entropy_coef, p.eps_clip,
+            creg,
         )
         return e.ppo_loss_bwd(
             mo, A, act, lse, ent, logp, behav.reshape(-1), adv, td,
             p.policy_loss_coef, p.value_loss_coef, p.entropy_coef, p.eps_clip,
+            creg,
         )
 
     def compute_grads_only(self, batch):
