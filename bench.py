@@ -102,6 +102,10 @@ def main():
     model = model_cls(obs_dim, n_act, params.seq_len, params.hidden_size)
     reducer = GradReducer() if world > 1 else None
     updater = updater_cls(model, params, device, grad_reducer=reducer)
+    if world == 1:  # eager-path algos: whole-step hipGraph capture
+        from pdrl_amd.ops.graphed import maybe_graph
+
+        updater = maybe_graph(updater, device)
 
     batch = make_synthetic_batch(params, device, seed=100 + rank)
 

@@ -119,6 +119,10 @@ class Learner:
         self.updater = updater_cls(model, params, self.device, grad_reducer=grad_reducer)
         if resume_path:
             self.updater.load(resume_path, map_location=self.device)
+        if world_size == 1:  # eager-path algos: whole-step hipGraph capture
+            from pdrl_amd.ops.graphed import maybe_graph
+
+            self.updater = maybe_graph(self.updater, self.device)
 
         self.on_policy = is_on_policy(params.algo)
         self.stager = BatchStager(self.device)

@@ -12,6 +12,9 @@ import torch
 import torch.nn.functional as F
 
 
+_soft_update_tables: dict = {}
+
+
 def _use_hip(*tensors) -> bool:
     if not all(t.is_cuda for t in tensors if isinstance(t, torch.Tensor)):
         return False
@@ -104,7 +107,21 @@ def soft_update(net: torch.nn.Module, target_net: torch.nn.Module, tau: float = 
     if params and _use_hip(*params, *tparams):
         from pdrl_amd import ops
 
-        ops.ext().soft_update([p.data for p in params], [t.data for t in tparams], float(tau))
+        key = (id(net), id(target_net))
+        tab = _soft_update_tables.get(key)
+        if tab is None or tab[3] != [p.data.data_ptr() for p in params]:
+            dev = params[0].device
+            sp = torch.tensor([p.data.data_ptr() for p in params],
+                              dtype=torch.int64).to(dev)
+            dp = torch.tensor([t.data.data_ptr() for t in tparams],
+                              dtype=torch.int64).to(dev)
+            ne = torch.tensor([p.numel() for p in params],
+                              dtype=torch.int64).to(dev)
+            mx = max(p.numel() for p in params)
+            tab = (sp, dp, ne, [p.data.data_ptr() for p in params], mx)
+            _soft_update_tables[key] = tab
+        ops.ext().soft_update_cached(tab[0], tab[1], tab[2], len(params),
+                                     tab[4], float(tau))
         return
     for p, tp in zip(params, tparams):
         tp.data.mul_(1.0 - tau).add_(p.data, alpha=tau)

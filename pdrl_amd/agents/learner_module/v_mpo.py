@@ -43,11 +43,13 @@ class VMPOUpdater(BaseUpdater):
     def extra_state(self):
         return {"log_eta": self.log_eta.detach().cpu(), "log_alpha": self.log_alpha.detach().cpu()}
 
-    def get_coef_alpha(self) -> float:
+    def get_coef_alpha(self):
         """KL-bound coefficient sampled uniformly in [below, upper]
-        (reference: learner.py:340-348)."""
+        (reference: learner.py:340-348). Sampled ON DEVICE so a hipGraph
+        replay of the step keeps re-sampling (host RNG would freeze)."""
         p = self.params
-        return float(self._rng.uniform(p.coef_alpha_below, p.coef_alpha_upper))
+        u = torch.rand((), device=self.device)
+        return p.coef_alpha_below + (p.coef_alpha_upper - p.coef_alpha_below) * u
 
     def compute_losses(self, batch: dict[str, torch.Tensor]):
         p = self.params

@@ -67,6 +67,8 @@ void adam_step_hip(at::Tensor&, const at::Tensor&, at::Tensor&, at::Tensor&,
                    double, double);
 void soft_update_hip(const std::vector<at::Tensor>&,
                      const std::vector<at::Tensor>&, double);
+void soft_update_cached_hip(const at::Tensor&, const at::Tensor&,
+                            const at::Tensor&, long, long, double);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("seq_lstm_forward", &seq_lstm_forward_hip,
@@ -99,4 +101,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsprop_step", &rmsprop_step_hip, "fused clip+RMSprop on flat buffers");
   m.def("adam_step", &adam_step_hip, "fused clip+Adam on flat buffers");
   m.def("soft_update", &soft_update_hip, "multi-tensor Polyak update");
+  m.def("soft_update_cached", &soft_update_cached_hip,
+        "Polyak update with prebuilt device pointer tables (graph-safe)");
 }

@@ -152,6 +152,20 @@ void adam_step_hip(at::Tensor& p, const at::Tensor& g, at::Tensor& m,
   HIP_CHECK_LAST();
 }
 
+void soft_update_cached_hip(const at::Tensor& src_ptrs,
+                            const at::Tensor& dst_ptrs,
+                            const at::Tensor& numels, long n, long max_numel,
+                            double tau) {
+  // pointer tables prebuilt ON DEVICE by the caller (graph-capture safe:
+  // no host→device transfer inside the step)
+  dim3 grid(grid_for(max_numel), (int)n);
+  hipLaunchKernelGGL(soft_update_kernel, grid, dim3(kThreads), 0,
+                     current_stream(), src_ptrs.data_ptr<long>(),
+                     dst_ptrs.data_ptr<long>(), numels.data_ptr<long>(),
+                     (int)n, (float)tau);
+  HIP_CHECK_LAST();
+}
+
 void soft_update_hip(const std::vector<at::Tensor>& src,
                      const std::vector<at::Tensor>& dst, double tau) {
   TORCH_CHECK(src.size() == dst.size(), "src/dst count mismatch");
