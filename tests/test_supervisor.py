@@ -35,7 +35,7 @@ def test_supervisor_restarts_dead_process():
 
     t = threading.Thread(target=sup.monitor, kwargs={"poll_s": 0.2}, daemon=True)
     t.start()
-    deadline = time.time() + 15
+    deadline = time.time() + 40
     while time.time() < deadline:
         if sup.specs["dier"]["proc"] is not first:
             break
@@ -75,7 +75,7 @@ def test_supervisor_restarts_stale_heartbeat(monkeypatch):
 
     t = threading.Thread(target=sup.monitor, kwargs={"poll_s": 0.2}, daemon=True)
     t.start()
-    deadline = time.time() + 15
+    deadline = time.time() + 40
     while time.time() < deadline:
         if sup.specs["stale"]["proc"] is not first:
             break
