@@ -195,6 +195,8 @@ class Supervisor:
             time.sleep(poll_s)
             now = time.time()
             for name, spec in self.specs.items():
+                if self.stop_event.is_set():
+                    return
                 p = spec["proc"]
                 hb = spec["heartbeat"]
                 stale = hb is not None and hb.value > 0 and (now - hb.value) > HEARTBEAT_TIMEOUT_S
@@ -206,20 +208,27 @@ class Supervisor:
                         p.join(5.0)
                     if hb is not None:
                         hb.value = time.time()
-                    spec["proc"] = mp.Process(
+                    fresh = mp.Process(
                         target=spec["target"], args=spec["args"], daemon=True, name=name
                     )
-                    spec["proc"].start()
+                    fresh.start()  # start BEFORE publishing (shutdown may join it)
+                    spec["proc"] = fresh
 
     def shutdown(self):
         self.stop_event.set()
         for spec in self.specs.values():
-            if spec["proc"].is_alive():
-                spec["proc"].terminate()
+            try:
+                if spec["proc"].is_alive():
+                    spec["proc"].terminate()
+            except (AssertionError, ValueError):
+                pass
         for spec in self.specs.values():
-            spec["proc"].join(5.0)
-            if spec["proc"].is_alive():
-                spec["proc"].kill()
+            try:
+                spec["proc"].join(5.0)
+                if spec["proc"].is_alive():
+                    spec["proc"].kill()
+            except (AssertionError, ValueError):
+                pass  # a respawn raced shutdown; daemon procs die with us
 
 
 # --------------------------------------------------------------------------- #
