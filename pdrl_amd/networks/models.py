@@ -198,7 +198,11 @@ class MlpLSTMContinuous(nn.Module):
         )
 
     def _dist(self, mu, std):
-        return torch.distributions.Normal(torch.tanh(mu), F.softplus(std) + 1e-4)
+        # validate_args=False: the validation reduction is a host sync, which
+        # both costs latency and forbids hipGraph capture of the step
+        return torch.distributions.Normal(
+            torch.tanh(mu), F.softplus(std) + 1e-4, validate_args=False
+        )
 
     @torch.no_grad()
     def act(self, obs, lstm_hxs):
@@ -265,7 +269,7 @@ class MlpLSTMActorContinuous(nn.Module):
     def _sample(self, mu, log_std, reparam: bool):
         log_std = torch.clamp(log_std, LOG_STD_MIN, LOG_STD_MAX)
         std = log_std.exp()
-        dist = torch.distributions.Normal(mu, std)
+        dist = torch.distributions.Normal(mu, std, validate_args=False)
         z = dist.rsample() if reparam else dist.sample()
         action = torch.tanh(z)
         log_prob = dist.log_prob(z) - torch.log(1.0 - action.pow(2) + 1e-7)
