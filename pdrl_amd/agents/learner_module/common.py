@@ -55,7 +55,7 @@ class BaseUpdater:
             return torch.optim.RMSprop(parameters, lr=lr, eps=kw.get("eps", 1e-5))
         return torch.optim.Adam(parameters, lr=lr)
 
-    def make_fused_step(self, algo: str, model, optimizer):
+    def make_fused_step(self, algo: str, model, optimizer, duals=None):
         """Whole-step fused HIP DAG (IMPALA/PPO, discrete policies) when the
         extension is loaded on GPU; None → the eager-autograd path runs."""
         from pdrl_amd import ops
@@ -79,7 +79,7 @@ class BaseUpdater:
         )
         return FusedOnPolicyStep(algo, core, self.params, optimizer,
                                  grad_reducer=self.grad_reducer,
-                                 use_graph=use_graph)
+                                 use_graph=use_graph, duals=duals)
 
     def apply_step(self, optimizer, parameters):
         """Gradient epilogue: all-reduce across ranks → clip → update.

@@ -32,6 +32,10 @@ class VMPOUpdater(BaseUpdater):
             lr=params.lr,
             eps=1e-5,
         )
+        self.fused_step = self.make_fused_step(
+            "V-MPO", self.model, self.optimizer,
+            duals=(self.log_eta, self.log_alpha),
+        )
         self._rng = np.random.default_rng()
 
     def trainable_modules(self):
@@ -110,6 +114,10 @@ class VMPOUpdater(BaseUpdater):
         return loss, stats
 
     def step(self, batch: dict[str, torch.Tensor]) -> dict:
+        if self.fused_step is not None and self.fused_step.fits(batch):
+            stats = self.fused_step.run(batch)
+            self.update_count += 1
+            return stats
         stats = {}
         for _ in range(self.params.K_epoch):
             loss, stats = self.compute_losses(batch)

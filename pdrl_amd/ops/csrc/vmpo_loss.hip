@@ -1,0 +1,343 @@
+// Fused V-MPO loss for CDNA4 (gfx950) — K10 of SURVEY.md §2.4.
+//
+// One single-block launch computes the ENTIRE V-MPO loss and its analytic
+// backward (reference math: agents/learner_module/v_mpo/learning.py:49-124):
+//   A. log-softmax stats per (b,t)
+//   B. GAE advantages + TD targets (per-row scan)
+//   C. top-half advantage selection: exact k-th-largest via monotonic
+//      float-bit binary search (32 count-reduce rounds over BT ≤ 2048)
+//   D. psi-softmax policy loss, eta temperature dual, alpha KL dual,
+//      smooth-L1 value loss, logit L2 — reduced into the stats vector
+//   E. analytic gradients: packed head grads gouts = [dlogits | dvalue],
+//      plus dlog_eta / dlog_alpha written to their flat-grad views.
+//
+// The KL-bound coefficient eps_alpha is sampled IN-KERNEL from an LCG whose
+// state lives in device memory (advances on every launch — graph-replay
+// keeps re-sampling, matching the reference's per-iteration host sampling,
+// learner.py:340-348).
+#include "common.h"
+
+namespace {
+
+__device__ __forceinline__ float huber_v(float d) {
+  const float a = fabsf(d);
+  return (a < 1.0f) ? 0.5f * d * d : a - 0.5f;
+}
+__device__ __forceinline__ float huber_grad_v(float d) {
+  return fminf(fmaxf(d, -1.0f), 1.0f);
+}
+
+// monotonic float<->uint mapping for order-preserving bit binary search
+__device__ __forceinline__ unsigned f2u(float f) {
+  unsigned u = __float_as_uint(f);
+  return (u & 0x80000000u) ? ~u : (u | 0x80000000u);
+}
+
+constexpr int kThreads = 256;
+
+__global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
+    const float* __restrict__ mo,      // (N,D) packed model out
+    const float* __restrict__ act,     // (N)
+    const float* __restrict__ behav,   // (N,A) behaviour logits
+    const float* __restrict__ rew,     // (B,S)
+    const float* __restrict__ fir,     // (B,S)
+    const float* __restrict__ log_eta_p,    // (1) current log_eta value
+    const float* __restrict__ log_alpha_p,  // (1) current log_alpha value
+    float* __restrict__ gouts,         // (B,S,D)
+    float* __restrict__ g_eta,         // (1) dloss/dlog_eta
+    float* __restrict__ g_alpha,       // (1) dloss/dlog_alpha
+    float* __restrict__ stats,         // {total, policy, value, eta, alpha, kl}
+    float* __restrict__ norm_sq,       // optional: zeroed + eta/alpha part
+    unsigned* __restrict__ rng_state,  // LCG state for eps_alpha
+    int B, int S, int A, float gamma, float lmbda, float rew_scale, float cp,
+    float cv, float creg, float eps_eta, float alpha_below, float alpha_upper) {
+  const int D = A + 1;
+  const int T = S - 1;
+  const int N = B * S;
+  const int BT = B * T;
+  const int tid = threadIdx.x;
+
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* s_lse = reinterpret_cast<float*>(smem_raw);  // (N)
+  float* s_logp = s_lse + N;                          // (N) log pi(a)
+  float* s_adv = s_logp + N;                          // (BT)
+  float* s_td = s_adv + BT;                           // (BT)
+  float* s_psi = s_td + BT;                           // (BT) psi or 0
+  __shared__ float red[4][kThreads];
+  __shared__ unsigned ured[kThreads];
+  __shared__ float s_scalars[8];  // {eps_alpha, thresh, m, Z, sumwa, kl, eta, alpha}
+
+  if (tid == 0) {
+    if (norm_sq != nullptr) *norm_sq = 0.f;
+    // LCG advance (Numerical Recipes constants); uniform in [below, upper]
+    unsigned st = *rng_state * 1664525u + 1013904223u;
+    *rng_state = st;
+    const float u = (st >> 8) * (1.0f / 16777216.0f);
+    s_scalars[0] = alpha_below + (alpha_upper - alpha_below) * u;
+    s_scalars[6] = __expf(*log_eta_p);
+    s_scalars[7] = __expf(*log_alpha_p);
+  }
+
+  // phase A: log-softmax stats
+  for (int i = tid; i < N; i += kThreads) {
+    const float* z = mo + (long)i * D;
+    float mx = z[0];
+    for (int j = 1; j < A; ++j) mx = fmaxf(mx, z[j]);
+    float sum = 0.f;
+    for (int j = 0; j < A; ++j) sum += __expf(z[j] - mx);
+    const float l = mx + __logf(sum);
+    s_lse[i] = l;
+    s_logp[i] = z[(int)act[i]] - l;
+  }
+  __syncthreads();
+
+  // phase B: GAE scan per batch row (value = col A of mo)
+  for (int b = tid; b < B; b += kThreads) {
+    const long sb = (long)b * S, tb = (long)b * T;
+    float run = 0.f;
+    for (int t = T - 1; t >= 0; --t) {
+      const float mask = 1.f - fir[sb + t + 1];
+      const float tdv = rew[sb + t] * rew_scale +
+                        gamma * mask * mo[(sb + t + 1) * D + A];
+      const float delta = tdv - mo[(sb + t) * D + A];
+      run = fmaf(gamma * lmbda * mask, run, delta);
+      s_td[tb + t] = tdv;
+      s_adv[tb + t] = run;
+    }
+  }
+  __syncthreads();
+
+  // phase C: k-th largest advantage via bit binary search (k = BT/2)
+  const int K = BT / 2 > 0 ? BT / 2 : 1;
+  {
+    unsigned lo = 0, hi = 0xFFFFFFFFu;
+    for (int it = 0; it < 32; ++it) {
+      const unsigned mid = lo + ((hi - lo) >> 1);
+      int cnt = 0;
+      for (int i = tid; i < BT; i += kThreads) {
+        if (f2u(s_adv[i]) > mid) ++cnt;
+      }
+      red[0][tid] = (float)cnt;
+      __syncthreads();
+      for (int off = kThreads / 2; off > 0; off >>= 1) {
+        if (tid < off) red[0][tid] += red[0][tid + off];
+        __syncthreads();
+      }
+      const int total_gt = (int)red[0][0];
+      __syncthreads();
+      if (total_gt >= K) {
+        lo = mid + 1;  // threshold is higher
+      } else {
+        hi = mid;
+      }
+      if (lo >= hi) break;
+    }
+    // hi = smallest u such that count(> u) < K → the K-th largest has bits hi
+    if (tid == 0) s_scalars[1] = __uint_as_float((hi & 0x80000000u) ? (hi & 0x7FFFFFFFu) : ~hi);
+    __syncthreads();
+  }
+  const float thresh = s_scalars[1];
+  const unsigned thresh_u = f2u(thresh);
+
+  // mark selected: strictly greater always; equal by ascending index to k
+  {
+    int cnt_gt = 0;
+    for (int i = tid; i < BT; i += kThreads)
+      if (f2u(s_adv[i]) > thresh_u) ++cnt_gt;
+    red[0][tid] = (float)cnt_gt;
+    __syncthreads();
+    for (int off = kThreads / 2; off > 0; off >>= 1) {
+      if (tid < off) red[0][tid] += red[0][tid + off];
+      __syncthreads();
+    }
+    const int n_gt = (int)red[0][0];
+    __syncthreads();
+    int need_eq = K - n_gt;  // ties to include, by lowest index
+    // serial-ish tie resolution (ties are rare): thread 0 marks
+    for (int i = tid; i < BT; i += kThreads) {
+      s_psi[i] = (f2u(s_adv[i]) > thresh_u) ? 1.f : 0.f;
+    }
+    __syncthreads();
+    if (tid == 0 && need_eq > 0) {
+      for (int i = 0; i < BT && need_eq > 0; ++i) {
+        if (f2u(s_adv[i]) == thresh_u) {
+          s_psi[i] = 1.f;
+          --need_eq;
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // phase C': psi softmax over selected (max = global max of selected advs)
+  {
+    float mx = -1e30f;
+    for (int i = tid; i < BT; i += kThreads)
+      if (s_psi[i] > 0.f) mx = fmaxf(mx, s_adv[i]);
+    red[0][tid] = mx;
+    __syncthreads();
+    for (int off = kThreads / 2; off > 0; off >>= 1) {
+      if (tid < off) red[0][tid] = fmaxf(red[0][tid], red[0][tid + off]);
+      __syncthreads();
+    }
+    if (tid == 0) s_scalars[2] = red[0][0];
+    __syncthreads();
+    const float m = s_scalars[2];
+    const float eta = s_scalars[6];
+    float z = 0.f, wa = 0.f;
+    for (int i = tid; i < BT; i += kThreads) {
+      if (s_psi[i] > 0.f) {
+        const float e = __expf((s_adv[i] - m) / eta);
+        z += e;
+        wa = fmaf(e, s_adv[i], wa);
+      }
+    }
+    red[0][tid] = z; red[1][tid] = wa;
+    __syncthreads();
+    for (int off = kThreads / 2; off > 0; off >>= 1) {
+      if (tid < off) {
+        red[0][tid] += red[0][tid + off];
+        red[1][tid] += red[1][tid + off];
+      }
+      __syncthreads();
+    }
+    if (tid == 0) { s_scalars[3] = red[0][0]; s_scalars[4] = red[1][0]; }
+    __syncthreads();
+    const float Z = s_scalars[3];
+    for (int i = tid; i < BT; i += kThreads) {
+      if (s_psi[i] > 0.f)
+        s_psi[i] = __expf((s_adv[i] - m) / eta) / Z;
+    }
+  }
+  __syncthreads();
+
+  // phase D: reductions — policy, value, KL, logit reg
+  {
+    const float eta = s_scalars[6];
+    float pl = 0.f, vl = 0.f, kl = 0.f, rg = 0.f;
+    for (int i = tid; i < BT; i += kThreads) {
+      const int b = i / T, t = i % T;
+      const long si = (long)b * S + t;
+      pl -= s_psi[i] * s_logp[si];
+      vl += huber_v(mo[si * D + A] - s_td[i]);
+      // KL(behav || target) per element
+      const float* zb = behav + si * A;
+      const float* zq = mo + si * D;
+      float mb = zb[0];
+      for (int j = 1; j < A; ++j) mb = fmaxf(mb, zb[j]);
+      float sb_ = 0.f;
+      for (int j = 0; j < A; ++j) sb_ += __expf(zb[j] - mb);
+      const float lb = mb + __logf(sb_);
+      for (int j = 0; j < A; ++j) {
+        const float lpb = zb[j] - lb;
+        const float lpq = zq[j] - s_lse[si];
+        kl += __expf(lpb) * (lpb - lpq);
+        rg = fmaf(zq[j], zq[j], rg);
+      }
+    }
+    red[0][tid] = pl; red[1][tid] = vl; red[2][tid] = kl; red[3][tid] = rg;
+    __syncthreads();
+    for (int off = kThreads / 2; off > 0; off >>= 1) {
+      if (tid < off)
+        for (int r = 0; r < 4; ++r) red[r][tid] += red[r][tid + off];
+      __syncthreads();
+    }
+    if (tid == 0) {
+      const float eta_v = s_scalars[6], alpha_v = s_scalars[7];
+      const float m = s_scalars[2], Z = s_scalars[3], wa = s_scalars[4];
+      const float eps_alpha = s_scalars[0];
+      const float kl_mean = red[2][0] / BT;
+      s_scalars[5] = kl_mean;
+      const float pl_v = red[0][0];
+      const float vl_v = red[1][0] / BT;
+      const float lse_sel = m / eta_v + __logf(Z);  // logsumexp(adv/eta)
+      const float eta_loss =
+          eta_v * eps_eta + eta_v * (lse_sel - __logf((float)K));
+      const float alpha_loss = alpha_v * (eps_alpha - kl_mean) + alpha_v * kl_mean;
+      // note: alpha*(eps-kl.detach()) + alpha.detach()*kl has VALUE
+      // alpha*eps (the kl terms cancel) but distinct gradients
+      const float reg_v = creg * red[3][0] / (BT * A);
+      stats[0] = cp * pl_v + cv * vl_v + eta_loss + (alpha_v * eps_alpha) + reg_v;
+      stats[1] = pl_v;
+      stats[2] = vl_v;
+      stats[3] = eta_v;
+      stats[4] = alpha_v;
+      stats[5] = kl_mean;
+      // duals: dlog_eta, dlog_alpha (derived + verified vs autograd)
+      const float deta = eps_eta + (lse_sel - __logf((float)K)) -
+                         (wa / Z) / eta_v;
+      g_eta[0] = deta * eta_v;
+      g_alpha[0] = alpha_v * (eps_alpha - kl_mean);
+      if (norm_sq != nullptr) {
+        atomicAdd(norm_sq, g_eta[0] * g_eta[0] + g_alpha[0] * g_alpha[0]);
+      }
+    }
+  }
+  __syncthreads();
+
+  // phase E: packed head grads
+  {
+    const float alpha_v = s_scalars[7];
+    const float invBT = 1.0f / BT;
+    const float dreg = 2.0f * creg * invBT / A;
+    for (int i = tid; i < N; i += kThreads) {
+      const int t = i % S, b = i / S;
+      float* g = gouts + (long)i * D;
+      if (t >= T) {
+        for (int j = 0; j < D; ++j) g[j] = 0.f;
+        continue;
+      }
+      const long ti = (long)b * T + t;
+      const float dlogp = -cp * s_psi[ti];  // zero for unselected
+      // q = softmax(logits), p_beh = softmax(behav)
+      const float* zq = mo + (long)i * D;
+      const float* zb = behav + (long)i * A;
+      float mb = zb[0];
+      for (int j = 1; j < A; ++j) mb = fmaxf(mb, zb[j]);
+      float sb_ = 0.f;
+      for (int j = 0; j < A; ++j) sb_ += __expf(zb[j] - mb);
+      const float lb = mb + __logf(sb_);
+      const int a = (int)act[i];
+      for (int j = 0; j < A; ++j) {
+        const float q = __expf(zq[j] - s_lse[i]);
+        const float pb = __expf(zb[j] - lb);
+        g[j] = dlogp * ((j == a ? 1.f : 0.f) - q)   // psi-weighted policy
+               + alpha_v * (q - pb) * invBT          // KL(P_beh||Q) wrt z_q
+               + dreg * zq[j];                       // logit L2
+      }
+      g[A] = cv * huber_grad_v(zq[A] - s_td[ti]) * invBT;
+    }
+  }
+}
+
+}  // namespace
+
+bool vmpo_loss_mega_hip(const at::Tensor& mo, const at::Tensor& act,
+                        const at::Tensor& behav, const at::Tensor& rew,
+                        const at::Tensor& fir, const at::Tensor& log_eta,
+                        const at::Tensor& log_alpha, at::Tensor& gouts,
+                        at::Tensor& g_eta, at::Tensor& g_alpha,
+                        at::Tensor& stats,
+                        const c10::optional<at::Tensor>& norm_sq,
+                        at::Tensor& rng_state, long A, double gamma,
+                        double lmbda, double rew_scale, double cp, double cv,
+                        double creg, double eps_eta, double alpha_below,
+                        double alpha_upper) {
+  const int B = mo.size(0), S = mo.size(1);
+  const int N = B * S, BT = B * (S - 1);
+  const long lds = (2L * N + 3L * BT) * sizeof(float);
+  if (lds > 56 * 1024) return false;
+  hipLaunchKernelGGL(
+      vmpo_loss_mega_kernel, dim3(1), dim3(256), lds, current_stream(),
+      mo.data_ptr<float>(), act.data_ptr<float>(), behav.data_ptr<float>(),
+      rew.data_ptr<float>(), fir.data_ptr<float>(),
+      log_eta.data_ptr<float>(), log_alpha.data_ptr<float>(),
+      gouts.data_ptr<float>(), g_eta.data_ptr<float>(),
+      g_alpha.data_ptr<float>(), stats.data_ptr<float>(),
+      norm_sq.has_value() ? norm_sq->data_ptr<float>() : nullptr,
+      (unsigned*)rng_state.data_ptr<int>(), B, S, (int)A, (float)gamma,
+      (float)lmbda, (float)rew_scale, (float)cp, (float)cv, (float)creg,
+      (float)eps_eta, (float)alpha_below, (float)alpha_upper);
+  HIP_CHECK_LAST();
+  return true;
+}

@@ -32,14 +32,15 @@ _PPO_STATS = [
     "loss-total", "loss-policy", "loss-value", "entropy",
     "ratio-avg", "ratio-min", "ratio-max",
 ]
+_VMPO_STATS = ["loss-total", "loss-policy", "loss-value", "eta", "alpha", "kl"]
 
 BATCH_FIELDS = ["obs", "act", "rew", "logits", "log_prob", "is_fir", "hx", "cx"]
 
 
 class FusedOnPolicyStep:
     def __init__(self, algo: str, core, params, optimizer, grad_reducer=None,
-                 use_graph: bool = True):
-        assert algo in ("IMPALA", "PPO")
+                 use_graph: bool = True, duals=None):
+        assert algo in ("IMPALA", "PPO", "V-MPO")
         self.algo = algo
         self.core = core
         self.params = params
@@ -48,7 +49,13 @@ class FusedOnPolicyStep:
         self.A = int(params.n_actions)
         dev = core.body_w.device
         self.stats_buf = torch.zeros(8, dtype=torch.float32, device=dev)
-        self.stat_names = _IMPALA_STATS if algo == "IMPALA" else _PPO_STATS
+        self.stat_names = {"IMPALA": _IMPALA_STATS, "PPO": _PPO_STATS,
+                           "V-MPO": _VMPO_STATS}[algo]
+        self.duals = duals  # (log_eta, log_alpha) params for V-MPO
+        if algo == "V-MPO":
+            assert duals is not None
+            self.rng_state = torch.randint(
+                1, 2**31 - 1, (1,), dtype=torch.int32, device=dev)
         self.use_graph = use_graph
         self._graph = None
         self._static: dict[str, torch.Tensor] | None = None
@@ -70,12 +77,35 @@ class FusedOnPolicyStep:
         # wgrad_out writes (dw_ih, dw_hh, dbody_w, dbody_b, db_g, dheads_w, dheads_b)
         return [gs[2], gs[3], gs[0], gs[1], gs[4], gs[5], gs[6]]
 
+    def fits(self, batch) -> bool:
+        """Whether the single-launch loss kernel covers this shape (V-MPO
+        has no multi-kernel fallback; IMPALA/PPO do)."""
+        if self.algo != "V-MPO":
+            return True
+        B, S, _ = batch["obs"].shape
+        return (2 * B * S + 3 * B * (S - 1)) * 4 <= 56 * 1024
+
     def _loss(self, e, mo, act, behav, rew, fir, B, S, A, p):
         """Loss stats + analytic head-grad buffer. One mega-kernel launch
         when the shape fits a CU's LDS; 4-kernel sequence otherwise."""
         gouts = torch.empty_like(mo)
         norm = self._norm_buf()
         creg = float(getattr(p, "logit_reg", 0.0))
+        if self.algo == "V-MPO":
+            log_eta, log_alpha = self.duals
+            ok = e.vmpo_loss_mega(
+                mo, act, self._behav_logits, rew, fir,
+                log_eta.data.view(1), log_alpha.data.view(1), gouts,
+                log_eta.grad.view(1), log_alpha.grad.view(1), self.stats_buf,
+                norm, self.rng_state, A, p.gamma, p.lmbda, p.reward_scale,
+                p.policy_loss_coef, p.value_loss_coef, creg, p.coef_eta,
+                p.coef_alpha_below, p.coef_alpha_upper,
+            )
+            assert ok, "V-MPO shape exceeds the fused-loss LDS budget"
+            if norm is None:
+                # multi-rank: eta/alpha grads are in the flat bucket already
+                pass
+            return gouts
         if self.algo == "IMPALA":
             if e.impala_loss_mega(
                 mo, act, behav, rew, fir, gouts, self.stats_buf, norm, A,
@@ -143,6 +173,8 @@ class FusedOnPolicyStep:
         rew = batch["rew"].reshape(B, S)
         behav = batch["log_prob"].reshape(B, S)
         fir = batch["is_fir"].reshape(B, S)
+        if self.algo == "V-MPO":
+            self._behav_logits = batch["logits"].reshape(B * S, self.A)
 
         mo, hS, cS, stash = e.seq_lstm_forward(
             x, hx0, cx0, c.body_w, c.body_b, c.w_ih, c.w_hh, c.b_g,
