@@ -187,7 +187,14 @@ class Learner:
         return None
 
     def publish_model(self):
+        """Broadcast actor weights. The reference publishes after EVERY
+        update (ppo/learning.py:108) — kept as the default; at multi-kHz
+        update rates ``model_publish_interval`` bounds the encode cost
+        (SURVEY.md §7 hard part (e))."""
         if self.pub is None:
+            return
+        interval = int(getattr(self.params, "model_publish_interval", 1) or 1)
+        if interval > 1 and self.updater.update_count % interval != 0:
             return
         header, payload = encode(Protocol.Model, self.updater.actor_state_dict())
         self.pub.send(header, payload)

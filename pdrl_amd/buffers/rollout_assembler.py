@@ -56,15 +56,23 @@ class RolloutAssembler:
         self.active: dict[str, Trajectory] = {}  # partial, still-running episodes
         self.parked_done: dict[str, Trajectory] = {}  # finished but short episodes
         self.out_queue = out_queue if out_queue is not None else asyncio.Queue(1024)
+        self._next_evict = 0.0  # staleness sweep is amortized (time-based)
 
     async def push(self, step: dict):
         missing = REQUIRED_KEYS - set(step)
         assert not missing, f"rollout step missing fields: {missing}"
         eid = step["id"]
 
-        # staleness eviction of partial trajectories
-        for k in [k for k, tr in self.active.items() if tr.age > self.stale_s]:
-            del self.active[k]
+        # staleness eviction of partial trajectories (swept at most every
+        # stale_s/4 — a per-push scan of the active dict measurably taxed
+        # the ingest loop at tens of kHz push rates)
+        import time as _time
+
+        now = _time.monotonic()
+        if now >= self._next_evict:
+            self._next_evict = now + self.stale_s / 4
+            for k in [k for k, tr in self.active.items() if tr.age > self.stale_s]:
+                del self.active[k]
 
         if eid in self.active:
             traj = self.active[eid]

@@ -10,6 +10,7 @@ from .utils import (  # noqa: F401
     load_machines,
     refresh_result_dirs,
     to_torch,
+    flatten,
     mul,
     make_gpu_batch,
     obs_preprocess,

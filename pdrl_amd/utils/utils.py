@@ -139,6 +139,17 @@ def to_torch(x: np.ndarray) -> torch.Tensor:
     return torch.from_numpy(np.ascontiguousarray(x)).float()
 
 
+def flatten(nested) -> list:
+    """Flatten arbitrarily nested lists/tuples (reference: utils.py:87-91)."""
+    out = []
+    for item in nested:
+        if isinstance(item, (list, tuple)):
+            out.extend(flatten(item))
+        else:
+            out.append(item)
+    return out
+
+
 def mul(shape) -> int:
     out = 1
     for s in shape:
