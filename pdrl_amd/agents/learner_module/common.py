@@ -66,8 +66,12 @@ class BaseUpdater:
             return None
         actor = getattr(model, "actor", model)
         core = getattr(actor, "core", None)
-        if core is None or core.head_names != ["logits", "value"]:
-            return None  # continuous-policy fused step not implemented yet
+        if core is None:
+            return None
+        if core.head_names == ["mu", "std", "value"] and algo == "PPO":
+            algo = "PPO-C"  # Gaussian-policy fused loss (K5)
+        elif core.head_names != ["logits", "value"]:
+            return None
         from pdrl_amd.ops.fused_step import FusedOnPolicyStep
 
         # graph replay is single-rank only for now: capturing the RCCL

@@ -77,7 +77,7 @@ class PPOUpdater(BaseUpdater):
         return loss, stats
 
     def step(self, batch: dict[str, torch.Tensor]) -> dict:
-        if self.fused_step is not None:
+        if self.fused_step is not None and self.fused_step.fits(batch):
             stats = self.fused_step.run(batch)
             self.update_count += 1
             return stats

@@ -67,6 +67,12 @@ bool vmpo_loss_mega_hip(const at::Tensor&, const at::Tensor&,
                         const c10::optional<at::Tensor>&, at::Tensor&, long,
                         double, double, double, double, double, double,
                         double, double, double);
+bool ppoc_loss_mega_hip(const at::Tensor&, const at::Tensor&,
+                        const at::Tensor&, const at::Tensor&,
+                        const at::Tensor&, at::Tensor&, at::Tensor&,
+                        const c10::optional<at::Tensor>&, long, double,
+                        double, double, double, double, double, double,
+                        double);
 void l2norm_sq_hip(const at::Tensor&, at::Tensor&);
 void rmsprop_step_hip(at::Tensor&, const at::Tensor&, at::Tensor&,
                       const at::Tensor&, double, double, double, double);
@@ -107,6 +113,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "single-launch PPO loss: stats+gae+reduce+bwd");
   m.def("vmpo_loss_mega", &vmpo_loss_mega_hip,
         "single-launch V-MPO loss: stats+gae+topk+duals+bwd");
+  m.def("ppoc_loss_mega", &ppoc_loss_mega_hip,
+        "single-launch Gaussian-policy PPO loss (K5)");
   m.def("l2norm_sq", &l2norm_sq_hip, "squared L2 norm into a device scalar");
   m.def("rmsprop_step", &rmsprop_step_hip, "fused clip+RMSprop on flat buffers");
   m.def("adam_step", &adam_step_hip, "fused clip+Adam on flat buffers");

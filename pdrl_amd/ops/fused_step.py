@@ -40,7 +40,7 @@ BATCH_FIELDS = ["obs", "act", "rew", "logits", "log_prob", "is_fir", "hx", "cx"]
 class FusedOnPolicyStep:
     def __init__(self, algo: str, core, params, optimizer, grad_reducer=None,
                  use_graph: bool = True, duals=None):
-        assert algo in ("IMPALA", "PPO", "V-MPO")
+        assert algo in ("IMPALA", "PPO", "V-MPO", "PPO-C")
         self.algo = algo
         self.core = core
         self.params = params
@@ -50,7 +50,7 @@ class FusedOnPolicyStep:
         dev = core.body_w.device
         self.stats_buf = torch.zeros(8, dtype=torch.float32, device=dev)
         self.stat_names = {"IMPALA": _IMPALA_STATS, "PPO": _PPO_STATS,
-                           "V-MPO": _VMPO_STATS}[algo]
+                           "V-MPO": _VMPO_STATS, "PPO-C": _PPO_STATS}[algo]
         self.duals = duals  # (log_eta, log_alpha) params for V-MPO
         if algo == "V-MPO":
             assert duals is not None
@@ -80,10 +80,12 @@ class FusedOnPolicyStep:
     def fits(self, batch) -> bool:
         """Whether the single-launch loss kernel covers this shape (V-MPO
         has no multi-kernel fallback; IMPALA/PPO do)."""
-        if self.algo != "V-MPO":
-            return True
         B, S, _ = batch["obs"].shape
-        return (2 * B * S + 3 * B * (S - 1)) * 4 <= 56 * 1024
+        if self.algo == "V-MPO":
+            return (2 * B * S + 3 * B * (S - 1)) * 4 <= 56 * 1024
+        if self.algo == "PPO-C":
+            return (2 * B * S + 2 * B * (S - 1)) * 4 <= 56 * 1024
+        return True
 
     def _loss(self, e, mo, act, behav, rew, fir, B, S, A, p):
         """Loss stats + analytic head-grad buffer. One mega-kernel launch
@@ -105,6 +107,14 @@ class FusedOnPolicyStep:
             if norm is None:
                 # multi-rank: eta/alpha grads are in the flat bucket already
                 pass
+            return gouts
+        if self.algo == "PPO-C":
+            ok = e.ppoc_loss_mega(
+                mo, act, behav, rew, fir, gouts, self.stats_buf, norm, A,
+                p.gamma, p.lmbda, p.reward_scale, p.policy_loss_coef,
+                p.value_loss_coef, p.entropy_coef, p.eps_clip, creg,
+            )
+            assert ok, "PPO-C shape exceeds the fused-loss LDS budget"
             return gouts
         if self.algo == "IMPALA":
             if e.impala_loss_mega(

@@ -19,6 +19,7 @@ SRC = [
     "pdrl_amd/ops/csrc/wgrad.hip",
     "pdrl_amd/ops/csrc/losses.hip",
     "pdrl_amd/ops/csrc/vmpo_loss.hip",
+    "pdrl_amd/ops/csrc/ppoc_loss.hip",
     "pdrl_amd/ops/csrc/scans.hip",
     "pdrl_amd/ops/csrc/multi_tensor.hip",
 ]

@@ -339,3 +339,36 @@ def test_vmpo_fused_grad_parity_vs_autograd():
     upd.fused_step._body(batch, update=False)
     torch.testing.assert_close(upd.fused_step.stats_buf[0], loss.detach(),
                                rtol=1e-3, atol=1e-4)
+
+
+def test_ppo_continuous_fused_grad_parity():
+    """Fused Gaussian-policy PPO loss (K5) vs GPU eager autograd."""
+    _ops()
+    from pdrl_amd.agents.learner_module import PPOUpdater
+    from pdrl_amd.networks import MlpLSTMSingleContinuous
+    from pdrl_amd.utils import load_params
+    from tests.conftest import make_batch
+
+    p = load_params()
+    p.algo = "PPO-Continuous"
+    p.batch_size, p.seq_len, p.obs_dim, p.n_actions = 16, 5, 2, 1
+    torch.manual_seed(9)
+    model = MlpLSTMSingleContinuous(2, 1, p.seq_len, p.hidden_size)
+    upd = PPOUpdater(model, p, DEV)
+    assert upd.fused_step is not None and upd.fused_step.algo == "PPO-C"
+    batch = make_batch(p, n_actions=1, continuous=True, device=DEV, seed=41)
+    batch["obs"] = torch.randn(p.batch_size, p.seq_len, 2, device=DEV)
+    assert upd.fused_step.fits(batch)
+
+    upd.fused_step.compute_grads_only(batch)
+    fused = {n: q.grad.detach().clone() for n, q in model.named_parameters()}
+
+    upd.optimizer.zero_grad()
+    loss, _ = upd.compute_losses(batch)
+    loss.backward()
+    for n, q in model.named_parameters():
+        torch.testing.assert_close(fused[n], q.grad, rtol=2e-4, atol=2e-6,
+                                   msg=lambda m: f"{n}: {m}")
+    upd.fused_step._body(batch, update=False)
+    torch.testing.assert_close(upd.fused_step.stats_buf[0], loss.detach(),
+                               rtol=1e-3, atol=1e-4)
