@@ -14,7 +14,7 @@ MI355X-first differences:
   batch from the ring and all-reduces gradients over RCCL/xGMI via
   GradReducer (reference has a single-GPU learner only);
 * batch staging goes through a pinned-host buffer + async H2D copy
-  (ops/staging.py) instead of per-field blocking copies.
+  (BatchStager below) instead of per-field blocking copies.
 """
 from __future__ import annotations
 

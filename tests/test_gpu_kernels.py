@@ -372,3 +372,23 @@ def test_ppo_continuous_fused_grad_parity():
     upd.fused_step._body(batch, update=False)
     torch.testing.assert_close(upd.fused_step.stats_buf[0], loss.detach(),
                                rtol=1e-3, atol=1e-4)
+
+
+def test_batch_stager_roundtrip_gpu():
+    """Packed pinned-buffer staging delivers exact field contents with
+    STABLE device tensors across calls (zero-copy graph replay contract)."""
+    _ops()
+    from pdrl_amd.agents import BatchStager
+
+    st = BatchStager(DEV)
+    rng = np.random.default_rng(0)
+    b1 = {k: rng.standard_normal((4, 5, d)).astype(np.float32)
+          for k, d in [("obs", 4), ("rew", 1), ("hx", 64)]}
+    out1 = st.stage(b1)
+    for k in b1:
+        torch.testing.assert_close(out1[k].cpu(), torch.from_numpy(b1[k]))
+    b2 = {k: rng.standard_normal(v.shape).astype(np.float32) for k, v in b1.items()}
+    out2 = st.stage(b2)
+    for k in b2:
+        assert out2[k].data_ptr() == out1[k].data_ptr()  # stable views
+        torch.testing.assert_close(out2[k].cpu(), torch.from_numpy(b2[k]))
