@@ -9,6 +9,7 @@ episodes before forwarding (51-83).
 """
 from __future__ import annotations
 
+import pickle
 import time
 
 import numpy as np
@@ -38,8 +39,6 @@ class Manager:
         if msg is None:
             return False
         header, payload = msg
-        import pickle
-
         protocol = pickle.loads(header)  # header-only inspect: rollouts are
         if protocol is Protocol.Rollout:  # forwarded without decompressing
             self.pub.send(header, payload)

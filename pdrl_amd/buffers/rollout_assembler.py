@@ -19,6 +19,7 @@ pushed to an asyncio queue.
 from __future__ import annotations
 
 import asyncio
+import time
 
 import torch
 
@@ -66,9 +67,7 @@ class RolloutAssembler:
         # staleness eviction of partial trajectories (swept at most every
         # stale_s/4 — a per-push scan of the active dict measurably taxed
         # the ingest loop at tens of kHz push rates)
-        import time as _time
-
-        now = _time.monotonic()
+        now = time.monotonic()
         if now >= self._next_evict:
             self._next_evict = now + self.stale_s / 4
             for k in [k for k, tr in self.active.items() if tr.age > self.stale_s]:
