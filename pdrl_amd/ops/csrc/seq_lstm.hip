@@ -10,10 +10,13 @@
 // framework's operating point (B=128, H=64, S=5) the step is latency-bound,
 // not FLOP-bound: batch-row parallelism fills 128 of 256 CUs with fully
 // independent work and zero inter-workgroup traffic (the recurrence is
-// row-local). An MFMA tiling of the (B×4H×2H) gate GEMM would concentrate
-// the work on ~4 CUs and serialize on the matrix pipe's 64-cycle f32 MFMA
-// issue — measured slower at this shape (see profiles/). Per the platform
-// guide's GEMV rule, per-row work streams weights straight into registers.
+// row-local). An MFMA tiling of the (B×4H×2H) gate GEMM would need ≥32
+// batch rows per workgroup to feed 32×32 fragments, concentrating the grid
+// on B/32 = 4 CUs, and the f32 MFMA (64-cycle issue) buys no rate over the
+// f32 VALU (guide §3: equal 157 TF rate) — so at this shape row-parallel
+// VALU with register-resident weights wins on occupancy, not on peak. The
+// framework's MFMA usage lives where the work IS GEMM-shaped with K≫16:
+// the weight-gradient kernels (wgrad.hip).
 //
 // Weight residency:
 //   forward  — each thread owns one gate column: w_ih[:,g] and w_hh[:,g]

@@ -392,3 +392,31 @@ def test_batch_stager_roundtrip_gpu():
     for k in b2:
         assert out2[k].data_ptr() == out1[k].data_ptr()  # stable views
         torch.testing.assert_close(out2[k].cpu(), torch.from_numpy(b2[k]))
+
+
+@pytest.mark.parametrize("B,S,A", [(1, 5, 2), (16, 2, 2), (7, 8, 5), (16, 5, 16)])
+def test_fused_impala_shape_sweep(B, S, A):
+    """Fused-step gradient parity across odd shapes (tail paths: B*S % 32,
+    single-row batches, minimum seq, wide action spaces)."""
+    _ops()
+    from pdrl_amd.agents.learner_module import ImpalaUpdater
+    from pdrl_amd.networks import MlpLSTMSingle
+    from pdrl_amd.utils import load_params
+    from tests.conftest import make_batch
+
+    p = load_params()
+    p.batch_size, p.seq_len, p.obs_dim, p.n_actions = B, S, 4, A
+    torch.manual_seed(B * 100 + S * 10 + A)
+    model = MlpLSTMSingle(4, A, S, p.hidden_size)
+    upd = ImpalaUpdater(model, p, DEV)
+    assert upd.fused_step is not None
+    batch = make_batch(p, n_actions=A, device=DEV, seed=B + S + A)
+
+    upd.fused_step.compute_grads_only(batch)
+    fused = {n: q.grad.detach().clone() for n, q in model.named_parameters()}
+    upd.optimizer.zero_grad()
+    loss, _ = upd.compute_losses(batch)
+    loss.backward()
+    for n, q in model.named_parameters():
+        torch.testing.assert_close(fused[n], q.grad, rtol=2e-4, atol=2e-6,
+                                   msg=lambda m: f"shape {(B,S,A)} {n}: {m}")
