@@ -46,6 +46,7 @@ class SACUpdater(BaseUpdater):
         self.critic_optimizer = self.make_optimizer("adam", self.critic.parameters(), lr=params.lr)
         self.alpha_optimizer = self.make_optimizer("adam", [self.log_alpha], lr=params.lr,
                                                    clip=False)
+        self.fused_step = self._make_sac_fused_step()
 
     def trainable_modules(self):
         return {"model": self.model, "target_critic": self.target_critic}
@@ -57,11 +58,35 @@ class SACUpdater(BaseUpdater):
             "alpha_optimizer": self.alpha_optimizer,
         }
 
+    def _make_sac_fused_step(self):
+        """Whole-step fused HIP DAG for discrete SAC (ops/sac_step.py)."""
+        from pdrl_amd import ops
+
+        if not (self.device.type == "cuda" and ops.available()):
+            return None
+        if not all(getattr(o, "is_fused", False) for o in
+                   (self.actor_optimizer, self.critic_optimizer,
+                    self.alpha_optimizer)):
+            return None
+        if getattr(self.actor, "core", None) is None or \
+                self.actor.core.head_names != ["logits"]:
+            return None
+        import os
+
+        from pdrl_amd.ops.sac_step import FusedSacStep
+
+        use_graph = bool(int(os.environ.get("PDRL_USE_GRAPH", "1")))
+        return FusedSacStep(self, use_graph=use_graph)
+
     @property
     def alpha(self):
         return self.log_alpha.exp()
 
     def step(self, batch: dict[str, torch.Tensor]) -> dict:
+        if self.fused_step is not None and self.fused_step.fits(batch):
+            stats = self.fused_step.run(batch)
+            self.update_count += 1
+            return stats
         p = self.params
         obs, act = batch["obs"], batch["act"]
         rew = batch["rew"] * p.reward_scale

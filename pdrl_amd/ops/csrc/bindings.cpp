@@ -73,6 +73,18 @@ bool ppoc_loss_mega_hip(const at::Tensor&, const at::Tensor&,
                         const c10::optional<at::Tensor>&, long, double,
                         double, double, double, double, double, double,
                         double);
+void sac_actor_loss_hip(const at::Tensor&, const at::Tensor&,
+                        const at::Tensor&, const at::Tensor&, at::Tensor&,
+                        at::Tensor&, at::Tensor&,
+                        const c10::optional<at::Tensor>&,
+                        const c10::optional<at::Tensor>&, double);
+void sac_critic_loss_hip(const at::Tensor&, const at::Tensor&,
+                         const at::Tensor&, const at::Tensor&,
+                         const at::Tensor&, const at::Tensor&,
+                         const at::Tensor&, const at::Tensor&,
+                         const at::Tensor&, at::Tensor&, at::Tensor&,
+                         at::Tensor&, const c10::optional<at::Tensor>&,
+                         double, double);
 void l2norm_sq_hip(const at::Tensor&, at::Tensor&);
 void rmsprop_step_hip(at::Tensor&, const at::Tensor&, at::Tensor&,
                       const at::Tensor&, double, double, double, double);
@@ -115,6 +127,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "single-launch V-MPO loss: stats+gae+topk+duals+bwd");
   m.def("ppoc_loss_mega", &ppoc_loss_mega_hip,
         "single-launch Gaussian-policy PPO loss (K5)");
+  m.def("sac_actor_loss", &sac_actor_loss_hip,
+        "SAC discrete actor+alpha loss with analytic grads");
+  m.def("sac_critic_loss", &sac_critic_loss_hip,
+        "SAC discrete soft-Q target + twin critic loss grads");
   m.def("l2norm_sq", &l2norm_sq_hip, "squared L2 norm into a device scalar");
   m.def("rmsprop_step", &rmsprop_step_hip, "fused clip+RMSprop on flat buffers");
   m.def("adam_step", &adam_step_hip, "fused clip+Adam on flat buffers");

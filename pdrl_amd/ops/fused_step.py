@@ -37,7 +37,57 @@ _VMPO_STATS = ["loss-total", "loss-policy", "loss-value", "eta", "alpha", "kl"]
 BATCH_FIELDS = ["obs", "act", "rew", "logits", "log_prob", "is_fir", "hx", "cx"]
 
 
-class FusedOnPolicyStep:
+class GraphableStep:
+    """Capture/replay scaffolding shared by the fused step DAGs: first run
+    captures the kernel sequence on the caller's tensors; later runs replay
+    (copying only fields whose storage moved)."""
+
+    use_graph: bool = True
+    _graph = None
+    _static = None
+    _graph_failed = False
+    stat_names: list = []
+    stats_buf = None
+    params = None
+
+    def _full(self, batch):  # implemented by subclasses
+        raise NotImplementedError
+
+    def _try_capture(self, batch):
+        self._static = {k: batch[k] for k in BATCH_FIELDS}
+        self._static_ptrs = {k: batch[k].data_ptr() for k in BATCH_FIELDS}
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):  # warmup (allocator + RCCL channels)
+                    self._full(self._static)
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._full(self._static)
+            self._graph = g
+        except Exception as exc:  # capture unsupported → stream-ordered path
+            warnings.warn(f"hipGraph capture failed ({exc}); running the fused "
+                          "step stream-ordered instead")
+            self._graph_failed = True
+            self._static = None
+
+    def run(self, batch) -> dict:
+        if self.use_graph and self._graph is None and not self._graph_failed:
+            self._try_capture(batch)
+        if self._graph is not None:
+            for k in BATCH_FIELDS:
+                if batch[k].data_ptr() != self._static_ptrs[k]:
+                    self._static[k].copy_(batch[k], non_blocking=True)
+            self._graph.replay()
+        else:
+            self._full(batch)
+        return {name: self.stats_buf[i] for i, name in enumerate(self.stat_names)}
+
+
+class FusedOnPolicyStep(GraphableStep):
     def __init__(self, algo: str, core, params, optimizer, grad_reducer=None,
                  use_graph: bool = True, duals=None):
         assert algo in ("IMPALA", "PPO", "V-MPO", "PPO-C")
@@ -212,38 +262,4 @@ class FusedOnPolicyStep:
         for _ in range(self.params.K_epoch):
             self._body(batch)
 
-    # ------------------------------------------------------------------ #
-    def _try_capture(self, batch):
-        # capture on the CALLER's tensors: when later batches reuse the same
-        # storage (bench; packed-staging learner) replay needs ZERO copies
-        self._static = {k: batch[k] for k in BATCH_FIELDS}
-        self._static_ptrs = {k: batch[k].data_ptr() for k in BATCH_FIELDS}
-        try:
-            side = torch.cuda.Stream()
-            side.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(side):
-                for _ in range(3):  # warmup (allocator + RCCL channels)
-                    self._full(self._static)
-            torch.cuda.current_stream().wait_stream(side)
-            torch.cuda.synchronize()
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
-                self._full(self._static)
-            self._graph = g
-        except Exception as exc:  # capture unsupported → stream-ordered path
-            warnings.warn(f"hipGraph capture failed ({exc}); running the fused "
-                          "step stream-ordered instead")
-            self._graph_failed = True
-            self._static = None
 
-    def run(self, batch) -> dict:
-        if self.use_graph and self._graph is None and not self._graph_failed:
-            self._try_capture(batch)
-        if self._graph is not None:
-            for k in BATCH_FIELDS:
-                if batch[k].data_ptr() != self._static_ptrs[k]:
-                    self._static[k].copy_(batch[k], non_blocking=True)
-            self._graph.replay()
-        else:
-            self._full(batch)
-        return {name: self.stats_buf[i] for i, name in enumerate(self.stat_names)}
