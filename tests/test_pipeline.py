@@ -162,3 +162,35 @@ def test_find_latest_checkpoint(tmp_path):
     assert find_latest_checkpoint(str(d), "IMPALA").endswith("IMPALA_20.pt")
     assert find_latest_checkpoint(str(d), "PPO") is None
     assert find_latest_checkpoint(str(tmp_path / "nope"), "PPO") is None
+
+
+def test_learner_resume_from_checkpoint(pipeline_params, tmp_path):
+    """Learner(resume_path=...) restores updater state (reference:
+    set_model_weight, main.py:128-146)."""
+    import multiprocessing
+
+    from pdrl_amd.agents import Learner
+    from pdrl_amd.agents.learner_module import PPOUpdater
+    from pdrl_amd.buffers import SharedRolloutRing, rollout_fields
+    from pdrl_amd.networks import MlpLSTMSingle
+    from tests.conftest import make_batch
+
+    p = pipeline_params
+    torch.manual_seed(3)
+    model = MlpLSTMSingle(p.obs_dim, p.n_actions, p.seq_len, p.hidden_size)
+    upd = PPOUpdater(model, p, "cpu")
+    upd.step(make_batch(p))
+    ckpt = tmp_path / "PPO_1.pt"
+    upd.save(ckpt)
+
+    fields = rollout_fields(p.obs_dim, p.n_actions, p.hidden_size, p.continuous)
+    ring = SharedRolloutRing(fields, p.seq_len, p.batch_size, True)
+    lrn_port = _free_port_pair()
+    lrn = Learner(ring, "127.0.0.1", lrn_port, p, device="cpu",
+                  shared_stat=multiprocessing.Array("d", 3),
+                  resume_path=str(ckpt))
+    assert lrn.updater.update_count == 1
+    for p1, p2 in zip(model.parameters(),
+                      lrn.updater.trainable_modules()["model"].parameters()):
+        torch.testing.assert_close(p1.detach(), p2.detach())
+    lrn.close()
