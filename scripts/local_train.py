@@ -69,6 +69,8 @@ def main():
     ap.add_argument("--entropy", type=float, default=None)
     ap.add_argument("--clip", type=float, default=None, help="max_grad_norm")
     ap.add_argument("--batch-size", type=int, default=None)
+    ap.add_argument("--set", action="append", default=[],
+                    help="extra param override key=value (repeatable)")
     args = ap.parse_args()
 
     from pdrl_amd.utils import load_params
@@ -86,6 +88,12 @@ def main():
         base["max_grad_norm"] = args.clip
     if args.batch_size is not None:
         base["batch_size"] = args.batch_size
+    for kv in args.set:
+        k, v = kv.split("=", 1)
+        try:
+            base[k] = json.loads(v)
+        except json.JSONDecodeError:
+            base[k] = v
     tmp = tempfile.NamedTemporaryFile("w", suffix=".json", delete=False)
     json.dump(base, tmp)
     tmp.close()
