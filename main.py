@@ -113,12 +113,14 @@ def worker_run(model, worker_idx, manager_ip, manager_port, learner_ip, learner_
         raise
 
 
-def manager_run(manager_ip, manager_port, learner_ip, learner_port, heartbeat, stop_event):
+def manager_run(manager_ip, manager_port, learner_ip, learner_port, heartbeat, stop_event,
+                storage_shards=1):
     from pdrl_amd.agents import Manager
 
     try:
         m = Manager(manager_ip, manager_port, learner_ip, learner_port,
-                    stop_event=stop_event, heartbeat=heartbeat)
+                    stop_event=stop_event, heartbeat=heartbeat,
+                    storage_shards=storage_shards)
         m.run()
     except Exception:
         save_error_log("manager", traceback.format_exc())
@@ -254,10 +256,18 @@ def learner_sub_process(learner_ip, learner_port, *_):
     sup = Supervisor(stop_event)
     _install_cleanup(sup)
 
-    hb_storage = mp.Value("d", time.time())
-    sup.spawn("storage", storage_run,
-              (ring, learner_ip, learner_port, params, shared_stat, hb_storage, stop_event),
-              heartbeat=hb_storage)
+    # sharded ingest: each shard is its own decode+assemble process binding
+    # its own data port; all shards write the one lock-guarded ring. The
+    # manager routes rollouts across shards by worker connection.
+    from pdrl_amd.agents import storage_shard_ports
+
+    n_shards = max(1, int(getattr(params, "storage_shards", 1) or 1))
+    for k, port in enumerate(storage_shard_ports(learner_port, n_shards)):
+        hb_storage = mp.Value("d", time.time())
+        sup.spawn(f"storage-{k}", storage_run,
+                  (ring, learner_ip, port, params,
+                   shared_stat if k == 0 else None, hb_storage, stop_event),
+                  heartbeat=hb_storage)
 
     world_size = int(getattr(params, "num_learner_gpus", 1) or 1)
     if torch.cuda.is_available():
@@ -279,9 +289,11 @@ def manager_sub_process(manager_ip, learner_ip, port, learner_port, *_):
     stop_event = mp.Event()
     sup = Supervisor(stop_event)
     _install_cleanup(sup)
+    n_shards = max(1, int(getattr(Params, "storage_shards", 1) or 1))
     hb = mp.Value("d", time.time())
     sup.spawn("manager", manager_run,
-              (manager_ip, int(port), learner_ip, int(learner_port), hb, stop_event),
+              (manager_ip, int(port), learner_ip, int(learner_port), hb, stop_event,
+               n_shards),
               heartbeat=hb)
     sup.monitor()
 

@@ -18,30 +18,51 @@ from pdrl_amd.transport import pub_connect, sub_bind
 from pdrl_amd.utils import Protocol, decode, encode
 
 
+def storage_shard_ports(learner_port: int, n_shards: int) -> list[int]:
+    """Data-plane ports of the learner-storage shards. Shard 0 keeps the
+    reference convention (learner_port); the weight plane is learner_port+1
+    and the DP rendezvous learner_port+2, so extra shards bind
+    learner_port+2+k (k >= 1)."""
+    return [learner_port] + [learner_port + 2 + k for k in range(1, n_shards)]
+
+
 class Manager:
     STAT_INTERVAL = 50
 
     def __init__(self, manager_ip, manager_port, learner_ip, learner_port, stop_event=None,
-                 heartbeat=None):
+                 heartbeat=None, storage_shards: int = 1):
         self.sub = sub_bind(manager_ip, manager_port)
-        self.pub = pub_connect(learner_ip, learner_port)
+        self.pubs = [pub_connect(learner_ip, port)
+                     for port in storage_shard_ports(learner_port, max(1, storage_shards))]
+        self.pub = self.pubs[0]  # stat plane + back-compat alias
         self.stop_event = stop_event
         self.heartbeat = heartbeat
         self.stat_q: list[float] = []
         self.game_count = 0
+        # worker-peer → shard. All steps of an episode uuid come from one
+        # worker connection, so routing by peer keeps every uuid's chunks on
+        # one shard's assembler without decoding payloads.
+        self._route: dict[int, int] = {}
 
     def _stopped(self) -> bool:
         return self.stop_event is not None and self.stop_event.is_set()
 
+    def _shard_of(self, peer_id: int) -> int:
+        shard = self._route.get(peer_id)
+        if shard is None:
+            shard = len(self._route) % len(self.pubs)  # round-robin first-seen
+            self._route[peer_id] = shard
+        return shard
+
     def relay_once(self, timeout: float = 0.5) -> bool:
         """Handle one inbound message. Returns True if one was processed."""
-        msg = self.sub.recv(timeout=timeout)
+        msg = self.sub.recv(timeout=timeout, with_peer=True)
         if msg is None:
             return False
-        header, payload = msg
+        peer_id, header, payload = msg
         protocol = pickle.loads(header)  # header-only inspect: rollouts are
         if protocol is Protocol.Rollout:  # forwarded without decompressing
-            self.pub.send(header, payload)
+            self.pubs[self._shard_of(peer_id)].send(header, payload)
         elif protocol is Protocol.Stat:
             _, data = decode(header, payload)
             self.game_count += 1
@@ -66,4 +87,5 @@ class Manager:
 
     def close(self):
         self.sub.close()
-        self.pub.close()
+        for p in self.pubs:
+            p.close()

@@ -48,10 +48,11 @@ _CHUNK = 1 << 18
 
 class _Peer:
     __slots__ = ("sock", "rx", "tx", "inflight", "inflight_off", "alive",
-                 "registered")
+                 "registered", "pid")
 
-    def __init__(self, sock: socket.socket):
+    def __init__(self, sock: socket.socket, pid: int):
         self.sock = sock
+        self.pid = pid  # stable endpoint-local id (fds are reused; this isn't)
         self.rx = bytearray()
         self.tx: deque = deque()  # droppable framed bytes objects
         self.inflight: bytes | None = None  # partially-sent frame — NEVER
@@ -88,6 +89,7 @@ class Endpoint:
         self._paused = False
         self._resume_req = False
         self._listener: socket.socket | None = None
+        self._peer_seq = 0
         self._connect_addr = connect
         self._connecting: socket.socket | None = None
         self._next_connect = 0.0
@@ -123,7 +125,8 @@ class Endpoint:
             sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
         except OSError:
             pass
-        peer = _Peer(sock)
+        self._peer_seq += 1
+        peer = _Peer(sock, self._peer_seq)
         with self._peers_lock:
             self._peers[sock.fileno()] = peer
         self._update_interest(peer)
@@ -279,7 +282,7 @@ class Endpoint:
             if len(buf) - off < total:
                 break
             h0 = off + _HDR.size
-            msgs.append((bytes(buf[h0 : h0 + hlen]),
+            msgs.append((peer.pid, bytes(buf[h0 : h0 + hlen]),
                          bytes(buf[h0 + hlen : h0 + hlen + plen])))
             off += total
         if off:
@@ -339,8 +342,11 @@ class Endpoint:
             self._pending_tx.append(frame)
         self._wake()
 
-    def recv(self, timeout: float | None = None):
-        """Pop one (header, payload) message, or None on timeout."""
+    def recv(self, timeout: float | None = None, with_peer: bool = False):
+        """Pop one (header, payload) message, or None on timeout. With
+        ``with_peer`` the tuple is (peer_id, header, payload) — peer_id is a
+        stable endpoint-local id of the originating connection (used by the
+        manager to shard rollout routing by worker)."""
         deadline = None if timeout is None else time.monotonic() + timeout
         with self._rx_cv:
             while not self._rx:
@@ -355,14 +361,15 @@ class Endpoint:
         if self._paused and depth <= self._rx_low:
             self._resume_req = True
             self._wake()
-        return out
+        return out if with_peer else out[1:]
 
-    def recv_many(self, max_n: int = 1024) -> list:
+    def recv_many(self, max_n: int = 1024, with_peer: bool = False) -> list:
         """Drain up to max_n queued messages without blocking."""
         out = []
         with self._rx_cv:
             while self._rx and len(out) < max_n:
-                out.append(self._rx.popleft())
+                m = self._rx.popleft()
+                out.append(m if with_peer else m[1:])
             depth = len(self._rx)
         if self._paused and depth <= self._rx_low:
             self._resume_req = True

@@ -16,12 +16,26 @@ REPO = Path(__file__).resolve().parent.parent
 sys.path.insert(0, str(REPO))
 
 
-def free_port() -> int:
-    s = socket.socket()
-    s.bind(("127.0.0.1", 0))
-    p = s.getsockname()[1]
-    s.close()
-    return p
+def free_port(span: int = 0) -> int:
+    """A port p with p..p+span all bindable (shards use p+2+k)."""
+    for _ in range(128):
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        p = s.getsockname()[1]
+        s.close()
+        socks = []
+        try:
+            for off in range(span + 1):
+                t = socket.socket()
+                t.bind(("127.0.0.1", p + off))
+                socks.append(t)
+        except OSError:
+            continue
+        finally:
+            for t in socks:
+                t.close()
+        return p
+    raise RuntimeError("no free port span")
 
 
 def _worker(model, idx, mport, seconds):
@@ -44,14 +58,14 @@ def _worker(model, idx, mport, seconds):
         w.collect(max_episodes=1)
 
 
-def _manager(mport, lport, seconds, relayed):
+def _manager(mport, lport, seconds, relayed, shards=1):
     sys.path.insert(0, str(REPO))
     import torch
 
     torch.set_num_threads(1)
     from pdrl_amd.agents import Manager
 
-    m = Manager("127.0.0.1", mport, "127.0.0.1", lport)
+    m = Manager("127.0.0.1", mport, "127.0.0.1", lport, storage_shards=shards)
     deadline = time.monotonic() + seconds
     n = 0
     while time.monotonic() < deadline:
@@ -100,6 +114,8 @@ def main():
 
     ap = argparse.ArgumentParser()
     ap.add_argument("--workers", type=int, default=8)
+    ap.add_argument("--shards", type=int, default=1,
+                    help="storage shard processes (manager routes by worker)")
     ap.add_argument("--seconds", type=float, default=15.0)
     ap.add_argument("--shared-model", action="store_true")
     args = ap.parse_args()
@@ -107,8 +123,10 @@ def main():
     p = load_params()
     p.algo, p.env = "IMPALA", "CartPole-v1"
     main_mod.probe_env_spaces(p)
+    from pdrl_amd.agents import storage_shard_ports
+
     ctx = mp.get_context("spawn")
-    mport, lport = free_port(), free_port()
+    mport, lport = free_port(), free_port(2 + args.shards)
     fields = rollout_fields(p.obs_dim, p.n_actions, p.hidden_size, False)
     # big off-policy ring so storage never blocks on a full ring
     ring = SharedRolloutRing(fields, p.seq_len, 65536, on_policy=False)
@@ -119,22 +137,30 @@ def main():
         model.share_memory()
 
     relayed = ctx.Value("q", 0)
-    ingested = ctx.Value("q", 0)
-    stored = ctx.Value("q", 0)
+    ingesteds = [ctx.Value("q", 0) for _ in range(args.shards)]
+    storeds = [ctx.Value("q", 0) for _ in range(args.shards)]
     run_s = args.seconds + 10  # children outlive the measure window
-    procs = [ctx.Process(target=_manager, args=(mport, lport, run_s, relayed)),
-             ctx.Process(target=_storage, args=(ring, lport, run_s, ingested, stored))]
+    procs = [ctx.Process(target=_manager,
+                         args=(mport, lport, run_s, relayed, args.shards))]
+    procs += [ctx.Process(target=_storage,
+                          args=(ring, port, run_s, ingesteds[k], storeds[k]))
+              for k, port in enumerate(storage_shard_ports(lport, args.shards))]
     procs += [ctx.Process(target=_worker, args=(model, i, mport, run_s))
               for i in range(args.workers)]
     for pr in procs:
         pr.start()
     time.sleep(5)  # spin-up
-    r0, i0, s0 = relayed.value, ingested.value, stored.value
+    r0 = relayed.value
+    i0 = sum(c.value for c in ingesteds)
+    s0 = sum(c.value for c in storeds)
     t0 = time.monotonic()
     time.sleep(args.seconds)
     dt = time.monotonic() - t0
-    r1, i1, s1 = relayed.value, ingested.value, stored.value
-    print(f"workers={args.workers} relayed={(r1-r0)/dt:8.0f}/s "
+    r1 = relayed.value
+    i1 = sum(c.value for c in ingesteds)
+    s1 = sum(c.value for c in storeds)
+    print(f"workers={args.workers} shards={args.shards} "
+          f"relayed={(r1-r0)/dt:8.0f}/s "
           f"ingested={(i1-i0)/dt:8.0f}/s stored={(s1-s0)/dt:8.0f} traj/s "
           f"(~{(s1-s0)*p.seq_len/dt:.0f} steps/s)")
     for pr in procs:
