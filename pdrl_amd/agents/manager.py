@@ -79,9 +79,42 @@ class Manager:
             raise AssertionError(f"unexpected protocol at manager: {protocol}")
         return True
 
+    def relay_batch(self, timeout: float = 0.5) -> int:
+        """Drain and forward a burst of inbound messages. Rollouts are grouped
+        per shard and queued with one send_many each (one lock + one reactor
+        wake per shard per burst instead of per message). Returns the number
+        of messages handled."""
+        first = self.sub.recv(timeout=timeout, with_peer=True)
+        if first is None:
+            return 0
+        msgs = [first]
+        msgs.extend(self.sub.recv_many(1023, with_peer=True))
+        out: list[list] = [[] for _ in self.pubs]
+        for peer_id, header, payload in msgs:
+            protocol = pickle.loads(header)
+            if protocol is Protocol.Rollout:
+                out[self._shard_of(peer_id)].append((header, payload))
+            elif protocol is Protocol.Stat:
+                _, data = decode(header, payload)
+                self.game_count += 1
+                self.stat_q.append(float(data["epi_rew"]))
+                if len(self.stat_q) >= self.STAT_INTERVAL:
+                    mean_rew = float(np.mean(self.stat_q))
+                    self.pubs[0].send(*encode(
+                        Protocol.Stat,
+                        {"game_count": self.game_count, "mean_stat": mean_rew},
+                    ))
+                    self.stat_q.clear()
+            else:
+                raise AssertionError(f"unexpected protocol at manager: {protocol}")
+        for shard, batch in enumerate(out):
+            if batch:
+                self.pubs[shard].send_many(batch)
+        return len(msgs)
+
     def run(self):
         while not self._stopped():
-            self.relay_once(timeout=0.5)
+            self.relay_batch(timeout=0.5)
             if self.heartbeat is not None:
                 self.heartbeat.value = time.time()
 

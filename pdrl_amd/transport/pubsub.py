@@ -342,6 +342,17 @@ class Endpoint:
             self._pending_tx.append(frame)
         self._wake()
 
+    def send_many(self, messages):
+        """Queue many (header, payload) messages with ONE lock acquisition and
+        ONE reactor wake — the relay fast path (per-send overhead capped a
+        24-worker manager at ~25K msg/s)."""
+        if not messages:
+            return
+        frames = [_HDR.pack(len(h), len(p)) + h + p for h, p in messages]
+        with self._pending_lock:
+            self._pending_tx.extend(frames)
+        self._wake()
+
     def recv(self, timeout: float | None = None, with_peer: bool = False):
         """Pop one (header, payload) message, or None on timeout. With
         ``with_peer`` the tuple is (peer_id, header, payload) — peer_id is a

@@ -69,8 +69,9 @@ def _manager(mport, lport, seconds, relayed, shards=1):
     deadline = time.monotonic() + seconds
     n = 0
     while time.monotonic() < deadline:
-        if m.relay_once(timeout=0.2):
-            n += 1
+        k = m.relay_batch(timeout=0.2)
+        if k:
+            n += k
             relayed.value = n
 
 

@@ -117,3 +117,19 @@ def test_pub_never_blocks_without_peer():
         pub.send(b"h", b"p")
     assert time.monotonic() - t0 < 1.0
     pub.close()
+
+def test_send_many_preserves_order_and_framing():
+    pub = Endpoint(bind=("127.0.0.1", 0))
+    sub = Endpoint(connect=("127.0.0.1", pub.bound_port))
+    assert _wait(lambda: pub.n_peers() == 1)
+    batch = [encode(Protocol.Stat, {"i": i}) for i in range(32)]
+    pub.send_many(batch)
+    vals = []
+    deadline = time.monotonic() + 10.0
+    while len(vals) < 32 and time.monotonic() < deadline:
+        msg = sub.recv(timeout=1.0)
+        if msg is not None:
+            vals.append(decode(*msg)[1]["i"])
+    assert vals == list(range(32))
+    pub.close()
+    sub.close()
