@@ -90,6 +90,10 @@ class Endpoint:
         self._resume_req = False
         self._listener: socket.socket | None = None
         self._peer_seq = 0
+        # diagnostic counters (reactor-thread writes; reads are racy-but-fine)
+        self._tx_dropped = 0  # frames shed from peer send queues (HWM)
+        self._rx_dropped = 0  # frames shed from the RX deque (HWM)
+        self._rx_total = 0
         self._connect_addr = connect
         self._connecting: socket.socket | None = None
         self._next_connect = 0.0
@@ -289,6 +293,10 @@ class Endpoint:
             del buf[:off]
         if msgs:
             with self._rx_cv:
+                overflow = len(self._rx) + len(msgs) - self._rx.maxlen
+                if overflow > 0:
+                    self._rx_dropped += min(overflow, len(self._rx) + len(msgs))
+                self._rx_total += len(msgs)
                 self._rx.extend(msgs)  # deque(maxlen) drops oldest
                 self._rx_cv.notify()
                 depth = len(self._rx)
@@ -332,6 +340,7 @@ class Endpoint:
                 if len(peer.tx) >= self._send_hwm:
                     peer.tx.popleft()  # drop oldest (PUB HWM semantics);
                     # the in-flight frame lives outside tx, so framing is safe
+                    self._tx_dropped += 1
                 peer.tx.append(f)
             self._write_ready(peer)
 
@@ -386,6 +395,17 @@ class Endpoint:
             self._resume_req = True
             self._wake()
         return out
+
+    def stats(self) -> dict:
+        """Diagnostic counters: frames received, and frames shed by the
+        send-side (tx_dropped, per-peer HWM) and receive-side (rx_dropped,
+        RX HWM) bounded queues. Localizes WHERE an overloaded pipeline is
+        shedding (see profiles/ingest_shards.md)."""
+        return {
+            "rx_total": self._rx_total,
+            "rx_dropped": self._rx_dropped,
+            "tx_dropped": self._tx_dropped,
+        }
 
     def n_peers(self) -> int:
         with self._peers_lock:

@@ -58,7 +58,7 @@ def _worker(model, idx, mport, seconds):
         w.collect(max_episodes=1)
 
 
-def _manager(mport, lport, seconds, relayed, shards=1):
+def _manager(mport, lport, seconds, relayed, shards=1, drops=None):
     sys.path.insert(0, str(REPO))
     import torch
 
@@ -73,6 +73,10 @@ def _manager(mport, lport, seconds, relayed, shards=1):
         if k:
             n += k
             relayed.value = n
+        if drops is not None:
+            # [rx drops at manager sub, tx drops across shard pubs]
+            drops[0] = m.sub.stats()["rx_dropped"]
+            drops[1] = sum(p.stats()["tx_dropped"] for p in m.pubs)
 
 
 def _storage(ring, lport, seconds, ingested, stored):
@@ -138,11 +142,12 @@ def main():
         model.share_memory()
 
     relayed = ctx.Value("q", 0)
+    drops = ctx.Array("q", 2)
     ingesteds = [ctx.Value("q", 0) for _ in range(args.shards)]
     storeds = [ctx.Value("q", 0) for _ in range(args.shards)]
     run_s = args.seconds + 10  # children outlive the measure window
     procs = [ctx.Process(target=_manager,
-                         args=(mport, lport, run_s, relayed, args.shards))]
+                         args=(mport, lport, run_s, relayed, args.shards, drops))]
     procs += [ctx.Process(target=_storage,
                           args=(ring, port, run_s, ingesteds[k], storeds[k]))
               for k, port in enumerate(storage_shard_ports(lport, args.shards))]
@@ -163,7 +168,8 @@ def main():
     print(f"workers={args.workers} shards={args.shards} "
           f"relayed={(r1-r0)/dt:8.0f}/s "
           f"ingested={(i1-i0)/dt:8.0f}/s stored={(s1-s0)/dt:8.0f} traj/s "
-          f"(~{(s1-s0)*p.seq_len/dt:.0f} steps/s)")
+          f"(~{(s1-s0)*p.seq_len/dt:.0f} steps/s) "
+          f"mgr_drops: rx={drops[0]} tx={drops[1]} (cumulative)")
     for pr in procs:
         pr.terminate()
     for pr in procs:
