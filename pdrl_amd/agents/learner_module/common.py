@@ -112,6 +112,12 @@ class BaseUpdater:
         actor = getattr(m, "actor", m)
         return {k: v.cpu() for k, v in actor.state_dict().items()}
 
+    def extra_params(self) -> dict:
+        """Loose nn.Parameters outside any module (SAC's log_alpha, V-MPO's
+        duals): they must be checkpointed explicitly or they silently reset
+        on resume."""
+        return {}
+
     def save(self, path):
         torch.save(
             {
@@ -119,6 +125,7 @@ class BaseUpdater:
                 "update_count": self.update_count,
                 "modules": {k: m.state_dict() for k, m in self.trainable_modules().items()},
                 "optimizers": {k: o.state_dict() for k, o in self.optimizers().items()},
+                "extras": {k: p.detach().cpu() for k, p in self.extra_params().items()},
             },
             path,
         )
@@ -128,6 +135,10 @@ class BaseUpdater:
         for k, m in self.trainable_modules().items():
             if k in ckpt.get("modules", {}):
                 m.load_state_dict(ckpt["modules"][k])
+        for k, p in self.extra_params().items():
+            if k in ckpt.get("extras", {}):
+                with torch.no_grad():
+                    p.data.copy_(ckpt["extras"][k].to(p.device))
         for k, o in self.optimizers().items():
             if k in ckpt.get("optimizers", {}):
                 try:

@@ -45,6 +45,9 @@ class SACContinuousUpdater(BaseUpdater):
     def trainable_modules(self):
         return {"model": self.model, "target_critic": self.target_critic}
 
+    def extra_params(self):
+        return {"log_alpha": self.log_alpha}
+
     def optimizers(self):
         return {
             "actor_optimizer": self.actor_optimizer,

@@ -44,8 +44,8 @@ class VMPOUpdater(BaseUpdater):
     def optimizers(self):
         return {"optimizer": self.optimizer}
 
-    def extra_state(self):
-        return {"log_eta": self.log_eta.detach().cpu(), "log_alpha": self.log_alpha.detach().cpu()}
+    def extra_params(self):
+        return {"log_eta": self.log_eta, "log_alpha": self.log_alpha}
 
     def get_coef_alpha(self):
         """KL-bound coefficient sampled uniformly in [below, upper]

@@ -129,3 +129,36 @@ def test_checkpoint_roundtrip(params, tmp_path):
     assert upd2.update_count == 1
     for p1, p2 in zip(model.parameters(), model2.parameters()):
         torch.testing.assert_close(p1, p2)
+
+
+def test_checkpoint_persists_loose_duals(params, tmp_path):
+    """SAC's log_alpha and V-MPO's log_eta/log_alpha live outside any module;
+    save/load must round-trip them or resume silently resets the temperature
+    and dual variables."""
+    from pdrl_amd.agents.learner_module import SACUpdater, VMPOUpdater
+    from pdrl_amd.networks import MlpLSTMSeperate, MlpLSTMSingle
+
+    torch.manual_seed(1)
+    params.obs_dim, params.n_actions = 4, 2
+
+    sac = SACUpdater(MlpLSTMSeperate(4, 2, params.seq_len, params.hidden_size),
+                     params, "cpu")
+    with torch.no_grad():
+        sac.log_alpha.fill_(-1.2345)
+    sac.save(tmp_path / "SAC_1.pt")
+    sac2 = SACUpdater(MlpLSTMSeperate(4, 2, params.seq_len, params.hidden_size),
+                      params, "cpu")
+    sac2.load(tmp_path / "SAC_1.pt")
+    assert float(sac2.log_alpha) == pytest.approx(-1.2345)
+
+    vmpo = VMPOUpdater(MlpLSTMSingle(4, 2, params.seq_len, params.hidden_size),
+                       params, "cpu")
+    with torch.no_grad():
+        vmpo.log_eta.fill_(0.777)
+        vmpo.log_alpha.fill_(-0.333)
+    vmpo.save(tmp_path / "V-MPO_1.pt")
+    vmpo2 = VMPOUpdater(MlpLSTMSingle(4, 2, params.seq_len, params.hidden_size),
+                        params, "cpu")
+    vmpo2.load(tmp_path / "V-MPO_1.pt")
+    assert float(vmpo2.log_eta) == pytest.approx(0.777)
+    assert float(vmpo2.log_alpha) == pytest.approx(-0.333)
