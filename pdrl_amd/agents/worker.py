@@ -55,6 +55,39 @@ class Worker:
         # ~10K steps/s; seq_len-sized chunks amortize pickle+zlib 5×)
         self.batch_steps = int(getattr(params, "worker_batch_steps", 0)) or params.seq_len
         self._step_buf: list[dict] = []
+        # Warmup exploration for continuous control (off by default = reference
+        # behavior). Per-step policy noise is temporally UNCORRELATED, which
+        # cannot build momentum on tasks like MountainCarContinuous; for the
+        # first ``explore_warmup_steps`` env steps actions come from an
+        # Ornstein-Uhlenbeck process instead (correlated bang-bang), with the
+        # record's log_prob still evaluated under the CURRENT policy.
+        self.explore_warmup_steps = int(getattr(params, "explore_warmup_steps", 0))
+        self._ou_theta = float(getattr(params, "explore_ou_theta", 0.15))
+        self._ou_sigma = float(getattr(params, "explore_ou_sigma", 0.6))
+        self._continuous = bool(getattr(params, "continuous", False))
+        self._total_steps = 0
+        self._ou_state: torch.Tensor | None = None
+        self._ou_gen = torch.Generator().manual_seed((seed or 0) * 9973 + 17)
+
+    # ------------------------------------------------------------------ #
+    def _ou_explore(self, action, logits):
+        """OU-noise action + its tanh-Gaussian log-prob under the policy."""
+        if self._ou_state is None:
+            self._ou_state = torch.zeros_like(action, dtype=torch.float32)
+        noise = torch.randn(action.shape, generator=self._ou_gen)
+        self._ou_state = (
+            self._ou_state - self._ou_theta * self._ou_state + self._ou_sigma * noise
+        )
+        a = torch.tanh(self._ou_state)
+        A = action.shape[-1]
+        mu, log_std = logits[..., :A], logits[..., A:]
+        std = log_std.clamp(-20.0, 2.0).exp()
+        z = self._ou_state
+        log_prob = (
+            -0.5 * ((z - mu) / std) ** 2 - log_std - 0.5 * torch.log(torch.tensor(2 * torch.pi))
+            - torch.log(1.0 - a.pow(2) + 1e-7)
+        ).sum(-1, keepdim=True)
+        return a, log_prob
 
     # ------------------------------------------------------------------ #
     def pub_rollout(self, step_data: dict, flush: bool = False):
@@ -96,6 +129,7 @@ class Worker:
             obs = self.env.reset()
             hx = torch.zeros(1, H)
             cx = torch.zeros(1, H)
+            self._ou_state = None  # OU noise restarts with each episode
             epi_rew = 0.0
             epi_id = uuid.uuid4().hex
             is_fir = 1.0
@@ -104,6 +138,9 @@ class Worker:
                     break
                 self.poll_model()
                 action, logits, log_prob, (next_hx, next_cx) = self.model.act(obs, (hx, cx))
+                if self._continuous and self._total_steps < self.explore_warmup_steps:
+                    action, log_prob = self._ou_explore(action, logits)
+                self._total_steps += 1
                 next_obs, rew, done, _ = self.env.step(action)
                 epi_rew += rew
                 step_data = {

@@ -90,3 +90,53 @@ def test_fake_env_deterministic():
 def test_make_unknown_env():
     with pytest.raises(ValueError):
         envs.make("Atari-Breakout")
+
+
+def test_worker_ou_warmup_exploration():
+    """With explore_warmup_steps set on a continuous env, actions come from a
+    temporally-correlated OU process (in [-1,1], correlated across steps) and
+    the rollout record schema is unchanged; after warmup the policy acts."""
+    import numpy as np
+    import torch
+
+    from pdrl_amd.agents import Worker
+    from pdrl_amd.networks import MlpLSTMSeperateContinuous
+    from pdrl_amd.transport import Endpoint
+    from pdrl_amd.utils import Protocol, decode, load_params
+
+    p = load_params()
+    p.env = "MountainCarContinuous-v0"
+    p.algo = "SAC-Continuous"
+    p.obs_dim, p.n_actions, p.continuous = 2, 1, True
+    p.seq_len = 5
+    p.explore_warmup_steps = 10_000
+    p.explore_ou_sigma = 0.6
+
+    mgr_sub = Endpoint(bind=("127.0.0.1", 0))
+    model = MlpLSTMSeperateContinuous(2, 1, p.seq_len, p.hidden_size).actor
+    # worker subscribes to learner_port+1; point it at an unused port
+    w = Worker(model, 0, "127.0.0.1", mgr_sub.bound_port, "127.0.0.1", 1, p, seed=3)
+    w.collect(max_episodes=1)
+
+    acts = []
+    while True:
+        msg = mgr_sub.recv(timeout=2.0)
+        if msg is None:
+            break
+        proto, data = decode(*msg)
+        if proto is Protocol.Rollout:
+            for step in data:
+                acts.append(float(step["act"][0]))
+                assert set(step) == {
+                    "obs", "act", "rew", "logits", "log_prob", "is_fir",
+                    "done", "hx", "cx", "id",
+                }
+                assert np.isfinite(step["log_prob"]).all()
+    assert len(acts) >= 50
+    a = np.array(acts)
+    assert (np.abs(a) <= 1.0).all()
+    # OU actions are temporally correlated — far beyond iid tanh-Gaussian
+    lag1 = np.corrcoef(a[:-1], a[1:])[0, 1]
+    assert lag1 > 0.5, f"expected correlated warmup actions, lag-1 r={lag1:.3f}"
+    w.close()
+    mgr_sub.close()
