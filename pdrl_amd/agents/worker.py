@@ -43,7 +43,15 @@ class Worker:
         self.params = params
         self.worker_idx = worker_idx
         self.model = model.cpu().eval()
-        self.env = EnvBase(params.env, seed=seed)
+        # M envs per worker process (default 1 = reference behavior): one
+        # batched model.act per tick steps all of them, amortizing the
+        # Python/inference overhead that bounds per-worker throughput.
+        self.num_envs = max(1, int(getattr(params, "num_envs_per_worker", 1) or 1))
+        self.envs = [
+            EnvBase(params.env, seed=None if seed is None else seed * 1000 + i)
+            for i in range(self.num_envs)
+        ]
+        self.env = self.envs[0]
         self.heartbeat = heartbeat
         self.stop_event = stop_event
         # data plane: PUB → manager; weight plane: SUB ← learner (port+1)
@@ -123,6 +131,8 @@ class Worker:
     # ------------------------------------------------------------------ #
     def collect(self, max_episodes: int | None = None):
         """Roll out episodes forever (or for max_episodes, for tests)."""
+        if self.num_envs > 1:
+            return self._collect_vec(max_episodes)
         H = self.params.hidden_size
         episodes = 0
         while not self._stopped():
@@ -166,6 +176,70 @@ class Worker:
                     break
             self.pub_stat(epi_rew)
             episodes += 1
+            if max_episodes is not None and episodes >= max_episodes:
+                break
+
+    def _collect_vec(self, max_episodes: int | None = None):
+        """Vectorized rollout: M envs, ONE batched model.act per tick. Each
+        env keeps its own episode uuid / recurrent-state row / reward
+        accumulator; records are identical to the scalar path (the storage
+        assembler routes per step by uuid, so mixed-uuid chunks are fine)."""
+        p = self.params
+        H, M = p.hidden_size, self.num_envs
+        obs = torch.cat([e.reset() for e in self.envs], dim=0)  # (M, F)
+        hx = torch.zeros(M, H)
+        cx = torch.zeros(M, H)
+        epi_rew = [0.0] * M
+        epi_id = [uuid.uuid4().hex for _ in range(M)]
+        is_fir = [1.0] * M
+        epi_steps = [0] * M
+        episodes = 0
+        while not self._stopped():
+            self.poll_model()
+            action, logits, log_prob, (next_hx, next_cx) = self.model.act(obs, (hx, cx))
+            if self._continuous and self._total_steps < self.explore_warmup_steps:
+                action, log_prob = self._ou_explore(action, logits)
+            self._total_steps += M
+            next_rows = []
+            for i, env in enumerate(self.envs):
+                next_obs, rew, done, _ = env.step(action[i])
+                epi_rew[i] += rew
+                step_data = {
+                    "obs": obs[i].numpy(),
+                    "act": action[i].reshape(-1).float().numpy(),
+                    "rew": rew,
+                    "logits": logits[i].numpy(),
+                    "log_prob": log_prob[i].reshape(-1).numpy(),
+                    "is_fir": is_fir[i],
+                    "done": float(done),
+                    "hx": hx[i].numpy(),
+                    "cx": cx[i].numpy(),
+                    "id": epi_id[i],
+                }
+                epi_steps[i] += 1
+                horizon = done or epi_steps[i] >= p.time_horizon
+                self.pub_rollout(step_data, flush=done)
+                if horizon:
+                    self.pub_stat(epi_rew[i])
+                    episodes += 1
+                    next_obs = env.reset()
+                    next_hx[i] = 0.0
+                    next_cx[i] = 0.0
+                    if self._ou_state is not None:
+                        self._ou_state[i] = 0.0
+                    epi_rew[i] = 0.0
+                    epi_id[i] = uuid.uuid4().hex
+                    is_fir[i] = 1.0
+                    epi_steps[i] = 0
+                else:
+                    is_fir[i] = 0.0
+                next_rows.append(next_obs)
+            obs = torch.cat(next_rows, dim=0)
+            hx, cx = next_hx, next_cx
+            if self.heartbeat is not None:
+                self.heartbeat.value = time.time()
+            if self.step_sleep > 0:
+                time.sleep(self.step_sleep)
             if max_episodes is not None and episodes >= max_episodes:
                 break
 
