@@ -196,7 +196,8 @@ class Learner:
         interval = int(getattr(self.params, "model_publish_interval", 1) or 1)
         if interval > 1 and self.updater.update_count % interval != 0:
             return
-        header, payload = encode(Protocol.Model, self.updater.actor_state_dict())
+        header, payload = encode(Protocol.Model, self.updater.actor_state_dict(),
+                                 compress=False)
         self.pub.send(header, payload)
 
     def log_stats(self, stats: dict):

@@ -13,6 +13,7 @@ import asyncio
 import time
 
 from pdrl_amd.buffers import RolloutAssembler, SharedRolloutRing
+from pdrl_amd.buffers.wire import is_packed, unpack_steps
 from pdrl_amd.transport import sub_bind
 from pdrl_amd.utils import Protocol, decode
 
@@ -53,7 +54,10 @@ class LearnerStorage:
             for msg in msgs:
                 protocol, data = decode(*msg)
                 if protocol is Protocol.Rollout:
-                    steps = data if isinstance(data, list) else [data]
+                    if is_packed(data):  # packed chunk: one matrix, n×field views
+                        steps = unpack_steps(data)
+                    else:  # plain step dict(s) — compatibility path
+                        steps = data if isinstance(data, list) else [data]
                     for step in steps:
                         await self.assembler.push(step)
                     self.n_ingested += len(steps)

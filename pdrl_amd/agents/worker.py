@@ -104,7 +104,10 @@ class Worker:
         assembly semantics are unchanged)."""
         self._step_buf.append(step_data)
         if len(self._step_buf) >= self.batch_steps or flush:
-            header, payload = encode(Protocol.Rollout, self._step_buf)
+            from pdrl_amd.buffers.wire import pack_steps
+
+            header, payload = encode(Protocol.Rollout, pack_steps(self._step_buf),
+                                     compress=False)
             self.pub.send(header, payload)
             self._step_buf = []
 

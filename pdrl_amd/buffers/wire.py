@@ -1,0 +1,62 @@
+"""Packed rollout wire format.
+
+A rollout chunk as a list of per-step dicts costs the storage shard one
+pickle object graph per step × 10 numpy arrays per dict — the decode is the
+measured per-shard ingest ceiling (profiles/ingest_shards.md). Packing the
+chunk into ONE (n_steps, row_width) float32 array + a uuid list makes the
+payload a single numpy buffer: decode is one allocation and n×10 array
+VIEWS. The storage side stays compatible with unpacked lists (tests and
+third-party producers can still send plain step dicts).
+
+Row layout (fixed field order; widths vary per env/model and ride along
+once per chunk): obs | act | rew | logits | log_prob | is_fir | done |
+hx | cx.
+"""
+from __future__ import annotations
+
+import numpy as np
+
+FIELD_ORDER = ("obs", "act", "rew", "logits", "log_prob", "is_fir", "done",
+               "hx", "cx")
+
+
+def pack_steps(steps: list[dict]) -> dict:
+    """[{field: array-like} × n] → {"ids", "widths", "pk"} (one float32 mat)."""
+    first = steps[0]
+    widths = [int(np.asarray(first[k], dtype=np.float32).reshape(-1).shape[0])
+              for k in FIELD_ORDER]
+    W = sum(widths)
+    pk = np.empty((len(steps), W), dtype=np.float32)
+    for i, s in enumerate(steps):
+        off = 0
+        row = pk[i]
+        for k, w in zip(FIELD_ORDER, widths):
+            row[off:off + w] = np.asarray(s[k], dtype=np.float32).reshape(-1)
+            off += w
+    return {"ids": [s["id"] for s in steps], "widths": widths, "pk": pk}
+
+
+def unpack_steps(obj: dict) -> list[dict]:
+    """Inverse of pack_steps. Field values are VIEWS into the chunk matrix
+    (zero copies; the assembler stacks them into owned trajectory tensors)."""
+    pk = obj["pk"]
+    widths = obj["widths"]
+    offs = np.cumsum([0] + list(widths))
+    out = []
+    for i, eid in enumerate(obj["ids"]):
+        row = pk[i]
+        step = {
+            k: row[offs[j]:offs[j + 1]]
+            for j, k in enumerate(FIELD_ORDER)
+        }
+        # scalar fields ride as 1-wide vectors; consumers index/flatten anyway
+        step["rew"] = float(step["rew"][0])
+        step["done"] = float(step["done"][0])
+        step["is_fir"] = float(step["is_fir"][0])
+        step["id"] = eid
+        out.append(step)
+    return out
+
+
+def is_packed(payload) -> bool:
+    return isinstance(payload, dict) and "pk" in payload and "ids" in payload

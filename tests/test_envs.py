@@ -125,7 +125,10 @@ def test_worker_ou_warmup_exploration():
             break
         proto, data = decode(*msg)
         if proto is Protocol.Rollout:
-            for step in data:
+            from pdrl_amd.buffers.wire import is_packed, unpack_steps
+
+            steps = unpack_steps(data) if is_packed(data) else data
+            for step in steps:
                 acts.append(float(step["act"][0]))
                 assert set(step) == {
                     "obs", "act", "rew", "logits", "log_prob", "is_fir",
@@ -179,7 +182,10 @@ def test_vectorized_worker_collects_complete_trajectories():
                 break
             proto, data = decode(*msg)
             if proto is Protocol.Rollout:
-                for step in data:
+                from pdrl_amd.buffers.wire import is_packed, unpack_steps
+
+                steps = unpack_steps(data) if is_packed(data) else data
+                for step in steps:
                     uuids.add(step["id"])
                     n_steps += 1
                     await asm.push(step)
