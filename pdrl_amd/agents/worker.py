@@ -70,6 +70,13 @@ class Worker:
         # Ornstein-Uhlenbeck process instead (correlated bang-bang), with the
         # record's log_prob still evaluated under the CURRENT policy.
         self.explore_warmup_steps = int(getattr(params, "explore_warmup_steps", 0))
+        if self.explore_warmup_steps:
+            from pdrl_amd.agents.learner_module import is_on_policy
+
+            # off-policy only: PPO-C/V-MPO importance ratios assume actions
+            # came from the recorded policy, which OU actions do not
+            if is_on_policy(getattr(params, "algo", "")):
+                self.explore_warmup_steps = 0
         self._ou_theta = float(getattr(params, "explore_ou_theta", 0.15))
         self._ou_sigma = float(getattr(params, "explore_ou_sigma", 0.6))
         self._continuous = bool(getattr(params, "continuous", False))
