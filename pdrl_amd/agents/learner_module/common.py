@@ -143,7 +143,12 @@ class BaseUpdater:
             if k in ckpt.get("optimizers", {}):
                 try:
                     o.load_state_dict(ckpt["optimizers"][k])
-                except Exception:
-                    pass
+                except Exception as e:  # e.g. fused↔eager optimizer switch
+                    import warnings
+
+                    warnings.warn(
+                        f"checkpoint optimizer state for '{k}' not restored "
+                        f"({type(e).__name__}: {e}); continuing with fresh "
+                        "optimizer state")
         self.update_count = int(ckpt.get("update_count", 0))
         return ckpt
