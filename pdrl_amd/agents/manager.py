@@ -54,31 +54,6 @@ class Manager:
             self._route[peer_id] = shard
         return shard
 
-    def relay_once(self, timeout: float = 0.5) -> bool:
-        """Handle one inbound message. Returns True if one was processed."""
-        msg = self.sub.recv(timeout=timeout, with_peer=True)
-        if msg is None:
-            return False
-        peer_id, header, payload = msg
-        protocol = pickle.loads(header)  # header-only inspect: rollouts are
-        if protocol is Protocol.Rollout:  # forwarded without decompressing
-            self.pubs[self._shard_of(peer_id)].send(header, payload)
-        elif protocol is Protocol.Stat:
-            _, data = decode(header, payload)
-            self.game_count += 1
-            self.stat_q.append(float(data["epi_rew"]))
-            if len(self.stat_q) >= self.STAT_INTERVAL:
-                mean_rew = float(np.mean(self.stat_q))
-                out = encode(
-                    Protocol.Stat,
-                    {"game_count": self.game_count, "mean_stat": mean_rew},
-                )
-                self.pub.send(*out)
-                self.stat_q.clear()
-        else:
-            raise AssertionError(f"unexpected protocol at manager: {protocol}")
-        return True
-
     def relay_batch(self, timeout: float = 0.5) -> int:
         """Drain and forward a burst of inbound messages. Rollouts are grouped
         per shard and queued with one send_many each (one lock + one reactor
