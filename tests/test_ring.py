@@ -93,3 +93,29 @@ def test_drain_new_off_policy():
         ring.put(make_traj(v))
     out = ring.drain_new()
     np.testing.assert_array_equal(out["rew"][:, 0, 0], list(range(12, 20)))
+
+
+def test_drain_new_handles_producer_lap():
+    """Off-policy conveyor: if the producer laps the consumer, drain_new
+    resumes from the oldest still-present slot instead of re-reading
+    overwritten data."""
+    import numpy as np
+    import torch
+
+    from pdrl_amd.buffers import SharedRolloutRing
+
+    ring = SharedRolloutRing({"x": 2}, seq_len=3, capacity=4, on_policy=False)
+    def traj(v):
+        return {"x": torch.full((3, 2), float(v))}
+
+    for v in range(10):  # laps the 4-slot ring twice without any drain
+        ring.put(traj(v))
+    out = ring.drain_new()
+    # only the newest capacity=4 trajectories (6, 7, 8, 9) survive
+    assert out["x"].shape == (4, 3, 2)
+    np.testing.assert_allclose(out["x"][:, 0, 0], [6.0, 7.0, 8.0, 9.0])
+    # nothing new since
+    assert ring.drain_new() is None
+    ring.put(traj(10))
+    out = ring.drain_new()
+    np.testing.assert_allclose(out["x"][:, 0, 0], [10.0])

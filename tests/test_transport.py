@@ -133,3 +133,31 @@ def test_send_many_preserves_order_and_framing():
     assert vals == list(range(32))
     pub.close()
     sub.close()
+
+
+def test_connect_side_reconnects_after_listener_restart():
+    """A connect-mode endpoint transparently reconnects when its peer dies
+    and comes back on the same port (worker behavior across a learner
+    restart — the supervisor respawn path depends on this)."""
+    import socket
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+
+    sub1 = Endpoint(bind=("127.0.0.1", port))
+    pub = Endpoint(connect=("127.0.0.1", port))
+    assert _wait(lambda: sub1.n_peers() == 1)
+    pub.send(*encode(Protocol.Stat, {"i": 1}))
+    assert sub1.recv(timeout=5.0) is not None
+    sub1.close()  # listener dies
+    assert _wait(lambda: pub.n_peers() == 0)
+
+    sub2 = Endpoint(bind=("127.0.0.1", port))  # listener returns
+    assert _wait(lambda: sub2.n_peers() == 1), "no reconnect"
+    pub.send(*encode(Protocol.Stat, {"i": 2}))
+    msg = sub2.recv(timeout=5.0)
+    assert msg is not None and decode(*msg)[1]["i"] == 2
+    pub.close()
+    sub2.close()
