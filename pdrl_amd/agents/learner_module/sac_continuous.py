@@ -32,7 +32,10 @@ class SACContinuousUpdater(BaseUpdater):
             p.requires_grad_(False)
 
         n_actions = self.actor.n_outputs
-        self.target_entropy = -float(n_actions)
+        # default -dim(A) (the SAC heuristic); overridable — on sparse
+        # exploration tasks a lower target lets alpha decay slower
+        self.target_entropy = float(
+            getattr(params, "target_entropy", None) or -float(n_actions))
         self.log_alpha = torch.nn.Parameter(
             torch.tensor(float(np.log(params.alpha)), device=self.device)
         )

@@ -70,6 +70,11 @@ class Worker:
         # Ornstein-Uhlenbeck process instead (correlated bang-bang), with the
         # record's log_prob still evaluated under the CURRENT policy.
         self.explore_warmup_steps = int(getattr(params, "explore_warmup_steps", 0))
+        # dedicated explorers: workers with idx < explore_ou_workers use OU
+        # actions FOREVER, keeping fresh success data in the replay even
+        # after the trained policy takes over the rest of the fleet
+        if worker_idx < int(getattr(params, "explore_ou_workers", 0) or 0):
+            self.explore_warmup_steps = 1 << 62
         if self.explore_warmup_steps:
             from pdrl_amd.agents.learner_module import is_on_policy
 
