@@ -44,6 +44,29 @@ class SACContinuousUpdater(BaseUpdater):
         self.critic_optimizer = self.make_optimizer("adam", self.critic.parameters(), lr=params.lr)
         self.alpha_optimizer = self.make_optimizer("adam", [self.log_alpha], lr=params.lr,
                                                    clip=False)
+        self.fused_step = self._make_fused_step()
+
+    def _make_fused_step(self):
+        """Whole-step fused HIP DAG (ops/sacc_step.py; kernel math verified
+        vs autograd in tests/test_sacc_analytic.py)."""
+        import os
+
+        from pdrl_amd import ops
+
+        if not (self.device.type == "cuda" and ops.available()):
+            return None
+        if not all(getattr(o, "is_fused", False) for o in
+                   (self.actor_optimizer, self.critic_optimizer,
+                    self.alpha_optimizer)):
+            return None
+        if self.actor.core.head_names != ["mu", "log_std"]:
+            return None
+        if not bool(int(os.environ.get("PDRL_SACC_FUSED", "1"))):
+            return None
+        from pdrl_amd.ops.sacc_step import FusedSacContinuousStep
+
+        use_graph = bool(int(os.environ.get("PDRL_USE_GRAPH", "1")))
+        return FusedSacContinuousStep(self, use_graph=use_graph)
 
     def trainable_modules(self):
         return {"model": self.model, "target_critic": self.target_critic}
@@ -63,6 +86,10 @@ class SACContinuousUpdater(BaseUpdater):
         return self.log_alpha.exp()
 
     def step(self, batch: dict[str, torch.Tensor]) -> dict:
+        if self.fused_step is not None and self.fused_step.fits(batch):
+            stats = self.fused_step.run(batch)
+            self.update_count += 1
+            return stats
         p = self.params
         obs, act = batch["obs"], batch["act"]
         rew = batch["rew"] * p.reward_scale

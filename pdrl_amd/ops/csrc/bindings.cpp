@@ -85,6 +85,20 @@ void sac_critic_loss_hip(const at::Tensor&, const at::Tensor&,
                          const at::Tensor&, at::Tensor&, at::Tensor&,
                          at::Tensor&, const c10::optional<at::Tensor>&,
                          double, double);
+void sacc_sample_hip(const at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
+                     at::Tensor&);
+void sacc_actor_grad_hip(const at::Tensor&, const at::Tensor&,
+                         const at::Tensor&, const at::Tensor&,
+                         const at::Tensor&, const at::Tensor&,
+                         const at::Tensor&, at::Tensor&, at::Tensor&,
+                         at::Tensor&, const c10::optional<at::Tensor>&,
+                         const c10::optional<at::Tensor>&, double);
+void sacc_critic_loss_hip(const at::Tensor&, const at::Tensor&,
+                          const at::Tensor&, const at::Tensor&,
+                          const at::Tensor&, const at::Tensor&,
+                          const at::Tensor&, const at::Tensor&, at::Tensor&,
+                          at::Tensor&, at::Tensor&,
+                          const c10::optional<at::Tensor>&, double, double);
 void l2norm_sq_hip(const at::Tensor&, at::Tensor&);
 void rmsprop_step_hip(at::Tensor&, const at::Tensor&, at::Tensor&,
                       const at::Tensor&, double, double, double, double);
@@ -131,6 +145,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "SAC discrete actor+alpha loss with analytic grads");
   m.def("sac_critic_loss", &sac_critic_loss_hip,
         "SAC discrete soft-Q target + twin critic loss grads");
+  m.def("sacc_sample", &sacc_sample_hip,
+        "reparameterized tanh-Gaussian sample + log-prob (graph-safe RNG)");
+  m.def("sacc_actor_grad", &sacc_actor_grad_hip,
+        "SAC-continuous actor+alpha loss with analytic dmu/dlog_std");
+  m.def("sacc_critic_loss", &sacc_critic_loss_hip,
+        "SAC-continuous soft-Q target + twin critic loss grads");
   m.def("l2norm_sq", &l2norm_sq_hip, "squared L2 norm into a device scalar");
   m.def("rmsprop_step", &rmsprop_step_hip, "fused clip+RMSprop on flat buffers");
   m.def("adam_step", &adam_step_hip, "fused clip+Adam on flat buffers");

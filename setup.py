@@ -21,6 +21,7 @@ SRC = [
     "pdrl_amd/ops/csrc/vmpo_loss.hip",
     "pdrl_amd/ops/csrc/ppoc_loss.hip",
     "pdrl_amd/ops/csrc/sac_loss.hip",
+    "pdrl_amd/ops/csrc/sacc_loss.hip",
     "pdrl_amd/ops/csrc/scans.hip",
     "pdrl_amd/ops/csrc/multi_tensor.hip",
 ]
