@@ -112,6 +112,12 @@ def main():
         procs.append(proc)
         return proc
 
+    # SIGTERM must run the finally-block child cleanup (timeout(1) sends it)
+    def _on_term(*_):
+        raise SystemExit(143)
+
+    signal.signal(signal.SIGTERM, _on_term)
+
     t0 = time.monotonic()
     spawn("learner_sub_process", "127.0.0.1", lrn_port)
     spawn("manager_sub_process", "127.0.0.1", "127.0.0.1", mgr_port, lrn_port)
