@@ -3,10 +3,10 @@
 Forward: ONE kernel computes body GEMM+ReLU, the whole LSTM sequence, and all
 heads (concatenated along the output dim), stashing activations for backward.
 Backward: ONE kernel runs BPTT through heads/recurrence/body per batch row
-(emitting per-step pre-activation gate grads), then the weight gradients are
-plain library GEMMs (hipBLASLt via torch.matmul) over the stashed
-activations — GEMM-shaped reductions belong on the matrix cores, and library
-GEMMs are the sanctioned path for plain GEMMs.
+(emitting per-step pre-activation gate grads), then the weight gradients run
+on the hand-written MFMA kernels (csrc/wgrad.hip: v_mfma_f32_16x16x4_f32
+tiles for the gate-weight GEMMs, wave-per-element reductions for the small
+grads) over the stashed activations.
 """
 from __future__ import annotations
 
