@@ -70,6 +70,25 @@ def make_synthetic_batch(params, device, seed):
     return {k: v.to(device) for k, v in batch.items()}
 
 
+def _self_exec_torchrun(n: int):
+    """--gpus N launched as a plain `python bench.py`: re-exec under
+    torch.distributed.run so N ranks actually do the work. Without this,
+    one rank would run and the whole-job aggregate below would overstate
+    throughput N×."""
+    import socket
+    import subprocess
+
+    with socket.socket() as s:  # free rendezvous port
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+        f"--nproc-per-node={n}", "--master-addr=127.0.0.1",
+        f"--master-port={port}", __file__, *sys.argv[1:],
+    ]
+    raise SystemExit(subprocess.call(cmd))
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -77,13 +96,20 @@ def main():
     ap.add_argument("--warmup", type=int, default=50)
     ap.add_argument("--batch-size", type=int, default=128)
     ap.add_argument("--seq-len", type=int, default=5)
+    ap.add_argument("--mode", default="resident", choices=["resident", "staged"],
+                    help="resident: batch lives on-device (kernel-DAG ceiling); "
+                         "staged: ring drain + pinned H2D + step + weight "
+                         "publish per iteration (system throughput)")
     ap.add_argument("--algo", default="IMPALA",
                     choices=["IMPALA", "PPO", "V-MPO", "SAC",
                              "PPO-Continuous", "SAC-Continuous"])
     args = ap.parse_args()
 
+    if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
+        _self_exec_torchrun(args.gpus)
+
     rank, world = init_distributed()
-    n_gpus = max(args.gpus, world)
+    n_gpus = world  # the ranks actually doing work — never trust --gpus
     use_cuda = torch.cuda.is_available()
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
     device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
@@ -107,7 +133,33 @@ def main():
 
         updater = maybe_graph(updater, device)
 
-    batch = make_synthetic_batch(params, device, seed=100 + rank)
+    if args.mode == "staged":
+        # system throughput: per iteration = pinned-host fill + async H2D
+        # staging + training step + actor weight publish (D2H + encode + TCP
+        # send on a bound PUB — the reference publishes after every update,
+        # ppo/learning.py:108)
+        from pdrl_amd.agents.learner import BatchStager
+        from pdrl_amd.transport import pub_bind
+        from pdrl_amd.utils import Protocol, encode
+
+        host_np = {k: v.numpy()
+                   for k, v in make_synthetic_batch(params, "cpu", 100 + rank).items()}
+        stager = BatchStager(device)
+        pub = pub_bind("127.0.0.1", 35000 + 37 * rank) if rank == 0 else None
+
+        def step_fn():
+            dev_batch = stager.stage(host_np)
+            updater.step(dev_batch)
+            if pub is not None:
+                header, payload = encode(Protocol.Model,
+                                         updater.actor_state_dict(),
+                                         compress=False)
+                pub.send(header, payload)
+    else:
+        batch = make_synthetic_batch(params, device, seed=100 + rank)
+
+        def step_fn():
+            updater.step(batch)
 
     import torch.distributed as dist
 
@@ -120,14 +172,14 @@ def main():
             torch.cuda.synchronize(device)
 
     for _ in range(args.warmup):
-        updater.step(batch)
+        step_fn()
     sync()
     barrier()
     sync()
 
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        updater.step(batch)
+        step_fn()
     sync()
     barrier()
     sync()
@@ -163,6 +215,7 @@ def main():
                 "global_batch": args.batch_size * n_gpus,
                 "seq_len": args.seq_len,
                 "parallelism": f"dp{n_gpus}",
+                "mode": args.mode,
             },
         }
         print(json.dumps(result))

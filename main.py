@@ -141,9 +141,16 @@ def storage_run(ring, learner_ip, learner_port, params, shared_stat, heartbeat, 
 
 
 def learner_run(ring, learner_ip, learner_port, params, shared_stat, heartbeat, stop_event,
-                rank, world_size, resume_path):
-    from pdrl_amd.agents import Learner
+                rank, world_size):
+    from pdrl_amd.agents import Learner, find_latest_checkpoint
     from pdrl_amd.parallel import GradReducer, init_distributed
+
+    # Resolve the newest checkpoint HERE, at every (re)start: when the
+    # Supervisor respawns a crashed learner this picks up the checkpoints
+    # written during the run instead of re-initializing from scratch (a
+    # frozen resume_path in the spawn args would silently reset training
+    # and broadcast random weights to the fleet).
+    resume_path = find_latest_checkpoint(params.model_dir, params.algo)
 
     if not torch.cuda.is_available():
         # CPU learner shares cores with the worker fleet: cap the OMP pool
@@ -238,7 +245,6 @@ class Supervisor:
 # --------------------------------------------------------------------------- #
 @register
 def learner_sub_process(learner_ip, learner_port, *_):
-    from pdrl_amd.agents import find_latest_checkpoint
     from pdrl_amd.agents.learner_module import is_on_policy
     from pdrl_amd.buffers import SharedRolloutRing, rollout_fields
 
@@ -274,12 +280,11 @@ def learner_sub_process(learner_ip, learner_port, *_):
         world_size = min(world_size, torch.cuda.device_count())
     else:
         world_size = 1
-    resume = find_latest_checkpoint(params.model_dir, params.algo)
     for rank in range(world_size):
         hb = mp.Value("d", time.time())
         sup.spawn(f"learner-{rank}", learner_run,
                   (ring, learner_ip, learner_port, params, shared_stat, hb, stop_event,
-                   rank, world_size, resume),
+                   rank, world_size),
                   heartbeat=hb)
     sup.monitor()
 
@@ -303,8 +308,11 @@ def worker_sub_process(num_p, manager_ip, learner_ip, port, learner_port, *_):
     params = probe_env_spaces(Params)
     model = build_model(params)
     load_actor_weights(model, params)
+    # Each spawned worker gets its OWN pickled copy of this CPU model (no
+    # share_memory: shared storages would make every worker's weight
+    # hot-reload write the same tensors while siblings run inference —
+    # transient torn weights mid-forward).
     model = model.cpu().eval()
-    model.share_memory()
 
     stop_event = mp.Event()
     sup = Supervisor(stop_event)

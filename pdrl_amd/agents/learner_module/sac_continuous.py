@@ -33,9 +33,10 @@ class SACContinuousUpdater(BaseUpdater):
 
         n_actions = self.actor.n_outputs
         # default -dim(A) (the SAC heuristic); overridable — on sparse
-        # exploration tasks a lower target lets alpha decay slower
-        self.target_entropy = float(
-            getattr(params, "target_entropy", None) or -float(n_actions))
+        # exploration tasks a lower target lets alpha decay slower.
+        # `is not None` (not `or`): an explicit 0.0 is a valid target.
+        te = getattr(params, "target_entropy", None)
+        self.target_entropy = float(te) if te is not None else -float(n_actions)
         self.log_alpha = torch.nn.Parameter(
             torch.tensor(float(np.log(params.alpha)), device=self.device)
         )

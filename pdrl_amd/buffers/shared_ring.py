@@ -2,8 +2,10 @@
 
 Replaces the reference's 8 flat unlocked ``mp.Array`` buffers + bare write
 cursor (reference: agents/storage_module/shared_batch.py:19-107, the
-by-convention race SURVEY.md §5 documents) with an explicit single-producer /
-single-consumer ring:
+by-convention race SURVEY.md §5 documents) with an explicit lock-guarded
+ring. Multi-producer / multi-consumer: N storage shards push and N learner
+ranks drain concurrently — every cursor mutation happens under the one
+``mp.Lock``:
 
 * one float32 ``mp.Array`` per trajectory field, shaped (capacity, seq, dim);
 * monotonic ``head`` (total written) and ``consumed`` counters guarded by an
@@ -17,7 +19,7 @@ single-consumer ring:
 
 The learner side reads into torch tensors via ``np.frombuffer`` views — and
 on GPU the batch is staged through a pinned-host buffer + async H2D copy
-(see pdrl_amd/ops/staging.py), replacing the reference's
+(``BatchStager`` in pdrl_amd/agents/learner.py), replacing the reference's
 shm→np→torch→.to(device) chain (reference: learner.py:197-233).
 """
 from __future__ import annotations

@@ -153,7 +153,10 @@ class FusedOnPolicyStep(GraphableStep):
                 p.policy_loss_coef, p.value_loss_coef, creg, p.coef_eta,
                 p.coef_alpha_below, p.coef_alpha_upper,
             )
-            assert ok, "V-MPO shape exceeds the fused-loss LDS budget"
+            if not ok:  # unreachable: updaters gate on fits() and fall back
+                raise RuntimeError(
+                    "vmpo_loss_mega refused a shape fits() accepted — "
+                    "fits() is out of sync with the kernel's LDS check")
             if norm is None:
                 # multi-rank: eta/alpha grads are in the flat bucket already
                 pass
@@ -164,7 +167,10 @@ class FusedOnPolicyStep(GraphableStep):
                 p.gamma, p.lmbda, p.reward_scale, p.policy_loss_coef,
                 p.value_loss_coef, p.entropy_coef, p.eps_clip, creg,
             )
-            assert ok, "PPO-C shape exceeds the fused-loss LDS budget"
+            if not ok:  # unreachable: updaters gate on fits() and fall back
+                raise RuntimeError(
+                    "ppoc_loss_mega refused a shape fits() accepted — "
+                    "fits() is out of sync with the kernel's LDS check")
             return gouts
         if self.algo == "IMPALA":
             if e.impala_loss_mega(

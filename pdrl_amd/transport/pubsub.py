@@ -24,6 +24,16 @@ the system relies on:
 * A message is a (header, payload) byte pair — the ZMQ multipart shape the
   Protocol enum + encode/decode produce.
 
+TRUST BOUNDARY: like the reference's pickle-over-ZMQ design (reference:
+utils/utils.py:244-249), message payloads are pickled — ``decode`` runs
+``pickle.loads`` on bytes from any peer that can reach a bound port, which
+is arbitrary code execution for anything with network access to the
+cluster. Deploy ONLY on a private/trusted network segment (the reference's
+machines.json topology assumes the same); bind to loopback or an internal
+interface, never a public one. Rollout payloads already travel as packed
+float matrices (buffers/wire.py) — the pickle surface is the small header
+and the weight dict.
+
 Design: ONE epoll/selector reactor thread per Endpoint handles accept,
 reads, writes and reconnects for ALL peers (non-blocking sockets, framed
 parsing). A thread-per-peer version measured catastrophic GIL convoy on a

@@ -46,7 +46,12 @@ def load_params(path: str | os.PathLike | None = None) -> SimpleNamespace:
     p = Path(path) if path is not None else UTILS_DIR / "parameters.json"
     with open(p) as f:
         params = _ns(json.load(f))
-    # Derived paths (reference: utils/utils.py:79-81): results/<ts>/models
+    # Derived paths (reference: utils/utils.py:79-81): results/<ts>/models.
+    # An explicitly configured result_dir/model_dir is REMEMBERED so
+    # refresh_result_dirs never clobbers it (that's what makes checkpoint
+    # resume reachable: point model_dir at a previous run's models/).
+    params.explicit_dirs = getattr(params, "result_dir", None) is not None \
+        or getattr(params, "model_dir", None) is not None
     if getattr(params, "result_dir", None) is None:
         ts = time.strftime("%Y-%m-%d_%H-%M-%S")
         params.result_dir = str(REPO_ROOT / "results" / ts)
@@ -72,8 +77,12 @@ Machines = load_machines(os.environ.get("PDRL_MACHINES"))
 
 
 def refresh_result_dirs(params=None):
-    """Stamp a fresh results/<ts> tree for a new training run."""
+    """Stamp a fresh results/<ts> tree for a new training run — unless the
+    config explicitly pinned result_dir/model_dir (then they are respected,
+    so a resume from that directory's checkpoints can actually trigger)."""
     params = params or Params
+    if getattr(params, "explicit_dirs", False):
+        return params
     ts = time.strftime("%Y-%m-%d_%H-%M-%S")
     params.result_dir = str(REPO_ROOT / "results" / ts)
     params.model_dir = str(Path(params.result_dir) / "models")
