@@ -62,10 +62,61 @@ def register(name):
     return deco
 
 
+def _convert_space(sp) -> Space:
+    """Translate a gymnasium space into the native Space types the rest of
+    the framework type-checks against (EnvBase / probe_env_spaces use
+    ``isinstance(..., Box)``)."""
+    if hasattr(sp, "n"):
+        return Discrete(int(sp.n))
+    if hasattr(sp, "low") and hasattr(sp, "high"):
+        return Box(sp.low, sp.high, shape=getattr(sp, "shape", None))
+    raise TypeError(f"unsupported gymnasium space type: {type(sp).__name__}")
+
+
+class GymnasiumAdapter:
+    """Any-gymnasium-env support behind the native registry (the reference
+    wraps arbitrary gym envs — agents/worker_module/env_maker.py:6-31).
+    Exposes the native API: ``seed()`` (applied at the next reset, the
+    gymnasium convention), native Space objects, and the standard
+    reset/step tuples, which gymnasium already produces."""
+
+    def __init__(self, env):
+        self.env = env
+        self._pending_seed: int | None = None
+        self.observation_space = _convert_space(env.observation_space)
+        self.action_space = _convert_space(env.action_space)
+
+    def seed(self, seed: int):
+        self._pending_seed = int(seed)
+
+    def reset(self):
+        if self._pending_seed is not None:
+            seed, self._pending_seed = self._pending_seed, None
+            return self.env.reset(seed=seed)
+        return self.env.reset()
+
+    def step(self, action):
+        return self.env.step(action)
+
+    def close(self):
+        if hasattr(self.env, "close"):
+            self.env.close()
+
+
 def make(env_name: str, **kwargs):
-    """gym.make-equivalent for the built-in environment set."""
+    """gym.make-equivalent: native envs first, then any gymnasium env when
+    gymnasium is importable (absent in this container — the native
+    CartPole/MountainCarContinuous implementations cover the reference's
+    target envs without it)."""
     from . import cartpole, mountain_car, fake  # noqa: F401  (populate registry)
 
-    if env_name not in _REGISTRY:
-        raise ValueError(f"unknown env '{env_name}'; available: {sorted(_REGISTRY)}")
-    return _REGISTRY[env_name](**kwargs)
+    if env_name in _REGISTRY:
+        return _REGISTRY[env_name](**kwargs)
+    try:
+        import gymnasium
+    except ImportError:
+        raise ValueError(
+            f"unknown env '{env_name}'; native envs: {sorted(_REGISTRY)} "
+            "(gymnasium not importable, so arbitrary gym envs are unavailable)"
+        ) from None
+    return GymnasiumAdapter(gymnasium.make(env_name, **kwargs))

@@ -92,6 +92,64 @@ def test_make_unknown_env():
         envs.make("Atari-Breakout")
 
 
+def test_gymnasium_adapter_wraps_gym_style_env():
+    """envs.make resolves arbitrary gymnasium envs when gymnasium is
+    importable (reference: agents/worker_module/env_maker.py:6-31). The
+    container has no gymnasium, so exercise the adapter directly with a
+    gym-API stub: space conversion to native types, seed-at-next-reset,
+    and reset/step passthrough."""
+    from pdrl_amd.envs.base import Box, Discrete, GymnasiumAdapter
+
+    class _GymDiscrete:
+        n = 3
+
+    class _GymBox:
+        low = np.array([-1.0, -2.0], dtype=np.float32)
+        high = np.array([1.0, 2.0], dtype=np.float32)
+        shape = (2,)
+
+    class _GymEnv:
+        observation_space = _GymBox()
+        action_space = _GymDiscrete()
+
+        def __init__(self):
+            self.reset_seeds = []
+
+        def reset(self, seed=None):
+            self.reset_seeds.append(seed)
+            return np.zeros(2, dtype=np.float32), {}
+
+        def step(self, action):
+            return np.ones(2, dtype=np.float32), 1.0, False, False, {"a": action}
+
+    inner = _GymEnv()
+    env = GymnasiumAdapter(inner)
+    assert isinstance(env.action_space, Discrete) and env.action_space.n == 3
+    assert isinstance(env.observation_space, Box)
+    assert env.observation_space.shape == (2,)
+
+    env.seed(42)
+    env.reset()
+    env.reset()
+    assert inner.reset_seeds == [42, None]  # seed applies once, at next reset
+
+    obs, rew, term, trunc, info = env.step(1)
+    assert rew == 1.0 and info["a"] == 1 and not (term or trunc)
+
+    # EnvBase composes with the adapter (done-merge, tensor obs)
+    from pdrl_amd.agents.env_maker import EnvBase
+
+    eb = EnvBase.__new__(EnvBase)
+    eb.env = env
+    eb.observation_space = env.observation_space
+    eb.action_space = env.action_space
+    eb.continuous = False
+    obs = eb.reset()
+    assert tuple(obs.shape)[-1] == 2  # obs_preprocess adds the batch dim
+    obs, rew, done, info = eb.step(1)
+    assert done is False and rew == 1.0
+
+
 def test_worker_ou_warmup_exploration():
     """With explore_warmup_steps set on a continuous env, actions come from a
     temporally-correlated OU process (in [-1,1], correlated across steps) and
