@@ -128,10 +128,9 @@ def main():
     model = model_cls(obs_dim, n_act, params.seq_len, params.hidden_size)
     reducer = GradReducer() if world > 1 else None
     updater = updater_cls(model, params, device, grad_reducer=reducer)
-    if world == 1:  # eager-path algos: whole-step hipGraph capture
-        from pdrl_amd.ops.graphed import maybe_graph
+    from pdrl_amd.ops.graphed import maybe_graph
 
-        updater = maybe_graph(updater, device)
+    updater = maybe_graph(updater, device)
 
     if args.mode == "staged":
         # system throughput: per iteration = pinned-host fill + async H2D

@@ -119,10 +119,12 @@ class Learner:
         self.updater = updater_cls(model, params, self.device, grad_reducer=grad_reducer)
         if resume_path:
             self.updater.load(resume_path, map_location=self.device)
-        if world_size == 1:  # eager-path algos: whole-step hipGraph capture
-            from pdrl_amd.ops.graphed import maybe_graph
+        # eager-path algos: whole-step hipGraph capture (any world size —
+        # the RCCL collective captures with the graph; failure on a rank
+        # falls back to stream-ordered with identical collective order)
+        from pdrl_amd.ops.graphed import maybe_graph
 
-            self.updater = maybe_graph(self.updater, self.device)
+        self.updater = maybe_graph(self.updater, self.device)
 
         self.on_policy = is_on_policy(params.algo)
         self.stager = BatchStager(self.device)

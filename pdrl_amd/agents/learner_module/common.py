@@ -74,13 +74,19 @@ class BaseUpdater:
             return None
         from pdrl_amd.ops.fused_step import FusedOnPolicyStep
 
-        # graph replay is single-rank only for now: capturing the RCCL
-        # all-reduce is unverified on this stack (stream-ordered fused path
-        # runs for multi-rank; correctness identical)
-        use_graph = (
-            bool(int(__import__("os").environ.get("PDRL_USE_GRAPH", "1")))
-            and self.grad_reducer is None
-        )
+        # multi-rank capture INCLUDES the RCCL all-reduce in the graph (the
+        # communicator is warmed up by the pre-capture side-stream runs). If
+        # capture fails on a rank it falls back to the stream-ordered fused
+        # path — correct either way, because graphed and non-graphed ranks
+        # issue the identical collective sequence. PDRL_GRAPH_RCCL=0 forces
+        # stream-ordered for all multi-rank steps.
+        import os
+
+        use_graph = bool(int(os.environ.get("PDRL_USE_GRAPH", "1")))
+        if self.grad_reducer is not None and not bool(
+            int(os.environ.get("PDRL_GRAPH_RCCL", "1"))
+        ):
+            use_graph = False
         return FusedOnPolicyStep(algo, core, self.params, optimizer,
                                  grad_reducer=self.grad_reducer,
                                  use_graph=use_graph, duals=duals)

@@ -78,6 +78,16 @@ class GradReducer:
         grads = [g for g in grads if g is not None]
         if not grads:
             return
+        if len(grads) == 1 and grads[0].is_contiguous():
+            # the fused optimizers hand over ONE flat grad buffer: reduce it
+            # in place, no staging copies (2 launches fewer per step)
+            flat = grads[0]
+            if hasattr(dist.ReduceOp, "AVG") and dist.get_backend(self.pg) == "nccl":
+                dist.all_reduce(flat, op=dist.ReduceOp.AVG, group=self.pg)
+            else:
+                dist.all_reduce(flat, op=dist.ReduceOp.SUM, group=self.pg)
+                flat.div_(self.world_size())
+            return
         numel = sum(g.numel() for g in grads)
         if self._flat is None or self._flat.numel() < numel or self._flat.device != grads[0].device:
             self._flat = torch.empty(numel, dtype=grads[0].dtype, device=grads[0].device)
