@@ -50,7 +50,8 @@ class BaseUpdater:
             if kind == "rmsprop":
                 return FusedRMSprop(parameters, lr=lr, max_norm=max_norm,
                                     eps=kw.get("eps", 1e-5))
-            return FusedAdam(parameters, lr=lr, max_norm=max_norm)
+            return FusedAdam(parameters, lr=lr, max_norm=max_norm,
+                             clock=kw.get("clock"))
         if kind == "rmsprop":
             return torch.optim.RMSprop(parameters, lr=lr, eps=kw.get("eps", 1e-5))
         return torch.optim.Adam(parameters, lr=lr)

@@ -48,10 +48,17 @@ class SACUpdater(BaseUpdater):
             torch.tensor(float(np.log(params.alpha)), device=self.device)
         )
 
-        self.actor_optimizer = self.make_optimizer("adam", self.actor.parameters(), lr=params.lr)
-        self.critic_optimizer = self.make_optimizer("adam", self.critic.parameters(), lr=params.lr)
-        self.alpha_optimizer = self.make_optimizer("adam", [self.log_alpha], lr=params.lr,
-                                                   clip=False)
+        # the three optimizers always step together once per iteration →
+        # they share ONE device step clock (t, bc1, bc2), advanced once
+        # per step; lets the fused DAG batch their updates into one launch
+        clock = torch.zeros(3, dtype=torch.float32, device=self.device) \
+            if self.device.type == "cuda" else None
+        self.actor_optimizer = self.make_optimizer(
+            "adam", self.actor.parameters(), lr=params.lr, clock=clock)
+        self.critic_optimizer = self.make_optimizer(
+            "adam", self.critic.parameters(), lr=params.lr, clock=clock)
+        self.alpha_optimizer = self.make_optimizer(
+            "adam", [self.log_alpha], lr=params.lr, clip=False, clock=clock)
         self.fused_step = self._make_sac_fused_step()
 
     def trainable_modules(self):

@@ -41,10 +41,16 @@ class SACContinuousUpdater(BaseUpdater):
             torch.tensor(float(np.log(params.alpha)), device=self.device)
         )
 
-        self.actor_optimizer = self.make_optimizer("adam", self.actor.parameters(), lr=params.lr)
-        self.critic_optimizer = self.make_optimizer("adam", self.critic.parameters(), lr=params.lr)
-        self.alpha_optimizer = self.make_optimizer("adam", [self.log_alpha], lr=params.lr,
-                                                   clip=False)
+        # one shared device Adam step clock across the three optimizers
+        # (they always step together): fused DAG ticks it once per step
+        clock = torch.zeros(3, dtype=torch.float32, device=self.device) \
+            if self.device.type == "cuda" else None
+        self.actor_optimizer = self.make_optimizer(
+            "adam", self.actor.parameters(), lr=params.lr, clock=clock)
+        self.critic_optimizer = self.make_optimizer(
+            "adam", self.critic.parameters(), lr=params.lr, clock=clock)
+        self.alpha_optimizer = self.make_optimizer(
+            "adam", [self.log_alpha], lr=params.lr, clip=False, clock=clock)
         self.fused_step = self._make_fused_step()
 
     def _make_fused_step(self):

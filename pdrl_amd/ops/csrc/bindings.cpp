@@ -104,7 +104,10 @@ void rmsprop_step_hip(at::Tensor&, const at::Tensor&, at::Tensor&,
                       const at::Tensor&, double, double, double, double);
 void adam_step_hip(at::Tensor&, const at::Tensor&, at::Tensor&, at::Tensor&,
                    at::Tensor&, const at::Tensor&, double, double, double,
-                   double, double);
+                   double, double, bool);
+void adam_prep_hip(at::Tensor&, double, double);
+void adam_multi_hip(const at::Tensor&, const at::Tensor&, const at::Tensor&,
+                    long, long, double, double, double);
 void soft_update_hip(const std::vector<at::Tensor>&,
                      const std::vector<at::Tensor>&, double);
 void soft_update_cached_hip(const at::Tensor&, const at::Tensor&,
@@ -153,7 +156,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "SAC-continuous soft-Q target + twin critic loss grads");
   m.def("l2norm_sq", &l2norm_sq_hip, "squared L2 norm into a device scalar");
   m.def("rmsprop_step", &rmsprop_step_hip, "fused clip+RMSprop on flat buffers");
-  m.def("adam_step", &adam_step_hip, "fused clip+Adam on flat buffers");
+  m.def("adam_step", &adam_step_hip, "fused clip+Adam on flat buffers",
+        py::arg("p"), py::arg("g"), py::arg("m"), py::arg("v"),
+        py::arg("state3"), py::arg("norm_sq"), py::arg("lr"), py::arg("beta1"),
+        py::arg("beta2"), py::arg("eps"), py::arg("max_norm"),
+        py::arg("do_prep") = true);
+  m.def("adam_prep", &adam_prep_hip,
+        "advance the shared device Adam step clock");
+  m.def("adam_multi", &adam_multi_hip,
+        "multi-group fused clip+Adam (shared step clock, pointer table)");
   m.def("soft_update", &soft_update_hip, "multi-tensor Polyak update");
   m.def("soft_update_cached", &soft_update_cached_hip,
         "Polyak update with prebuilt device pointer tables (graph-safe)");

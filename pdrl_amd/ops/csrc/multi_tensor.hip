@@ -90,6 +90,39 @@ __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
   }
 }
 
+// Multi-group Adam: one launch updates N flat param groups (pointer table
+// prebuilt on device, graph-safe). All groups share one device step clock
+// (state3) — valid because the framework always steps its optimizers
+// together once per iteration. Replaces SAC's 3×(prep+adam)=6 launches
+// with prep + ≤2 multi-launches.
+__global__ void adam_multi_kernel(
+    const long* __restrict__ ptrs,   // [G][5]: p, g, m, v, norm_sq(or 0)
+    const float* __restrict__ cfg,   // [G][3]: numel, lr, max_norm
+    const float* __restrict__ state3, int n_groups, float beta1, float beta2,
+    float eps) {
+  const int gi = blockIdx.y;
+  if (gi >= n_groups) return;
+  float* p = reinterpret_cast<float*>(ptrs[gi * 5 + 0]);
+  const float* g = reinterpret_cast<const float*>(ptrs[gi * 5 + 1]);
+  float* m = reinterpret_cast<float*>(ptrs[gi * 5 + 2]);
+  float* v = reinterpret_cast<float*>(ptrs[gi * 5 + 3]);
+  const float* nsq = reinterpret_cast<const float*>(ptrs[gi * 5 + 4]);
+  const long n = (long)cfg[gi * 3 + 0];
+  const float lr = cfg[gi * 3 + 1];
+  const float max_norm = cfg[gi * 3 + 2];
+  const float scale = (nsq != nullptr) ? clip_scale(nsq, max_norm) : 1.0f;
+  const float bc1 = state3[1], bc2 = state3[2];
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    const float gi_ = g[i] * scale;
+    const float mi = beta1 * m[i] + (1.0f - beta1) * gi_;
+    const float vi = beta2 * v[i] + (1.0f - beta2) * gi_ * gi_;
+    m[i] = mi;
+    v[i] = vi;
+    p[i] -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+  }
+}
+
 // chunk table: src/dst pointers packed as int64 in a device tensor
 __global__ void soft_update_kernel(const long* __restrict__ src_ptrs,
                                    const long* __restrict__ dst_ptrs,
@@ -139,16 +172,36 @@ void rmsprop_step_hip(at::Tensor& p, const at::Tensor& g, at::Tensor& sq_avg,
 void adam_step_hip(at::Tensor& p, const at::Tensor& g, at::Tensor& m,
                    at::Tensor& v, at::Tensor& state3,
                    const at::Tensor& norm_sq, double lr, double beta1,
-                   double beta2, double eps, double max_norm) {
+                   double beta2, double eps, double max_norm, bool do_prep) {
   CHECK_IN(p); CHECK_IN(g); CHECK_IN(m); CHECK_IN(v);
-  hipLaunchKernelGGL(adam_prep_kernel, dim3(1), dim3(1), 0, current_stream(),
-                     state3.data_ptr<float>(), (float)beta1, (float)beta2);
+  if (do_prep) {
+    hipLaunchKernelGGL(adam_prep_kernel, dim3(1), dim3(1), 0, current_stream(),
+                       state3.data_ptr<float>(), (float)beta1, (float)beta2);
+  }
   hipLaunchKernelGGL(adam_kernel, dim3(grid_for(p.numel())), dim3(kThreads), 0,
                      current_stream(), p.data_ptr<float>(),
                      g.data_ptr<float>(), m.data_ptr<float>(),
                      v.data_ptr<float>(), state3.data_ptr<float>(),
                      norm_sq.data_ptr<float>(), p.numel(), (float)lr,
                      (float)beta1, (float)beta2, (float)eps, (float)max_norm);
+  HIP_CHECK_LAST();
+}
+
+void adam_prep_hip(at::Tensor& state3, double beta1, double beta2) {
+  hipLaunchKernelGGL(adam_prep_kernel, dim3(1), dim3(1), 0, current_stream(),
+                     state3.data_ptr<float>(), (float)beta1, (float)beta2);
+  HIP_CHECK_LAST();
+}
+
+void adam_multi_hip(const at::Tensor& ptrs, const at::Tensor& cfg,
+                    const at::Tensor& state3, long n_groups, long max_numel,
+                    double beta1, double beta2, double eps) {
+  CHECK_GPU(ptrs); CHECK_GPU(cfg);
+  dim3 grid(grid_for(max_numel), (unsigned)n_groups);
+  hipLaunchKernelGGL(adam_multi_kernel, grid, dim3(kThreads), 0,
+                     current_stream(), ptrs.data_ptr<long>(),
+                     cfg.data_ptr<float>(), state3.data_ptr<float>(),
+                     (int)n_groups, (float)beta1, (float)beta2, (float)eps);
   HIP_CHECK_LAST();
 }
 

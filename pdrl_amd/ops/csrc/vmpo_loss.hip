@@ -34,6 +34,44 @@ __device__ __forceinline__ unsigned f2u(float f) {
 }
 
 constexpr int kThreads = 256;
+constexpr int kWaves = kThreads / kWave;
+
+// Wave-level reductions (no barriers): 6 shfl steps across the 64 lanes.
+__device__ __forceinline__ float wred_sum(float v) {
+#pragma unroll
+  for (int off = kWave / 2; off > 0; off >>= 1) v += __shfl_down(v, off, kWave);
+  return v;
+}
+__device__ __forceinline__ float wred_max(float v) {
+#pragma unroll
+  for (int off = kWave / 2; off > 0; off >>= 1)
+    v = fmaxf(v, __shfl_down(v, off, kWave));
+  return v;
+}
+
+// Block-wide sum broadcast to every thread: 2 barriers total (vs the 8-step
+// shared-memory tree = 9 barriers that made the original kernel sync-bound
+// at ~350 barriers/launch).
+__device__ __forceinline__ float block_sum(float v, float* s4) {
+  v = wred_sum(v);
+  __syncthreads();  // s4 may still be read from a previous reduction
+  if ((threadIdx.x & (kWave - 1)) == 0) s4[threadIdx.x >> 6] = v;
+  __syncthreads();
+  float r = 0.f;
+#pragma unroll
+  for (int w = 0; w < kWaves; ++w) r += s4[w];
+  return r;
+}
+__device__ __forceinline__ float block_max(float v, float* s4) {
+  v = wred_max(v);
+  __syncthreads();
+  if ((threadIdx.x & (kWave - 1)) == 0) s4[threadIdx.x >> 6] = v;
+  __syncthreads();
+  float r = -1e30f;
+#pragma unroll
+  for (int w = 0; w < kWaves; ++w) r = fmaxf(r, s4[w]);
+  return r;
+}
 
 __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
     const float* __restrict__ mo,      // (N,D) packed model out
@@ -63,8 +101,8 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
   float* s_adv = s_logp + N;                          // (BT)
   float* s_td = s_adv + BT;                           // (BT)
   float* s_psi = s_td + BT;                           // (BT) psi or 0
-  __shared__ float red[4][kThreads];
-  __shared__ unsigned ured[kThreads];
+  __shared__ float s4[kWaves];      // cross-wave combine scratch
+  __shared__ float s4b[3][kWaves];  // multi-value combine scratch
   __shared__ float s_scalars[8];  // {eps_alpha, thresh, m, Z, sumwa, kl, eta, alpha}
 
   if (tid == 0) {
@@ -107,57 +145,45 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
   }
   __syncthreads();
 
-  // phase C: k-th largest advantage via bit binary search (k = BT/2)
+  // phase C: k-th largest advantage via bit binary search (k = BT/2).
+  // Per-thread advantage bits live in REGISTERS (the LDS budget caps
+  // BT ≤ 2867, i.e. ≤ 12 elements/thread), and every reduction is
+  // wave-shfl + one 4-slot LDS combine — 2 barriers per search iteration
+  // instead of the 9-barrier shared-memory tree.
   const int K = BT / 2 > 0 ? BT / 2 : 1;
+  constexpr int kMaxPer = 12;
+  unsigned myu[kMaxPer];
+  int mycnt = 0;
+  for (int i = tid; i < BT; i += kThreads) myu[mycnt++] = f2u(s_adv[i]);
+  unsigned thresh_u;
   {
     unsigned lo = 0, hi = 0xFFFFFFFFu;
-    for (int it = 0; it < 32; ++it) {
+    while (lo < hi) {
       const unsigned mid = lo + ((hi - lo) >> 1);
       int cnt = 0;
-      for (int i = tid; i < BT; i += kThreads) {
-        if (f2u(s_adv[i]) > mid) ++cnt;
-      }
-      red[0][tid] = (float)cnt;
-      __syncthreads();
-      for (int off = kThreads / 2; off > 0; off >>= 1) {
-        if (tid < off) red[0][tid] += red[0][tid + off];
-        __syncthreads();
-      }
-      const int total_gt = (int)red[0][0];
-      __syncthreads();
+      for (int k = 0; k < mycnt; ++k) cnt += (myu[k] > mid);
+      const int total_gt = (int)block_sum((float)cnt, s4);
       if (total_gt >= K) {
         lo = mid + 1;  // threshold is higher
       } else {
         hi = mid;
       }
-      if (lo >= hi) break;
     }
     // hi = smallest u such that count(> u) < K → the K-th largest has bits hi
-    if (tid == 0) s_scalars[1] = __uint_as_float((hi & 0x80000000u) ? (hi & 0x7FFFFFFFu) : ~hi);
-    __syncthreads();
+    thresh_u = hi;  // identical on every thread: no broadcast needed
   }
-  const float thresh = s_scalars[1];
-  const unsigned thresh_u = f2u(thresh);
 
   // mark selected: strictly greater always; equal by ascending index to k
   {
     int cnt_gt = 0;
-    for (int i = tid; i < BT; i += kThreads)
-      if (f2u(s_adv[i]) > thresh_u) ++cnt_gt;
-    red[0][tid] = (float)cnt_gt;
-    __syncthreads();
-    for (int off = kThreads / 2; off > 0; off >>= 1) {
-      if (tid < off) red[0][tid] += red[0][tid + off];
-      __syncthreads();
-    }
-    const int n_gt = (int)red[0][0];
-    __syncthreads();
+    for (int k = 0; k < mycnt; ++k) cnt_gt += (myu[k] > thresh_u);
+    const int n_gt = (int)block_sum((float)cnt_gt, s4);
     int need_eq = K - n_gt;  // ties to include, by lowest index
-    // serial-ish tie resolution (ties are rare): thread 0 marks
     for (int i = tid; i < BT; i += kThreads) {
       s_psi[i] = (f2u(s_adv[i]) > thresh_u) ? 1.f : 0.f;
     }
     __syncthreads();
+    // serial-ish tie resolution (ties are rare): thread 0 marks
     if (tid == 0 && need_eq > 0) {
       for (int i = 0; i < BT && need_eq > 0; ++i) {
         if (f2u(s_adv[i]) == thresh_u) {
@@ -174,15 +200,7 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
     float mx = -1e30f;
     for (int i = tid; i < BT; i += kThreads)
       if (s_psi[i] > 0.f) mx = fmaxf(mx, s_adv[i]);
-    red[0][tid] = mx;
-    __syncthreads();
-    for (int off = kThreads / 2; off > 0; off >>= 1) {
-      if (tid < off) red[0][tid] = fmaxf(red[0][tid], red[0][tid + off]);
-      __syncthreads();
-    }
-    if (tid == 0) s_scalars[2] = red[0][0];
-    __syncthreads();
-    const float m = s_scalars[2];
+    const float m = block_max(mx, s4);
     const float eta = s_scalars[6];
     float z = 0.f, wa = 0.f;
     for (int i = tid; i < BT; i += kThreads) {
@@ -192,18 +210,22 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
         wa = fmaf(e, s_adv[i], wa);
       }
     }
-    red[0][tid] = z; red[1][tid] = wa;
+    // paired sum: 2 barriers for both values
+    z = wred_sum(z);
+    wa = wred_sum(wa);
     __syncthreads();
-    for (int off = kThreads / 2; off > 0; off >>= 1) {
-      if (tid < off) {
-        red[0][tid] += red[0][tid + off];
-        red[1][tid] += red[1][tid + off];
-      }
-      __syncthreads();
+    if ((tid & (kWave - 1)) == 0) {
+      s4b[0][tid >> 6] = z;
+      s4b[1][tid >> 6] = wa;
     }
-    if (tid == 0) { s_scalars[3] = red[0][0]; s_scalars[4] = red[1][0]; }
     __syncthreads();
-    const float Z = s_scalars[3];
+    float Z = 0.f, WA = 0.f;
+#pragma unroll
+    for (int w = 0; w < kWaves; ++w) {
+      Z += s4b[0][w];
+      WA += s4b[1][w];
+    }
+    if (tid == 0) { s_scalars[2] = m; s_scalars[3] = Z; s_scalars[4] = WA; }
     for (int i = tid; i < BT; i += kThreads) {
       if (s_psi[i] > 0.f)
         s_psi[i] = __expf((s_adv[i] - m) / eta) / Z;
@@ -213,7 +235,6 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
 
   // phase D: reductions — policy, value, KL, logit reg
   {
-    const float eta = s_scalars[6];
     float pl = 0.f, vl = 0.f, kl = 0.f, rg = 0.f;
     for (int i = tid; i < BT; i += kThreads) {
       const int b = i / T, t = i % T;
@@ -235,13 +256,21 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
         rg = fmaf(zq[j], zq[j], rg);
       }
     }
-    red[0][tid] = pl; red[1][tid] = vl; red[2][tid] = kl; red[3][tid] = rg;
+    // 4-value sum: 2 barriers
+    pl = wred_sum(pl); vl = wred_sum(vl); kl = wred_sum(kl); rg = wred_sum(rg);
     __syncthreads();
-    for (int off = kThreads / 2; off > 0; off >>= 1) {
-      if (tid < off)
-        for (int r = 0; r < 4; ++r) red[r][tid] += red[r][tid + off];
-      __syncthreads();
+    if ((tid & (kWave - 1)) == 0) {
+      const int w = tid >> 6;
+      s4[w] = pl; s4b[0][w] = vl; s4b[1][w] = kl; s4b[2][w] = rg;
     }
+    __syncthreads();
+    float red0 = 0.f, red1 = 0.f, red2 = 0.f, red3 = 0.f;
+#pragma unroll
+    for (int w = 0; w < kWaves; ++w) {
+      red0 += s4[w]; red1 += s4b[0][w]; red2 += s4b[1][w]; red3 += s4b[2][w];
+    }
+    float red[4][1];
+    red[0][0] = red0; red[1][0] = red1; red[2][0] = red2; red[3][0] = red3;
     if (tid == 0) {
       const float eta_v = s_scalars[6], alpha_v = s_scalars[7];
       const float m = s_scalars[2], Z = s_scalars[3], wa = s_scalars[4];

@@ -109,24 +109,67 @@ class FusedRMSprop(_FusedOptimizer):
 
 class FusedAdam(_FusedOptimizer):
     """torch.optim.Adam(lr, betas, eps) + clip_grad_norm_ fused.
-    Step count + bias corrections live on device (graph-replay safe)."""
+    Step count + bias corrections live on device (graph-replay safe).
 
-    def __init__(self, params, lr, betas=(0.9, 0.999), eps=1e-8, max_norm=None):
+    ``clock``: an optional SHARED state3 tensor (t, bc1, bc2). Optimizers
+    that always step together once per iteration can share one clock; the
+    owner advances it with ``tick()`` (or ``ext().adam_prep``) once per
+    step and each member updates with ``_update(prep=False)``."""
+
+    def __init__(self, params, lr, betas=(0.9, 0.999), eps=1e-8, max_norm=None,
+                 clock: torch.Tensor | None = None):
         super().__init__(params, lr, max_norm)
         self.beta1, self.beta2 = float(betas[0]), float(betas[1])
         self.eps = float(eps)
         self.exp_avg = torch.zeros_like(self.space.flat_param)
         self.exp_avg_sq = torch.zeros_like(self.space.flat_param)
-        self.state3 = torch.zeros(3, dtype=torch.float32,
-                                  device=self.space.flat_param.device)
+        self.shared_clock = clock is not None
+        self.state3 = clock if clock is not None else torch.zeros(
+            3, dtype=torch.float32, device=self.space.flat_param.device)
 
     def _state(self):
         return {"exp_avg": self.exp_avg, "exp_avg_sq": self.exp_avg_sq,
                 "state3": self.state3}
 
-    def _update(self):
+    def tick(self):
+        """Advance the device step clock (once per iteration when shared)."""
+        ext().adam_prep(self.state3, self.beta1, self.beta2)
+
+    def _update(self, prep: bool | None = None):
         ext().adam_step(
             self.space.flat_param, self.space.flat_grad, self.exp_avg,
             self.exp_avg_sq, self.state3, self.norm_sq, self.lr, self.beta1,
             self.beta2, self.eps, self.max_norm,
+            do_prep=(not self.shared_clock) if prep is None else prep,
         )
+
+
+class AdamMultiGroup:
+    """ONE launch updating several FusedAdam groups that share a step clock
+    (pointer + config tables prebuilt on device; graph-capture safe)."""
+
+    def __init__(self, opts: list[FusedAdam]):
+        assert opts and all(o.shared_clock for o in opts)
+        assert all(o.state3 is opts[0].state3 for o in opts)
+        assert all((o.beta1, o.beta2, o.eps) ==
+                   (opts[0].beta1, opts[0].beta2, opts[0].eps) for o in opts)
+        self.opts = opts
+        self.state3 = opts[0].state3
+        self.beta1, self.beta2, self.eps = (
+            opts[0].beta1, opts[0].beta2, opts[0].eps)
+        dev = opts[0].flat_param.device
+        rows = []
+        cfg = []
+        for o in opts:
+            nsq = o.norm_sq.data_ptr() if o.max_norm > 0 else 0
+            rows.append([o.flat_param.data_ptr(), o.flat_grad.data_ptr(),
+                         o.exp_avg.data_ptr(), o.exp_avg_sq.data_ptr(), nsq])
+            cfg.append([float(o.space.numel), o.lr, o.max_norm])
+        self.ptrs = torch.tensor(rows, dtype=torch.int64).to(dev)
+        self.cfg = torch.tensor(cfg, dtype=torch.float32).to(dev)
+        self.max_numel = max(o.space.numel for o in opts)
+
+    def update(self):
+        """Apply all member updates (clock must already be ticked)."""
+        ext().adam_multi(self.ptrs, self.cfg, self.state3, len(self.opts),
+                         self.max_numel, self.beta1, self.beta2, self.eps)

@@ -43,6 +43,15 @@ class FusedSacStep(GraphableStep):
         self.stats_buf = torch.zeros(8, dtype=torch.float32, device=dev)
         self.stat_names = _SAC_STATS
         self.use_graph = use_graph and self.grad_reducer is None
+        # actor+alpha Adam updates happen back-to-back at the same DAG
+        # point → one multi-group launch (shared device step clock)
+        self.adam_aa = None
+        if self.grad_reducer is None and getattr(
+                updater.actor_optimizer, "shared_clock", False):
+            from .optim import AdamMultiGroup
+
+            self.adam_aa = AdamMultiGroup(
+                [updater.actor_optimizer, updater.alpha_optimizer])
 
     def fits(self, batch) -> bool:
         return True  # the loss kernels grid-stride; no LDS shape limit
@@ -94,6 +103,8 @@ class FusedSacStep(GraphableStep):
         fir = batch["is_fir"].reshape(B, S)
         log_alpha = u.log_alpha.data.view(1)
         single = self.grad_reducer is None
+        if getattr(u.actor_optimizer, "shared_clock", False):
+            u.actor_optimizer.tick()  # one device clock for all 3 Adams
 
         # 1. actor + twin critic forwards
         moA1, stA = self._fwd(self.cores["actor"], x, hx0, cx0)
@@ -110,10 +121,13 @@ class FusedSacStep(GraphableStep):
         # 3-4. actor + alpha updates
         self._bwd_wgrad(self.cores["actor"], gA, stA, x, hx0, cx0,
                         u.actor_optimizer.norm_sq if single else None)
-        self._opt(u.actor_optimizer)
-        if self.grad_reducer is not None:
-            self.grad_reducer.all_reduce([u.alpha_optimizer.flat_grad])
-        u.alpha_optimizer._update()  # no clip: norm unused
+        if self.adam_aa is not None:
+            self.adam_aa.update()  # actor + alpha in ONE launch
+        else:
+            self._opt(u.actor_optimizer)
+            if self.grad_reducer is not None:
+                self.grad_reducer.all_reduce([u.alpha_optimizer.flat_grad])
+            u.alpha_optimizer._update()  # no clip: norm unused
 
         # 5-6. post-update actor + target critics
         moA2, _ = self._fwd(self.cores["actor"], x, hx0, cx0)

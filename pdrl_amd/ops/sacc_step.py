@@ -52,6 +52,14 @@ class FusedSacContinuousStep(GraphableStep):
         self.stat_names = _SACC_STATS
         self.rng = torch.randint(1, 1 << 30, (1,), dtype=torch.int32, device=dev)
         self.use_graph = use_graph and self.grad_reducer is None
+        # actor+alpha Adam updates batch into one launch (shared clock)
+        self.adam_aa = None
+        if self.grad_reducer is None and getattr(
+                updater.actor_optimizer, "shared_clock", False):
+            from .optim import AdamMultiGroup
+
+            self.adam_aa = AdamMultiGroup(
+                [updater.actor_optimizer, updater.alpha_optimizer])
 
     def fits(self, batch) -> bool:
         return True  # loss kernels grid-stride; no LDS shape limit
@@ -170,6 +178,8 @@ class FusedSacContinuousStep(GraphableStep):
         log_alpha = u.log_alpha.data.view(1)
         single = self.grad_reducer is None
         dev = x.device
+        if getattr(u.actor_optimizer, "shared_clock", False):
+            u.actor_optimizer.tick()  # one device clock for all 3 Adams
 
         # 1. actor fwd + reparameterized sample
         moA, stA = self._actor_fwd(x, hx0, cx0)
@@ -198,10 +208,13 @@ class FusedSacContinuousStep(GraphableStep):
         # 4. actor + alpha updates
         self._actor_bwd_wgrad(dmoA, stA, x, hx0, cx0,
                               u.actor_optimizer.norm_sq if single else None)
-        self._opt(u.actor_optimizer)
-        if self.grad_reducer is not None:
-            self.grad_reducer.all_reduce([u.alpha_optimizer.flat_grad])
-        u.alpha_optimizer._update()  # no clip: norm unused
+        if self.adam_aa is not None:
+            self.adam_aa.update()  # actor + alpha in ONE launch
+        else:
+            self._opt(u.actor_optimizer)
+            if self.grad_reducer is not None:
+                self.grad_reducer.all_reduce([u.alpha_optimizer.flat_grad])
+            u.alpha_optimizer._update()  # no clip: norm unused
 
         # 5. post-update sample
         moA2, _ = self._actor_fwd(x, hx0, cx0)
