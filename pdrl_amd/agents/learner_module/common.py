@@ -51,7 +51,8 @@ class BaseUpdater:
                 return FusedRMSprop(parameters, lr=lr, max_norm=max_norm,
                                     eps=kw.get("eps", 1e-5))
             return FusedAdam(parameters, lr=lr, max_norm=max_norm,
-                             clock=kw.get("clock"))
+                             clock=kw.get("clock"),
+                             clock_owner=kw.get("clock_owner", False))
         if kind == "rmsprop":
             return torch.optim.RMSprop(parameters, lr=lr, eps=kw.get("eps", 1e-5))
         return torch.optim.Adam(parameters, lr=lr)

@@ -49,12 +49,15 @@ class SACUpdater(BaseUpdater):
         )
 
         # the three optimizers always step together once per iteration →
-        # they share ONE device step clock (t, bc1, bc2), advanced once
-        # per step; lets the fused DAG batch their updates into one launch
+        # they share ONE device step clock (t, bc1, bc2). The actor
+        # optimizer owns the clock and steps FIRST in every path (eager
+        # order actor→alpha→critic; fused DAG ticks at the actor update),
+        # so alpha/critic always see a freshly advanced bias correction.
         clock = torch.zeros(3, dtype=torch.float32, device=self.device) \
             if self.device.type == "cuda" else None
         self.actor_optimizer = self.make_optimizer(
-            "adam", self.actor.parameters(), lr=params.lr, clock=clock)
+            "adam", self.actor.parameters(), lr=params.lr, clock=clock,
+            clock_owner=True)
         self.critic_optimizer = self.make_optimizer(
             "adam", self.critic.parameters(), lr=params.lr, clock=clock)
         self.alpha_optimizer = self.make_optimizer(

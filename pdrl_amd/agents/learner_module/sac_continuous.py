@@ -42,11 +42,13 @@ class SACContinuousUpdater(BaseUpdater):
         )
 
         # one shared device Adam step clock across the three optimizers
-        # (they always step together): fused DAG ticks it once per step
+        # (they always step together); the actor optimizer owns the clock
+        # and steps first in both the eager and the fused ordering
         clock = torch.zeros(3, dtype=torch.float32, device=self.device) \
             if self.device.type == "cuda" else None
         self.actor_optimizer = self.make_optimizer(
-            "adam", self.actor.parameters(), lr=params.lr, clock=clock)
+            "adam", self.actor.parameters(), lr=params.lr, clock=clock,
+            clock_owner=True)
         self.critic_optimizer = self.make_optimizer(
             "adam", self.critic.parameters(), lr=params.lr, clock=clock)
         self.alpha_optimizer = self.make_optimizer(

@@ -38,8 +38,10 @@ namespace {
 // stash layout per (b, t): [xb(H) | gates i,f,g,o (4H) | c(H) | h(H)] = 7H
 constexpr int kStashFields = 7;
 
+// Whole-sequence forward for ONE batch row (the workgroup). Shared by the
+// single-core kernel and the multi-core (pointer-table) kernel below.
 template <int H>
-__global__ __launch_bounds__(4 * H) void seq_lstm_fwd_kernel(
+__device__ __forceinline__ void seq_lstm_fwd_row(
     const float* __restrict__ x,       // (B,S,F)
     const float* __restrict__ h0,      // (B,H)
     const float* __restrict__ c0,      // (B,H)
@@ -54,12 +56,10 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_fwd_kernel(
     float* __restrict__ hS,            // (B,H)
     float* __restrict__ cS,            // (B,H)
     float* __restrict__ stash,         // (B,S,7H)
-    int S, int F, int D, long h0s) {   // h0s: row stride of h0/c0
+    int b, int S, int F, int D, long h0s, char* smem_raw) {
   constexpr int G = 4 * H;
-  const int b = blockIdx.x;
   const int tid = threadIdx.x;
 
-  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   float* xb = reinterpret_cast<float*>(smem_raw);  // (S,H)
   float* hs = xb + S * H;                          // (S,H)
   float* gates = hs + S * H;                       // (4H)
@@ -154,6 +154,49 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_fwd_kernel(
     }
     outs[((long)b * S + t) * D + d] = acc;
   }
+}
+
+template <int H>
+__global__ __launch_bounds__(4 * H) void seq_lstm_fwd_kernel(
+    const float* __restrict__ x, const float* __restrict__ h0,
+    const float* __restrict__ c0, const float* __restrict__ body_w,
+    const float* __restrict__ body_b, const float* __restrict__ w_ih,
+    const float* __restrict__ w_hh, const float* __restrict__ b_g,
+    const float* __restrict__ heads_w, const float* __restrict__ heads_b,
+    float* __restrict__ outs, float* __restrict__ hS, float* __restrict__ cS,
+    float* __restrict__ stash, int S, int F, int D, long h0s) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  seq_lstm_fwd_row<H>(x, h0, c0, body_w, body_b, w_ih, w_hh, b_g, heads_w,
+                      heads_b, outs, hS, cS, stash, blockIdx.x, S, F, D, h0s,
+                      smem_raw);
+}
+
+// Multi-core forward: blockIdx.y picks the network. One launch evaluates C
+// same-shaped cores (different weights) on the SAME input — SAC's
+// actor + twin critics / twin target critics collapse from 3 launches to 1
+// (the step is launch-latency bound: profiles/algo_breakdown_r02a.md).
+// Weight/output pointers come from device int64 tables (graph-safe).
+template <int H>
+__global__ __launch_bounds__(4 * H) void seq_lstm_fwd_multi_kernel(
+    const float* __restrict__ x, const float* __restrict__ h0,
+    const float* __restrict__ c0,
+    const long* __restrict__ core_tab,  // [C][7] body_w..heads_b
+    const long* __restrict__ out_tab,   // [C][4] outs,hS,cS,stash
+    int S, int F, int D, long h0s) {
+  const long* ct = core_tab + (long)blockIdx.y * 7;
+  const long* ot = out_tab + (long)blockIdx.y * 4;
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  seq_lstm_fwd_row<H>(
+      x, h0, c0, reinterpret_cast<const float*>(ct[0]),
+      reinterpret_cast<const float*>(ct[1]),
+      reinterpret_cast<const float*>(ct[2]),
+      reinterpret_cast<const float*>(ct[3]),
+      reinterpret_cast<const float*>(ct[4]),
+      reinterpret_cast<const float*>(ct[5]),
+      reinterpret_cast<const float*>(ct[6]),
+      reinterpret_cast<float*>(ot[0]), reinterpret_cast<float*>(ot[1]),
+      reinterpret_cast<float*>(ot[2]), reinterpret_cast<float*>(ot[3]),
+      blockIdx.x, S, F, D, h0s, smem_raw);
 }
 
 // Backward through heads + recurrence + body for one batch row.

@@ -112,18 +112,21 @@ class FusedAdam(_FusedOptimizer):
     Step count + bias corrections live on device (graph-replay safe).
 
     ``clock``: an optional SHARED state3 tensor (t, bc1, bc2). Optimizers
-    that always step together once per iteration can share one clock; the
-    owner advances it with ``tick()`` (or ``ext().adam_prep``) once per
-    step and each member updates with ``_update(prep=False)``."""
+    that always step together once per iteration can share one clock;
+    exactly one of them is the ``clock_owner`` and advances it — inside its
+    own ``step()``, so the EAGER path stays correct as long as the owner
+    steps first each iteration (the updaters' fixed actor→alpha→critic
+    order guarantees that). Fused DAGs tick via AdamMultiGroup/step()."""
 
     def __init__(self, params, lr, betas=(0.9, 0.999), eps=1e-8, max_norm=None,
-                 clock: torch.Tensor | None = None):
+                 clock: torch.Tensor | None = None, clock_owner: bool = False):
         super().__init__(params, lr, max_norm)
         self.beta1, self.beta2 = float(betas[0]), float(betas[1])
         self.eps = float(eps)
         self.exp_avg = torch.zeros_like(self.space.flat_param)
         self.exp_avg_sq = torch.zeros_like(self.space.flat_param)
         self.shared_clock = clock is not None
+        self.clock_owner = clock_owner or clock is None
         self.state3 = clock if clock is not None else torch.zeros(
             3, dtype=torch.float32, device=self.space.flat_param.device)
 
@@ -135,12 +138,21 @@ class FusedAdam(_FusedOptimizer):
         """Advance the device step clock (once per iteration when shared)."""
         ext().adam_prep(self.state3, self.beta1, self.beta2)
 
+    def step(self):
+        if self.shared_clock and self.clock_owner:
+            self.tick()
+        super().step()
+
     def _update(self, prep: bool | None = None):
+        if prep is None:
+            # shared clock: the owner ticks it (step()/AdamMultiGroup);
+            # private clock: prep inline as before
+            prep = not self.shared_clock
         ext().adam_step(
             self.space.flat_param, self.space.flat_grad, self.exp_avg,
             self.exp_avg_sq, self.state3, self.norm_sq, self.lr, self.beta1,
             self.beta2, self.eps, self.max_norm,
-            do_prep=(not self.shared_clock) if prep is None else prep,
+            do_prep=prep,
         )
 
 
@@ -168,8 +180,10 @@ class AdamMultiGroup:
         self.ptrs = torch.tensor(rows, dtype=torch.int64).to(dev)
         self.cfg = torch.tensor(cfg, dtype=torch.float32).to(dev)
         self.max_numel = max(o.space.numel for o in opts)
+        self.owner = next((o for o in opts if o.clock_owner), opts[0])
 
     def update(self):
-        """Apply all member updates (clock must already be ticked)."""
+        """Tick the shared clock (owner) + apply all member updates."""
+        self.owner.tick()
         ext().adam_multi(self.ptrs, self.cfg, self.state3, len(self.opts),
                          self.max_numel, self.beta1, self.beta2, self.eps)

@@ -103,8 +103,6 @@ class FusedSacStep(GraphableStep):
         fir = batch["is_fir"].reshape(B, S)
         log_alpha = u.log_alpha.data.view(1)
         single = self.grad_reducer is None
-        if getattr(u.actor_optimizer, "shared_clock", False):
-            u.actor_optimizer.tick()  # one device clock for all 3 Adams
 
         # 1. actor + twin critic forwards
         moA1, stA = self._fwd(self.cores["actor"], x, hx0, cx0)

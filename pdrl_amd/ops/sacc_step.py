@@ -178,8 +178,6 @@ class FusedSacContinuousStep(GraphableStep):
         log_alpha = u.log_alpha.data.view(1)
         single = self.grad_reducer is None
         dev = x.device
-        if getattr(u.actor_optimizer, "shared_clock", False):
-            u.actor_optimizer.tick()  # one device clock for all 3 Adams
 
         # 1. actor fwd + reparameterized sample
         moA, stA = self._actor_fwd(x, hx0, cx0)
