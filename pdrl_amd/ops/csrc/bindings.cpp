@@ -21,6 +21,14 @@ void seq_lstm_wgrad_out_hip(const at::Tensor&, const at::Tensor&,
                             at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
                             at::Tensor&, at::Tensor&,
                             const c10::optional<at::Tensor>&);
+void seq_lstm_forward_multi_hip(const at::Tensor&, const at::Tensor&,
+                                const at::Tensor&, const at::Tensor&,
+                                const at::Tensor&, long, long);
+void seq_lstm_backward_multi_hip(const at::Tensor&, const at::Tensor&,
+                                 const at::Tensor&, const at::Tensor&, long,
+                                 long);
+void seq_lstm_wgrad_multi_hip(const at::Tensor&, const at::Tensor&,
+                              const at::Tensor&, long, long);
 at::Tensor gae_hip(const at::Tensor&, double, double, const at::Tensor&);
 std::vector<at::Tensor> vtrace_hip(const at::Tensor&, const at::Tensor&,
                                    const at::Tensor&, const at::Tensor&,
@@ -122,6 +130,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MFMA weight-gradient GEMMs + wave-per-element small grads");
   m.def("seq_lstm_wgrad_out", &seq_lstm_wgrad_out_hip,
         "wgrad writing into caller buffers (flat grad views)");
+  m.def("seq_lstm_forward_multi", &seq_lstm_forward_multi_hip,
+        "multi-network fused forward (device pointer tables)");
+  m.def("seq_lstm_backward_multi", &seq_lstm_backward_multi_hip,
+        "multi-network fused backward (leaf inputs; dgates/dxb only)");
+  m.def("seq_lstm_wgrad_multi", &seq_lstm_wgrad_multi_hip,
+        "multi-network MFMA weight grads (device pointer table)");
   m.def("gae", &gae_hip, "GAE reverse scan");
   m.def("vtrace", &vtrace_hip, "fused V-trace scan",
         pybind11::arg("behav_lp"), pybind11::arg("target_lp"),

@@ -204,7 +204,7 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_fwd_multi_kernel(
 // (dxb) consumed by the MFMA weight-gradient kernels (wgrad.hip), plus
 // dx / dh0 / dc0.
 template <int H>
-__global__ __launch_bounds__(4 * H) void seq_lstm_bwd_kernel(
+__device__ __forceinline__ void seq_lstm_bwd_row(
     const float* __restrict__ gouts,   // (B,S,D) head-output grads
     const float* __restrict__ ghS,     // (B,H) or nullptr
     const float* __restrict__ gcS,     // (B,H) or nullptr
@@ -215,17 +215,15 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_kernel(
     const float* __restrict__ w_ih,    // (H,4H)
     const float* __restrict__ w_hh,    // (H,4H)
     const float* __restrict__ heads_w, // (H,D)
-    float* __restrict__ dx,            // (B,S,F)
-    float* __restrict__ dh0,           // (B,H)
-    float* __restrict__ dc0,           // (B,H)
+    float* __restrict__ dx,            // (B,S,F) or nullptr (leaf input)
+    float* __restrict__ dh0,           // (B,H)   or nullptr
+    float* __restrict__ dc0,           // (B,H)   or nullptr
     float* __restrict__ dgates,        // (B,S,4H) pre-activation
     float* __restrict__ dxb,           // (B,S,H) pre-ReLU
-    int S, int F, int D, long h0s) {
+    int b, int S, int F, int D, long h0s, char* smem_raw) {
   constexpr int G = 4 * H;
-  const int b = blockIdx.x;
   const int tid = threadIdx.x;
 
-  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   float* dhh = reinterpret_cast<float*>(smem_raw);    // (S, H) head-grad dh
   float* dg4 = dhh + S * H;                           // (4H)
   float* dxb_s = dg4 + G;                             // (S, H)
@@ -322,10 +320,11 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_kernel(
     __syncthreads();  // dg4 reused next iteration
   }
 
-  if (tid < H) {
+  if (tid < H && dh0 != nullptr) {
     dh0[(long)b * H + tid] = dh_rec;
     dc0[(long)b * H + tid] = dc_rec;
   }
+  if (dx == nullptr) return;  // leaf input: skip the dx GEMM entirely
   __syncthreads();
 
   // dx[t][f] = sum_j dxb[t][j] * body_w[f][j]
@@ -338,6 +337,46 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_kernel(
     for (int j = 0; j < H; ++j) acc = fmaf(dr[j], wr[j], acc);
     dx[((long)b * S + t) * F + f] = acc;
   }
+}
+
+template <int H>
+__global__ __launch_bounds__(4 * H) void seq_lstm_bwd_kernel(
+    const float* __restrict__ gouts, const float* __restrict__ ghS,
+    const float* __restrict__ gcS, const float* __restrict__ stash,
+    const float* __restrict__ x, const float* __restrict__ c0,
+    const float* __restrict__ body_w, const float* __restrict__ w_ih,
+    const float* __restrict__ w_hh, const float* __restrict__ heads_w,
+    float* __restrict__ dx, float* __restrict__ dh0, float* __restrict__ dc0,
+    float* __restrict__ dgates, float* __restrict__ dxb, int S, int F, int D,
+    long h0s) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  seq_lstm_bwd_row<H>(gouts, ghS, gcS, stash, x, c0, body_w, w_ih, w_hh,
+                      heads_w, dx, dh0, dc0, dgates, dxb, blockIdx.x, S, F, D,
+                      h0s, smem_raw);
+}
+
+// Multi-core backward: blockIdx.y picks the network (per-core gouts, stash,
+// weights, dgates/dxb outputs from device pointer tables). dx/dh0/dc0 are
+// skipped — the multi path serves the twin-critic backward, whose inputs
+// are leaves. One launch replaces C.
+template <int H>
+__global__ __launch_bounds__(4 * H) void seq_lstm_bwd_multi_kernel(
+    const float* __restrict__ x, const float* __restrict__ c0,
+    const long* __restrict__ in_tab,   // [C][6] gouts,stash,w_ih,w_hh,heads_w,body_w
+    const long* __restrict__ out_tab,  // [C][2] dgates,dxb
+    int S, int F, int D, long h0s) {
+  const long* it = in_tab + (long)blockIdx.y * 6;
+  const long* ot = out_tab + (long)blockIdx.y * 2;
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  seq_lstm_bwd_row<H>(
+      reinterpret_cast<const float*>(it[0]), nullptr, nullptr,
+      reinterpret_cast<const float*>(it[1]), x, c0,
+      reinterpret_cast<const float*>(it[5]),
+      reinterpret_cast<const float*>(it[2]),
+      reinterpret_cast<const float*>(it[3]),
+      reinterpret_cast<const float*>(it[4]), nullptr, nullptr, nullptr,
+      reinterpret_cast<float*>(ot[0]), reinterpret_cast<float*>(ot[1]),
+      blockIdx.x, S, F, D, h0s, smem_raw);
 }
 
 template <int H>
@@ -422,6 +461,67 @@ std::vector<at::Tensor> seq_lstm_forward_hip(
       TORCH_CHECK(false, "hidden size ", H, " unsupported (32/64/128)");
   }
   return {outs, hS, cS, stash};
+}
+
+void seq_lstm_forward_multi_hip(const at::Tensor& x, const at::Tensor& h0,
+                                const at::Tensor& c0,
+                                const at::Tensor& core_tab,
+                                const at::Tensor& out_tab, long C, long D) {
+  CHECK_IN(x);
+  CHECK_GPU(h0); CHECK_F32(h0); CHECK_GPU(c0); CHECK_F32(c0);
+  CHECK_GPU(core_tab); CHECK_GPU(out_tab);
+  TORCH_CHECK(h0.stride(1) == 1 && c0.stride(1) == 1, "h0/c0 inner stride");
+  TORCH_CHECK(h0.stride(0) == c0.stride(0), "h0/c0 stride mismatch");
+  const int B = x.size(0), S = x.size(1), F = x.size(2);
+  const int H = h0.size(1);
+  TORCH_CHECK(S >= 1 && S <= 32, "seq_len must be in [1, 32]");
+  const long h0s = (long)h0.stride(0);
+  dim3 grid(B, (unsigned)C);
+#define PDRL_LAUNCH_FWD_MULTI(HH)                                             \
+  hipLaunchKernelGGL((seq_lstm_fwd_multi_kernel<HH>), grid, dim3(4 * HH),     \
+                     (2 * S * HH + 4 * HH + 2 * HH) * sizeof(float),          \
+                     current_stream(), x.data_ptr<float>(),                   \
+                     h0.data_ptr<float>(), c0.data_ptr<float>(),              \
+                     core_tab.data_ptr<long>(), out_tab.data_ptr<long>(), S,  \
+                     F, (int)D, h0s)
+  switch (H) {
+    case 32: PDRL_LAUNCH_FWD_MULTI(32); break;
+    case 64: PDRL_LAUNCH_FWD_MULTI(64); break;
+    case 128: PDRL_LAUNCH_FWD_MULTI(128); break;
+    default: TORCH_CHECK(false, "hidden size ", H, " unsupported");
+  }
+#undef PDRL_LAUNCH_FWD_MULTI
+  HIP_CHECK_LAST();
+}
+
+void seq_lstm_backward_multi_hip(const at::Tensor& x, const at::Tensor& c0,
+                                 const at::Tensor& in_tab,
+                                 const at::Tensor& out_tab, long C, long D) {
+  CHECK_IN(x);
+  CHECK_GPU(c0); CHECK_F32(c0);
+  CHECK_GPU(in_tab); CHECK_GPU(out_tab);
+  TORCH_CHECK(c0.stride(1) == 1, "c0 inner stride must be 1");
+  const int B = x.size(0), S = x.size(1), F = x.size(2);
+  const int H = c0.size(1);
+  const long h0s = (long)c0.stride(0);
+  dim3 grid(B, (unsigned)C);
+#define PDRL_LAUNCH_BWD_MULTI(HH)                                             \
+  do {                                                                        \
+    const int G = 4 * HH;                                                     \
+    const int lds = (S * HH + G + S * HH + 2 * G) * sizeof(float);            \
+    hipLaunchKernelGGL((seq_lstm_bwd_multi_kernel<HH>), grid, dim3(G), lds,   \
+                       current_stream(), x.data_ptr<float>(),                 \
+                       c0.data_ptr<float>(), in_tab.data_ptr<long>(),         \
+                       out_tab.data_ptr<long>(), S, F, (int)D, h0s);          \
+  } while (0)
+  switch (H) {
+    case 32: PDRL_LAUNCH_BWD_MULTI(32); break;
+    case 64: PDRL_LAUNCH_BWD_MULTI(64); break;
+    case 128: PDRL_LAUNCH_BWD_MULTI(128); break;
+    default: TORCH_CHECK(false, "hidden size ", H, " unsupported");
+  }
+#undef PDRL_LAUNCH_BWD_MULTI
+  HIP_CHECK_LAST();
 }
 
 std::vector<at::Tensor> seq_lstm_backward_core_hip(

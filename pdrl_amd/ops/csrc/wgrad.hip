@@ -50,7 +50,7 @@ __device__ void wgrad_small_body(const float*, const float*, const float*,
 // blocks [0, gemm_blocks) × y∈{0,1} run the two GEMMs; blocks beyond that on
 // y==0 run the wave-per-element small-grad reductions concurrently.
 template <int H>
-__global__ __launch_bounds__(256) void wgrad_gates_mfma_kernel(
+__device__ __forceinline__ void wgrad_gates_body(
     const float* __restrict__ stash,   // (B,S,7H)
     const float* __restrict__ h0,      // (B,H)
     const float* __restrict__ dgates,  // (N,4H)
@@ -63,7 +63,7 @@ __global__ __launch_bounds__(256) void wgrad_gates_mfma_kernel(
     float* __restrict__ dbody_w, float* __restrict__ dbody_b,
     float* __restrict__ db_g, float* __restrict__ dheads_w,
     float* __restrict__ dheads_b, int F, int D,
-    int N, int S, long h0s) {
+    int N, int S, long h0s, char* smem_raw) {
   constexpr int G = 4 * H;
   const int g_blocks = G / kWave;  // 64-wide g blocks
   const int gemm_blocks = (H / 16) * g_blocks;
@@ -87,7 +87,6 @@ __global__ __launch_bounds__(256) void wgrad_gates_mfma_kernel(
   // A-row pointer table, built ONCE per block: the per-row n%S / n/S integer
   // divisions in the load loop (640 of them per lane) serialized the K sweep
   // — with the table each load is ptr[n] (LDS broadcast) + one global load.
-  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   const float** tab = reinterpret_cast<const float**>(smem_raw);
   for (int n = threadIdx.x; n < N; n += 256) {
     tab[n] = gate_a_row<H>(stash, h0, n, S, sel, h0s);
@@ -165,6 +164,44 @@ __global__ __launch_bounds__(256) void wgrad_gates_mfma_kernel(
     if (threadIdx.x == 0)
       atomicAdd(norm_sq, red[0] + red[1] + red[2] + red[3]);
   }
+}
+
+template <int H>
+__global__ __launch_bounds__(256) void wgrad_gates_mfma_kernel(
+    const float* __restrict__ stash, const float* __restrict__ h0,
+    const float* __restrict__ dgates, float* __restrict__ dw_ih,
+    float* __restrict__ dw_hh, float* __restrict__ norm_sq,
+    const float* __restrict__ x, const float* __restrict__ dxb,
+    const float* __restrict__ gouts, float* __restrict__ dbody_w,
+    float* __restrict__ dbody_b, float* __restrict__ db_g,
+    float* __restrict__ dheads_w, float* __restrict__ dheads_b, int F, int D,
+    int N, int S, long h0s) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  wgrad_gates_body<H>(stash, h0, dgates, dw_ih, dw_hh, norm_sq, x, dxb, gouts,
+                      dbody_w, dbody_b, db_g, dheads_w, dheads_b, F, D, N, S,
+                      h0s, smem_raw);
+}
+
+// Multi-core weight grads: blockIdx.z picks the network; per-core pointers
+// (stash/dgates/dxb/gouts inputs, 7 grad outputs, optional norm
+// accumulator) come from a device int64 table. x/h0 are shared.
+template <int H>
+__global__ __launch_bounds__(256) void wgrad_gates_mfma_multi_kernel(
+    const float* __restrict__ x, const float* __restrict__ h0,
+    const long* __restrict__ tab,  // [C][12]
+    int F, int D, int N, int S, long h0s) {
+  const long* ct = tab + (long)blockIdx.z * 12;
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  wgrad_gates_body<H>(
+      reinterpret_cast<const float*>(ct[0]),   // stash
+      h0, reinterpret_cast<const float*>(ct[1]),  // dgates
+      reinterpret_cast<float*>(ct[4]), reinterpret_cast<float*>(ct[5]),
+      reinterpret_cast<float*>(ct[11]),  // norm_sq or 0
+      x, reinterpret_cast<const float*>(ct[2]),   // dxb
+      reinterpret_cast<const float*>(ct[3]),      // gouts
+      reinterpret_cast<float*>(ct[6]), reinterpret_cast<float*>(ct[7]),
+      reinterpret_cast<float*>(ct[8]), reinterpret_cast<float*>(ct[9]),
+      reinterpret_cast<float*>(ct[10]), F, D, N, S, h0s, smem_raw);
 }
 
 // One wave per output element; lanes stride the K=N reduction.
@@ -268,6 +305,38 @@ void launch_wgrad(const at::Tensor& x, const at::Tensor& h0,
 }
 
 }  // namespace
+
+void seq_lstm_wgrad_multi_hip(const at::Tensor& x, const at::Tensor& h0,
+                              const at::Tensor& tab, long C, long D) {
+  CHECK_IN(x);
+  CHECK_GPU(h0); CHECK_F32(h0); CHECK_GPU(tab);
+  TORCH_CHECK(h0.stride(1) == 1, "h0 inner stride must be 1");
+  const int B = x.size(0), S = x.size(1), F = x.size(2);
+  const int H = h0.size(1);
+  const int N = B * S;
+  const int tab_lds = N * sizeof(const float*);
+  TORCH_CHECK(tab_lds <= 64 * 1024, "wgrad row table exceeds LDS");
+#define PDRL_LAUNCH_WG_MULTI(HH)                                              \
+  do {                                                                        \
+    constexpr int G = 4 * HH;                                                 \
+    const int gemm_blocks = (HH / 16) * (G / kWave);                          \
+    const int total_waves = F * HH + HH + G + HH * (int)D + (int)D;           \
+    const int small_blocks = (total_waves * kWave + 255) / 256;               \
+    dim3 grid(gemm_blocks + small_blocks, 2, (unsigned)C);                    \
+    hipLaunchKernelGGL((wgrad_gates_mfma_multi_kernel<HH>), grid, dim3(256),  \
+                       tab_lds, current_stream(), x.data_ptr<float>(),        \
+                       h0.data_ptr<float>(), tab.data_ptr<long>(), F, (int)D, \
+                       N, S, (long)h0.stride(0));                             \
+  } while (0)
+  switch (H) {
+    case 32: PDRL_LAUNCH_WG_MULTI(32); break;
+    case 64: PDRL_LAUNCH_WG_MULTI(64); break;
+    case 128: PDRL_LAUNCH_WG_MULTI(128); break;
+    default: TORCH_CHECK(false, "hidden size ", H, " unsupported");
+  }
+#undef PDRL_LAUNCH_WG_MULTI
+  HIP_CHECK_LAST();
+}
 
 void seq_lstm_wgrad_out_hip(const at::Tensor& x, const at::Tensor& h0,
                             const at::Tensor& stash, const at::Tensor& dgates,

@@ -2,15 +2,16 @@
 DAG — the discrete half of K11, hipGraph-capturable.
 
 Follows the EAGER reference ordering exactly
-(agents/learner_module/sac/learning.py):
-  1.  actor fwd, twin-critic fwds
+(agents/learner_module/sac/learning.py), in 12 launches (launch latency IS
+the step time at this model size — profiles/algo_breakdown_r02a.md):
+  1.  actor + twin-critic forwards — ONE multi-network launch
   2.  sac_actor_loss      — analytic dlogits + dlog_alpha + stats
-  3.  actor BPTT + MFMA wgrad → flat actor grads (+ norm) → fused Adam
-  4.  alpha fused Adam (no clip)
-  5.  actor fwd AGAIN (post-update policy, as the reference does)
-  6.  target-critic fwds
+  3.  actor BPTT + MFMA wgrad → flat actor grads (+ norm)
+  4.  shared-clock tick + actor/alpha Adam — ONE multi-group launch
+  5.  actor fwd AGAIN (post-update policy, as the reference does) +
+      target-critic fwds — ONE multi-network launch
   7.  sac_critic_loss     — soft-Q TD target, twin huber, analytic dq1/dq2
-  8.  q1/q2 BPTT + wgrads → flat critic grads (+ norm) → fused Adam
+  8.  q1+q2 BPTT (one launch) + q1+q2 MFMA wgrads (one launch) → Adam
   9.  Polyak soft_update (cached device pointer tables)
 
 The twin-critic outputs from step 1 are reused for the value loss — the
@@ -57,13 +58,86 @@ class FusedSacStep(GraphableStep):
         return True  # the loss kernels grid-stride; no LDS shape limit
 
     # ------------------------------------------------------------------ #
-    def _fwd(self, core, x, hx0, cx0):
-        e = ext()
-        mo, hS, cS, stash = e.seq_lstm_forward(
-            x, hx0, cx0, core.body_w, core.body_b, core.w_ih, core.w_hh,
-            core.b_g, core.heads_w, core.heads_b,
-        )
-        return mo, stash
+    def _multi_setup(self, batch):
+        """Preallocate per-core output buffers and build the device pointer
+        tables for the multi-network launches (all five cores share one
+        head width here: logits(A) for the actor, Q(A) for the critics).
+        Tables reference persistent storage only (parameters/grads are
+        views into the flat optimizer buffers; activations below are owned
+        here), so graph capture replays them safely."""
+        c = self.cores
+        u = self.u
+        x = batch["obs"]
+        B, S, _ = x.shape
+        dev = x.device
+        H = c["actor"].w_ih.size(0)
+        D = c["actor"].heads_w.size(1)
+        assert all(cc.heads_w.size(1) == D for cc in c.values())
+        self._D = D
+
+        def mk(*shape):
+            return torch.empty(*shape, device=dev)
+
+        buf = {}
+        for name in ("actor", "q1", "q2", "actor2", "t1", "t2"):
+            buf[name] = {"outs": mk(B, S, D), "hS": mk(B, H), "cS": mk(B, H),
+                         "stash": mk(B, S, 7 * H)}
+        for name in ("q1", "q2"):
+            buf[name]["dgates"] = mk(B, S, 4 * H)
+            buf[name]["dxb"] = mk(B, S, H)
+            buf[name]["gq"] = mk(B, S, D)
+        self.buf = buf
+
+        def t64(rows):
+            return torch.tensor(rows, dtype=torch.int64).to(dev)
+
+        def wrow(core):
+            return [core.body_w.data_ptr(), core.body_b.data_ptr(),
+                    core.w_ih.data_ptr(), core.w_hh.data_ptr(),
+                    core.b_g.data_ptr(), core.heads_w.data_ptr(),
+                    core.heads_b.data_ptr()]
+
+        def orow(b):
+            return [b["outs"].data_ptr(), b["hS"].data_ptr(),
+                    b["cS"].data_ptr(), b["stash"].data_ptr()]
+
+        self.fwd1_cores = t64([wrow(c["actor"]), wrow(c["q1"]), wrow(c["q2"])])
+        self.fwd1_outs = t64([orow(buf["actor"]), orow(buf["q1"]),
+                              orow(buf["q2"])])
+        self.fwd2_cores = t64([wrow(c["actor"]), wrow(c["t1"]), wrow(c["t2"])])
+        self.fwd2_outs = t64([orow(buf["actor2"]), orow(buf["t1"]),
+                              orow(buf["t2"])])
+
+        def birow(name):
+            core, b = c[name], buf[name]
+            return [b["gq"].data_ptr(), b["stash"].data_ptr(),
+                    core.w_ih.data_ptr(), core.w_hh.data_ptr(),
+                    core.heads_w.data_ptr(), core.body_w.data_ptr()]
+
+        def borow(name):
+            b = buf[name]
+            return [b["dgates"].data_ptr(), b["dxb"].data_ptr()]
+
+        self.bwd_in = t64([birow("q1"), birow("q2")])
+        self.bwd_out = t64([borow("q1"), borow("q2")])
+
+        nrm = (u.critic_optimizer.norm_sq.data_ptr()
+               if self.grad_reducer is None else 0)
+
+        def grow(name):
+            core, b = c[name], buf[name]
+            g = [core.body_w.grad, core.body_b.grad, core.w_ih.grad,
+                 core.w_hh.grad, core.b_g.grad, core.heads_w.grad,
+                 core.heads_b.grad]
+            assert all(t is not None for t in g)
+            return [b["stash"].data_ptr(), b["dgates"].data_ptr(),
+                    b["dxb"].data_ptr(), b["gq"].data_ptr(),
+                    g[2].data_ptr(), g[3].data_ptr(), g[0].data_ptr(),
+                    g[1].data_ptr(), g[4].data_ptr(), g[5].data_ptr(),
+                    g[6].data_ptr(), nrm]
+
+        self.wg_tab = t64([grow("q1"), grow("q2")])
+        self._mshape = (B, S)
 
     def _bwd_wgrad(self, core, gouts, stash, x, hx0, cx0, norm):
         e = ext()
@@ -103,11 +177,15 @@ class FusedSacStep(GraphableStep):
         fir = batch["is_fir"].reshape(B, S)
         log_alpha = u.log_alpha.data.view(1)
         single = self.grad_reducer is None
+        if getattr(self, "_mshape", None) != (B, S):
+            self._multi_setup(batch)
+        buf, D = self.buf, self._D
 
-        # 1. actor + twin critic forwards
-        moA1, stA = self._fwd(self.cores["actor"], x, hx0, cx0)
-        mq1, st1 = self._fwd(self.cores["q1"], x, hx0, cx0)
-        mq2, st2 = self._fwd(self.cores["q2"], x, hx0, cx0)
+        # 1. actor + twin critic forwards — ONE multi-network launch
+        e.seq_lstm_forward_multi(x, hx0, cx0, self.fwd1_cores,
+                                 self.fwd1_outs, 3, D)
+        moA1 = buf["actor"]["outs"]
+        mq1, mq2 = buf["q1"]["outs"], buf["q2"]["outs"]
 
         # 2. actor + temperature losses (analytic grads)
         gA = torch.empty_like(moA1)
@@ -117,34 +195,34 @@ class FusedSacStep(GraphableStep):
             None, u.target_entropy,
         )
         # 3-4. actor + alpha updates
-        self._bwd_wgrad(self.cores["actor"], gA, stA, x, hx0, cx0,
+        self._bwd_wgrad(self.cores["actor"], gA, buf["actor"]["stash"], x,
+                        hx0, cx0,
                         u.actor_optimizer.norm_sq if single else None)
         if self.adam_aa is not None:
-            self.adam_aa.update()  # actor + alpha in ONE launch
+            self.adam_aa.update()  # tick + actor + alpha in 2 launches
         else:
             self._opt(u.actor_optimizer)
             if self.grad_reducer is not None:
                 self.grad_reducer.all_reduce([u.alpha_optimizer.flat_grad])
             u.alpha_optimizer._update()  # no clip: norm unused
 
-        # 5-6. post-update actor + target critics
-        moA2, _ = self._fwd(self.cores["actor"], x, hx0, cx0)
-        mt1, _ = self._fwd(self.cores["t1"], x, hx0, cx0)
-        mt2, _ = self._fwd(self.cores["t2"], x, hx0, cx0)
+        # 5-6. post-update actor + target critics — ONE launch
+        e.seq_lstm_forward_multi(x, hx0, cx0, self.fwd2_cores,
+                                 self.fwd2_outs, 3, D)
+        moA2 = buf["actor2"]["outs"]
+        mt1, mt2 = buf["t1"]["outs"], buf["t2"]["outs"]
 
-        # 7. critic losses
-        gq1 = torch.empty_like(mq1)
-        gq2 = torch.empty_like(mq2)
+        # 7. critic losses (head grads land in the persistent gq buffers
+        #    the bwd/wgrad pointer tables reference)
+        gq1, gq2 = buf["q1"]["gq"], buf["q2"]["gq"]
         e.sac_critic_loss(
             moA2, mq1, mq2, mt1, mt2, act, rew, fir, log_alpha, gq1, gq2,
             self.stats_buf[4:5], u.critic_optimizer.norm_sq if single else None,
             p.gamma, p.reward_scale,
         )
-        # 8. critic updates (both cores accumulate into one flat space)
-        self._bwd_wgrad(self.cores["q1"], gq1, st1, x, hx0, cx0,
-                        u.critic_optimizer.norm_sq if single else None)
-        self._bwd_wgrad(self.cores["q2"], gq2, st2, x, hx0, cx0,
-                        u.critic_optimizer.norm_sq if single else None)
+        # 8. twin-critic backward + MFMA weight grads — one launch each
+        e.seq_lstm_backward_multi(x, cx0, self.bwd_in, self.bwd_out, 2, D)
+        e.seq_lstm_wgrad_multi(x, hx0, self.wg_tab, 2, D)
         self._opt(u.critic_optimizer)
 
         # 9. Polyak target update
