@@ -85,7 +85,8 @@ void sac_actor_loss_hip(const at::Tensor&, const at::Tensor&,
                         const at::Tensor&, const at::Tensor&, at::Tensor&,
                         at::Tensor&, at::Tensor&,
                         const c10::optional<at::Tensor>&,
-                        const c10::optional<at::Tensor>&, double);
+                        const c10::optional<at::Tensor>&, double,
+                        const c10::optional<at::Tensor>&, double, double);
 void sac_critic_loss_hip(const at::Tensor&, const at::Tensor&,
                          const at::Tensor&, const at::Tensor&,
                          const at::Tensor&, const at::Tensor&,
@@ -100,7 +101,8 @@ void sacc_actor_grad_hip(const at::Tensor&, const at::Tensor&,
                          const at::Tensor&, const at::Tensor&,
                          const at::Tensor&, at::Tensor&, at::Tensor&,
                          at::Tensor&, const c10::optional<at::Tensor>&,
-                         const c10::optional<at::Tensor>&, double);
+                         const c10::optional<at::Tensor>&, double,
+                         const c10::optional<at::Tensor>&, double, double);
 void sacc_critic_loss_hip(const at::Tensor&, const at::Tensor&,
                           const at::Tensor&, const at::Tensor&,
                           const at::Tensor&, const at::Tensor&,
@@ -159,13 +161,24 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ppoc_loss_mega", &ppoc_loss_mega_hip,
         "single-launch Gaussian-policy PPO loss (K5)");
   m.def("sac_actor_loss", &sac_actor_loss_hip,
-        "SAC discrete actor+alpha loss with analytic grads");
+        "SAC discrete actor+alpha loss with analytic grads",
+        py::arg("moA"), py::arg("q1"), py::arg("q2"), py::arg("log_alpha"),
+        py::arg("gouts"), py::arg("g_alpha"), py::arg("stats"),
+        py::arg("actor_norm"), py::arg("alpha_norm"),
+        py::arg("target_entropy"), py::arg("clock") = c10::nullopt,
+        py::arg("beta1") = 0.9, py::arg("beta2") = 0.999);
   m.def("sac_critic_loss", &sac_critic_loss_hip,
         "SAC discrete soft-Q target + twin critic loss grads");
   m.def("sacc_sample", &sacc_sample_hip,
         "reparameterized tanh-Gaussian sample + log-prob (graph-safe RNG)");
   m.def("sacc_actor_grad", &sacc_actor_grad_hip,
-        "SAC-continuous actor+alpha loss with analytic dmu/dlog_std");
+        "SAC-continuous actor+alpha loss with analytic dmu/dlog_std",
+        py::arg("moA"), py::arg("eps"), py::arg("act"), py::arg("g"),
+        py::arg("q1"), py::arg("q2"), py::arg("log_alpha"), py::arg("dmoA"),
+        py::arg("g_alpha"), py::arg("stats"), py::arg("actor_norm"),
+        py::arg("alpha_norm"), py::arg("target_entropy"),
+        py::arg("clock") = c10::nullopt, py::arg("beta1") = 0.9,
+        py::arg("beta2") = 0.999);
   m.def("sacc_critic_loss", &sacc_critic_loss_hip,
         "SAC-continuous soft-Q target + twin critic loss grads");
   m.def("l2norm_sq", &l2norm_sq_hip, "squared L2 norm into a device scalar");

@@ -39,10 +39,19 @@ __global__ __launch_bounds__(kThreads) void sac_actor_loss_kernel(
     float* __restrict__ stats,       // (4)
     float* __restrict__ actor_norm,  // optional: zeroed here
     float* __restrict__ alpha_norm,  // optional: zeroed + g_alpha² added
-    int N, int A, float target_entropy) {
+    float* __restrict__ clock,       // optional shared Adam clock: prepped
+    int N, int A, float target_entropy, float beta1, float beta2) {
   const int tid = threadIdx.x;
   const float alpha = __expf(*log_alpha);
   if (tid == 0 && actor_norm != nullptr) *actor_norm = 0.f;
+  if (tid == 1 && clock != nullptr) {
+    // fold the Adam step-clock advance into this (single-block) kernel:
+    // saves the separate 1-thread prep launch (~4.4 µs of pure dispatch)
+    const float t = clock[0] + 1.0f;
+    clock[0] = t;
+    clock[1] = 1.0f - __powf(beta1, t);
+    clock[2] = 1.0f - __powf(beta2, t);
+  }
 
   float l_sum = 0.f, ent_sum = 0.f;
   for (int i = tid; i < N; i += kThreads) {
@@ -166,7 +175,9 @@ void sac_actor_loss_hip(const at::Tensor& moA, const at::Tensor& q1,
                         at::Tensor& stats,
                         const c10::optional<at::Tensor>& actor_norm,
                         const c10::optional<at::Tensor>& alpha_norm,
-                        double target_entropy) {
+                        double target_entropy,
+                        const c10::optional<at::Tensor>& clock,
+                        double beta1, double beta2) {
   const int A = moA.size(-1);
   const long N = moA.numel() / A;
   hipLaunchKernelGGL(sac_actor_loss_kernel, dim3(1), dim3(256), 0,
@@ -176,7 +187,9 @@ void sac_actor_loss_hip(const at::Tensor& moA, const at::Tensor& q1,
                      g_alpha.data_ptr<float>(), stats.data_ptr<float>(),
                      actor_norm.has_value() ? actor_norm->data_ptr<float>() : nullptr,
                      alpha_norm.has_value() ? alpha_norm->data_ptr<float>() : nullptr,
-                     (int)N, A, (float)target_entropy);
+                     clock.has_value() ? clock->data_ptr<float>() : nullptr,
+                     (int)N, A, (float)target_entropy, (float)beta1,
+                     (float)beta2);
   HIP_CHECK_LAST();
 }
 

@@ -152,16 +152,25 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
   // instead of the 9-barrier shared-memory tree.
   const int K = BT / 2 > 0 ? BT / 2 : 1;
   constexpr int kMaxPer = 12;
+  // FULLY STATIC indexing: a runtime-bounded fill/scan (`myu[mycnt++]`)
+  // demotes the array to scratch (= global memory), and the 32 sequential
+  // search rounds then pay global latency per element — measured 46 µs for
+  // this kernel. Static indices keep all 12 lanes in VGPRs (guide rule 20);
+  // the 0u padding is the order-map of -FLT_MAX, never counted as greater.
   unsigned myu[kMaxPer];
-  int mycnt = 0;
-  for (int i = tid; i < BT; i += kThreads) myu[mycnt++] = f2u(s_adv[i]);
+#pragma unroll
+  for (int k = 0; k < kMaxPer; ++k) {
+    const int i = tid + k * kThreads;
+    myu[k] = (i < BT) ? f2u(s_adv[i]) : 0u;
+  }
   unsigned thresh_u;
   {
     unsigned lo = 0, hi = 0xFFFFFFFFu;
     while (lo < hi) {
       const unsigned mid = lo + ((hi - lo) >> 1);
       int cnt = 0;
-      for (int k = 0; k < mycnt; ++k) cnt += (myu[k] > mid);
+#pragma unroll
+      for (int k = 0; k < kMaxPer; ++k) cnt += (myu[k] > mid);
       const int total_gt = (int)block_sum((float)cnt, s4);
       if (total_gt >= K) {
         lo = mid + 1;  // threshold is higher
@@ -176,7 +185,8 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
   // mark selected: strictly greater always; equal by ascending index to k
   {
     int cnt_gt = 0;
-    for (int k = 0; k < mycnt; ++k) cnt_gt += (myu[k] > thresh_u);
+#pragma unroll
+    for (int k = 0; k < kMaxPer; ++k) cnt_gt += (myu[k] > thresh_u);
     const int n_gt = (int)block_sum((float)cnt_gt, s4);
     int need_eq = K - n_gt;  // ties to include, by lowest index
     for (int i = tid; i < BT; i += kThreads) {

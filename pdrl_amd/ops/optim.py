@@ -182,8 +182,11 @@ class AdamMultiGroup:
         self.max_numel = max(o.space.numel for o in opts)
         self.owner = next((o for o in opts if o.clock_owner), opts[0])
 
-    def update(self):
-        """Tick the shared clock (owner) + apply all member updates."""
-        self.owner.tick()
+    def update(self, tick: bool = True):
+        """Apply all member updates; ``tick=False`` when an earlier kernel
+        in the DAG already advanced the shared clock (the loss kernels can
+        fold the prep in — saves the 1-thread launch)."""
+        if tick:
+            self.owner.tick()
         ext().adam_multi(self.ptrs, self.cfg, self.state3, len(self.opts),
                          self.max_numel, self.beta1, self.beta2, self.eps)

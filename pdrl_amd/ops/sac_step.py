@@ -187,19 +187,24 @@ class FusedSacStep(GraphableStep):
         moA1 = buf["actor"]["outs"]
         mq1, mq2 = buf["q1"]["outs"], buf["q2"]["outs"]
 
-        # 2. actor + temperature losses (analytic grads)
+        # 2. actor + temperature losses (analytic grads); the shared Adam
+        #    clock is prepped inside this kernel (one less launch)
         gA = torch.empty_like(moA1)
+        clk = u.actor_optimizer if self.adam_aa is not None else None
         e.sac_actor_loss(
             moA1, mq1, mq2, log_alpha, gA, u.log_alpha.grad.view(1),
             self.stats_buf[:4], u.actor_optimizer.norm_sq if single else None,
             None, u.target_entropy,
+            clock=clk.state3 if clk is not None else None,
+            beta1=clk.beta1 if clk is not None else 0.9,
+            beta2=clk.beta2 if clk is not None else 0.999,
         )
         # 3-4. actor + alpha updates
         self._bwd_wgrad(self.cores["actor"], gA, buf["actor"]["stash"], x,
                         hx0, cx0,
                         u.actor_optimizer.norm_sq if single else None)
         if self.adam_aa is not None:
-            self.adam_aa.update()  # tick + actor + alpha in 2 launches
+            self.adam_aa.update(tick=False)  # clock prepped by the loss kernel
         else:
             self._opt(u.actor_optimizer)
             if self.grad_reducer is not None:

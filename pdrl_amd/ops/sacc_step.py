@@ -195,19 +195,23 @@ class FusedSacContinuousStep(GraphableStep):
         g = self._critic_bwd_dact(self.q["q1"], gq1p, st1p, enc1p, cx0) \
             + self._critic_bwd_dact(self.q["q2"], gq2p, st2p, enc2p, cx0)
 
-        # 3. analytic actor + temperature gradients
+        # 3. analytic actor + temperature gradients (+ Adam clock prep)
         dmoA = torch.empty_like(moA)
+        clk = u.actor_optimizer if self.adam_aa is not None else None
         e.sacc_actor_grad(
             moA, eps1, a1, g, qp1.reshape(-1), qp2.reshape(-1), log_alpha,
             dmoA, u.log_alpha.grad.view(1), self.stats_buf[:4],
             u.actor_optimizer.norm_sq if single else None,
             None, u.target_entropy,
+            clock=clk.state3 if clk is not None else None,
+            beta1=clk.beta1 if clk is not None else 0.9,
+            beta2=clk.beta2 if clk is not None else 0.999,
         )
         # 4. actor + alpha updates
         self._actor_bwd_wgrad(dmoA, stA, x, hx0, cx0,
                               u.actor_optimizer.norm_sq if single else None)
         if self.adam_aa is not None:
-            self.adam_aa.update()  # actor + alpha in ONE launch
+            self.adam_aa.update(tick=False)  # clock prepped by actor_grad
         else:
             self._opt(u.actor_optimizer)
             if self.grad_reducer is not None:
