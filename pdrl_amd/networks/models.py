@@ -46,17 +46,36 @@ class SeqLSTMCore(nn.Module):
       body_w (F, H), body_b (H)
       w_ih (H, 4H), w_hh (H, 4H), b_g (4H)   # gate order i, f, g, o
       per head k: head_w[k] (H, D_k), head_b[k] (D_k)
+
+    Dual-body mode (``input2_dim``): the LSTM input is the concatenation
+    [relu(x·body_w+body_b) | relu(x2·body2_w+body2_b)], each encoder to
+    H/2 — the reference's continuous-critic topology (obs encoder ‖ action
+    encoder straight into the LSTM, reference: networks/models.py:273-322).
+    Then body_w is (F, H/2) and body2_w (F2, H/2).
     """
 
-    def __init__(self, input_dim: int, hidden: int, heads: dict[str, int]):
+    def __init__(self, input_dim: int, hidden: int, heads: dict[str, int],
+                 input2_dim: int | None = None):
         super().__init__()
         self.input_dim = input_dim
+        self.input2_dim = input2_dim
         self.hidden = hidden
         self.head_names = list(heads.keys())
         self.head_dims = dict(heads)
         H = hidden
-        self.body_w = nn.Parameter(torch.empty(input_dim, H))
-        self.body_b = nn.Parameter(torch.empty(H))
+        if input2_dim is None:
+            self.body_w = nn.Parameter(torch.empty(input_dim, H))
+            self.body_b = nn.Parameter(torch.empty(H))
+            self.body2_w = None
+            self.body2_b = None
+        else:
+            assert H % 2 == 0, "dual body needs an even hidden size"
+            half = H // 2
+            self.body_w = nn.Parameter(torch.empty(input_dim, half))
+            self.body_b = nn.Parameter(torch.empty(half))
+            self.body2_w = nn.Parameter(torch.empty(input2_dim, half))
+            self.body2_b = nn.Parameter(torch.empty(half))
+            _init_linear_t(self.body2_w, self.body2_b, input2_dim)
         self.w_ih = nn.Parameter(torch.empty(H, 4 * H))
         self.w_hh = nn.Parameter(torch.empty(H, 4 * H))
         self.b_g = nn.Parameter(torch.empty(4 * H))
@@ -88,20 +107,29 @@ class SeqLSTMCore(nn.Module):
 
     # ------------------------------------------------------------------ #
     def forward(
-        self, x: torch.Tensor, hx: torch.Tensor, cx: torch.Tensor
+        self, x: torch.Tensor, hx: torch.Tensor, cx: torch.Tensor,
+        x2: torch.Tensor | None = None,
     ) -> tuple[dict[str, torch.Tensor], torch.Tensor, torch.Tensor]:
-        """x: (B, S, F); hx/cx: (B, H). Returns ({head: (B,S,D)}, h_S, c_S)."""
+        """x: (B, S, F); hx/cx: (B, H). Returns ({head: (B,S,D)}, h_S, c_S).
+        Dual-body cores additionally take x2: (B, S, F2)."""
+        assert (x2 is None) == (self.input2_dim is None), "dual-body mismatch"
         if x.is_cuda:
             from pdrl_amd import ops
 
             if ops.available():
-                return self._forward_fused(x, hx, cx)
-        return self._forward_eager(x, hx, cx)
+                return self._forward_fused(x, hx, cx, x2)
+        return self._forward_eager(x, hx, cx, x2)
 
-    def _forward_eager(self, x, hx, cx):
+    def _forward_eager(self, x, hx, cx, x2=None):
         B, S, Fdim = x.shape
         H = self.hidden
-        xb = F.relu(x.reshape(B * S, Fdim) @ self.body_w + self.body_b).view(B, S, H)
+        if x2 is None:
+            xb = F.relu(x.reshape(B * S, Fdim) @ self.body_w + self.body_b)
+            xb = xb.view(B, S, H)
+        else:
+            o = F.relu(x.reshape(B * S, Fdim) @ self.body_w + self.body_b)
+            a = F.relu(x2.reshape(B * S, -1) @ self.body2_w + self.body2_b)
+            xb = torch.cat([o, a], dim=-1).view(B, S, H)
         h, c = hx, cx
         hs = []
         for t in range(S):
@@ -118,15 +146,17 @@ class SeqLSTMCore(nn.Module):
         )
         return self.split_heads(outs_cat), h, c
 
-    def _forward_fused(self, x, hx, cx):
+    def _forward_fused(self, x, hx, cx, x2=None):
         from pdrl_amd import ops
 
-        return ops.seq_lstm_forward(self, x, hx, cx)
+        return ops.seq_lstm_forward(self, x, hx, cx, x2)
 
     @torch.no_grad()
-    def step(self, x: torch.Tensor, hx: torch.Tensor, cx: torch.Tensor):
+    def step(self, x: torch.Tensor, hx: torch.Tensor, cx: torch.Tensor,
+             x2: torch.Tensor | None = None):
         """Single-step inference (actor-side). x: (B, F) → ({head:(B,D)}, h, c)."""
-        outs, h, c = self._forward_eager(x.unsqueeze(1), hx, cx)
+        outs, h, c = self._forward_eager(
+            x.unsqueeze(1), hx, cx, x2.unsqueeze(1) if x2 is not None else None)
         return {k: v.squeeze(1) for k, v in outs.items()}, h, c
 
 
@@ -309,7 +339,9 @@ class MlpLSTMCritic(nn.Module):
 
 class MlpLSTMCriticContinuous(nn.Module):
     """SAC-continuous critic Q(s, a): obs and action encoded to half-hidden
-    each, concatenated into the LSTM (reference: networks/models.py:273-322)."""
+    each, concatenated straight into the LSTM — the reference topology
+    (networks/models.py:273-322), realized as a dual-body SeqLSTMCore so
+    the whole critic forward is ONE fused kernel launch on GPU."""
 
     def __init__(self, f: int, n_outputs: int, seq_len: int, hidden_size: int):
         super().__init__()
@@ -318,22 +350,10 @@ class MlpLSTMCriticContinuous(nn.Module):
         self.n_outputs = n_outputs
         self.seq_len = seq_len
         self.hidden_size = hidden_size
-        half = hidden_size // 2
-        self.obs_enc_w = nn.Parameter(torch.empty(f, half))
-        self.obs_enc_b = nn.Parameter(torch.empty(half))
-        self.act_enc_w = nn.Parameter(torch.empty(n_outputs, half))
-        self.act_enc_b = nn.Parameter(torch.empty(half))
-        _init_linear_t(self.obs_enc_w, self.obs_enc_b, f)
-        _init_linear_t(self.act_enc_w, self.act_enc_b, n_outputs)
-        # core body consumes the concatenated encoding (identity-sized body)
-        self.core = SeqLSTMCore(hidden_size, hidden_size, {"q": 1})
+        self.core = SeqLSTMCore(f, hidden_size, {"q": 1}, input2_dim=n_outputs)
 
     def forward(self, obs, act, lstm_hxs):
-        B, S, _ = obs.shape
-        o = F.relu(obs.reshape(B * S, -1) @ self.obs_enc_w + self.obs_enc_b)
-        a = F.relu(act.reshape(B * S, -1) @ self.act_enc_w + self.act_enc_b)
-        enc = torch.cat([o, a], dim=-1).view(B, S, self.hidden_size)
-        outs, _, _ = self.core(enc, *lstm_hxs)
+        outs, _, _ = self.core(obs, *lstm_hxs, x2=act)
         return outs["q"]
 
 

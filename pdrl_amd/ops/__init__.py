@@ -48,8 +48,9 @@ def ext():
     return e
 
 
-def seq_lstm_forward(core, x, hx, cx):
-    """Fused body+LSTM+heads forward via the HIP kernel (autograd-capable)."""
+def seq_lstm_forward(core, x, hx, cx, x2=None):
+    """Fused body+LSTM+heads forward via the HIP kernel (autograd-capable).
+    Dual-body cores (continuous critic) pass their second input as x2."""
     from .fused_core import seq_lstm_apply
 
-    return seq_lstm_apply(core, x, hx, cx)
+    return seq_lstm_apply(core, x, hx, cx, x2)

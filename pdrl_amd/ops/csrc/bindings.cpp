@@ -6,29 +6,34 @@
 std::vector<at::Tensor> seq_lstm_forward_hip(
     const at::Tensor&, const at::Tensor&, const at::Tensor&, const at::Tensor&,
     const at::Tensor&, const at::Tensor&, const at::Tensor&, const at::Tensor&,
-    const at::Tensor&, const at::Tensor&);
+    const at::Tensor&, const at::Tensor&, const c10::optional<at::Tensor>&,
+    const c10::optional<at::Tensor>&, const c10::optional<at::Tensor>&);
 std::vector<at::Tensor> seq_lstm_backward_core_hip(
     const at::Tensor&, const c10::optional<at::Tensor>&,
     const c10::optional<at::Tensor>&, const at::Tensor&, const at::Tensor&,
     const at::Tensor&, const at::Tensor&, const at::Tensor&, const at::Tensor&,
-    const at::Tensor&);
+    const at::Tensor&, const c10::optional<at::Tensor>&);
 std::vector<at::Tensor> seq_lstm_wgrad_hip(
     const at::Tensor&, const at::Tensor&, const at::Tensor&, const at::Tensor&,
-    const at::Tensor&, const at::Tensor&);
+    const at::Tensor&, const at::Tensor&, const c10::optional<at::Tensor>&);
 void seq_lstm_wgrad_out_hip(const at::Tensor&, const at::Tensor&,
                             const at::Tensor&, const at::Tensor&,
                             const at::Tensor&, const at::Tensor&, at::Tensor&,
                             at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
                             at::Tensor&, at::Tensor&,
+                            const c10::optional<at::Tensor>&,
+                            const c10::optional<at::Tensor>&,
+                            const c10::optional<at::Tensor>&,
                             const c10::optional<at::Tensor>&);
 void seq_lstm_forward_multi_hip(const at::Tensor&, const at::Tensor&,
                                 const at::Tensor&, const at::Tensor&,
-                                const at::Tensor&, long, long);
+                                const at::Tensor&, long, long, long, long);
 void seq_lstm_backward_multi_hip(const at::Tensor&, const at::Tensor&,
                                  const at::Tensor&, const at::Tensor&, long,
-                                 long);
+                                 long, long, long, bool);
 void seq_lstm_wgrad_multi_hip(const at::Tensor&, const at::Tensor&,
-                              const at::Tensor&, long, long);
+                              const at::Tensor&, long, long,
+                              const c10::optional<at::Tensor>&, long, long);
 at::Tensor gae_hip(const at::Tensor&, double, double, const at::Tensor&);
 std::vector<at::Tensor> vtrace_hip(const at::Tensor&, const at::Tensor&,
                                    const at::Tensor&, const at::Tensor&,
@@ -103,6 +108,8 @@ void sacc_actor_grad_hip(const at::Tensor&, const at::Tensor&,
                          at::Tensor&, const c10::optional<at::Tensor>&,
                          const c10::optional<at::Tensor>&, double,
                          const c10::optional<at::Tensor>&, double, double);
+void sacc_min_mask_hip(const at::Tensor&, const at::Tensor&, at::Tensor&,
+                       at::Tensor&, at::Tensor&);
 void sacc_critic_loss_hip(const at::Tensor&, const at::Tensor&,
                           const at::Tensor&, const at::Tensor&,
                           const at::Tensor&, const at::Tensor&,
@@ -125,19 +132,46 @@ void soft_update_cached_hip(const at::Tensor&, const at::Tensor&,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("seq_lstm_forward", &seq_lstm_forward_hip,
-        "fused body+LSTM+heads forward (gfx950)");
+        "fused body+LSTM+heads forward (gfx950); dual-body via x2/body2",
+        py::arg("x"), py::arg("h0"), py::arg("c0"), py::arg("body_w"),
+        py::arg("body_b"), py::arg("w_ih"), py::arg("w_hh"), py::arg("b_g"),
+        py::arg("heads_w"), py::arg("heads_b"), py::arg("x2") = c10::nullopt,
+        py::arg("body2_w") = c10::nullopt, py::arg("body2_b") = c10::nullopt);
   m.def("seq_lstm_backward_core", &seq_lstm_backward_core_hip,
-        "fused BPTT backward core (gfx950)");
+        "fused BPTT backward core (gfx950); dual-body adds dx2",
+        py::arg("gouts"), py::arg("ghS"), py::arg("gcS"), py::arg("stash"),
+        py::arg("x"), py::arg("c0"), py::arg("body_w"), py::arg("w_ih"),
+        py::arg("w_hh"), py::arg("heads_w"),
+        py::arg("body2_w") = c10::nullopt);
   m.def("seq_lstm_wgrad", &seq_lstm_wgrad_hip,
-        "MFMA weight-gradient GEMMs + wave-per-element small grads");
+        "MFMA weight-gradient GEMMs + wave-per-element small grads",
+        py::arg("x"), py::arg("h0"), py::arg("stash"), py::arg("dgates"),
+        py::arg("dxb"), py::arg("gouts"), py::arg("x2") = c10::nullopt);
   m.def("seq_lstm_wgrad_out", &seq_lstm_wgrad_out_hip,
-        "wgrad writing into caller buffers (flat grad views)");
+        "wgrad writing into caller buffers (flat grad views)",
+        py::arg("x"), py::arg("h0"), py::arg("stash"), py::arg("dgates"),
+        py::arg("dxb"), py::arg("gouts"), py::arg("dw_ih"), py::arg("dw_hh"),
+        py::arg("dbody_w"), py::arg("dbody_b"), py::arg("db_g"),
+        py::arg("dheads_w"), py::arg("dheads_b"), py::arg("norm_sq"),
+        py::arg("x2") = c10::nullopt, py::arg("dbody2_w") = c10::nullopt,
+        py::arg("dbody2_b") = c10::nullopt);
   m.def("seq_lstm_forward_multi", &seq_lstm_forward_multi_hip,
-        "multi-network fused forward (device pointer tables)");
+        "multi-network fused forward (device pointer tables; dual-body "
+        "rows carry body2/x2 pointers)",
+        py::arg("x"), py::arg("h0"), py::arg("c0"), py::arg("core_tab"),
+        py::arg("out_tab"), py::arg("C"), py::arg("D"), py::arg("F2") = 0,
+        py::arg("half") = 0);
   m.def("seq_lstm_backward_multi", &seq_lstm_backward_multi_hip,
-        "multi-network fused backward (leaf inputs; dgates/dxb only)");
+        "multi-network fused backward (leaf inputs; dgates/dxb [+ dx2])",
+        py::arg("x"), py::arg("c0"), py::arg("in_tab"), py::arg("out_tab"),
+        py::arg("C"), py::arg("D"), py::arg("F2") = 0, py::arg("half") = 0,
+        py::arg("accum_dx2") = false);
   m.def("seq_lstm_wgrad_multi", &seq_lstm_wgrad_multi_hip,
-        "multi-network MFMA weight grads (device pointer table)");
+        "multi-network MFMA weight grads (device pointer table; dual-body "
+        "encoder grad segments)",
+        py::arg("x"), py::arg("h0"), py::arg("tab"), py::arg("C"),
+        py::arg("D"), py::arg("x2") = c10::nullopt, py::arg("F2") = 0,
+        py::arg("half") = 0);
   m.def("gae", &gae_hip, "GAE reverse scan");
   m.def("vtrace", &vtrace_hip, "fused V-trace scan",
         pybind11::arg("behav_lp"), pybind11::arg("target_lp"),
@@ -181,6 +215,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("beta2") = 0.999);
   m.def("sacc_critic_loss", &sacc_critic_loss_hip,
         "SAC-continuous soft-Q target + twin critic loss grads");
+  m.def("sacc_min_mask", &sacc_min_mask_hip,
+        "min-critic selection masks for dE[-minQ]/dq + zero the dQ/da "
+        "accumulator");
   m.def("l2norm_sq", &l2norm_sq_hip, "squared L2 norm into a device scalar");
   m.def("rmsprop_step", &rmsprop_step_hip, "fused clip+RMSprop on flat buffers");
   m.def("adam_step", &adam_step_hip, "fused clip+Adam on flat buffers",

@@ -263,3 +263,43 @@ void sacc_critic_loss_hip(const at::Tensor& q1, const at::Tensor& q2,
                      B, S, (float)gamma, (float)rew_scale);
   HIP_CHECK_LAST();
 }
+
+namespace {
+
+// Min-critic selection for the actor objective: the gradient of
+// -E[min(Q1,Q2)] routes 1/N to whichever critic is lower per element
+// (reference eager: torch.min + mask — sac_continuous/learning.py:44-55).
+// Also zeroes the shared dQ/da accumulator the twin-critic backward
+// atomically adds into (one launch replaces mask/mul/zero eager chain).
+__global__ void sacc_min_mask_kernel(const float* __restrict__ qp1,
+                                     const float* __restrict__ qp2,
+                                     float* __restrict__ gq1,
+                                     float* __restrict__ gq2,
+                                     float* __restrict__ dact, int N,
+                                     int n_dact) {
+  const float inv = -1.0f / (float)N;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < N;
+       i += gridDim.x * blockDim.x) {
+    const bool m = qp1[i] <= qp2[i];
+    gq1[i] = m ? inv : 0.f;
+    gq2[i] = m ? 0.f : inv;
+  }
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < n_dact;
+       i += gridDim.x * blockDim.x) {
+    dact[i] = 0.f;
+  }
+}
+
+}  // namespace
+
+void sacc_min_mask_hip(const at::Tensor& qp1, const at::Tensor& qp2,
+                       at::Tensor& gq1, at::Tensor& gq2, at::Tensor& dact) {
+  const int N = (int)qp1.numel();
+  const int blocks = std::min(2048, (N + kThreads - 1) / kThreads + 1);
+  hipLaunchKernelGGL(sacc_min_mask_kernel, dim3(blocks), dim3(kThreads), 0,
+                     current_stream(), qp1.data_ptr<float>(),
+                     qp2.data_ptr<float>(), gq1.data_ptr<float>(),
+                     gq2.data_ptr<float>(), dact.data_ptr<float>(), N,
+                     (int)dact.numel());
+  HIP_CHECK_LAST();
+}

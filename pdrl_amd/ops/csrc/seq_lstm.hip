@@ -40,6 +40,13 @@ constexpr int kStashFields = 7;
 
 // Whole-sequence forward for ONE batch row (the workgroup). Shared by the
 // single-core kernel and the multi-core (pointer-table) kernel below.
+//
+// Dual-body mode (x2 != nullptr): the LSTM input is the concatenation
+// [relu(x·body_w + body_b) | relu(x2·body2_w + body2_b)] with the split at
+// ``half`` — the continuous-critic topology (obs encoder ‖ action encoder,
+// reference: networks/models.py MlpLSTMCriticContinuous 273-322). Single
+// body passes half == H and the extra pointers null; body_w's column
+// stride is ``half`` in both modes (== H for single).
 template <int H>
 __device__ __forceinline__ void seq_lstm_fwd_row(
     const float* __restrict__ x,       // (B,S,F)
@@ -56,7 +63,11 @@ __device__ __forceinline__ void seq_lstm_fwd_row(
     float* __restrict__ hS,            // (B,H)
     float* __restrict__ cS,            // (B,H)
     float* __restrict__ stash,         // (B,S,7H)
-    int b, int S, int F, int D, long h0s, char* smem_raw) {
+    int b, int S, int F, int D, long h0s, char* smem_raw,
+    const float* __restrict__ x2 = nullptr,       // (B,S,F2) dual body
+    const float* __restrict__ body2_w = nullptr,  // (F2,half)
+    const float* __restrict__ body2_b = nullptr,  // (half)
+    int F2 = 0, int half = H) {
   constexpr int G = 4 * H;
   const int tid = threadIdx.x;
 
@@ -76,12 +87,23 @@ __device__ __forceinline__ void seq_lstm_fwd_row(
   PDRL_PIN_REGS(whh, H);
   const float bias = b_g[tid];
 
-  // Body GEMM + ReLU for all S steps of this row (K1).
+  // Body GEMM + ReLU for all S steps of this row (K1); dual mode computes
+  // the [obs-enc | act-enc] split in the same pass.
   for (int idx = tid; idx < S * H; idx += G) {
     const int t = idx / H, j = idx % H;
-    float acc = body_b[j];
-    const float* xr = x + ((long)b * S + t) * F;
-    for (int k = 0; k < F; ++k) acc = fmaf(xr[k], body_w[k * H + j], acc);
+    float acc;
+    if (j < half) {
+      acc = body_b[j];
+      const float* xr = x + ((long)b * S + t) * F;
+      for (int k = 0; k < F; ++k) acc = fmaf(xr[k], body_w[k * half + j], acc);
+    } else {
+      const int jj = j - half;
+      const int w2s = H - half;  // body2_w column count (its row stride)
+      acc = body2_b[jj];
+      const float* xr = x2 + ((long)b * S + t) * F2;
+      for (int k = 0; k < F2; ++k)
+        acc = fmaf(xr[k], body2_w[k * w2s + jj], acc);
+    }
     acc = fmaxf(acc, 0.0f);
     xb[t * H + j] = acc;
     stash[(((long)b * S + t) * kStashFields) * H + j] = acc;
@@ -180,12 +202,13 @@ template <int H>
 __global__ __launch_bounds__(4 * H) void seq_lstm_fwd_multi_kernel(
     const float* __restrict__ x, const float* __restrict__ h0,
     const float* __restrict__ c0,
-    const long* __restrict__ core_tab,  // [C][7] body_w..heads_b
+    const long* __restrict__ core_tab,  // [C][10] body_w..heads_b,body2_w,body2_b,x2
     const long* __restrict__ out_tab,   // [C][4] outs,hS,cS,stash
-    int S, int F, int D, long h0s) {
-  const long* ct = core_tab + (long)blockIdx.y * 7;
+    int S, int F, int D, long h0s, int F2, int half) {
+  const long* ct = core_tab + (long)blockIdx.y * 10;
   const long* ot = out_tab + (long)blockIdx.y * 4;
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  const float* x2 = reinterpret_cast<const float*>(ct[9]);
   seq_lstm_fwd_row<H>(
       x, h0, c0, reinterpret_cast<const float*>(ct[0]),
       reinterpret_cast<const float*>(ct[1]),
@@ -196,7 +219,10 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_fwd_multi_kernel(
       reinterpret_cast<const float*>(ct[6]),
       reinterpret_cast<float*>(ot[0]), reinterpret_cast<float*>(ot[1]),
       reinterpret_cast<float*>(ot[2]), reinterpret_cast<float*>(ot[3]),
-      blockIdx.x, S, F, D, h0s, smem_raw);
+      blockIdx.x, S, F, D, h0s, smem_raw, x2,
+      reinterpret_cast<const float*>(ct[7]),
+      reinterpret_cast<const float*>(ct[8]),
+      x2 != nullptr ? F2 : 0, x2 != nullptr ? half : H);
 }
 
 // Backward through heads + recurrence + body for one batch row.
@@ -218,9 +244,12 @@ __device__ __forceinline__ void seq_lstm_bwd_row(
     float* __restrict__ dx,            // (B,S,F) or nullptr (leaf input)
     float* __restrict__ dh0,           // (B,H)   or nullptr
     float* __restrict__ dc0,           // (B,H)   or nullptr
-    float* __restrict__ dgates,        // (B,S,4H) pre-activation
+    float* __restrict__ dgates,        // (B,S,4H) pre-activation, or nullptr
     float* __restrict__ dxb,           // (B,S,H) pre-ReLU
-    int b, int S, int F, int D, long h0s, char* smem_raw) {
+    int b, int S, int F, int D, long h0s, char* smem_raw,
+    const float* __restrict__ body2_w = nullptr,  // (F2,half) dual body
+    float* __restrict__ dx2 = nullptr,            // (B,S,F2) second-input grad
+    int F2 = 0, int half = H, bool accum_dx2 = false) {
   constexpr int G = 4 * H;
   const int tid = threadIdx.x;
 
@@ -283,7 +312,7 @@ __device__ __forceinline__ void seq_lstm_bwd_row(
     }
     __syncthreads();
     // persist pre-activation gate grads for the weight GEMMs
-    dgates[((long)b * S + t) * G + tid] = dg4[tid];
+    if (dgates != nullptr) dgates[((long)b * S + t) * G + tid] = dg4[tid];
     {
       // recurrent + body back-projection, split over all 4H threads on
       // register-resident weight rows (dg4 reads broadcast from LDS)
@@ -324,18 +353,37 @@ __device__ __forceinline__ void seq_lstm_bwd_row(
     dh0[(long)b * H + tid] = dh_rec;
     dc0[(long)b * H + tid] = dc_rec;
   }
-  if (dx == nullptr) return;  // leaf input: skip the dx GEMM entirely
+  if (dx == nullptr && dx2 == nullptr) return;  // leaf inputs: skip dx GEMMs
   __syncthreads();
 
-  // dx[t][f] = sum_j dxb[t][j] * body_w[f][j]
-  for (int idx = tid; idx < S * F; idx += G) {
-    const int t = idx / F, f = idx % F;
-    float acc = 0.0f;
-    const float* dr = dxb_s + t * H;
-    const float* wr = body_w + f * H;
-#pragma unroll
-    for (int j = 0; j < H; ++j) acc = fmaf(dr[j], wr[j], acc);
-    dx[((long)b * S + t) * F + f] = acc;
+  // dx[t][f] = sum_{j<half} dxb[t][j] * body_w[f][j]  (half == H for single)
+  if (dx != nullptr) {
+    for (int idx = tid; idx < S * F; idx += G) {
+      const int t = idx / F, f = idx % F;
+      float acc = 0.0f;
+      const float* dr = dxb_s + t * H;
+      const float* wr = body_w + f * half;
+      for (int j = 0; j < half; ++j) acc = fmaf(dr[j], wr[j], acc);
+      dx[((long)b * S + t) * F + f] = acc;
+    }
+  }
+  // dual body: dx2[t][f] = sum_{j} dxb[t][half+j] * body2_w[f][j] — the
+  // cross-network dQ/da path; accum mode atomically adds (twin critics
+  // sum their action grads into one buffer, caller zeroes it first)
+  if (dx2 != nullptr) {
+    for (int idx = tid; idx < S * F2; idx += G) {
+      const int t = idx / F2, f = idx % F2;
+      float acc = 0.0f;
+      const float* dr = dxb_s + t * H + half;
+      const float* wr = body2_w + f * (H - half);
+      for (int j = 0; j < H - half; ++j) acc = fmaf(dr[j], wr[j], acc);
+      float* out = dx2 + ((long)b * S + t) * F2 + f;
+      if (accum_dx2) {
+        atomicAdd(out, acc);
+      } else {
+        *out = acc;
+      }
+    }
   }
 }
 
@@ -362,11 +410,11 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_kernel(
 template <int H>
 __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_multi_kernel(
     const float* __restrict__ x, const float* __restrict__ c0,
-    const long* __restrict__ in_tab,   // [C][6] gouts,stash,w_ih,w_hh,heads_w,body_w
-    const long* __restrict__ out_tab,  // [C][2] dgates,dxb
-    int S, int F, int D, long h0s) {
-  const long* it = in_tab + (long)blockIdx.y * 6;
-  const long* ot = out_tab + (long)blockIdx.y * 2;
+    const long* __restrict__ in_tab,   // [C][7] gouts,stash,w_ih,w_hh,heads_w,body_w,body2_w
+    const long* __restrict__ out_tab,  // [C][3] dgates,dxb,dx2
+    int S, int F, int D, long h0s, int F2, int half, int accum_dx2) {
+  const long* it = in_tab + (long)blockIdx.y * 7;
+  const long* ot = out_tab + (long)blockIdx.y * 3;
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   seq_lstm_bwd_row<H>(
       reinterpret_cast<const float*>(it[0]), nullptr, nullptr,
@@ -376,7 +424,26 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_multi_kernel(
       reinterpret_cast<const float*>(it[3]),
       reinterpret_cast<const float*>(it[4]), nullptr, nullptr, nullptr,
       reinterpret_cast<float*>(ot[0]), reinterpret_cast<float*>(ot[1]),
-      blockIdx.x, S, F, D, h0s, smem_raw);
+      blockIdx.x, S, F, D, h0s, smem_raw,
+      reinterpret_cast<const float*>(it[6]),
+      reinterpret_cast<float*>(ot[2]), F2, half, accum_dx2 != 0);
+}
+
+template <int H>
+__global__ __launch_bounds__(4 * H) void seq_lstm_fwd_dual_kernel(
+    const float* __restrict__ x, const float* __restrict__ h0,
+    const float* __restrict__ c0, const float* __restrict__ body_w,
+    const float* __restrict__ body_b, const float* __restrict__ w_ih,
+    const float* __restrict__ w_hh, const float* __restrict__ b_g,
+    const float* __restrict__ heads_w, const float* __restrict__ heads_b,
+    float* __restrict__ outs, float* __restrict__ hS, float* __restrict__ cS,
+    float* __restrict__ stash, int S, int F, int D, long h0s,
+    const float* __restrict__ x2, const float* __restrict__ body2_w,
+    const float* __restrict__ body2_b, int F2, int half) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  seq_lstm_fwd_row<H>(x, h0, c0, body_w, body_b, w_ih, w_hh, b_g, heads_w,
+                      heads_b, outs, hS, cS, stash, blockIdx.x, S, F, D, h0s,
+                      smem_raw, x2, body2_w, body2_b, F2, half);
 }
 
 template <int H>
@@ -385,9 +452,29 @@ void launch_fwd(const at::Tensor& x, const at::Tensor& h0, const at::Tensor& c0,
                 const at::Tensor& w_ih, const at::Tensor& w_hh,
                 const at::Tensor& b_g, const at::Tensor& heads_w,
                 const at::Tensor& heads_b, at::Tensor& outs, at::Tensor& hS,
-                at::Tensor& cS, at::Tensor& stash, int B, int S, int F, int D) {
+                at::Tensor& cS, at::Tensor& stash, int B, int S, int F, int D,
+                const c10::optional<at::Tensor>& x2 = c10::nullopt,
+                const c10::optional<at::Tensor>& body2_w = c10::nullopt,
+                const c10::optional<at::Tensor>& body2_b = c10::nullopt) {
   const int lds =
       (2 * S * H + 4 * H + 2 * H) * sizeof(float);
+  if (x2.has_value()) {
+    const int F2 = x2->size(2);
+    const int half = body_w.size(1);
+    hipLaunchKernelGGL((seq_lstm_fwd_dual_kernel<H>), dim3(B), dim3(4 * H),
+                       lds, current_stream(), x.data_ptr<float>(),
+                       h0.data_ptr<float>(), c0.data_ptr<float>(),
+                       body_w.data_ptr<float>(), body_b.data_ptr<float>(),
+                       w_ih.data_ptr<float>(), w_hh.data_ptr<float>(),
+                       b_g.data_ptr<float>(), heads_w.data_ptr<float>(),
+                       heads_b.data_ptr<float>(), outs.data_ptr<float>(),
+                       hS.data_ptr<float>(), cS.data_ptr<float>(),
+                       stash.data_ptr<float>(), S, F, D, (long)h0.stride(0),
+                       x2->data_ptr<float>(), body2_w->data_ptr<float>(),
+                       body2_b->data_ptr<float>(), F2, half);
+    HIP_CHECK_LAST();
+    return;
+  }
   hipLaunchKernelGGL((seq_lstm_fwd_kernel<H>), dim3(B), dim3(4 * H), lds,
                      current_stream(), x.data_ptr<float>(),
                      h0.data_ptr<float>(), c0.data_ptr<float>(),
@@ -403,6 +490,23 @@ void launch_fwd(const at::Tensor& x, const at::Tensor& h0, const at::Tensor& c0,
 }
 
 template <int H>
+__global__ __launch_bounds__(4 * H) void seq_lstm_bwd_dual_kernel(
+    const float* __restrict__ gouts, const float* __restrict__ ghS,
+    const float* __restrict__ gcS, const float* __restrict__ stash,
+    const float* __restrict__ x, const float* __restrict__ c0,
+    const float* __restrict__ body_w, const float* __restrict__ w_ih,
+    const float* __restrict__ w_hh, const float* __restrict__ heads_w,
+    float* __restrict__ dx, float* __restrict__ dh0, float* __restrict__ dc0,
+    float* __restrict__ dgates, float* __restrict__ dxb, int S, int F, int D,
+    long h0s, const float* __restrict__ body2_w, float* __restrict__ dx2,
+    int F2, int half) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  seq_lstm_bwd_row<H>(gouts, ghS, gcS, stash, x, c0, body_w, w_ih, w_hh,
+                      heads_w, dx, dh0, dc0, dgates, dxb, blockIdx.x, S, F, D,
+                      h0s, smem_raw, body2_w, dx2, F2, half, false);
+}
+
+template <int H>
 void launch_bwd(const at::Tensor& gouts, const c10::optional<at::Tensor>& ghS,
                 const c10::optional<at::Tensor>& gcS, const at::Tensor& stash,
                 const at::Tensor& x, const at::Tensor& c0,
@@ -410,10 +514,30 @@ void launch_bwd(const at::Tensor& gouts, const c10::optional<at::Tensor>& ghS,
                 const at::Tensor& w_hh, const at::Tensor& heads_w,
                 at::Tensor& dx, at::Tensor& dh0, at::Tensor& dc0,
                 at::Tensor& dgates, at::Tensor& dxb, int B, int S, int F,
-                int D) {
+                int D,
+                const c10::optional<at::Tensor>& body2_w = c10::nullopt,
+                at::Tensor* dx2 = nullptr) {
   const int G = 4 * H;
   const int lds = (S * H + G + S * H + 2 * G) * sizeof(float);
   TORCH_CHECK(lds <= 160 * 1024, "backward LDS footprint exceeds 160 KiB");
+  if (body2_w.has_value()) {
+    const int half = body_w.size(1);
+    const int F2 = body2_w->size(0);
+    hipLaunchKernelGGL(
+        (seq_lstm_bwd_dual_kernel<H>), dim3(B), dim3(G), lds,
+        current_stream(), gouts.data_ptr<float>(),
+        ghS.has_value() ? ghS->data_ptr<float>() : nullptr,
+        gcS.has_value() ? gcS->data_ptr<float>() : nullptr,
+        stash.data_ptr<float>(), x.data_ptr<float>(), c0.data_ptr<float>(),
+        body_w.data_ptr<float>(), w_ih.data_ptr<float>(),
+        w_hh.data_ptr<float>(), heads_w.data_ptr<float>(),
+        dx.data_ptr<float>(), dh0.data_ptr<float>(), dc0.data_ptr<float>(),
+        dgates.data_ptr<float>(), dxb.data_ptr<float>(), S, F, D,
+        (long)c0.stride(0), body2_w->data_ptr<float>(),
+        dx2->data_ptr<float>(), F2, half);
+    HIP_CHECK_LAST();
+    return;
+  }
   hipLaunchKernelGGL(
       (seq_lstm_bwd_kernel<H>), dim3(B), dim3(G), lds, current_stream(),
       gouts.data_ptr<float>(),
@@ -434,7 +558,9 @@ std::vector<at::Tensor> seq_lstm_forward_hip(
     const at::Tensor& x, const at::Tensor& h0, const at::Tensor& c0,
     const at::Tensor& body_w, const at::Tensor& body_b, const at::Tensor& w_ih,
     const at::Tensor& w_hh, const at::Tensor& b_g, const at::Tensor& heads_w,
-    const at::Tensor& heads_b) {
+    const at::Tensor& heads_b, const c10::optional<at::Tensor>& x2,
+    const c10::optional<at::Tensor>& body2_w,
+    const c10::optional<at::Tensor>& body2_b) {
   CHECK_IN(x); CHECK_IN(body_w); CHECK_IN(body_b);
   CHECK_IN(w_ih); CHECK_IN(w_hh); CHECK_IN(b_g); CHECK_IN(heads_w);
   CHECK_IN(heads_b);
@@ -443,7 +569,16 @@ std::vector<at::Tensor> seq_lstm_forward_hip(
   TORCH_CHECK(h0.stride(0) == c0.stride(0), "h0/c0 stride mismatch");
   const int B = x.size(0), S = x.size(1), F = x.size(2);
   const int H = h0.size(1), D = heads_w.size(1);
-  TORCH_CHECK(body_w.size(0) == F && body_w.size(1) == H, "body_w shape");
+  const bool dual = x2.has_value();
+  if (dual) {
+    CHECK_IN((*x2)); CHECK_IN((*body2_w)); CHECK_IN((*body2_b));
+    TORCH_CHECK(body_w.size(0) == F, "body_w rows");
+    TORCH_CHECK(body_w.size(1) + body2_w->size(1) == H,
+                "dual body halves must sum to H");
+    TORCH_CHECK(body2_w->size(0) == x2->size(2), "body2_w rows");
+  } else {
+    TORCH_CHECK(body_w.size(0) == F && body_w.size(1) == H, "body_w shape");
+  }
   TORCH_CHECK(w_ih.size(0) == H && w_ih.size(1) == 4 * H, "w_ih shape");
   TORCH_CHECK(S >= 1 && S <= 32, "seq_len must be in [1, 32]");
 
@@ -454,9 +589,9 @@ std::vector<at::Tensor> seq_lstm_forward_hip(
   auto stash = at::empty({B, S, kStashFields * H}, opt);
 
   switch (H) {
-    case 32: launch_fwd<32>(x, h0, c0, body_w, body_b, w_ih, w_hh, b_g, heads_w, heads_b, outs, hS, cS, stash, B, S, F, D); break;
-    case 64: launch_fwd<64>(x, h0, c0, body_w, body_b, w_ih, w_hh, b_g, heads_w, heads_b, outs, hS, cS, stash, B, S, F, D); break;
-    case 128: launch_fwd<128>(x, h0, c0, body_w, body_b, w_ih, w_hh, b_g, heads_w, heads_b, outs, hS, cS, stash, B, S, F, D); break;
+    case 32: launch_fwd<32>(x, h0, c0, body_w, body_b, w_ih, w_hh, b_g, heads_w, heads_b, outs, hS, cS, stash, B, S, F, D, x2, body2_w, body2_b); break;
+    case 64: launch_fwd<64>(x, h0, c0, body_w, body_b, w_ih, w_hh, b_g, heads_w, heads_b, outs, hS, cS, stash, B, S, F, D, x2, body2_w, body2_b); break;
+    case 128: launch_fwd<128>(x, h0, c0, body_w, body_b, w_ih, w_hh, b_g, heads_w, heads_b, outs, hS, cS, stash, B, S, F, D, x2, body2_w, body2_b); break;
     default:
       TORCH_CHECK(false, "hidden size ", H, " unsupported (32/64/128)");
   }
@@ -466,15 +601,18 @@ std::vector<at::Tensor> seq_lstm_forward_hip(
 void seq_lstm_forward_multi_hip(const at::Tensor& x, const at::Tensor& h0,
                                 const at::Tensor& c0,
                                 const at::Tensor& core_tab,
-                                const at::Tensor& out_tab, long C, long D) {
+                                const at::Tensor& out_tab, long C, long D,
+                                long F2, long half) {
   CHECK_IN(x);
   CHECK_GPU(h0); CHECK_F32(h0); CHECK_GPU(c0); CHECK_F32(c0);
   CHECK_GPU(core_tab); CHECK_GPU(out_tab);
   TORCH_CHECK(h0.stride(1) == 1 && c0.stride(1) == 1, "h0/c0 inner stride");
   TORCH_CHECK(h0.stride(0) == c0.stride(0), "h0/c0 stride mismatch");
+  TORCH_CHECK(core_tab.size(-1) == 10, "core_tab rows must be 10-wide");
   const int B = x.size(0), S = x.size(1), F = x.size(2);
   const int H = h0.size(1);
   TORCH_CHECK(S >= 1 && S <= 32, "seq_len must be in [1, 32]");
+  if (half <= 0) half = H;
   const long h0s = (long)h0.stride(0);
   dim3 grid(B, (unsigned)C);
 #define PDRL_LAUNCH_FWD_MULTI(HH)                                             \
@@ -483,7 +621,7 @@ void seq_lstm_forward_multi_hip(const at::Tensor& x, const at::Tensor& h0,
                      current_stream(), x.data_ptr<float>(),                   \
                      h0.data_ptr<float>(), c0.data_ptr<float>(),              \
                      core_tab.data_ptr<long>(), out_tab.data_ptr<long>(), S,  \
-                     F, (int)D, h0s)
+                     F, (int)D, h0s, (int)F2, (int)half)
   switch (H) {
     case 32: PDRL_LAUNCH_FWD_MULTI(32); break;
     case 64: PDRL_LAUNCH_FWD_MULTI(64); break;
@@ -496,13 +634,17 @@ void seq_lstm_forward_multi_hip(const at::Tensor& x, const at::Tensor& h0,
 
 void seq_lstm_backward_multi_hip(const at::Tensor& x, const at::Tensor& c0,
                                  const at::Tensor& in_tab,
-                                 const at::Tensor& out_tab, long C, long D) {
+                                 const at::Tensor& out_tab, long C, long D,
+                                 long F2, long half, bool accum_dx2) {
   CHECK_IN(x);
   CHECK_GPU(c0); CHECK_F32(c0);
   CHECK_GPU(in_tab); CHECK_GPU(out_tab);
   TORCH_CHECK(c0.stride(1) == 1, "c0 inner stride must be 1");
+  TORCH_CHECK(in_tab.size(-1) == 7 && out_tab.size(-1) == 3,
+              "bwd_multi tables must be 7/3-wide");
   const int B = x.size(0), S = x.size(1), F = x.size(2);
   const int H = c0.size(1);
+  if (half <= 0) half = H;
   const long h0s = (long)c0.stride(0);
   dim3 grid(B, (unsigned)C);
 #define PDRL_LAUNCH_BWD_MULTI(HH)                                             \
@@ -512,7 +654,8 @@ void seq_lstm_backward_multi_hip(const at::Tensor& x, const at::Tensor& c0,
     hipLaunchKernelGGL((seq_lstm_bwd_multi_kernel<HH>), grid, dim3(G), lds,   \
                        current_stream(), x.data_ptr<float>(),                 \
                        c0.data_ptr<float>(), in_tab.data_ptr<long>(),         \
-                       out_tab.data_ptr<long>(), S, F, (int)D, h0s);          \
+                       out_tab.data_ptr<long>(), S, F, (int)D, h0s, (int)F2,  \
+                       (int)half, accum_dx2 ? 1 : 0);                         \
   } while (0)
   switch (H) {
     case 32: PDRL_LAUNCH_BWD_MULTI(32); break;
@@ -529,13 +672,14 @@ std::vector<at::Tensor> seq_lstm_backward_core_hip(
     const c10::optional<at::Tensor>& gcS, const at::Tensor& stash,
     const at::Tensor& x, const at::Tensor& c0, const at::Tensor& body_w,
     const at::Tensor& w_ih, const at::Tensor& w_hh,
-    const at::Tensor& heads_w) {
+    const at::Tensor& heads_w, const c10::optional<at::Tensor>& body2_w) {
   CHECK_IN(gouts); CHECK_IN(stash); CHECK_IN(x);
   CHECK_IN(body_w); CHECK_IN(w_ih); CHECK_IN(w_hh); CHECK_IN(heads_w);
   CHECK_GPU(c0); CHECK_F32(c0);
   TORCH_CHECK(c0.stride(1) == 1, "c0 inner stride must be 1");
   const int B = x.size(0), S = x.size(1), F = x.size(2);
   const int H = c0.size(1), D = heads_w.size(1);
+  const bool dual = body2_w.has_value();
 
   auto opt = x.options();
   auto dx = at::empty({B, S, F}, opt);
@@ -543,13 +687,20 @@ std::vector<at::Tensor> seq_lstm_backward_core_hip(
   auto dc0 = at::empty({B, H}, opt);
   auto dgates = at::empty({B, S, 4 * H}, opt);
   auto dxb = at::empty({B, S, H}, opt);
+  at::Tensor dx2;
+  at::Tensor* dx2p = nullptr;
+  if (dual) {
+    dx2 = at::empty({B, S, (long)body2_w->size(0)}, opt);
+    dx2p = &dx2;
+  }
 
   switch (H) {
-    case 32: launch_bwd<32>(gouts, ghS, gcS, stash, x, c0, body_w, w_ih, w_hh, heads_w, dx, dh0, dc0, dgates, dxb, B, S, F, D); break;
-    case 64: launch_bwd<64>(gouts, ghS, gcS, stash, x, c0, body_w, w_ih, w_hh, heads_w, dx, dh0, dc0, dgates, dxb, B, S, F, D); break;
-    case 128: launch_bwd<128>(gouts, ghS, gcS, stash, x, c0, body_w, w_ih, w_hh, heads_w, dx, dh0, dc0, dgates, dxb, B, S, F, D); break;
+    case 32: launch_bwd<32>(gouts, ghS, gcS, stash, x, c0, body_w, w_ih, w_hh, heads_w, dx, dh0, dc0, dgates, dxb, B, S, F, D, body2_w, dx2p); break;
+    case 64: launch_bwd<64>(gouts, ghS, gcS, stash, x, c0, body_w, w_ih, w_hh, heads_w, dx, dh0, dc0, dgates, dxb, B, S, F, D, body2_w, dx2p); break;
+    case 128: launch_bwd<128>(gouts, ghS, gcS, stash, x, c0, body_w, w_ih, w_hh, heads_w, dx, dh0, dc0, dgates, dxb, B, S, F, D, body2_w, dx2p); break;
     default:
       TORCH_CHECK(false, "hidden size ", H, " unsupported (32/64/128)");
   }
+  if (dual) return {dx, dh0, dc0, dgates, dxb, dx2};
   return {dx, dh0, dc0, dgates, dxb};
 }

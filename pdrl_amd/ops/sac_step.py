@@ -92,10 +92,12 @@ class FusedSacStep(GraphableStep):
             return torch.tensor(rows, dtype=torch.int64).to(dev)
 
         def wrow(core):
+            # 10-wide row; last three (body2_w, body2_b, x2) are the
+            # dual-body slots, unused for the discrete (single-body) cores
             return [core.body_w.data_ptr(), core.body_b.data_ptr(),
                     core.w_ih.data_ptr(), core.w_hh.data_ptr(),
                     core.b_g.data_ptr(), core.heads_w.data_ptr(),
-                    core.heads_b.data_ptr()]
+                    core.heads_b.data_ptr(), 0, 0, 0]
 
         def orow(b):
             return [b["outs"].data_ptr(), b["hS"].data_ptr(),
@@ -112,11 +114,11 @@ class FusedSacStep(GraphableStep):
             core, b = c[name], buf[name]
             return [b["gq"].data_ptr(), b["stash"].data_ptr(),
                     core.w_ih.data_ptr(), core.w_hh.data_ptr(),
-                    core.heads_w.data_ptr(), core.body_w.data_ptr()]
+                    core.heads_w.data_ptr(), core.body_w.data_ptr(), 0]
 
         def borow(name):
             b = buf[name]
-            return [b["dgates"].data_ptr(), b["dxb"].data_ptr()]
+            return [b["dgates"].data_ptr(), b["dxb"].data_ptr(), 0]
 
         self.bwd_in = t64([birow("q1"), birow("q2")])
         self.bwd_out = t64([borow("q1"), borow("q2")])
@@ -134,7 +136,7 @@ class FusedSacStep(GraphableStep):
                     b["dxb"].data_ptr(), b["gq"].data_ptr(),
                     g[2].data_ptr(), g[3].data_ptr(), g[0].data_ptr(),
                     g[1].data_ptr(), g[4].data_ptr(), g[5].data_ptr(),
-                    g[6].data_ptr(), nrm]
+                    g[6].data_ptr(), nrm, 0, 0]
 
         self.wg_tab = t64([grow("q1"), grow("q2")])
         self._mshape = (B, S)

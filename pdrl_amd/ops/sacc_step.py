@@ -1,29 +1,30 @@
 """FusedSacContinuousStep: the SAC-Continuous training iteration as a fixed
 HIP kernel DAG — the continuous half of K11,
-hipGraph-capturable.
+hipGraph-capturable, ~18 launches per step (the round-1 version ran ~95:
+eager encoder GEMMs/cat/clamp/pow around the fused kernels —
+profiles/algo_breakdown_r02a.md).
 
 Follows the EAGER reference ordering exactly
 (agents/learner_module/sac_continuous/learning.py, reference
 sac_continuous/learning.py:13-151):
   1.  actor fwd → reparameterized tanh-Gaussian sample (sacc_sample:
-      in-kernel counter RNG, graph-replay safe) → a_new, log pi
-  2.  twin critics on (obs, a_new); d minQ/da via the critic cores'
-      backward dx routed back through the action encoder (the
-      cross-network gradient)
-  3.  sacc_actor_grad — analytic dmu/dlog_std (verified vs autograd,
-      tests/test_sacc_analytic.py) + dlog_alpha + stats
-  4.  actor BPTT + MFMA wgrad → fused Adam; alpha Adam
-  5.  actor fwd AGAIN (post-update sample a', log pi')
-  6.  target critics on (obs, a'); behavior critics on (obs, batch act)
-  7.  sacc_critic_loss — soft-Q target, twin huber, analytic dq1/dq2
-  8.  critic BPTT + wgrads (cores via MFMA kernels; the obs/act encoders
-      via library GEMMs into the flat-grad views) → fused Adam
-  9.  Polyak soft_update (cached device pointer tables)
-
-The critic stacks are MlpLSTMCriticContinuous: enc = [relu(obs@Wo+bo) |
-relu(act@Wa+ba)] feeding the LSTM core (networks/models.py) — the encoder
-fwd/bwd are single library GEMMs and stay as torch ops inside the captured
-graph.
+      in-kernel counter RNG, graph-replay safe) → a1, log pi
+  2.  twin critics on (obs, a1) — ONE dual-body multi-network launch (the
+      critic is a dual-body SeqLSTMCore: obs/action encoders feed the LSTM
+      directly, matching the reference topology)
+  3.  sacc_min_mask — min-critic selection grads + zero the dQ/da buffer
+  4.  twin-critic input-grad backward — ONE multi launch, both critics
+      atomically accumulate dminQ/da into the shared buffer
+  5.  sacc_actor_grad — analytic dmu/dlog_std + dlog_alpha + stats + the
+      shared Adam clock prep (all in-kernel)
+  6.  actor BPTT + MFMA wgrad → actor+alpha Adam in ONE multi-group launch
+  7.  actor fwd AGAIN (post-update sample a2, log pi2)
+  8.  target critics on (obs,a2) + behavior critics on (obs,act) — ONE
+      4-network dual-body launch (per-core x2 pointers)
+  9.  sacc_critic_loss — soft-Q target, twin huber, analytic dq1/dq2
+  10. twin-critic BPTT (one launch) + MFMA wgrads incl. encoder segments
+      (one launch) → critic Adam
+  11. Polyak soft_update (cached device pointer tables)
 """
 from __future__ import annotations
 
@@ -42,10 +43,10 @@ class FusedSacContinuousStep(GraphableStep):
         self.grad_reducer = updater.grad_reducer
         self.actor_core = updater.actor.core
         self.q = {
-            "q1": updater.critic.q1,
-            "q2": updater.critic.q2,
-            "t1": updater.target_critic.q1,
-            "t2": updater.target_critic.q2,
+            "q1": updater.critic.q1.core,
+            "q2": updater.critic.q2.core,
+            "t1": updater.target_critic.q1.core,
+            "t2": updater.target_critic.q2.core,
         }
         dev = self.actor_core.body_w.device
         self.stats_buf = torch.zeros(8, dtype=torch.float32, device=dev)
@@ -60,83 +61,126 @@ class FusedSacContinuousStep(GraphableStep):
 
             self.adam_aa = AdamMultiGroup(
                 [updater.actor_optimizer, updater.alpha_optimizer])
+        self._mshape = None
 
     def fits(self, batch) -> bool:
         return True  # loss kernels grid-stride; no LDS shape limit
 
     # ------------------------------------------------------------------ #
+    def _multi_setup(self, batch):
+        """Preallocate activation/grad buffers and device pointer tables for
+        the multi-network dual-body launches. Tables reference persistent
+        storage only (graph-replay safe): parameters/grads are flat-space
+        views; a1/a2/actb sample buffers are owned here."""
+        u = self.u
+        x = batch["obs"]
+        B, S, _ = x.shape
+        dev = x.device
+        ac = self.actor_core
+        H = ac.w_ih.size(0)
+        A = u.actor.n_outputs
+        self._A = A
+        q1 = self.q["q1"]
+        assert q1.body2_w is not None, "continuous critic must be dual-body"
+        self._half = q1.body_w.size(1)
+
+        def mk(*shape):
+            return torch.empty(*shape, device=dev)
+
+        buf = {
+            "a1": mk(B, S, A), "eps1": mk(B, S, A), "logpi1": mk(B, S, 1),
+            "a2": mk(B, S, A), "eps2": mk(B, S, A), "logpi2": mk(B, S, 1),
+            "actb": mk(B, S, A),
+            "dact": mk(B, S, A),           # accumulated dminQ/da
+            "gq1p": mk(B, S, 1), "gq2p": mk(B, S, 1),
+        }
+        # per-network activation slots (critics D=1)
+        for name in ("q1s", "q2s", "t1", "t2", "q1b", "q2b"):
+            buf[name] = {"outs": mk(B, S, 1), "hS": mk(B, H), "cS": mk(B, H),
+                         "stash": mk(B, S, 7 * H)}
+        for name in ("q1s", "q2s", "q1b", "q2b"):
+            buf[name]["dgates"] = mk(B, S, 4 * H)
+            buf[name]["dxb"] = mk(B, S, H)
+        for name in ("q1b", "q2b"):
+            buf[name]["gq"] = mk(B, S, 1)
+        self.buf = buf
+
+        def t64(rows):
+            return torch.tensor(rows, dtype=torch.int64).to(dev)
+
+        def wrow(core, x2_buf):
+            return [core.body_w.data_ptr(), core.body_b.data_ptr(),
+                    core.w_ih.data_ptr(), core.w_hh.data_ptr(),
+                    core.b_g.data_ptr(), core.heads_w.data_ptr(),
+                    core.heads_b.data_ptr(), core.body2_w.data_ptr(),
+                    core.body2_b.data_ptr(), x2_buf.data_ptr()]
+
+        def orow(b):
+            return [b["outs"].data_ptr(), b["hS"].data_ptr(),
+                    b["cS"].data_ptr(), b["stash"].data_ptr()]
+
+        q = self.q
+        # sample-pass critics on (x, a1)
+        self.fwdS_cores = t64([wrow(q["q1"], buf["a1"]),
+                               wrow(q["q2"], buf["a1"])])
+        self.fwdS_outs = t64([orow(buf["q1s"]), orow(buf["q2s"])])
+        # target critics on (x, a2) + behavior critics on (x, actb)
+        self.fwdT_cores = t64([
+            wrow(q["t1"], buf["a2"]), wrow(q["t2"], buf["a2"]),
+            wrow(q["q1"], buf["actb"]), wrow(q["q2"], buf["actb"]),
+        ])
+        self.fwdT_outs = t64([orow(buf["t1"]), orow(buf["t2"]),
+                              orow(buf["q1b"]), orow(buf["q2b"])])
+
+        def birow(core, gq, b, want_dgates=True):
+            return [gq.data_ptr(), b["stash"].data_ptr(),
+                    core.w_ih.data_ptr(), core.w_hh.data_ptr(),
+                    core.heads_w.data_ptr(), core.body_w.data_ptr(),
+                    core.body2_w.data_ptr()]
+
+        # input-grad pass: dminQ/da accumulated into dact (dgates skipped)
+        self.bwdP_in = t64([birow(q["q1"], buf["gq1p"], buf["q1s"]),
+                            birow(q["q2"], buf["gq2p"], buf["q2s"])])
+        self.bwdP_out = t64([
+            [buf["q1s"]["dgates"].data_ptr(), buf["q1s"]["dxb"].data_ptr(),
+             buf["dact"].data_ptr()],
+            [buf["q2s"]["dgates"].data_ptr(), buf["q2s"]["dxb"].data_ptr(),
+             buf["dact"].data_ptr()],
+        ])
+        # loss pass: dgates/dxb for the weight grads, no input grads
+        self.bwdB_in = t64([birow(q["q1"], buf["q1b"]["gq"], buf["q1b"]),
+                            birow(q["q2"], buf["q2b"]["gq"], buf["q2b"])])
+        self.bwdB_out = t64([
+            [buf["q1b"]["dgates"].data_ptr(), buf["q1b"]["dxb"].data_ptr(), 0],
+            [buf["q2b"]["dgates"].data_ptr(), buf["q2b"]["dxb"].data_ptr(), 0],
+        ])
+
+        nrm = (u.critic_optimizer.norm_sq.data_ptr()
+               if self.grad_reducer is None else 0)
+
+        def grow(core, b):
+            g = [core.body_w.grad, core.body_b.grad, core.w_ih.grad,
+                 core.w_hh.grad, core.b_g.grad, core.heads_w.grad,
+                 core.heads_b.grad, core.body2_w.grad, core.body2_b.grad]
+            assert all(t is not None for t in g)
+            return [b["stash"].data_ptr(), b["dgates"].data_ptr(),
+                    b["dxb"].data_ptr(), b["gq"].data_ptr(),
+                    g[2].data_ptr(), g[3].data_ptr(), g[0].data_ptr(),
+                    g[1].data_ptr(), g[4].data_ptr(), g[5].data_ptr(),
+                    g[6].data_ptr(), nrm, g[7].data_ptr(), g[8].data_ptr()]
+
+        self.wgB_tab = t64([grow(q["q1"], buf["q1b"]),
+                            grow(q["q2"], buf["q2b"])])
+        self._mshape = (B, S)
+
     def _actor_fwd(self, x, hx0, cx0):
         e = ext()
+        c = self.actor_core
         mo, _, _, stash = e.seq_lstm_forward(
-            x, hx0, cx0, self.actor_core.body_w, self.actor_core.body_b,
-            self.actor_core.w_ih, self.actor_core.w_hh, self.actor_core.b_g,
-            self.actor_core.heads_w, self.actor_core.heads_b,
+            x, hx0, cx0, c.body_w, c.body_b, c.w_ih, c.w_hh, c.b_g,
+            c.heads_w, c.heads_b,
         )
         return mo, stash
-
-    def _critic_fwd(self, qmod, obs, act, hx0, cx0):
-        """Critic-continuous forward: torch-GEMM encoders + fused core."""
-        B, S, _ = obs.shape
-        e = ext()
-        o = torch.relu(
-            obs.reshape(B * S, -1) @ qmod.obs_enc_w + qmod.obs_enc_b)
-        a = torch.relu(
-            act.reshape(B * S, -1) @ qmod.act_enc_w + qmod.act_enc_b)
-        enc = torch.cat([o, a], dim=-1).view(B, S, qmod.hidden_size)
-        mo, _, _, stash = e.seq_lstm_forward(
-            enc, hx0, cx0, qmod.core.body_w, qmod.core.body_b,
-            qmod.core.w_ih, qmod.core.w_hh, qmod.core.b_g,
-            qmod.core.heads_w, qmod.core.heads_b,
-        )
-        return mo, stash, enc, o, a
-
-    def _critic_bwd_dact(self, qmod, gq, stash, enc, cx0):
-        """Input-gradient-only critic backward: dminQ/da for the actor."""
-        e = ext()
-        dx, _, _, _, _ = e.seq_lstm_backward_core(
-            gq, None, None, stash, enc, cx0, qmod.core.body_w,
-            qmod.core.w_ih, qmod.core.w_hh, qmod.core.heads_w,
-        )
-        half = qmod.hidden_size // 2
-        B, S, _ = enc.shape
-        aenc = enc.reshape(B * S, -1)[:, half:]
-        dpre_a = dx.reshape(B * S, -1)[:, half:] * (aenc > 0).float()
-        return (dpre_a @ qmod.act_enc_w.t()).view(B, S, -1)
-
-    def _critic_bwd_wgrad(self, qmod, gq, stash, enc, obs, act, hx0, cx0,
-                          norm):
-        """Full critic backward: core wgrads via the MFMA kernels, encoder
-        wgrads via library GEMMs written into the flat-grad views."""
-        e = ext()
-        core = qmod.core
-        dx, _, _, dgates, dxb = e.seq_lstm_backward_core(
-            gq, None, None, stash, enc, cx0, core.body_w, core.w_ih,
-            core.w_hh, core.heads_w,
-        )
-        gs = [core.body_w.grad, core.body_b.grad, core.w_ih.grad,
-              core.w_hh.grad, core.b_g.grad, core.heads_w.grad,
-              core.heads_b.grad]
-        assert all(g is not None for g in gs)
-        e.seq_lstm_wgrad_out(enc, hx0, stash, dgates, dxb, gq,
-                             gs[2], gs[3], gs[0], gs[1], gs[4], gs[5], gs[6],
-                             norm)
-        B, S, _ = enc.shape
-        half = qmod.hidden_size // 2
-        encf = enc.reshape(B * S, -1)
-        dxf = dx.reshape(B * S, -1)
-        dpre_o = dxf[:, :half] * (encf[:, :half] > 0).float()
-        dpre_a = dxf[:, half:] * (encf[:, half:] > 0).float()
-        of = obs.reshape(B * S, -1)
-        af = act.reshape(B * S, -1)
-        qmod.obs_enc_w.grad.copy_(of.t() @ dpre_o)
-        qmod.obs_enc_b.grad.copy_(dpre_o.sum(0))
-        qmod.act_enc_w.grad.copy_(af.t() @ dpre_a)
-        qmod.act_enc_b.grad.copy_(dpre_a.sum(0))
-        if norm is not None:
-            norm.add_(qmod.obs_enc_w.grad.pow(2).sum()
-                      + qmod.obs_enc_b.grad.pow(2).sum()
-                      + qmod.act_enc_w.grad.pow(2).sum()
-                      + qmod.act_enc_b.grad.pow(2).sum())
 
     def _actor_bwd_wgrad(self, gouts, stash, x, hx0, cx0, norm):
         e = ext()
@@ -168,46 +212,49 @@ class FusedSacContinuousStep(GraphableStep):
         e = ext()
         x = batch["obs"]
         B, S, _ = x.shape
-        N = B * S
-        A = u.actor.n_outputs
+        if self._mshape != (B, S):
+            self._multi_setup(batch)
+        buf = self.buf
+        A = self._A
+        half = self._half
         hx0 = batch["hx"][:, 0]
         cx0 = batch["cx"][:, 0]
-        act_b = batch["act"].reshape(B, S, A)
         rew = batch["rew"].reshape(B, S)
         fir = batch["is_fir"].reshape(B, S)
         log_alpha = u.log_alpha.data.view(1)
         single = self.grad_reducer is None
-        dev = x.device
 
         # 1. actor fwd + reparameterized sample
         moA, stA = self._actor_fwd(x, hx0, cx0)
-        eps1 = torch.empty(B, S, A, device=dev)
-        a1 = torch.empty(B, S, A, device=dev)
-        logpi1 = torch.empty(B, S, 1, device=dev)
-        e.sacc_sample(moA, self.rng, eps1, a1, logpi1)
+        e.sacc_sample(moA, self.rng, buf["eps1"], buf["a1"], buf["logpi1"])
 
-        # 2. critics on the fresh sample; cross-network dminQ/da
-        qp1, st1p, enc1p, _, _ = self._critic_fwd(self.q["q1"], x, a1, hx0, cx0)
-        qp2, st2p, enc2p, _, _ = self._critic_fwd(self.q["q2"], x, a1, hx0, cx0)
-        m1 = (qp1 <= qp2).float()
-        gq1p = -m1 / N
-        gq2p = -(1.0 - m1) / N
-        g = self._critic_bwd_dact(self.q["q1"], gq1p, st1p, enc1p, cx0) \
-            + self._critic_bwd_dact(self.q["q2"], gq2p, st2p, enc2p, cx0)
+        # 2. twin critics on the fresh sample — ONE dual-body launch
+        e.seq_lstm_forward_multi(x, hx0, cx0, self.fwdS_cores, self.fwdS_outs,
+                                 2, 1, F2=A, half=half)
+        qp1 = buf["q1s"]["outs"]
+        qp2 = buf["q2s"]["outs"]
 
-        # 3. analytic actor + temperature gradients (+ Adam clock prep)
+        # 3. min-critic selection grads + zero the dQ/da accumulator
+        e.sacc_min_mask(qp1, qp2, buf["gq1p"], buf["gq2p"], buf["dact"])
+
+        # 4. cross-network dminQ/da — ONE multi launch, atomic accumulate
+        e.seq_lstm_backward_multi(x, cx0, self.bwdP_in, self.bwdP_out, 2, 1,
+                                  F2=A, half=half, accum_dx2=True)
+
+        # 5. analytic actor + temperature gradients (+ Adam clock prep)
         dmoA = torch.empty_like(moA)
         clk = u.actor_optimizer if self.adam_aa is not None else None
         e.sacc_actor_grad(
-            moA, eps1, a1, g, qp1.reshape(-1), qp2.reshape(-1), log_alpha,
-            dmoA, u.log_alpha.grad.view(1), self.stats_buf[:4],
+            moA, buf["eps1"], buf["a1"], buf["dact"], qp1.reshape(-1),
+            qp2.reshape(-1), log_alpha, dmoA, u.log_alpha.grad.view(1),
+            self.stats_buf[:4],
             u.actor_optimizer.norm_sq if single else None,
             None, u.target_entropy,
             clock=clk.state3 if clk is not None else None,
             beta1=clk.beta1 if clk is not None else 0.9,
             beta2=clk.beta2 if clk is not None else 0.999,
         )
-        # 4. actor + alpha updates
+        # 6. actor + alpha updates
         self._actor_bwd_wgrad(dmoA, stA, x, hx0, cx0,
                               u.actor_optimizer.norm_sq if single else None)
         if self.adam_aa is not None:
@@ -218,38 +265,37 @@ class FusedSacContinuousStep(GraphableStep):
                 self.grad_reducer.all_reduce([u.alpha_optimizer.flat_grad])
             u.alpha_optimizer._update()  # no clip: norm unused
 
-        # 5. post-update sample
+        # 7. post-update sample
         moA2, _ = self._actor_fwd(x, hx0, cx0)
-        eps2 = torch.empty(B, S, A, device=dev)
-        a2 = torch.empty(B, S, A, device=dev)
-        logpi2 = torch.empty(B, S, 1, device=dev)
-        e.sacc_sample(moA2, self.rng, eps2, a2, logpi2)
+        e.sacc_sample(moA2, self.rng, buf["eps2"], buf["a2"], buf["logpi2"])
 
-        # 6. target critics on a'; behavior critics on the batch actions
-        tq1, _, _, _, _ = self._critic_fwd(self.q["t1"], x, a2, hx0, cx0)
-        tq2, _, _, _, _ = self._critic_fwd(self.q["t2"], x, a2, hx0, cx0)
-        qb1, st1b, enc1b, _, _ = self._critic_fwd(self.q["q1"], x, act_b, hx0, cx0)
-        qb2, st2b, enc2b, _, _ = self._critic_fwd(self.q["q2"], x, act_b, hx0, cx0)
+        # 8. target critics on a2 + behavior critics on the batch actions —
+        #    ONE 4-network launch (per-core x2 pointers in the table; the
+        #    batch actions are copied into the persistent actb buffer the
+        #    table references, so non-graph callers stay correct)
+        buf["actb"].copy_(batch["act"].reshape(B, S, A), non_blocking=True)
+        e.seq_lstm_forward_multi(x, hx0, cx0, self.fwdT_cores, self.fwdT_outs,
+                                 4, 1, F2=A, half=half)
 
-        # 7. critic losses
-        gq1 = torch.empty_like(qb1)
-        gq2 = torch.empty_like(qb2)
+        # 9. critic losses (head grads land in the persistent gq buffers)
         e.sacc_critic_loss(
-            qb1.reshape(-1), qb2.reshape(-1), tq1.reshape(-1),
-            tq2.reshape(-1), logpi2.reshape(-1), rew, fir, log_alpha,
-            gq1.reshape(-1), gq2.reshape(-1), self.stats_buf[4:5],
+            buf["q1b"]["outs"].reshape(-1), buf["q2b"]["outs"].reshape(-1),
+            buf["t1"]["outs"].reshape(-1), buf["t2"]["outs"].reshape(-1),
+            buf["logpi2"].reshape(-1), rew, fir, log_alpha,
+            buf["q1b"]["gq"].reshape(-1), buf["q2b"]["gq"].reshape(-1),
+            self.stats_buf[4:5],
             u.critic_optimizer.norm_sq if single else None,
             p.gamma, p.reward_scale,
         )
-        # 8. critic updates (both stacks accumulate into one flat space)
-        cn = u.critic_optimizer.norm_sq if single else None
-        self._critic_bwd_wgrad(self.q["q1"], gq1, st1b, enc1b, x, act_b,
-                               hx0, cx0, cn)
-        self._critic_bwd_wgrad(self.q["q2"], gq2, st2b, enc2b, x, act_b,
-                               hx0, cx0, cn)
+        # 10. twin-critic backward + MFMA wgrads (encoder grads ride the
+        #     same launch as dual-body segments)
+        e.seq_lstm_backward_multi(x, cx0, self.bwdB_in, self.bwdB_out, 2, 1,
+                                  F2=A, half=half)
+        e.seq_lstm_wgrad_multi(x, hx0, self.wgB_tab, 2, 1, x2=buf["actb"],
+                               F2=A, half=half)
         self._opt(u.critic_optimizer)
 
-        # 9. Polyak target update
+        # 11. Polyak target update
         soft_update(u.critic, u.target_critic, u.TAU)
 
     def _full(self, batch):

@@ -45,6 +45,45 @@ def test_seq_lstm_forward_parity():
         torch.testing.assert_close(outs_f[k], outs_e[k], rtol=1e-5, atol=1e-5)
 
 
+def test_dual_body_core_parity():
+    """Dual-body SeqLSTMCore (continuous-critic topology: obs/action
+    encoders straight into the LSTM): fused forward AND full autograd
+    backward — dx, dx2, both encoder weight grads — vs the eager oracle."""
+    _ops()
+    from pdrl_amd.networks import SeqLSTMCore
+
+    torch.manual_seed(5)
+    core_f = SeqLSTMCore(3, H, {"q": 1}, input2_dim=2).to(DEV)
+    torch.manual_seed(5)
+    core_e = SeqLSTMCore(3, H, {"q": 1}, input2_dim=2).to(DEV)
+
+    torch.manual_seed(9)
+    x = torch.randn(B, S, 3, device=DEV)
+    x2 = torch.randn(B, S, 2, device=DEV)
+    hx = torch.randn(B, H, device=DEV) * 0.2
+    cx = torch.randn(B, H, device=DEV) * 0.2
+    w = torch.randn(B, S, 1, device=DEV)
+
+    xf = x.clone().requires_grad_()
+    x2f = x2.clone().requires_grad_()
+    xe = x.clone().requires_grad_()
+    x2e = x2.clone().requires_grad_()
+
+    outs_f, hf, cf = core_f._forward_fused(xf, hx, cx, x2f)
+    outs_e, he, ce = core_e._forward_eager(xe, hx, cx, x2e)
+    torch.testing.assert_close(outs_f["q"], outs_e["q"], rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(hf, he, rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(cf, ce, rtol=1e-5, atol=1e-5)
+
+    ((outs_f["q"] * w).sum() + 0.1 * hf.sum()).backward()
+    ((outs_e["q"] * w).sum() + 0.1 * he.sum()).backward()
+    torch.testing.assert_close(xf.grad, xe.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(x2f.grad, x2e.grad, rtol=1e-4, atol=1e-5)
+    for (n, pf), pe in zip(core_f.named_parameters(), core_e.parameters()):
+        torch.testing.assert_close(pf.grad, pe.grad, rtol=2e-4, atol=1e-5,
+                                   msg=lambda m: f"{n}: {m}")
+
+
 def test_seq_lstm_backward_parity():
     _ops()
     for seed in (0, 7):
