@@ -34,6 +34,18 @@ void seq_lstm_backward_multi_hip(const at::Tensor&, const at::Tensor&,
 void seq_lstm_wgrad_multi_hip(const at::Tensor&, const at::Tensor&,
                               const at::Tensor&, long, long,
                               const c10::optional<at::Tensor>&, long, long);
+bool megastep_onpolicy_hip(
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, at::Tensor&, at::Tensor&,
+    at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
+    at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
+    at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
+    at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, long,
+    double, double, double, double, double, double, double, double, double,
+    double, double, double, double, double, double, bool);
 at::Tensor gae_hip(const at::Tensor&, double, double, const at::Tensor&);
 std::vector<at::Tensor> vtrace_hip(const at::Tensor&, const at::Tensor&,
                                    const at::Tensor&, const at::Tensor&,
@@ -172,6 +184,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("h0"), py::arg("tab"), py::arg("C"),
         py::arg("D"), py::arg("x2") = c10::nullopt, py::arg("F2") = 0,
         py::arg("half") = 0);
+  m.def("megastep_onpolicy", &megastep_onpolicy_hip,
+        "ENTIRE IMPALA/PPO training step in one launch (fwd+loss+bwd+wgrad"
+        "+RMSprop between grid barriers); false => shape not co-resident");
   m.def("gae", &gae_hip, "GAE reverse scan");
   m.def("vtrace", &vtrace_hip, "fused V-trace scan",
         pybind11::arg("behav_lp"), pybind11::arg("target_lp"),

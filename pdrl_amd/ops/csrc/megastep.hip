@@ -1,0 +1,431 @@
+// Whole-training-step mega-kernel for IMPALA / PPO on CDNA4 (gfx950).
+//
+// ONE launch runs the ENTIRE iteration:
+//   phase 1  per-row blocks: fused fwd (body+LSTM+heads) — row-local
+//   phase 2  per-row blocks: loss (cat stats, V-trace/GAE scan, analytic
+//            head grads, atomic loss-stat partials) + BPTT backward —
+//            ALL of it is row-local for these losses (the only cross-row
+//            coupling in IMPALA/PPO is the monitoring means)
+//   phase 3  re-mapped blocks: MFMA gate-GEMM weight grads + small grads
+//            (+ stats finalization on the last block)
+//   phase 4  all blocks: fused clip+RMSprop over the flat parameter buffer
+// with device-scope generation barriers between phases. All blocks are
+// co-resident by construction (grid ≤ maxActiveBlocks, checked on the
+// host), so the spin barrier cannot deadlock.
+//
+// Why: at B=128/S=5 the stream-ordered fused DAG is 5 dependent launches
+// of 4-14 µs each — the step time IS launch+ramp latency
+// (profiles/impala_step_breakdown_r01.md). Collapsing the step into one
+// kernel removes four kernel boundaries; the optimizer update costs one
+// grid barrier instead of a full launch.
+//
+// Reference math: agents/learner_module/impala/learning.py:48-94 and
+// ppo/learning.py:59-106 (identical numerics to losses.hip's mega kernels,
+// which remain the multi-rank / large-shape path).
+#include "common.h"
+#include "core_rows.h"
+#include "wgrad_body.h"
+
+#include <vector>
+
+namespace {
+
+constexpr int kAlgoImpala = 0;
+constexpr int kAlgoPpo = 1;
+constexpr int kMsThreads = 256;
+
+__device__ __forceinline__ float ms_huber(float d) {
+  const float a = fabsf(d);
+  return (a < 1.0f) ? 0.5f * d * d : a - 0.5f;
+}
+__device__ __forceinline__ float ms_huber_grad(float d) {
+  return fminf(fmaxf(d, -1.0f), 1.0f);
+}
+
+// Generation-count grid barrier; safe because every block of the grid is
+// resident (host checks occupancy before choosing this kernel).
+__device__ __forceinline__ void grid_barrier(unsigned* bar, int nblocks) {
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    __threadfence();
+    volatile unsigned* gen = bar + 1;
+    const unsigned g = *gen;
+    if (atomicAdd(bar, 1u) == (unsigned)nblocks - 1) {
+      bar[0] = 0u;
+      __threadfence();
+      atomicAdd((unsigned*)(bar + 1), 1u);
+    } else {
+      while (*gen == g) {
+        __builtin_amdgcn_s_sleep(8);
+      }
+    }
+    __threadfence();
+  }
+  __syncthreads();
+}
+
+// Row-local IMPALA/PPO loss for batch row b: categorical stats, V-trace or
+// TD+GAE scan, analytic packed head grads, atomic loss-stat partials.
+// stats_acc: [pl, vl, es, ravg/rsum, rg]; mm: ordered-int {min,max} of the
+// PPO ratio (ratios are positive, so int-bit order == float order).
+__device__ void onpolicy_loss_row(
+    int algo, const float* __restrict__ outs, const float* __restrict__ act,
+    const float* __restrict__ behav, const float* __restrict__ rew,
+    const float* __restrict__ fir, float* __restrict__ gouts,
+    float* __restrict__ stats_acc, int* __restrict__ mm, int b, int B, int S,
+    int A, float gamma, float lmbda, float rho_bar, float rho_min,
+    float c_bar, float rew_scale, float cp, float cv, float ce,
+    float eps_clip, float creg, char* smem) {
+  const int D = A + 1;
+  const int T = S - 1;
+  const int BT = B * T;
+  const int tid = threadIdx.x;
+  float* s_logp = reinterpret_cast<float*>(smem);  // (S)
+  float* s_lse = s_logp + S;                       // (S)
+  float* s_ent = s_lse + S;                        // (S)
+  float* s_w = s_ent + S;                          // (T) rho (IMPALA)
+  float* s_adv = s_w + S;                          // (T)
+  float* s_ret = s_adv + S;                        // (T) vs (IMPALA) | td (PPO)
+  const long sb = (long)b * S;
+
+  if (tid < S) {
+    const float* z = outs + (sb + tid) * D;
+    float m = z[0];
+    for (int j = 1; j < A; ++j) m = fmaxf(m, z[j]);
+    float s = 0.f;
+    for (int j = 0; j < A; ++j) s += __expf(z[j] - m);
+    const float l = m + __logf(s);
+    float h = 0.f;
+    for (int j = 0; j < A; ++j) {
+      const float lp = z[j] - l;
+      h -= __expf(lp) * lp;
+    }
+    s_logp[tid] = z[(int)act[sb + tid]] - l;
+    s_lse[tid] = l;
+    s_ent[tid] = h;
+  }
+  __syncthreads();
+
+  if (tid == 0) {
+    // scans + this row's loss partials (T is tiny: serial on one lane)
+    float pl = 0.f, vl = 0.f, es = 0.f, rs = 0.f, rg = 0.f;
+    if (algo == kAlgoImpala) {
+      float acc = 0.f;
+      for (int t = T - 1; t >= 0; --t) {
+        const float ratio = __expf(s_logp[t] - behav[sb + t]);
+        const float rho = fminf(fmaxf(ratio, rho_min), rho_bar);
+        const float c = fminf(ratio, c_bar);
+        const float mask = 1.f - fir[sb + t + 1];
+        const float vt = outs[(sb + t) * D + A];
+        const float vn = outs[(sb + t + 1) * D + A];
+        const float delta =
+            rho * (rew[sb + t] * rew_scale + gamma * mask * vn - vt);
+        acc = fmaf(gamma * mask * c, acc, delta);
+        s_w[t] = rho;
+        s_ret[t] = vt + acc;
+      }
+      for (int t = 0; t < T; ++t) {
+        const float mask = 1.f - fir[sb + t + 1];
+        const float vnext = (t + 1 < T) ? s_ret[t + 1] : outs[(sb + T) * D + A];
+        s_adv[t] = s_w[t] * (rew[sb + t] * rew_scale + gamma * mask * vnext -
+                             outs[(sb + t) * D + A]);
+      }
+      for (int t = 0; t < T; ++t) {
+        pl -= s_logp[t] * s_adv[t];
+        vl += ms_huber(outs[(sb + t) * D + A] - s_ret[t]);
+        es += s_ent[t];
+        rs += s_w[t];
+        for (int j = 0; j < A; ++j) {
+          const float z = outs[(sb + t) * D + j];
+          rg = fmaf(z, z, rg);
+        }
+      }
+    } else {  // PPO: TD target + GAE, clipped surrogate
+      float run = 0.f;
+      for (int t = T - 1; t >= 0; --t) {
+        const float mask = 1.f - fir[sb + t + 1];
+        const float tdv = rew[sb + t] * rew_scale +
+                          gamma * mask * outs[(sb + t + 1) * D + A];
+        const float delta = tdv - outs[(sb + t) * D + A];
+        run = fmaf(gamma * lmbda * mask, run, delta);
+        s_ret[t] = tdv;
+        s_adv[t] = run;
+      }
+      float rmin = 1e30f, rmax = -1e30f;
+      for (int t = 0; t < T; ++t) {
+        const float r = __expf(s_logp[t] - behav[sb + t]);
+        const float a = s_adv[t];
+        const float s1 = r * a;
+        const float s2 =
+            fminf(fmaxf(r, 1.f - eps_clip), 1.f + eps_clip) * a;
+        pl -= fminf(s1, s2);
+        vl += ms_huber(outs[(sb + t) * D + A] - s_ret[t]);
+        es += s_ent[t];
+        rs += r;
+        rmin = fminf(rmin, r);
+        rmax = fmaxf(rmax, r);
+        for (int j = 0; j < A; ++j) {
+          const float z = outs[(sb + t) * D + j];
+          rg = fmaf(z, z, rg);
+        }
+      }
+      atomicMin(mm, __float_as_int(rmin));
+      atomicMax(mm + 1, __float_as_int(rmax));
+    }
+    atomicAdd(stats_acc + 0, pl);
+    atomicAdd(stats_acc + 1, vl);
+    atomicAdd(stats_acc + 2, es);
+    atomicAdd(stats_acc + 3, rs);
+    atomicAdd(stats_acc + 4, rg);
+  }
+  __syncthreads();
+
+  // analytic packed head grads, elementwise over this row
+  const float invN = 1.0f / BT;
+  const float dreg = 2.0f * creg * invN / A;
+  for (int idx = tid; idx < S * D; idx += kMsThreads) {
+    const int t = idx / D, j = idx % D;
+    float* g = gouts + (sb + t) * D;
+    if (t >= T) {
+      g[j] = 0.f;
+      continue;
+    }
+    const float* z = outs + (sb + t) * D;
+    if (j == A) {
+      g[A] = cv * ms_huber_grad(z[A] - s_ret[t]) * invN;
+      continue;
+    }
+    float dlogp;
+    if (algo == kAlgoImpala) {
+      dlogp = -cp * s_adv[t] * invN;
+    } else {
+      const float r = __expf(s_logp[t] - behav[sb + t]);
+      const float a_v = s_adv[t];
+      const bool inside = (r > 1.f - eps_clip) && (r < 1.f + eps_clip);
+      const float s1 = r * a_v;
+      const float s2 = fminf(fmaxf(r, 1.f - eps_clip), 1.f + eps_clip) * a_v;
+      const float gr = (inside || s1 < s2) ? a_v * r : 0.f;
+      dlogp = -cp * gr * invN;
+    }
+    const float dH = -ce * invN;
+    const float H = s_ent[t];
+    const float lp = z[j] - s_lse[t];
+    const float pj = __expf(lp);
+    const int a = (int)act[sb + t];
+    g[j] = dlogp * ((j == a ? 1.f : 0.f) - pj) + dH * (-pj * (lp + H)) +
+           dreg * z[j];
+  }
+}
+
+template <int H>
+__global__ __launch_bounds__(kMsThreads) void megastep_kernel(
+    // batch
+    const float* __restrict__ x,      // (B,S,F)
+    const float* __restrict__ h0,     // (B,H) strided
+    const float* __restrict__ c0,     // (B,H) strided
+    const float* __restrict__ act,    // (N)
+    const float* __restrict__ behav,  // (B,S)
+    const float* __restrict__ rew,    // (B,S)
+    const float* __restrict__ fir,    // (B,S)
+    // weights (views into flat_param)
+    const float* __restrict__ body_w, const float* __restrict__ body_b,
+    const float* __restrict__ w_ih, const float* __restrict__ w_hh,
+    const float* __restrict__ b_g, const float* __restrict__ heads_w,
+    const float* __restrict__ heads_b,
+    // workspaces
+    float* __restrict__ outs, float* __restrict__ hS, float* __restrict__ cS,
+    float* __restrict__ stash, float* __restrict__ gouts,
+    float* __restrict__ dgates, float* __restrict__ dxb,
+    float* __restrict__ stats, float* __restrict__ stats_acc,
+    int* __restrict__ mm, unsigned* __restrict__ bar,
+    // grad views (into flat_grad) + optimizer state
+    float* __restrict__ dw_ih, float* __restrict__ dw_hh,
+    float* __restrict__ dbody_w, float* __restrict__ dbody_b,
+    float* __restrict__ db_g, float* __restrict__ dheads_w,
+    float* __restrict__ dheads_b, float* __restrict__ norm_sq,
+    float* __restrict__ flat_param, float* __restrict__ flat_grad,
+    float* __restrict__ sq_avg, long numel,
+    // config
+    int algo, int B, int S, int F, int D, long h0s, float gamma, float lmbda,
+    float rho_bar, float rho_min, float c_bar, float rew_scale, float cp,
+    float cv, float ce, float eps_clip, float creg, float lr, float alpha,
+    float eps, float max_norm, int include_opt) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  const int b = blockIdx.x;
+  const int nblocks = gridDim.x;
+  const int A = D - 1;
+  const int N = B * S;
+
+  if (b == 0 && threadIdx.x == 0) {
+    *norm_sq = 0.f;
+    for (int i = 0; i < 5; ++i) stats_acc[i] = 0.f;
+    mm[0] = __float_as_int(1e30f);
+    mm[1] = __float_as_int(-1e30f);
+  }
+
+  // phase 1: forward, one block per batch row
+  if (b < B) {
+    seq_lstm_fwd_row<H>(x, h0, c0, body_w, body_b, w_ih, w_hh, b_g, heads_w,
+                        heads_b, outs, hS, cS, stash, b, S, F, D, h0s,
+                        smem_raw);
+  }
+  grid_barrier(bar, nblocks);
+
+  // phase 2: row-local loss + BPTT backward
+  if (b < B) {
+    onpolicy_loss_row(algo, outs, act, behav, rew, fir, gouts, stats_acc, mm,
+                      b, B, S, A, gamma, lmbda, rho_bar, rho_min, c_bar,
+                      rew_scale, cp, cv, ce, eps_clip, creg, smem_raw);
+    __syncthreads();
+    seq_lstm_bwd_row<H>(gouts, nullptr, nullptr, stash, x, c0, body_w, w_ih,
+                        w_hh, heads_w, nullptr, nullptr, nullptr, dgates, dxb,
+                        b, S, F, D, h0s, smem_raw);
+  }
+  grid_barrier(bar, nblocks);
+
+  // phase 3: weight grads (blocks re-mapped onto GEMM tiles + small grads);
+  // the last block's lane 0 finalizes the loss stats concurrently
+  if (b == nblocks - 1 && threadIdx.x == 0) {
+    const float inv = 1.0f / (B * (S - 1));
+    const float p = stats_acc[0] * inv, v = stats_acc[1] * inv,
+                e = stats_acc[2] * inv;
+    stats[0] = cp * p + cv * v - ce * e + creg * stats_acc[4] * inv / A;
+    stats[1] = p;
+    stats[2] = v;
+    stats[3] = e;
+    stats[4] = stats_acc[3] * inv;
+    if (algo == kAlgoPpo) {
+      stats[4] = stats_acc[3] * inv;
+      stats[5] = __int_as_float(mm[0]);
+      stats[6] = __int_as_float(mm[1]);
+    }
+  }
+  {
+    constexpr int G = 4 * H;
+    const int gemm_blocks = (H / 16) * (G / kWave);
+    const int total_waves = F * H + H + G + H * D + D;
+    const int small_blocks = (total_waves * kWave + kMsThreads - 1) / kMsThreads;
+    const int wg_total = 2 * gemm_blocks + small_blocks;
+    int bx = -1, by = 0;
+    if (b < 2 * gemm_blocks) {
+      bx = b % gemm_blocks;
+      by = b / gemm_blocks;
+    } else if (b < wg_total) {
+      bx = gemm_blocks + (b - 2 * gemm_blocks);
+      by = 0;
+    }
+    if (bx >= 0) {
+      wgrad_gates_body<H>(stash, h0, dgates, dw_ih, dw_hh, norm_sq, x, dxb,
+                          gouts, dbody_w, dbody_b, db_g, dheads_w, dheads_b,
+                          F, D, N, S, h0s, smem_raw, bx, by);
+    }
+  }
+  if (!include_opt) return;
+  grid_barrier(bar, nblocks);
+
+  // phase 4: fused clip + RMSprop over the flat buffers, all blocks
+  {
+    float scale = 1.0f;
+    if (max_norm > 0.0f) {
+      const float norm = sqrtf(*norm_sq) + 1e-6f;
+      scale = (norm > max_norm) ? (max_norm / norm) : 1.0f;
+    }
+    for (long i = (long)b * kMsThreads + threadIdx.x; i < numel;
+         i += (long)nblocks * kMsThreads) {
+      const float gi = flat_grad[i] * scale;
+      const float sa = alpha * sq_avg[i] + (1.0f - alpha) * gi * gi;
+      sq_avg[i] = sa;
+      flat_param[i] -= lr * gi / (sqrtf(sa) + eps);
+    }
+  }
+}
+
+int max_resident_blocks(const void* kernel, int lds_bytes) {
+  int per_cu = 0;
+  hipError_t e = hipOccupancyMaxActiveBlocksPerMultiprocessor(
+      &per_cu, kernel, kMsThreads, lds_bytes);
+  if (e != hipSuccess) return 0;
+  hipDeviceProp_t prop;
+  int dev = 0;
+  hipGetDevice(&dev);
+  hipGetDeviceProperties(&prop, dev);
+  return per_cu * prop.multiProcessorCount;
+}
+
+}  // namespace
+
+// Returns false when the shape cannot run co-resident (caller falls back to
+// the multi-launch fused path).
+bool megastep_onpolicy_hip(
+    const at::Tensor& x, const at::Tensor& h0, const at::Tensor& c0,
+    const at::Tensor& act, const at::Tensor& behav, const at::Tensor& rew,
+    const at::Tensor& fir, const at::Tensor& body_w, const at::Tensor& body_b,
+    const at::Tensor& w_ih, const at::Tensor& w_hh, const at::Tensor& b_g,
+    const at::Tensor& heads_w, const at::Tensor& heads_b, at::Tensor& outs,
+    at::Tensor& hS, at::Tensor& cS, at::Tensor& stash, at::Tensor& gouts,
+    at::Tensor& dgates, at::Tensor& dxb, at::Tensor& stats,
+    at::Tensor& stats_acc, at::Tensor& mm, at::Tensor& bar,
+    at::Tensor& dw_ih, at::Tensor& dw_hh, at::Tensor& dbody_w,
+    at::Tensor& dbody_b, at::Tensor& db_g, at::Tensor& dheads_w,
+    at::Tensor& dheads_b, at::Tensor& norm_sq, at::Tensor& flat_param,
+    at::Tensor& flat_grad, at::Tensor& sq_avg, long algo, double gamma,
+    double lmbda, double rho_bar, double rho_min, double c_bar,
+    double rew_scale, double cp, double cv, double ce, double eps_clip,
+    double creg, double lr, double alpha, double eps, double max_norm,
+    bool include_opt) {
+  CHECK_IN(x);
+  const int B = x.size(0), S = x.size(1), F = x.size(2);
+  const int H = h0.size(1), D = heads_w.size(1);
+  const int N = B * S;
+  constexpr int kH = 64;
+  if (H != kH) return false;  // specialized for the framework's H=64
+  const int G = 4 * kH;
+  const int gemm_blocks = (kH / 16) * (G / kWave);
+  const int total_waves = F * kH + kH + G + kH * D + D;
+  const int small_blocks = (total_waves * kWave + kMsThreads - 1) / kMsThreads;
+  const int wg_total = 2 * gemm_blocks + small_blocks;
+  const int nblocks = std::max(B, wg_total);
+
+  // dynamic LDS: max over phases (fwd | loss-row | bwd | wgrad row table)
+  const int lds_fwd = (2 * S * kH + 4 * kH + 2 * kH) * (int)sizeof(float);
+  const int lds_bwd = (S * kH + G + S * kH + 2 * G) * (int)sizeof(float);
+  const int lds_tab = N * (int)sizeof(const float*);
+  const int lds = std::max(std::max(lds_fwd, lds_bwd),
+                           std::max(lds_tab, 6 * S * (int)sizeof(float)));
+  if (lds > 64 * 1024) return false;
+
+  static int cached_max_blocks = -1;
+  static int cached_lds = -1;
+  if (cached_max_blocks < 0 || cached_lds != lds) {
+    cached_max_blocks =
+        max_resident_blocks((const void*)megastep_kernel<kH>, lds);
+    cached_lds = lds;
+  }
+  if (cached_max_blocks < nblocks) return false;  // spin barrier would hang
+
+  hipLaunchKernelGGL(
+      (megastep_kernel<kH>), dim3(nblocks), dim3(kMsThreads), lds,
+      current_stream(), x.data_ptr<float>(), h0.data_ptr<float>(),
+      c0.data_ptr<float>(), act.data_ptr<float>(), behav.data_ptr<float>(),
+      rew.data_ptr<float>(), fir.data_ptr<float>(), body_w.data_ptr<float>(),
+      body_b.data_ptr<float>(), w_ih.data_ptr<float>(),
+      w_hh.data_ptr<float>(), b_g.data_ptr<float>(),
+      heads_w.data_ptr<float>(), heads_b.data_ptr<float>(),
+      outs.data_ptr<float>(), hS.data_ptr<float>(), cS.data_ptr<float>(),
+      stash.data_ptr<float>(), gouts.data_ptr<float>(),
+      dgates.data_ptr<float>(), dxb.data_ptr<float>(),
+      stats.data_ptr<float>(), stats_acc.data_ptr<float>(),
+      mm.data_ptr<int>(), (unsigned*)bar.data_ptr<int>(),
+      dw_ih.data_ptr<float>(), dw_hh.data_ptr<float>(),
+      dbody_w.data_ptr<float>(), dbody_b.data_ptr<float>(),
+      db_g.data_ptr<float>(), dheads_w.data_ptr<float>(),
+      dheads_b.data_ptr<float>(), norm_sq.data_ptr<float>(),
+      flat_param.data_ptr<float>(), flat_grad.data_ptr<float>(),
+      sq_avg.data_ptr<float>(), (long)flat_param.numel(), (int)algo, B, S, F,
+      D, (long)h0.stride(0), (float)gamma, (float)lmbda, (float)rho_bar,
+      (float)rho_min, (float)c_bar, (float)rew_scale, (float)cp, (float)cv,
+      (float)ce, (float)eps_clip, (float)creg, (float)lr, (float)alpha,
+      (float)eps, (float)max_norm, include_opt ? 1 : 0);
+  HIP_CHECK_LAST();
+  return true;
+}

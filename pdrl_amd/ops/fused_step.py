@@ -226,6 +226,77 @@ class FusedOnPolicyStep(GraphableStep):
         buffer without touching the optimizer."""
         self._body(batch, update=False)
 
+    # ------------------------------------------------------------------ #
+    def _mega_setup(self, batch):
+        """Workspaces for the whole-step mega-kernel (megastep.hip): all
+        intermediates + the grid-barrier state live in persistent device
+        buffers so ONE launch replays per step."""
+        c = self.core
+        x = batch["obs"]
+        B, S, _ = x.shape
+        dev = x.device
+        H = c.w_ih.size(0)
+        D = c.heads_w.size(1)
+
+        def mk(*shape):
+            return torch.empty(*shape, device=dev)
+
+        self._ws = {
+            "outs": mk(B, S, D), "hS": mk(B, H), "cS": mk(B, H),
+            "stash": mk(B, S, 7 * H), "gouts": mk(B, S, D),
+            "dgates": mk(B, S, 4 * H), "dxb": mk(B, S, H),
+            "stats_acc": torch.zeros(8, dtype=torch.float32, device=dev),
+            "mm": torch.zeros(2, dtype=torch.int32, device=dev),
+            "bar": torch.zeros(2, dtype=torch.int32, device=dev),
+        }
+        self._mega_shape = (B, S)
+
+    def _try_megastep(self, e, batch, x, act, behav, rew, fir, hx0, cx0,
+                      B, S, A, p) -> bool:
+        """ONE kernel for the entire step (fwd+loss+bwd+wgrad+RMSprop with
+        in-kernel grid barriers). Single K_epoch, RMSprop-flat optimizers,
+        H=64 shapes that fit co-resident. Multi-rank runs the same kernel
+        WITHOUT the optimizer phase, then all-reduces + steps."""
+        import os
+
+        if self.algo not in ("IMPALA", "PPO"):
+            return False
+        if p.K_epoch != 1:
+            return False
+        if not bool(int(os.environ.get("PDRL_MEGASTEP", "1"))):
+            return False
+        opt = self.optimizer
+        if not hasattr(opt, "sq_avg"):  # FusedRMSprop only
+            return False
+        if getattr(self, "_mega_shape", None) != (B, S):
+            self._mega_setup(batch)
+        ws = self._ws
+        c = self.core
+        gv = self._grad_views()  # [dw_ih, dw_hh, dbody_w, dbody_b, db_g, dheads_w, dheads_b]
+        single = self.grad_reducer is None
+        ok = e.megastep_onpolicy(
+            x, hx0, cx0, act, behav, rew, fir,
+            c.body_w, c.body_b, c.w_ih, c.w_hh, c.b_g, c.heads_w, c.heads_b,
+            ws["outs"], ws["hS"], ws["cS"], ws["stash"], ws["gouts"],
+            ws["dgates"], ws["dxb"], self.stats_buf, ws["stats_acc"],
+            ws["mm"], ws["bar"],
+            gv[0], gv[1], gv[2], gv[3], gv[4], gv[5], gv[6], opt.norm_sq,
+            opt.flat_param, opt.flat_grad, opt.sq_avg,
+            0 if self.algo == "IMPALA" else 1,
+            p.gamma, p.lmbda, 0.8, 0.1, 1.0, p.reward_scale,
+            p.policy_loss_coef, p.value_loss_coef, p.entropy_coef,
+            p.eps_clip, float(getattr(p, "logit_reg", 0.0)),
+            opt.lr, opt.alpha, opt.eps, opt.max_norm,
+            single,  # include_opt: single-rank updates in-kernel
+        )
+        if not ok:
+            return False
+        if not single:
+            # norm of the AVERAGED grads: all-reduce then clip+update
+            self.grad_reducer.all_reduce([opt.flat_grad])
+            opt.step()
+        return True
+
     def _body(self, batch, update: bool = True):
         c, p, A = self.core, self.params, self.A
         e = ext()
@@ -241,6 +312,10 @@ class FusedOnPolicyStep(GraphableStep):
         fir = batch["is_fir"].reshape(B, S)
         if self.algo == "V-MPO":
             self._behav_logits = batch["logits"].reshape(B * S, self.A)
+
+        if update and self._try_megastep(e, batch, x, act, behav, rew, fir,
+                                         hx0, cx0, B, S, A, p):
+            return
 
         mo, hS, cS, stash = e.seq_lstm_forward(
             x, hx0, cx0, c.body_w, c.body_b, c.w_ih, c.w_hh, c.b_g,

@@ -24,6 +24,7 @@ SRC = [
     "pdrl_amd/ops/csrc/sacc_loss.hip",
     "pdrl_amd/ops/csrc/scans.hip",
     "pdrl_amd/ops/csrc/multi_tensor.hip",
+    "pdrl_amd/ops/csrc/megastep.hip",
 ]
 
 setup(

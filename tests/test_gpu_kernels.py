@@ -45,6 +45,52 @@ def test_seq_lstm_forward_parity():
         torch.testing.assert_close(outs_f[k], outs_e[k], rtol=1e-5, atol=1e-5)
 
 
+def test_megastep_matches_multilaunch():
+    """The whole-step mega-kernel (ONE launch: fwd+loss+bwd+wgrad+RMSprop
+    between grid barriers) must produce the same parameter trajectory and
+    stats as the multi-launch fused DAG, for both IMPALA and PPO."""
+    _ops()
+    import os
+
+    from pdrl_amd.agents.learner_module import switch_module
+    from pdrl_amd.utils import load_params
+    from tests.conftest import make_batch
+
+    for algo in ("IMPALA", "PPO"):
+        p = load_params()
+        p.algo = algo
+        p.batch_size, p.seq_len, p.obs_dim, p.n_actions = 32, 5, 4, 2
+        upd_cls, model_cls = switch_module(algo)
+
+        torch.manual_seed(21)
+        model_m = model_cls(4, 2, p.seq_len, p.hidden_size)
+        torch.manual_seed(21)
+        model_s = model_cls(4, 2, p.seq_len, p.hidden_size)
+
+        os.environ["PDRL_MEGASTEP"] = "1"
+        upd_m = upd_cls(model_m, p, DEV)
+        assert upd_m.fused_step is not None
+        upd_s = upd_cls(model_s, p, DEV)
+
+        batch = make_batch(p, seed=31, device=DEV)
+        try:
+            for _ in range(3):
+                os.environ["PDRL_MEGASTEP"] = "1"
+                sm = upd_m.step(batch)
+                os.environ["PDRL_MEGASTEP"] = "0"
+                ss = upd_s.step(batch)
+        finally:
+            os.environ.pop("PDRL_MEGASTEP", None)
+
+        fm = upd_m.optimizer.flat_param
+        fs = upd_s.optimizer.flat_param
+        torch.testing.assert_close(fm, fs, rtol=1e-4, atol=1e-6,
+                                   msg=lambda m: f"{algo} params: {m}")
+        for k in sm:
+            assert abs(float(sm[k]) - float(ss[k])) < 1e-3, (
+                algo, k, float(sm[k]), float(ss[k]))
+
+
 def test_dual_body_core_parity():
     """Dual-body SeqLSTMCore (continuous-critic topology: obs/action
     encoders straight into the LSTM): fused forward AND full autograd
