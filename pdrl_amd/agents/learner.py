@@ -86,6 +86,51 @@ class BatchStager:
         return dict(self._views)
 
 
+class WeightPublisher:
+    """Fast actor-weight broadcast: one fused device gather + ONE async D2H
+    into a reusable pinned buffer → a packed-weights payload
+    (buffers/wire.py pack format). Replaces the per-tensor blocking .cpu()
+    walk of state_dict (7 syncs + slow tensor pickling — the measured
+    staged-mode bottleneck, profiles/algo_breakdown_r02a.md)."""
+
+    def __init__(self, actor: torch.nn.Module, device):
+        self.device = torch.device(device)
+        self.use_cuda = self.device.type == "cuda"
+        sd = actor.state_dict(keep_vars=True)
+        self.schema = [(k, tuple(v.shape)) for k, v in sd.items()]
+        self.params = [v.detach() for v in sd.values()]
+        self.numel = sum(p.numel() for p in self.params)
+        if not self.use_cuda:
+            return
+        self._dev = torch.empty(self.numel, dtype=torch.float32,
+                                device=self.device)
+        self._views = []
+        off = 0
+        for p in self.params:
+            n = p.numel()
+            self._views.append(self._dev[off:off + n].view_as(p))
+            off += n
+        self._pinned = torch.empty(self.numel, dtype=torch.float32,
+                                   pin_memory=True)
+        self._host = self._pinned.numpy()
+
+    def payload(self) -> dict:
+        from pdrl_amd.buffers.wire import pack_weights
+
+        if not self.use_cuda:
+            return {
+                "wschema": self.schema,
+                "wbuf": np.concatenate(
+                    [p.detach().cpu().numpy().reshape(-1) for p in self.params]
+                ).astype(np.float32, copy=False),
+            }
+        torch._foreach_copy_(self._views, self.params)
+        self._pinned.copy_(self._dev, non_blocking=True)
+        torch.cuda.current_stream(self.device).synchronize()
+        # encode() pickles (copies) the buffer before the next publish reuses it
+        return {"wschema": self.schema, "wbuf": self._host}
+
+
 class Learner:
     def __init__(
         self,
@@ -144,6 +189,11 @@ class Learner:
 
         # weight plane: PUB bound at learner_port + 1 (rank 0 only)
         self.pub = pub_bind(learner_ip, learner_port + 1) if self.is_root else None
+        self.weight_pub = None
+        if self.is_root:
+            mods = self.updater.trainable_modules()
+            m = mods.get("model") or next(iter(mods.values()))
+            self.weight_pub = WeightPublisher(getattr(m, "actor", m), self.device)
         self.writer = SummaryWriter(params.result_dir) if self.is_root else None
         self.timer = ExecutionTimer(num_transition=params.seq_len * params.batch_size * world_size)
         self._replay_ready = False  # latching (reference: learner.py:385)
@@ -198,7 +248,7 @@ class Learner:
         interval = int(getattr(self.params, "model_publish_interval", 1) or 1)
         if interval > 1 and self.updater.update_count % interval != 0:
             return
-        header, payload = encode(Protocol.Model, self.updater.actor_state_dict(),
+        header, payload = encode(Protocol.Model, self.weight_pub.payload(),
                                  compress=False)
         self.pub.send(header, payload)
 

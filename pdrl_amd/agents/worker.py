@@ -137,6 +137,10 @@ class Worker:
         if newest is not None:
             protocol, state_dict = decode(*newest)
             if protocol is Protocol.Model:
+                from pdrl_amd.buffers.wire import is_packed_weights, unpack_weights
+
+                if is_packed_weights(state_dict):
+                    state_dict = unpack_weights(state_dict)
                 actor = getattr(self.model, "actor", self.model)
                 actor.load_state_dict(state_dict)
 

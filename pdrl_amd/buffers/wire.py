@@ -60,3 +60,44 @@ def unpack_steps(obj: dict) -> list[dict]:
 
 def is_packed(payload) -> bool:
     return isinstance(payload, dict) and "pk" in payload and "ids" in payload
+
+
+# --------------------------------------------------------------------------- #
+# Packed weight broadcast (learner → workers, Protocol.Model)
+# --------------------------------------------------------------------------- #
+def pack_weights(named: "dict[str, np.ndarray]") -> dict:
+    """{name: array} → {"wschema": [(name, shape)...], "wbuf": flat fp32}.
+
+    The reference pickles a whole state_dict of tensors per broadcast
+    (reference: agents/learner.py:85-93); packing makes the payload ONE
+    numpy buffer — one memcpy to serialize, and the learner can fill it
+    with a single async D2H instead of per-tensor blocking .cpu() copies.
+    """
+    schema = [(k, tuple(v.shape)) for k, v in named.items()]
+    total = sum(int(np.prod(s)) for _, s in schema)
+    buf = np.empty(total, dtype=np.float32)
+    off = 0
+    for k, v in named.items():
+        n = int(np.asarray(v).size)
+        buf[off:off + n] = np.asarray(v, dtype=np.float32).reshape(-1)
+        off += n
+    return {"wschema": schema, "wbuf": buf}
+
+
+def unpack_weights(obj: dict):
+    """Inverse of pack_weights → {name: torch.Tensor} (tensor views of the
+    decoded buffer; load_state_dict copies them into the model)."""
+    import torch
+
+    buf = torch.from_numpy(np.asarray(obj["wbuf"], dtype=np.float32))
+    out = {}
+    off = 0
+    for name, shape in obj["wschema"]:
+        n = int(np.prod(shape)) if shape else 1
+        out[name] = buf[off:off + n].view(shape)
+        off += n
+    return out
+
+
+def is_packed_weights(payload) -> bool:
+    return isinstance(payload, dict) and "wbuf" in payload and "wschema" in payload

@@ -45,7 +45,8 @@ bool megastep_onpolicy_hip(
     at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
     at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, long,
     double, double, double, double, double, double, double, double, double,
-    double, double, double, double, double, double, bool);
+    double, double, double, double, double, double, bool, long);
+void barrier_bench_hip(at::Tensor&, long, long);
 at::Tensor gae_hip(const at::Tensor&, double, double, const at::Tensor&);
 std::vector<at::Tensor> vtrace_hip(const at::Tensor&, const at::Tensor&,
                                    const at::Tensor&, const at::Tensor&,
@@ -187,6 +188,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("megastep_onpolicy", &megastep_onpolicy_hip,
         "ENTIRE IMPALA/PPO training step in one launch (fwd+loss+bwd+wgrad"
         "+RMSprop between grid barriers); false => shape not co-resident");
+  m.def("barrier_bench", &barrier_bench_hip,
+        "grid-barrier microbenchmark (N back-to-back barriers)");
   m.def("gae", &gae_hip, "GAE reverse scan");
   m.def("vtrace", &vtrace_hip, "fused V-trace scan",
         pybind11::arg("behav_lp"), pybind11::arg("target_lp"),

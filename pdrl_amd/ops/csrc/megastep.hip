@@ -249,7 +249,7 @@ __global__ __launch_bounds__(kMsThreads) void megastep_kernel(
     int algo, int B, int S, int F, int D, long h0s, float gamma, float lmbda,
     float rho_bar, float rho_min, float c_bar, float rew_scale, float cp,
     float cv, float ce, float eps_clip, float creg, float lr, float alpha,
-    float eps, float max_norm, int include_opt) {
+    float eps, float max_norm, int include_opt, int max_phase) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   const int b = blockIdx.x;
   const int nblocks = gridDim.x;
@@ -269,6 +269,7 @@ __global__ __launch_bounds__(kMsThreads) void megastep_kernel(
                         heads_b, outs, hS, cS, stash, b, S, F, D, h0s,
                         smem_raw);
   }
+  if (max_phase <= 1) return;
   grid_barrier(bar, nblocks);
 
   // phase 2: row-local loss + BPTT backward
@@ -281,6 +282,7 @@ __global__ __launch_bounds__(kMsThreads) void megastep_kernel(
                         w_hh, heads_w, nullptr, nullptr, nullptr, dgates, dxb,
                         b, S, F, D, h0s, smem_raw);
   }
+  if (max_phase <= 2) return;
   grid_barrier(bar, nblocks);
 
   // phase 3: weight grads (blocks re-mapped onto GEMM tiles + small grads);
@@ -320,7 +322,7 @@ __global__ __launch_bounds__(kMsThreads) void megastep_kernel(
                           F, D, N, S, h0s, smem_raw, bx, by);
     }
   }
-  if (!include_opt) return;
+  if (!include_opt || max_phase <= 3) return;
   grid_barrier(bar, nblocks);
 
   // phase 4: fused clip + RMSprop over the flat buffers, all blocks
@@ -338,6 +340,11 @@ __global__ __launch_bounds__(kMsThreads) void megastep_kernel(
       flat_param[i] -= lr * gi / (sqrtf(sa) + eps);
     }
   }
+}
+
+// Microbenchmark: N grid barriers back-to-back (to price the spin barrier).
+__global__ void barrier_bench_kernel(unsigned* bar, int iters) {
+  for (int i = 0; i < iters; ++i) grid_barrier(bar, gridDim.x);
 }
 
 int max_resident_blocks(const void* kernel, int lds_bytes) {
@@ -372,7 +379,7 @@ bool megastep_onpolicy_hip(
     double lmbda, double rho_bar, double rho_min, double c_bar,
     double rew_scale, double cp, double cv, double ce, double eps_clip,
     double creg, double lr, double alpha, double eps, double max_norm,
-    bool include_opt) {
+    bool include_opt, long max_phase) {
   CHECK_IN(x);
   const int B = x.size(0), S = x.size(1), F = x.size(2);
   const int H = h0.size(1), D = heads_w.size(1);
@@ -425,7 +432,14 @@ bool megastep_onpolicy_hip(
       D, (long)h0.stride(0), (float)gamma, (float)lmbda, (float)rho_bar,
       (float)rho_min, (float)c_bar, (float)rew_scale, (float)cp, (float)cv,
       (float)ce, (float)eps_clip, (float)creg, (float)lr, (float)alpha,
-      (float)eps, (float)max_norm, include_opt ? 1 : 0);
+      (float)eps, (float)max_norm, include_opt ? 1 : 0, (int)max_phase);
   HIP_CHECK_LAST();
   return true;
+}
+
+void barrier_bench_hip(at::Tensor& bar, long nblocks, long iters) {
+  hipLaunchKernelGGL(barrier_bench_kernel, dim3((unsigned)nblocks), dim3(256),
+                     0, current_stream(), (unsigned*)bar.data_ptr<int>(),
+                     (int)iters);
+  HIP_CHECK_LAST();
 }

@@ -288,6 +288,7 @@ class FusedOnPolicyStep(GraphableStep):
             p.eps_clip, float(getattr(p, "logit_reg", 0.0)),
             opt.lr, opt.alpha, opt.eps, opt.max_norm,
             single,  # include_opt: single-rank updates in-kernel
+            int(os.environ.get("PDRL_MEGA_PHASE", "99")),  # profiling knob
         )
         if not ok:
             return False
