@@ -44,21 +44,39 @@ __device__ __forceinline__ float ms_huber_grad(float d) {
 
 // Generation-count grid barrier; safe because every block of the grid is
 // resident (host checks occupancy before choosing this kernel).
+//
+// TWO-LEVEL arrival: a flat single-counter barrier serializes one atomic
+// RMW per block on ONE cache line — measured 7.5 µs at 128 blocks
+// (gpurun_out/mega_probe.log). Groups of 16 blocks first arrive on their
+// own group line (groups proceed in parallel), then one representative
+// per group arrives at the root: ~n/16 serialized root atomics instead
+// of n. Layout: bar[0] root counter, bar[1] generation, bar[2+g] group
+// counters.
+constexpr int kBarGroup = 16;
+
 __device__ __forceinline__ void grid_barrier(unsigned* bar, int nblocks) {
   __syncthreads();
   if (threadIdx.x == 0) {
     __threadfence();
     volatile unsigned* gen = bar + 1;
     const unsigned g = *gen;
-    if (atomicAdd(bar, 1u) == (unsigned)nblocks - 1) {
-      bar[0] = 0u;
-      __threadfence();
-      atomicAdd((unsigned*)(bar + 1), 1u);
-    } else {
-      while (*gen == g) {
-        __builtin_amdgcn_s_sleep(8);
+    const int grp = (int)blockIdx.x / kBarGroup;
+    const int ngroups = (nblocks + kBarGroup - 1) / kBarGroup;
+    const int gsize =
+        min(kBarGroup, nblocks - grp * kBarGroup);  // last group is ragged
+    if ((int)atomicAdd(bar + 2 + grp, 1u) == gsize - 1) {
+      atomicExch(bar + 2 + grp, 0u);  // atomics: always at the coherence point
+      if ((int)atomicAdd(bar, 1u) == ngroups - 1) {
+        atomicExch(bar, 0u);
+        __threadfence();
+        atomicAdd((unsigned*)(bar + 1), 1u);
+        goto released;
       }
     }
+    while (*gen == g) {
+      __builtin_amdgcn_s_sleep(8);
+    }
+  released:
     __threadfence();
   }
   __syncthreads();
@@ -270,6 +288,9 @@ __global__ __launch_bounds__(kMsThreads) void megastep_kernel(
                         smem_raw);
   }
   if (max_phase <= 1) return;
+  // barrier needed only to order block 0's accumulator zeroing before the
+  // other blocks' phase-2 atomics (the loss/backward data itself is
+  // row-local to each block)
   grid_barrier(bar, nblocks);
 
   // phase 2: row-local loss + BPTT backward

@@ -25,7 +25,7 @@ def main():
 
     e = ext()
     dev = torch.device("cuda")
-    bar = torch.zeros(2, dtype=torch.int32, device=dev)
+    bar = torch.zeros(64, dtype=torch.int32, device=dev)
     for nb in (81, 128, 192, 256):
         e.barrier_bench(bar, nb, 10)
         sync()
