@@ -92,7 +92,7 @@ bool vmpo_loss_mega_hip(const at::Tensor&, const at::Tensor&,
                         at::Tensor&, at::Tensor&,
                         const c10::optional<at::Tensor>&, at::Tensor&, long,
                         double, double, double, double, double, double,
-                        double, double, double);
+                        double, double, double, long);
 bool ppoc_loss_mega_hip(const at::Tensor&, const at::Tensor&,
                         const at::Tensor&, const at::Tensor&,
                         const at::Tensor&, at::Tensor&, at::Tensor&,
@@ -209,7 +209,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ppo_loss_mega", &ppo_loss_mega_hip,
         "single-launch PPO loss: stats+gae+reduce+bwd");
   m.def("vmpo_loss_mega", &vmpo_loss_mega_hip,
-        "single-launch V-MPO loss: stats+gae+topk+duals+bwd");
+        "single-launch V-MPO loss: stats+gae+topk+duals+bwd",
+        py::arg("mo"), py::arg("act"), py::arg("behav"), py::arg("rew"),
+        py::arg("fir"), py::arg("log_eta"), py::arg("log_alpha"),
+        py::arg("gouts"), py::arg("g_eta"), py::arg("g_alpha"),
+        py::arg("stats"), py::arg("norm_sq"), py::arg("rng_state"),
+        py::arg("A"), py::arg("gamma"), py::arg("lmbda"),
+        py::arg("rew_scale"), py::arg("cp"), py::arg("cv"), py::arg("creg"),
+        py::arg("eps_eta"), py::arg("alpha_below"), py::arg("alpha_upper"),
+        py::arg("max_phase") = 99);
   m.def("ppoc_loss_mega", &ppoc_loss_mega_hip,
         "single-launch Gaussian-policy PPO loss (K5)");
   m.def("sac_actor_loss", &sac_actor_loss_hip,

@@ -88,7 +88,8 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
     float* __restrict__ norm_sq,       // optional: zeroed + eta/alpha part
     unsigned* __restrict__ rng_state,  // LCG state for eps_alpha
     int B, int S, int A, float gamma, float lmbda, float rew_scale, float cp,
-    float cv, float creg, float eps_eta, float alpha_below, float alpha_upper) {
+    float cv, float creg, float eps_eta, float alpha_below, float alpha_upper,
+    int max_phase) {
   const int D = A + 1;
   const int T = S - 1;
   const int N = B * S;
@@ -128,6 +129,7 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
     s_logp[i] = z[(int)act[i]] - l;
   }
   __syncthreads();
+  if (max_phase <= 1) return;
 
   // phase B: GAE scan per batch row (value = col A of mo)
   for (int b = tid; b < B; b += kThreads) {
@@ -144,6 +146,7 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
     }
   }
   __syncthreads();
+  if (max_phase <= 2) return;
 
   // phase C: k-th largest advantage via bit binary search (k = BT/2).
   // Per-thread advantage bits live in REGISTERS (the LDS budget caps
@@ -205,6 +208,7 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
     __syncthreads();
   }
 
+  if (max_phase <= 3) return;
   // phase C': psi softmax over selected (max = global max of selected advs)
   {
     float mx = -1e30f;
@@ -243,6 +247,7 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
   }
   __syncthreads();
 
+  if (max_phase <= 4) return;
   // phase D: reductions — policy, value, KL, logit reg
   {
     float pl = 0.f, vl = 0.f, kl = 0.f, rg = 0.f;
@@ -314,6 +319,7 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
   }
   __syncthreads();
 
+  if (max_phase <= 5) return;
   // phase E: packed head grads
   {
     const float alpha_v = s_scalars[7];
@@ -361,7 +367,7 @@ bool vmpo_loss_mega_hip(const at::Tensor& mo, const at::Tensor& act,
                         at::Tensor& rng_state, long A, double gamma,
                         double lmbda, double rew_scale, double cp, double cv,
                         double creg, double eps_eta, double alpha_below,
-                        double alpha_upper) {
+                        double alpha_upper, long max_phase) {
   const int B = mo.size(0), S = mo.size(1);
   const int N = B * S, BT = B * (S - 1);
   const long lds = (2L * N + 3L * BT) * sizeof(float);
@@ -376,7 +382,8 @@ bool vmpo_loss_mega_hip(const at::Tensor& mo, const at::Tensor& act,
       norm_sq.has_value() ? norm_sq->data_ptr<float>() : nullptr,
       (unsigned*)rng_state.data_ptr<int>(), B, S, (int)A, (float)gamma,
       (float)lmbda, (float)rew_scale, (float)cp, (float)cv, (float)creg,
-      (float)eps_eta, (float)alpha_below, (float)alpha_upper);
+      (float)eps_eta, (float)alpha_below, (float)alpha_upper,
+      (int)max_phase);
   HIP_CHECK_LAST();
   return true;
 }

@@ -152,6 +152,8 @@ class FusedOnPolicyStep(GraphableStep):
                 norm, self.rng_state, A, p.gamma, p.lmbda, p.reward_scale,
                 p.policy_loss_coef, p.value_loss_coef, creg, p.coef_eta,
                 p.coef_alpha_below, p.coef_alpha_upper,
+                max_phase=int(__import__("os").environ.get(
+                    "PDRL_VMPO_PHASE", "99")),  # profiling knob
             )
             if not ok:  # unreachable: updaters gate on fits() and fall back
                 raise RuntimeError(

@@ -75,5 +75,41 @@ def main():
     print(f"multi-launch DAG (no graph): {(time.perf_counter() - t0) / 2000 * 1e6:.2f} us/step")
 
 
+
+
+def vmpo_phase_probe():
+    """Truncate the V-MPO mega loss kernel after each phase to locate its
+    cost (A catstats, B GAE, C search, C' softmax, D reductions, E grads)."""
+    import os
+
+    import torch
+    from bench import env_shape, make_synthetic_batch
+    from pdrl_amd.agents.learner_module import switch_module
+    from pdrl_amd.utils import load_params
+
+    dev = torch.device("cuda")
+    params = load_params()
+    params.algo = "V-MPO"
+    params.obs_dim, params.n_actions, _ = env_shape("V-MPO")
+    upd_cls, model_cls = switch_module("V-MPO")
+    for phase in (1, 2, 3, 4, 5, 99):
+        os.environ["PDRL_VMPO_PHASE"] = str(phase)
+        torch.manual_seed(0)
+        model = model_cls(4, 2, params.seq_len, params.hidden_size)
+        upd = upd_cls(model, params, dev)
+        batch = make_synthetic_batch(params, dev, seed=7)
+        for _ in range(200):
+            upd.step(batch)
+        sync()
+        t0 = time.perf_counter()
+        for _ in range(2000):
+            upd.step(batch)
+        sync()
+        print(f"vmpo max_phase={phase}: "
+              f"{(time.perf_counter() - t0) / 2000 * 1e6:.2f} us/step")
+    os.environ["PDRL_VMPO_PHASE"] = "99"
+
+
 if __name__ == "__main__":
     main()
+    vmpo_phase_probe()
