@@ -50,9 +50,12 @@ __device__ __forceinline__ float ms_huber_grad(float d) {
 // (gpurun_out/mega_probe.log). Groups of 16 blocks first arrive on their
 // own group line (groups proceed in parallel), then one representative
 // per group arrives at the root: ~n/16 serialized root atomics instead
-// of n. Layout: bar[0] root counter, bar[1] generation, bar[2+g] group
-// counters.
+// of n. Group counters are strided ONE CACHE LINE apart (32 ints) — on the
+// first cut they shared a line, which serialized exactly like the flat
+// version. Layout: bar[0] root counter, bar[1] generation,
+// bar[32 + 32*g] group counters.
 constexpr int kBarGroup = 16;
+constexpr int kBarStride = 32;  // ints per 128 B cache line
 
 __device__ __forceinline__ void grid_barrier(unsigned* bar, int nblocks) {
   __syncthreads();
@@ -64,8 +67,9 @@ __device__ __forceinline__ void grid_barrier(unsigned* bar, int nblocks) {
     const int ngroups = (nblocks + kBarGroup - 1) / kBarGroup;
     const int gsize =
         min(kBarGroup, nblocks - grp * kBarGroup);  // last group is ragged
-    if ((int)atomicAdd(bar + 2 + grp, 1u) == gsize - 1) {
-      atomicExch(bar + 2 + grp, 0u);  // atomics: always at the coherence point
+    unsigned* gctr = bar + kBarStride + (long)grp * kBarStride;
+    if ((int)atomicAdd(gctr, 1u) == gsize - 1) {
+      atomicExch(gctr, 0u);  // atomics: always at the coherence point
       if ((int)atomicAdd(bar, 1u) == ngroups - 1) {
         atomicExch(bar, 0u);
         __threadfence();

@@ -249,7 +249,7 @@ class FusedOnPolicyStep(GraphableStep):
             "dgates": mk(B, S, 4 * H), "dxb": mk(B, S, H),
             "stats_acc": torch.zeros(8, dtype=torch.float32, device=dev),
             "mm": torch.zeros(2, dtype=torch.int32, device=dev),
-            "bar": torch.zeros(64, dtype=torch.int32, device=dev),
+            "bar": torch.zeros(1024, dtype=torch.int32, device=dev),
         }
         self._mega_shape = (B, S)
 

@@ -25,7 +25,7 @@ def main():
 
     e = ext()
     dev = torch.device("cuda")
-    bar = torch.zeros(64, dtype=torch.int32, device=dev)
+    bar = torch.zeros(1024, dtype=torch.int32, device=dev)
     for nb in (81, 128, 192, 256):
         e.barrier_bench(bar, nb, 10)
         sync()
@@ -45,6 +45,7 @@ def main():
     torch.manual_seed(0)
     upd_cls, model_cls = switch_module("IMPALA")
 
+    os.environ["PDRL_MEGASTEP"] = "1"
     for phase in (1, 2, 3, 99):
         os.environ["PDRL_MEGA_PHASE"] = str(phase)
         model = model_cls(4, 2, params.seq_len, params.hidden_size)
