@@ -137,7 +137,7 @@ def main():
         # staging + training step + actor weight publish (D2H + encode + TCP
         # send on a bound PUB — the reference publishes after every update,
         # ppo/learning.py:108)
-        from pdrl_amd.agents.learner import BatchStager
+        from pdrl_amd.agents.learner import BatchStager, WeightPublisher
         from pdrl_amd.transport import pub_bind
         from pdrl_amd.utils import Protocol, encode
 
@@ -145,13 +145,14 @@ def main():
                    for k, v in make_synthetic_batch(params, "cpu", 100 + rank).items()}
         stager = BatchStager(device)
         pub = pub_bind("127.0.0.1", 35000 + 37 * rank) if rank == 0 else None
+        actor = getattr(model, "actor", model)
+        wpub = WeightPublisher(actor, device) if pub is not None else None
 
         def step_fn():
             dev_batch = stager.stage(host_np)
             updater.step(dev_batch)
             if pub is not None:
-                header, payload = encode(Protocol.Model,
-                                         updater.actor_state_dict(),
+                header, payload = encode(Protocol.Model, wpub.payload(),
                                          compress=False)
                 pub.send(header, payload)
     else:

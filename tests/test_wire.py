@@ -72,3 +72,25 @@ def test_mixed_uuid_chunk_assembles(params):
     assert len(trajs) == 2
     for tr in trajs:
         assert tr["obs"].shape == (S, 4)
+
+
+def test_packed_weights_roundtrip():
+    """WeightPublisher payload → encode/decode → unpack → load_state_dict
+    reproduces the exact actor weights (the learner's fast broadcast path)."""
+    import torch
+
+    from pdrl_amd.agents.learner import WeightPublisher
+    from pdrl_amd.buffers.wire import is_packed_weights, unpack_weights
+    from pdrl_amd.networks import MlpLSTMSingle
+    from pdrl_amd.utils import Protocol, decode, encode
+
+    torch.manual_seed(4)
+    src = MlpLSTMSingle(4, 2, 5, 64)
+    dst = MlpLSTMSingle(4, 2, 5, 64)
+    pub = WeightPublisher(src.actor, "cpu")
+    header, payload = encode(Protocol.Model, pub.payload(), compress=False)
+    proto, obj = decode(header, payload)
+    assert proto is Protocol.Model and is_packed_weights(obj)
+    dst.actor.load_state_dict(unpack_weights(obj))
+    for a, b in zip(src.actor.parameters(), dst.actor.parameters()):
+        torch.testing.assert_close(a, b)
