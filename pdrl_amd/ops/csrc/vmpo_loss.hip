@@ -174,26 +174,33 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
   // of the selected bucket publishes the narrowed prefix.
   unsigned thresh_u;
   {
-    __shared__ float s_hist[256];  // dedicated: BT may be < 256
+    // WAVE-PRIVATE histograms: advantages cluster into a handful of
+    // exponent buckets, so a single shared histogram serializes thousands
+    // of same-address LDS atomics; four private copies cut the contention
+    // 4× and merge for free during the scan read.
+    __shared__ float s_hist[kWaves][256];  // dedicated: BT may be < 256
     __shared__ unsigned s_sel[4];  // {prefix, count_above}
     unsigned prefix = 0;           // high bits fixed so far
     int above = 0;                 // elements strictly greater than the zone
+    const int mywave = tid >> 6;
     for (int shift = 24; shift >= 0; shift -= 8) {
       const unsigned pmask = (shift == 24) ? 0u : (0xFFFFFFFFu << (shift + 8));
-      s_hist[tid] = 0.f;
+#pragma unroll
+      for (int w = 0; w < kWaves; ++w) s_hist[w][tid] = 0.f;
       __syncthreads();
 #pragma unroll
       for (int k = 0; k < kMaxPer; ++k) {
         if ((myu[k] & pmask) == prefix && myu[k] != 0u) {
-          atomicAdd(&s_hist[(myu[k] >> shift) & 255u], 1.0f);
+          atomicAdd(&s_hist[mywave][(myu[k] >> shift) & 255u], 1.0f);
         }
       }
       __syncthreads();
-      // suffix sum over the 256 bins: intra-wave via shfl_up on reversed
-      // lanes is fiddly — do it with one value per thread: each thread
-      // needs sum of bins STRICTLY ABOVE its own (suf_gt) and including
-      // its own (suf_ge).
-      float mine = s_hist[tid];
+      // suffix sum over the 256 bins: each thread owns bin == tid and
+      // needs the count of bins STRICTLY ABOVE its own (suf_gt) and
+      // including its own (suf_ge).
+      float mine = 0.f;
+#pragma unroll
+      for (int w = 0; w < kWaves; ++w) mine += s_hist[w][tid];
       // wave-level inclusive suffix scan (lane i sums lanes >= i)
       float suf = mine;
 #pragma unroll
