@@ -306,6 +306,64 @@ class FusedOnPolicyStep(GraphableStep):
             opt.step()
         return True
 
+    # -- multi-rank split-graph: capture [fwd..wgrad] and [optimizer] as
+    # two graphs with the RCCL all-reduce stream-ordered between them.
+    # Full capture WITH the collective (PDRL_GRAPH_RCCL=1) is faster but
+    # cannot be validated on a single-GPU box; a bad replay would hang the
+    # driver's one multi-GPU run, so the split is the default.
+    def _try_capture_split(self, batch):
+        import os
+
+        self._split_graphs = None
+        self._split_failed = True
+        if self.params.K_epoch != 1:
+            return
+        try:
+            self._static = {k: batch[k] for k in BATCH_FIELDS}
+            self._static_ptrs = {k: batch[k].data_ptr() for k in BATCH_FIELDS}
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):  # warmup incl. RCCL channel setup
+                    self._full(self._static)
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            g_pre = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g_pre):
+                self._body(self._static, update=False)
+            g_post = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g_post):
+                self.optimizer.step()
+            self._split_graphs = (g_pre, g_post)
+            self._split_failed = False
+        except Exception as exc:
+            warnings.warn(
+                f"split-graph capture failed ({exc}); running the fused "
+                "multi-rank step stream-ordered instead")
+            self._static = None
+
+    def run(self, batch) -> dict:
+        if self.grad_reducer is not None and self.use_graph and not bool(
+            int(__import__("os").environ.get("PDRL_GRAPH_RCCL", "0"))
+        ):
+            if getattr(self, "_split_graphs", None) is None and not getattr(
+                    self, "_split_failed", False):
+                self._try_capture_split(batch)
+            if getattr(self, "_split_graphs", None) is not None:
+                for k in BATCH_FIELDS:
+                    if batch[k].data_ptr() != self._static_ptrs[k]:
+                        self._static[k].copy_(batch[k], non_blocking=True)
+                g_pre, g_post = self._split_graphs
+                g_pre.replay()
+                self.grad_reducer.all_reduce([self.optimizer.flat_grad])
+                g_post.replay()
+                return {name: self.stats_buf[i]
+                        for i, name in enumerate(self.stat_names)}
+            self._full(batch)
+            return {name: self.stats_buf[i]
+                    for i, name in enumerate(self.stat_names)}
+        return super().run(batch)
+
     def _body(self, batch, update: bool = True):
         c, p, A = self.core, self.params, self.A
         e = ext()

@@ -45,6 +45,45 @@ def test_seq_lstm_forward_parity():
         torch.testing.assert_close(outs_f[k], outs_e[k], rtol=1e-5, atol=1e-5)
 
 
+def test_split_graph_multirank_path():
+    """The multi-rank split-graph machinery (capture [fwd..wgrad] and
+    [optimizer] separately, collective stream-ordered between) must produce
+    the single-rank trajectory when the reducer is a no-op — exercised on
+    one GPU so the driver's first real 8-GPU run isn't its first run."""
+    _ops()
+    from pdrl_amd.agents.learner_module import switch_module
+    from pdrl_amd.parallel import GradReducer
+    from pdrl_amd.utils import load_params
+    from tests.conftest import make_batch
+
+    p = load_params()
+    p.algo = "IMPALA"
+    p.batch_size, p.seq_len, p.obs_dim, p.n_actions = 32, 5, 4, 2
+    upd_cls, model_cls = switch_module("IMPALA")
+
+    torch.manual_seed(13)
+    model_a = model_cls(4, 2, p.seq_len, p.hidden_size)
+    torch.manual_seed(13)
+    model_b = model_cls(4, 2, p.seq_len, p.hidden_size)
+
+    reducer = GradReducer()  # dist not initialized → enabled=False (no-op)
+    assert not reducer.enabled
+    upd_a = upd_cls(model_a, p, DEV, grad_reducer=reducer)
+    upd_b = upd_cls(model_b, p, DEV)
+
+    batch = make_batch(p, seed=41, device=DEV)
+    for _ in range(3):
+        sa = upd_a.step(batch)
+        sb = upd_b.step(batch)
+    assert getattr(upd_a.fused_step, "_split_graphs", None) is not None, \
+        "split-graph capture must engage on the multi-rank path"
+    torch.testing.assert_close(upd_a.optimizer.flat_param,
+                               upd_b.optimizer.flat_param,
+                               rtol=1e-4, atol=1e-6)
+    for k in sa:
+        assert abs(float(sa[k]) - float(sb[k])) < 1e-3, (k,)
+
+
 def test_megastep_matches_multilaunch():
     """The whole-step mega-kernel (ONE launch: fwd+loss+bwd+wgrad+RMSprop
     between grid barriers) must produce the same parameter trajectory and
