@@ -137,7 +137,8 @@ def main():
         # staging + training step + actor weight publish (D2H + encode + TCP
         # send on a bound PUB — the reference publishes after every update,
         # ppo/learning.py:108)
-        from pdrl_amd.agents.learner import BatchStager, WeightPublisher
+        from pdrl_amd.agents.learner import (
+            AsyncWeightPublisher, BatchStager, WeightPublisher)
         from pdrl_amd.transport import pub_bind
         from pdrl_amd.utils import Protocol, encode
 
@@ -146,12 +147,23 @@ def main():
         stager = BatchStager(device)
         pub = pub_bind("127.0.0.1", 35000 + 37 * rank) if rank == 0 else None
         actor = getattr(model, "actor", model)
-        wpub = WeightPublisher(actor, device) if pub is not None else None
+        wpub = None
+        awpub = None
+        if pub is not None:
+            if use_cuda:
+                # pipelined broadcast: gather + async D2H on the hot loop,
+                # pack/encode/send on the publisher thread (every update
+                # still broadcast — same as the learner process)
+                awpub = AsyncWeightPublisher(actor, device, pub.send)
+            else:
+                wpub = WeightPublisher(actor, device)
 
         def step_fn():
             dev_batch = stager.stage(host_np)
             updater.step(dev_batch)
-            if pub is not None:
+            if awpub is not None:
+                awpub.publish()
+            elif wpub is not None:
                 header, payload = encode(Protocol.Model, wpub.payload(),
                                          compress=False)
                 pub.send(header, payload)

@@ -131,6 +131,81 @@ class WeightPublisher:
         return {"wschema": self.schema, "wbuf": self._host}
 
 
+class AsyncWeightPublisher:
+    """Pipelined weight broadcast: the hot loop only issues the fused
+    device gather + an async D2H into one of two pinned snapshots and
+    records an event; a publisher thread waits on the event, packs,
+    encodes and sends. The broadcast still happens for every update —
+    it just overlaps the next training step instead of stalling it
+    (double-buffered; if both snapshots are in flight the publish is
+    coalesced into the next one, which is drop-oldest semantics — the
+    wire already has those). GPU-only; CPU callers use WeightPublisher."""
+
+    def __init__(self, actor: torch.nn.Module, device, send_fn):
+        import queue
+        import threading
+
+        self.device = torch.device(device)
+        sd = actor.state_dict(keep_vars=True)
+        self.schema = [(k, tuple(v.shape)) for k, v in sd.items()]
+        self.params = [v.detach() for v in sd.values()]
+        self.numel = sum(p.numel() for p in self.params)
+        self.send_fn = send_fn
+        self._slots = []
+        for _ in range(2):
+            dev = torch.empty(self.numel, dtype=torch.float32,
+                              device=self.device)
+            views = []
+            off = 0
+            for p in self.params:
+                n = p.numel()
+                views.append(dev[off:off + n].view_as(p))
+                off += n
+            pinned = torch.empty(self.numel, dtype=torch.float32,
+                                 pin_memory=True)
+            self._slots.append({
+                "dev": dev, "views": views, "pinned": pinned,
+                "host": pinned.numpy(), "ev": torch.cuda.Event(),
+            })
+        self._free = queue.Queue()
+        for i in range(2):
+            self._free.put(i)
+        self._ready: "queue.Queue[int | None]" = queue.Queue()
+        self._thread = threading.Thread(target=self._drain, daemon=True)
+        self._thread.start()
+
+    def publish(self):
+        try:
+            i = self._free.get_nowait()
+        except Exception:
+            return  # both snapshots in flight: coalesce into the next one
+        s = self._slots[i]
+        torch._foreach_copy_(s["views"], self.params)
+        s["pinned"].copy_(s["dev"], non_blocking=True)
+        s["ev"].record()
+        self._ready.put(i)
+
+    def _drain(self):
+        from pdrl_amd.utils import Protocol, encode
+
+        while True:
+            i = self._ready.get()
+            if i is None:
+                return
+            s = self._slots[i]
+            s["ev"].synchronize()
+            payload = {"wschema": self.schema, "wbuf": s["host"]}
+            header, body = encode(Protocol.Model, payload, compress=False)
+            try:
+                self.send_fn(header, body)
+            finally:
+                self._free.put(i)
+
+    def close(self):
+        self._ready.put(None)
+        self._thread.join(timeout=5.0)
+
+
 class Learner:
     def __init__(
         self,
@@ -190,10 +265,16 @@ class Learner:
         # weight plane: PUB bound at learner_port + 1 (rank 0 only)
         self.pub = pub_bind(learner_ip, learner_port + 1) if self.is_root else None
         self.weight_pub = None
+        self.weight_pub_async = None
         if self.is_root:
             mods = self.updater.trainable_modules()
             m = mods.get("model") or next(iter(mods.values()))
-            self.weight_pub = WeightPublisher(getattr(m, "actor", m), self.device)
+            actor = getattr(m, "actor", m)
+            if self.device.type == "cuda" and self.pub is not None:
+                self.weight_pub_async = AsyncWeightPublisher(
+                    actor, self.device, self.pub.send)
+            else:
+                self.weight_pub = WeightPublisher(actor, self.device)
         self.writer = SummaryWriter(params.result_dir) if self.is_root else None
         self.timer = ExecutionTimer(num_transition=params.seq_len * params.batch_size * world_size)
         self._replay_ready = False  # latching (reference: learner.py:385)
@@ -248,6 +329,9 @@ class Learner:
         interval = int(getattr(self.params, "model_publish_interval", 1) or 1)
         if interval > 1 and self.updater.update_count % interval != 0:
             return
+        if self.weight_pub_async is not None:
+            self.weight_pub_async.publish()
+            return
         header, payload = encode(Protocol.Model, self.weight_pub.payload(),
                                  compress=False)
         self.pub.send(header, payload)
@@ -301,6 +385,8 @@ class Learner:
                 break
 
     def close(self):
+        if self.weight_pub_async is not None:
+            self.weight_pub_async.close()
         if self.pub is not None:
             self.pub.close()
         if self.writer is not None:
