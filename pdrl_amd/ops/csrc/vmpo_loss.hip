@@ -148,36 +148,22 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
   __syncthreads();
   if (max_phase <= 2) return;
 
-  // phase C: k-th largest advantage via bit binary search (k = BT/2).
-  // Per-thread advantage bits live in REGISTERS (the LDS budget caps
-  // BT ≤ 2867, i.e. ≤ 12 elements/thread), and every reduction is
-  // wave-shfl + one 4-slot LDS combine — 2 barriers per search iteration
-  // instead of the 9-barrier shared-memory tree.
+  // phase C: k-th largest advantage via RADIX-256 selection, MSB-first —
+  // 4 rounds instead of the 32-round bit binary search (each search round
+  // was a full block-reduce LATENCY chain ≈ 0.7 µs → 23 µs total,
+  // measured via PDRL_VMPO_PHASE). CODE SIZE MATTERS MORE THAN ALU HERE:
+  // register-array + unrolled variants of this phase made even the
+  // UNCHANGED earlier phases ~10% slower (single-CU kernel, instruction
+  // fetch bound), so the loops below deliberately stay compact: plain
+  // strided re-reads of s_adv from LDS, no unrolled per-thread arrays.
   const int K = BT / 2 > 0 ? BT / 2 : 1;
-  constexpr int kMaxPer = 12;
-  // FULLY STATIC indexing: a runtime-bounded fill/scan (`myu[mycnt++]`)
-  // demotes the array to scratch (= global memory), and the 32 sequential
-  // search rounds then pay global latency per element — measured 46 µs for
-  // this kernel. Static indices keep all 12 lanes in VGPRs (guide rule 20);
-  // the 0u padding is the order-map of -FLT_MAX, never counted as greater.
-  unsigned myu[kMaxPer];
-#pragma unroll
-  for (int k = 0; k < kMaxPer; ++k) {
-    const int i = tid + k * kThreads;
-    myu[k] = (i < BT) ? f2u(s_adv[i]) : 0u;
-  }
-  // RADIX-256 selection, MSB-first: 4 rounds instead of the 32-round bit
-  // binary search (each search round was a full block-reduce LATENCY chain
-  // ≈ 0.7 µs → 23 µs total, measured via PDRL_VMPO_PHASE). Per round:
-  // one 256-bin LDS histogram of the elements still matching the prefix,
-  // a wave-level suffix scan (shfl, barrier-light), and the owning thread
-  // of the selected bucket publishes the narrowed prefix.
   unsigned thresh_u;
+  int n_gt;  // elements strictly greater than the threshold (radix output)
   {
-    // WAVE-PRIVATE histograms: advantages cluster into a handful of
-    // exponent buckets, so a single shared histogram serializes thousands
-    // of same-address LDS atomics; four private copies cut the contention
-    // 4× and merge for free during the scan read.
+    // wave-private histograms: advantages cluster into a handful of
+    // buckets, so one shared histogram would serialize same-address
+    // LDS atomics; four private copies cut that 4× and merge during the
+    // scan read.
     __shared__ float s_hist[kWaves][256];  // dedicated: BT may be < 256
     __shared__ unsigned s_sel[4];  // {prefix, count_above}
     unsigned prefix = 0;           // high bits fixed so far
@@ -188,25 +174,10 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
 #pragma unroll
       for (int w = 0; w < kWaves; ++w) s_hist[w][tid] = 0.f;
       __syncthreads();
-      // WAVE-AGGREGATED histogram adds: advantages cluster into 1-2
-      // buckets per round, so plain per-lane atomics serialize thousands
-      // of same-address RMWs (measured ~18 µs for the 4 rounds). Lanes
-      // sharing a bucket combine via ballot and the leader issues ONE
-      // atomic per distinct bucket per slot — the inner loop runs
-      // #distinct-buckets times (1-2 when clustered).
-#pragma unroll
-      for (int k = 0; k < kMaxPer; ++k) {
-        const bool valid = (myu[k] & pmask) == prefix && myu[k] != 0u;
-        const unsigned bucket = (myu[k] >> shift) & 255u;
-        unsigned long long todo = __ballot(valid);
-        while (todo) {
-          const int leader = __ffsll((long long)todo) - 1;
-          const unsigned b = __shfl(bucket, leader, kWave);
-          const unsigned long long same = __ballot(valid && bucket == b);
-          if ((tid & (kWave - 1)) == leader) {
-            atomicAdd(&s_hist[mywave][b], (float)__popcll(same));
-          }
-          todo &= ~same;
+      for (int i = tid; i < BT; i += kThreads) {
+        const unsigned u = f2u(s_adv[i]);
+        if ((u & pmask) == prefix && u != 0u) {
+          atomicAdd(&s_hist[mywave][(u >> shift) & 255u], 1.0f);
         }
       }
       __syncthreads();
@@ -244,19 +215,14 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
       __syncthreads();  // s_hist/s4 reused next round
     }
     thresh_u = prefix;
-    // special case: all-padding (BT < 1) can't happen (K >= 1 and real
-    // elements exist); elements equal to 0u (== f2u(-FLT_MAX)) are excluded
-    // from histograms, matching the search's "never greater" semantics —
-    // if the K-th largest were the minimum representable, ties resolve in
-    // the marking phase below exactly as before.
+    n_gt = above;  // after the last byte, `above` counts u > thresh exactly
+    // elements equal to 0u (== order-map of a NaN pattern) are excluded
+    // from histograms — they can only appear as padding/garbage and are
+    // never selected.
   }
 
   // mark selected: strictly greater always; equal by ascending index to k
   {
-    int cnt_gt = 0;
-#pragma unroll
-    for (int k = 0; k < kMaxPer; ++k) cnt_gt += (myu[k] > thresh_u);
-    const int n_gt = (int)block_sum((float)cnt_gt, s4);
     int need_eq = K - n_gt;  // ties to include, by lowest index
     for (int i = tid; i < BT; i += kThreads) {
       s_psi[i] = (f2u(s_adv[i]) > thresh_u) ? 1.f : 0.f;
