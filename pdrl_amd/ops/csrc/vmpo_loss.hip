@@ -169,7 +169,12 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
     unsigned prefix = 0;           // high bits fixed so far
     int above = 0;                 // elements strictly greater than the zone
     const int mywave = tid >> 6;
+    // probe codes 31..34: stop after that many radix rounds (timing only)
+    const int max_rounds = (max_phase >= 31 && max_phase <= 34)
+                               ? max_phase - 30 : 4;
+    int round_i = 0;
     for (int shift = 24; shift >= 0; shift -= 8) {
+      if (round_i++ >= max_rounds) return;
       const unsigned pmask = (shift == 24) ? 0u : (0xFFFFFFFFu << (shift + 8));
 #pragma unroll
       for (int w = 0; w < kWaves; ++w) s_hist[w][tid] = 0.f;
@@ -220,6 +225,7 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
     // from histograms — they can only appear as padding/garbage and are
     // never selected.
   }
+  if (max_phase >= 31 && max_phase <= 34) return;  // radix probe codes
 
   // mark selected: strictly greater always; equal by ascending index to k
   {

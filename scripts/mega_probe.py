@@ -93,7 +93,7 @@ def vmpo_phase_probe():
     params.algo = "V-MPO"
     params.obs_dim, params.n_actions, _ = env_shape("V-MPO")
     upd_cls, model_cls = switch_module("V-MPO")
-    for phase in (1, 2, 3, 4, 5, 99):
+    for phase in (1, 2, 31, 32, 33, 34, 3, 4, 5, 99):
         os.environ["PDRL_VMPO_PHASE"] = str(phase)
         torch.manual_seed(0)
         model = model_cls(4, 2, params.seq_len, params.hidden_size)
