@@ -45,6 +45,45 @@ def test_seq_lstm_forward_parity():
         torch.testing.assert_close(outs_f[k], outs_e[k], rtol=1e-5, atol=1e-5)
 
 
+def test_async_weight_publisher_gpu():
+    """AsyncWeightPublisher (the learner's pipelined broadcast): payloads
+    decode back to the exact actor weights; close() joins cleanly."""
+    _ops()
+    import time
+
+    from pdrl_amd.agents.learner import AsyncWeightPublisher
+    from pdrl_amd.buffers.wire import unpack_weights
+    from pdrl_amd.networks import MlpLSTMSingle
+    from pdrl_amd.utils import Protocol, decode
+
+    torch.manual_seed(2)
+    model = MlpLSTMSingle(4, 2, 5, 64).to(DEV)
+    got = []
+    pub = AsyncWeightPublisher(model.actor, DEV,
+                               lambda h, p: got.append((h, p)))
+    for _ in range(5):
+        with torch.no_grad():
+            for q in model.actor.parameters():
+                q.add_(0.01)
+        pub.publish()  # may coalesce (double-buffered) — that's the design
+    # quiesce, then publish the FINAL weights once and wait for it
+    deadline = time.monotonic() + 10
+    while pub._free.qsize() < 2 and time.monotonic() < deadline:
+        time.sleep(0.02)
+    n_before = len(got)
+    pub.publish()
+    while len(got) <= n_before and time.monotonic() < deadline:
+        time.sleep(0.02)
+    pub.close()
+    assert len(got) > n_before, "final broadcast did not arrive"
+    proto, obj = decode(*got[-1])
+    assert proto is Protocol.Model
+    state = unpack_weights(obj)
+    sd = model.actor.state_dict()
+    for k, v in state.items():
+        torch.testing.assert_close(v.to(DEV), sd[k], rtol=0, atol=0)
+
+
 def test_split_graph_multirank_path():
     """The multi-rank split-graph machinery (capture [fwd..wgrad] and
     [optimizer] separately, collective stream-ordered between) must produce
