@@ -227,19 +227,43 @@ __global__ __launch_bounds__(kThreads) void vmpo_loss_mega_kernel(
   }
   if (max_phase >= 31 && max_phase <= 34) return;  // radix probe codes
 
-  // mark selected: strictly greater always; equal by ascending index to k
+  // mark selected: strictly greater always; equal by ascending index to k.
+  // Tie marking is PARALLEL: the original thread-0 serial walk over BT
+  // elements cost ~17 µs whenever ties existed — and under the fixed-batch
+  // bench the advantages collapse into exact ties after a few hundred
+  // updates, so this path dominated the whole kernel (PDRL_VMPO_PHASE
+  // probes 31-34 vs 3). Each thread owns a contiguous chunk; a wave
+  // prefix scan of per-chunk equal-counts gives every tied element its
+  // ascending-index rank among equals.
   {
-    int need_eq = K - n_gt;  // ties to include, by lowest index
+    const int need_eq = K - n_gt;  // ties to include, by lowest index
     for (int i = tid; i < BT; i += kThreads) {
       s_psi[i] = (f2u(s_adv[i]) > thresh_u) ? 1.f : 0.f;
     }
     __syncthreads();
-    // serial-ish tie resolution (ties are rare): thread 0 marks
-    if (tid == 0 && need_eq > 0) {
-      for (int i = 0; i < BT && need_eq > 0; ++i) {
+    if (need_eq > 0) {  // uniform branch (need_eq broadcast from the radix)
+      const int chunk = (BT + kThreads - 1) / kThreads;
+      const int lo = tid * chunk;
+      const int hi = min(lo + chunk, BT);
+      int cnt = 0;
+      for (int i = lo; i < hi; ++i) cnt += (f2u(s_adv[i]) == thresh_u);
+      // exclusive prefix of the 256 per-thread counts: wave inclusive
+      // scan (shfl_up) + cross-wave base offsets
+      float inc = (float)cnt;
+#pragma unroll
+      for (int off = 1; off < kWave; off <<= 1) {
+        const float dn = __shfl_up(inc, off, kWave);
+        if ((tid & (kWave - 1)) >= off) inc += dn;
+      }
+      if ((tid & (kWave - 1)) == kWave - 1) s4[tid >> 6] = inc;
+      __syncthreads();
+      float base = 0.f;
+      for (int w = 0; w < (tid >> 6); ++w) base += s4[w];
+      int rank = (int)(base + inc) - cnt;  // my chunk's first equal's rank
+      for (int i = lo; i < hi && rank < need_eq; ++i) {
         if (f2u(s_adv[i]) == thresh_u) {
           s_psi[i] = 1.f;
-          --need_eq;
+          ++rank;
         }
       }
     }
