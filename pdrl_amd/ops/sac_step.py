@@ -2,14 +2,16 @@
 DAG — the discrete half of K11, hipGraph-capturable.
 
 Follows the EAGER reference ordering exactly
-(agents/learner_module/sac/learning.py), in 12 launches (launch latency IS
+(agents/learner_module/sac/learning.py), in 11 launches (launch latency IS
 the step time at this model size — profiles/algo_breakdown_r02a.md):
-  1.  actor + twin-critic forwards — ONE multi-network launch
+  1.  actor + twin critics + twin TARGET critics — ONE 5-network launch
+      (target params are constant until step 9, so their forward commutes
+      with the actor update)
   2.  sac_actor_loss      — analytic dlogits + dlog_alpha + stats
+                            (+ the shared Adam clock prep, in-kernel)
   3.  actor BPTT + MFMA wgrad → flat actor grads (+ norm)
-  4.  shared-clock tick + actor/alpha Adam — ONE multi-group launch
-  5.  actor fwd AGAIN (post-update policy, as the reference does) +
-      target-critic fwds — ONE multi-network launch
+  4.  actor/alpha Adam — ONE multi-group launch
+  5.  actor fwd AGAIN (post-update policy, as the reference does)
   7.  sac_critic_loss     — soft-Q TD target, twin huber, analytic dq1/dq2
   8.  q1+q2 BPTT (one launch) + q1+q2 MFMA wgrads (one launch) → Adam
   9.  Polyak soft_update (cached device pointer tables)
@@ -79,7 +81,7 @@ class FusedSacStep(GraphableStep):
             return torch.empty(*shape, device=dev)
 
         buf = {}
-        for name in ("actor", "q1", "q2", "actor2", "t1", "t2"):
+        for name in ("actor", "q1", "q2", "t1", "t2"):
             buf[name] = {"outs": mk(B, S, D), "hS": mk(B, H), "cS": mk(B, H),
                          "stash": mk(B, S, 7 * H)}
         for name in ("q1", "q2"):
@@ -103,11 +105,15 @@ class FusedSacStep(GraphableStep):
             return [b["outs"].data_ptr(), b["hS"].data_ptr(),
                     b["cS"].data_ptr(), b["stash"].data_ptr()]
 
-        self.fwd1_cores = t64([wrow(c["actor"]), wrow(c["q1"]), wrow(c["q2"])])
+        # fwd1 carries FIVE networks: actor + twin critics + twin TARGET
+        # critics — the targets' parameters only move at the Polyak update
+        # at the END of the step, so their forward commutes with the actor
+        # update and joins the first launch. Only the post-update actor
+        # re-forward remains separate.
+        self.fwd1_cores = t64([wrow(c["actor"]), wrow(c["q1"]), wrow(c["q2"]),
+                               wrow(c["t1"]), wrow(c["t2"])])
         self.fwd1_outs = t64([orow(buf["actor"]), orow(buf["q1"]),
-                              orow(buf["q2"])])
-        self.fwd2_cores = t64([wrow(c["actor"]), wrow(c["t1"]), wrow(c["t2"])])
-        self.fwd2_outs = t64([orow(buf["actor2"]), orow(buf["t1"]),
+                              orow(buf["q2"]), orow(buf["t1"]),
                               orow(buf["t2"])])
 
         def birow(name):
@@ -183,9 +189,11 @@ class FusedSacStep(GraphableStep):
             self._multi_setup(batch)
         buf, D = self.buf, self._D
 
-        # 1. actor + twin critic forwards — ONE multi-network launch
+        # 1. actor + twin critics + twin TARGET critics — ONE 5-network
+        #    launch (the targets' params are constant until the Polyak
+        #    update at step 9, so their forward commutes with steps 2-5)
         e.seq_lstm_forward_multi(x, hx0, cx0, self.fwd1_cores,
-                                 self.fwd1_outs, 3, D)
+                                 self.fwd1_outs, 5, D)
         moA1 = buf["actor"]["outs"]
         mq1, mq2 = buf["q1"]["outs"], buf["q2"]["outs"]
 
@@ -213,10 +221,13 @@ class FusedSacStep(GraphableStep):
                 self.grad_reducer.all_reduce([u.alpha_optimizer.flat_grad])
             u.alpha_optimizer._update()  # no clip: norm unused
 
-        # 5-6. post-update actor + target critics — ONE launch
-        e.seq_lstm_forward_multi(x, hx0, cx0, self.fwd2_cores,
-                                 self.fwd2_outs, 3, D)
-        moA2 = buf["actor2"]["outs"]
+        # 5. post-update actor forward (single network; the targets were
+        #    already evaluated in the 5-network launch)
+        ac = self.cores["actor"]
+        moA2, _, _, _ = e.seq_lstm_forward(
+            x, hx0, cx0, ac.body_w, ac.body_b, ac.w_ih, ac.w_hh, ac.b_g,
+            ac.heads_w, ac.heads_b,
+        )
         mt1, mt2 = buf["t1"]["outs"], buf["t2"]["outs"]
 
         # 7. critic losses (head grads land in the persistent gq buffers
