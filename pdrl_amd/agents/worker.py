@@ -88,6 +88,43 @@ class Worker:
         self._total_steps = 0
         self._ou_state: torch.Tensor | None = None
         self._ou_gen = torch.Generator().manual_seed((seed or 0) * 9973 + 17)
+        # C++ batched act (one native call per tick) when the actor is a
+        # standard discrete single-body core and the extension is built;
+        # otherwise the eager model.act path runs (continuous policies,
+        # missing extension)
+        self._act = self._make_fast_act(seed) or self.model.act
+
+    def _make_fast_act(self, seed):
+        """One-call C++ act (body+LSTM+logits+sample): removes the ~15
+        small-op eager dispatches per tick that bound worker throughput
+        (ops/csrc/cpu_actor.cpp). Reads the live parameter tensors, so
+        weight hot-reloads (in-place load_state_dict) apply immediately."""
+        if self._continuous:
+            return None
+        try:
+            from pdrl_amd.ops import _cpu_actor
+        except ImportError:
+            return None
+        actor = getattr(self.model, "actor", self.model)
+        core = getattr(actor, "core", None)
+        if core is None or getattr(core, "input2_dim", None) is not None:
+            return None
+        if not core.head_names or core.head_names[0] != "logits":
+            return None
+        A = core.head_dims["logits"]
+        rng = torch.tensor([((seed or 0) * 0x9E3779B97F4A7C15 + 0x2545F491) &
+                            ((1 << 63) - 1) | 1], dtype=torch.int64)
+
+        def fast(obs, hxs):
+            hx, cx = hxs
+            a, lg, lp, h, c = _cpu_actor.act_batch_discrete(
+                obs.contiguous(), hx.contiguous(), cx.contiguous(),
+                core.body_w.detach(), core.body_b.detach(),
+                core.w_ih.detach(), core.w_hh.detach(), core.b_g.detach(),
+                core.heads_w.detach(), core.heads_b.detach(), A, rng)
+            return a, lg, lp, (h, c)
+
+        return fast
 
     # ------------------------------------------------------------------ #
     def _ou_explore(self, action, logits):
@@ -166,7 +203,7 @@ class Worker:
                 if self._stopped():
                     break
                 self.poll_model()
-                action, logits, log_prob, (next_hx, next_cx) = self.model.act(obs, (hx, cx))
+                action, logits, log_prob, (next_hx, next_cx) = self._act(obs, (hx, cx))
                 if self._continuous and self._total_steps < self.explore_warmup_steps:
                     action, log_prob = self._ou_explore(action, logits)
                 self._total_steps += 1
@@ -215,7 +252,7 @@ class Worker:
         episodes = 0
         while not self._stopped():
             self.poll_model()
-            action, logits, log_prob, (next_hx, next_cx) = self.model.act(obs, (hx, cx))
+            action, logits, log_prob, (next_hx, next_cx) = self._act(obs, (hx, cx))
             if self._continuous and self._total_steps < self.explore_warmup_steps:
                 action, log_prob = self._ou_explore(action, logits)
             self._total_steps += M

@@ -1,0 +1,153 @@
+// CPU batched actor inference for the worker processes.
+//
+// The actor plane is CPU worker processes (reference topology kept:
+// SURVEY.md §2.5 — actor parallelism is the reference's only parallelism).
+// Its hot loop is `model.act` per tick: in eager PyTorch that is ~15 small
+// op dispatches on (M≤8, H=64) tensors — dispatch-overhead bound, and the
+// measured single-box ingest ceiling. This extension runs the whole act —
+// body GEMM+ReLU, one LSTM cell step, the logits head, categorical
+// sampling and log-prob — as ONE C++ call over the batch of envs, on the
+// same transposed weight layout as the HIP kernels / eager core
+// (networks/models.py SeqLSTMCore), so the worker swaps it in without any
+// model surgery. Discrete policies only (the continuous actor also needs
+// tanh-Gaussian sampling; its workers keep the eager path).
+//
+// Vector-friendly loop order: inner loops stream rows of the transposed
+// weights (unit stride) so -O3 autovectorizes them; no torch ops inside.
+#include <torch/extension.h>
+
+#include <cmath>
+#include <cstdint>
+#include <vector>
+
+namespace {
+
+inline float sigf(float x) { return 1.0f / (1.0f + std::exp(-x)); }
+
+inline uint64_t xorshift64(uint64_t& s) {
+  s ^= s << 13;
+  s ^= s >> 7;
+  s ^= s << 17;
+  return s;
+}
+
+}  // namespace
+
+// obs (M,F), hx/cx (M,H) fp32; weights in the SeqLSTMCore transposed layout;
+// logits live in head columns [0, A). rng: int64[1] state (advanced here).
+// Returns (action (M,1) int64, logits (M,A), log_prob (M,1), h (M,H), c (M,H)).
+std::vector<at::Tensor> act_batch_discrete(
+    const at::Tensor& obs, const at::Tensor& hx, const at::Tensor& cx,
+    const at::Tensor& body_w, const at::Tensor& body_b,
+    const at::Tensor& w_ih, const at::Tensor& w_hh, const at::Tensor& b_g,
+    const at::Tensor& heads_w, const at::Tensor& heads_b, int64_t A,
+    at::Tensor& rng) {
+  TORCH_CHECK(obs.device().is_cpu() && obs.dtype() == at::kFloat);
+  TORCH_CHECK(obs.is_contiguous() && hx.is_contiguous() && cx.is_contiguous());
+  const int M = obs.size(0), F = obs.size(1), H = hx.size(1);
+  const int G = 4 * H, D = heads_w.size(1);
+  TORCH_CHECK(body_w.size(0) == F && body_w.size(1) == H,
+              "single-body discrete core expected");
+  TORCH_CHECK(A <= D);
+
+  auto opt = obs.options();
+  auto action = at::empty({M, 1}, opt.dtype(at::kLong));
+  auto logits = at::empty({M, (long)A}, opt);
+  auto logp = at::empty({M, 1}, opt);
+  auto h_out = at::empty({M, H}, opt);
+  auto c_out = at::empty({M, H}, opt);
+
+  const float* ob = obs.data_ptr<float>();
+  const float* hp = hx.data_ptr<float>();
+  const float* cp = cx.data_ptr<float>();
+  const float* bw = body_w.data_ptr<float>();
+  const float* bb = body_b.data_ptr<float>();
+  const float* wih = w_ih.data_ptr<float>();
+  const float* whh = w_hh.data_ptr<float>();
+  const float* bg = b_g.data_ptr<float>();
+  const float* hw = heads_w.data_ptr<float>();
+  const float* hb = heads_b.data_ptr<float>();
+  int64_t* act_p = action.data_ptr<int64_t>();
+  float* lg_p = logits.data_ptr<float>();
+  float* lp_p = logp.data_ptr<float>();
+  float* ho_p = h_out.data_ptr<float>();
+  float* co_p = c_out.data_ptr<float>();
+
+  std::vector<float> xb(H), gates(G), hnew(H), prob(A);
+  uint64_t s = (uint64_t)rng.data_ptr<int64_t>()[0];
+
+  for (int m = 0; m < M; ++m) {
+    // body GEMM + ReLU (stream rows of body_w: unit stride over j)
+    for (int j = 0; j < H; ++j) xb[j] = bb[j];
+    for (int k = 0; k < F; ++k) {
+      const float x = ob[(long)m * F + k];
+      const float* row = bw + (long)k * H;
+      for (int j = 0; j < H; ++j) xb[j] += x * row[j];
+    }
+    for (int j = 0; j < H; ++j) xb[j] = xb[j] > 0.f ? xb[j] : 0.f;
+
+    // gates = xb @ w_ih + h @ w_hh + b_g
+    for (int g = 0; g < G; ++g) gates[g] = bg[g];
+    for (int k = 0; k < H; ++k) {
+      const float xk = xb[k];
+      const float* row = wih + (long)k * G;
+      for (int g = 0; g < G; ++g) gates[g] += xk * row[g];
+    }
+    for (int k = 0; k < H; ++k) {
+      const float hk = hp[(long)m * H + k];
+      const float* row = whh + (long)k * G;
+      for (int g = 0; g < G; ++g) gates[g] += hk * row[g];
+    }
+
+    // cell update (gate order i, f, g, o — matches the eager core)
+    for (int j = 0; j < H; ++j) {
+      const float gi = sigf(gates[j]);
+      const float gf = sigf(gates[H + j]);
+      const float gg = std::tanh(gates[2 * H + j]);
+      const float go = sigf(gates[3 * H + j]);
+      const float c = gf * cp[(long)m * H + j] + gi * gg;
+      const float h = go * std::tanh(c);
+      co_p[(long)m * H + j] = c;
+      ho_p[(long)m * H + j] = h;
+      hnew[j] = h;
+    }
+
+    // logits head (columns [0, A) of the packed heads)
+    float* lg = lg_p + (long)m * A;
+    for (int a = 0; a < A; ++a) lg[a] = hb[a];
+    for (int k = 0; k < H; ++k) {
+      const float hk = hnew[k];
+      const float* row = hw + (long)k * D;
+      for (int a = 0; a < A; ++a) lg[a] += hk * row[a];
+    }
+
+    // categorical sample + log-prob
+    float mx = lg[0];
+    for (int a = 1; a < A; ++a) mx = std::max(mx, lg[a]);
+    float Z = 0.f;
+    for (int a = 0; a < A; ++a) {
+      prob[a] = std::exp(lg[a] - mx);
+      Z += prob[a];
+    }
+    const float u = (float)((xorshift64(s) >> 11) *
+                            (1.0 / 9007199254740992.0));  // [0,1)
+    float acc = 0.f;
+    int chosen = A - 1;
+    for (int a = 0; a < A; ++a) {
+      acc += prob[a] / Z;
+      if (u < acc) {
+        chosen = a;
+        break;
+      }
+    }
+    act_p[m] = chosen;
+    lp_p[m] = lg[chosen] - mx - std::log(Z);
+  }
+  rng.data_ptr<int64_t>()[0] = (int64_t)s;
+  return {action, logits, logp, h_out, c_out};
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("act_batch_discrete", &act_batch_discrete,
+        "batched CPU actor step: body+LSTM+logits+sample in one call");
+}

@@ -11,7 +11,11 @@ from setuptools import setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
-from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+from torch.utils.cpp_extension import (  # noqa: E402
+    BuildExtension,
+    CppExtension,
+    CUDAExtension,
+)
 
 SRC = [
     "pdrl_amd/ops/csrc/bindings.cpp",
@@ -37,7 +41,16 @@ setup(
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17"],
             },
-        )
+        ),
+        # CPU batched actor for the worker processes (x86-64-v3 = AVX2:
+        # portable across the EPYC hosts; -march=native would pin to the
+        # build box)
+        CppExtension(
+            name="pdrl_amd.ops._cpu_actor",
+            sources=["pdrl_amd/ops/csrc/cpu_actor.cpp"],
+            extra_compile_args=["-O3", "-std=c++17", "-march=x86-64-v3",
+                                "-ffast-math"],
+        ),
     ],
     cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)},
 )
