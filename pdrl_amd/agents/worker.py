@@ -99,8 +99,6 @@ class Worker:
         small-op eager dispatches per tick that bound worker throughput
         (ops/csrc/cpu_actor.cpp). Reads the live parameter tensors, so
         weight hot-reloads (in-place load_state_dict) apply immediately."""
-        if self._continuous:
-            return None
         try:
             from pdrl_amd.ops import _cpu_actor
         except ImportError:
@@ -109,22 +107,38 @@ class Worker:
         core = getattr(actor, "core", None)
         if core is None or getattr(core, "input2_dim", None) is not None:
             return None
-        if not core.head_names or core.head_names[0] != "logits":
-            return None
-        A = core.head_dims["logits"]
+        heads = list(core.head_names or [])
         rng = torch.tensor([((seed or 0) * 0x9E3779B97F4A7C15 + 0x2545F491) &
                             ((1 << 63) - 1) | 1], dtype=torch.int64)
+        wargs = lambda: (core.body_w.detach(), core.body_b.detach(),  # noqa: E731
+                         core.w_ih.detach(), core.w_hh.detach(),
+                         core.b_g.detach(), core.heads_w.detach(),
+                         core.heads_b.detach())
 
-        def fast(obs, hxs):
-            hx, cx = hxs
-            a, lg, lp, h, c = _cpu_actor.act_batch_discrete(
-                obs.contiguous(), hx.contiguous(), cx.contiguous(),
-                core.body_w.detach(), core.body_b.detach(),
-                core.w_ih.detach(), core.w_hh.detach(), core.b_g.detach(),
-                core.heads_w.detach(), core.heads_b.detach(), A, rng)
-            return a, lg, lp, (h, c)
+        if not self._continuous and heads and heads[0] == "logits":
+            A = core.head_dims["logits"]
 
-        return fast
+            def fast(obs, hxs):
+                hx, cx = hxs
+                a, lg, lp, h, c = _cpu_actor.act_batch_discrete(
+                    obs.contiguous(), hx.contiguous(), cx.contiguous(),
+                    *wargs(), A, rng)
+                return a, lg, lp, (h, c)
+
+            return fast
+        if self._continuous and heads[:2] in (["mu", "std"], ["mu", "log_std"]):
+            A = core.head_dims["mu"]
+            mode = 0 if heads[1] == "std" else 1  # PPO-C | SAC-C sampling
+
+            def fastc(obs, hxs):
+                hx, cx = hxs
+                a, lg, lp, h, c = _cpu_actor.act_batch_gaussian(
+                    obs.contiguous(), hx.contiguous(), cx.contiguous(),
+                    *wargs(), A, mode, rng)
+                return a, lg, lp, (h, c)
+
+            return fastc
+        return None
 
     # ------------------------------------------------------------------ #
     def _ou_explore(self, action, logits):

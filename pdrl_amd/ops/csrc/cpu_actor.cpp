@@ -147,7 +147,128 @@ std::vector<at::Tensor> act_batch_discrete(
   return {action, logits, logp, h_out, c_out};
 }
 
+// Continuous policies. mode 0 = PPO-C scheme: Normal(tanh(mu),
+// softplus(std)+1e-4), unbounded sample (reference models.py:103-118);
+// mode 1 = SAC-C scheme: tanh-squashed reparameterized Gaussian with
+// log_std clamped to [-20, 2] (reference models.py:162-231). Heads:
+// mu = cols [0,A), second head = cols [A,2A) (a value head may follow).
+// Returns (action (M,A), logits (M,2A) = [mu|second_raw], logp (M,1), h, c).
+std::vector<at::Tensor> act_batch_gaussian(
+    const at::Tensor& obs, const at::Tensor& hx, const at::Tensor& cx,
+    const at::Tensor& body_w, const at::Tensor& body_b,
+    const at::Tensor& w_ih, const at::Tensor& w_hh, const at::Tensor& b_g,
+    const at::Tensor& heads_w, const at::Tensor& heads_b, int64_t A,
+    int64_t mode, at::Tensor& rng) {
+  TORCH_CHECK(obs.device().is_cpu() && obs.dtype() == at::kFloat);
+  TORCH_CHECK(obs.is_contiguous() && hx.is_contiguous() && cx.is_contiguous());
+  const int M = obs.size(0), F = obs.size(1), H = hx.size(1);
+  const int G = 4 * H, D = heads_w.size(1);
+  TORCH_CHECK(2 * A <= D);
+  constexpr float kHalfLog2Pi = 0.91893853320467274f;
+
+  auto opt = obs.options();
+  auto action = at::empty({M, (long)A}, opt);
+  auto logits = at::empty({M, (long)(2 * A)}, opt);
+  auto logp = at::empty({M, 1}, opt);
+  auto h_out = at::empty({M, H}, opt);
+  auto c_out = at::empty({M, H}, opt);
+
+  const float* ob = obs.data_ptr<float>();
+  const float* hp = hx.data_ptr<float>();
+  const float* cp = cx.data_ptr<float>();
+  const float* bw = body_w.data_ptr<float>();
+  const float* bb = body_b.data_ptr<float>();
+  const float* wih = w_ih.data_ptr<float>();
+  const float* whh = w_hh.data_ptr<float>();
+  const float* bg = b_g.data_ptr<float>();
+  const float* hw = heads_w.data_ptr<float>();
+  const float* hb = heads_b.data_ptr<float>();
+  float* act_p = action.data_ptr<float>();
+  float* lg_p = logits.data_ptr<float>();
+  float* lp_p = logp.data_ptr<float>();
+  float* ho_p = h_out.data_ptr<float>();
+  float* co_p = c_out.data_ptr<float>();
+
+  std::vector<float> xb(H), gates(G), hnew(H), head(2 * A);
+  uint64_t s = (uint64_t)rng.data_ptr<int64_t>()[0];
+  auto uni = [&]() {
+    return (float)(((xorshift64(s) >> 11) + 1) * (1.0 / 9007199254740993.0));
+  };
+
+  for (int m = 0; m < M; ++m) {
+    for (int j = 0; j < H; ++j) xb[j] = bb[j];
+    for (int k = 0; k < F; ++k) {
+      const float x = ob[(long)m * F + k];
+      const float* row = bw + (long)k * H;
+      for (int j = 0; j < H; ++j) xb[j] += x * row[j];
+    }
+    for (int j = 0; j < H; ++j) xb[j] = xb[j] > 0.f ? xb[j] : 0.f;
+
+    for (int g = 0; g < G; ++g) gates[g] = bg[g];
+    for (int k = 0; k < H; ++k) {
+      const float xk = xb[k];
+      const float* row = wih + (long)k * G;
+      for (int g = 0; g < G; ++g) gates[g] += xk * row[g];
+    }
+    for (int k = 0; k < H; ++k) {
+      const float hk = hp[(long)m * H + k];
+      const float* row = whh + (long)k * G;
+      for (int g = 0; g < G; ++g) gates[g] += hk * row[g];
+    }
+    for (int j = 0; j < H; ++j) {
+      const float gi = sigf(gates[j]);
+      const float gf = sigf(gates[H + j]);
+      const float gg = std::tanh(gates[2 * H + j]);
+      const float go = sigf(gates[3 * H + j]);
+      const float c = gf * cp[(long)m * H + j] + gi * gg;
+      const float h = go * std::tanh(c);
+      co_p[(long)m * H + j] = c;
+      ho_p[(long)m * H + j] = h;
+      hnew[j] = h;
+    }
+
+    for (int a = 0; a < 2 * A; ++a) head[a] = hb[a];
+    for (int k = 0; k < H; ++k) {
+      const float hk = hnew[k];
+      const float* row = hw + (long)k * D;
+      for (int a = 0; a < 2 * A; ++a) head[a] += hk * row[a];
+    }
+    for (int a = 0; a < 2 * A; ++a) lg_p[(long)m * 2 * A + a] = head[a];
+
+    float lp = 0.f;
+    for (int a = 0; a < A; ++a) {
+      const float mu_raw = head[a];
+      const float second = head[A + a];
+      // standard normal via Box-Muller
+      const float n =
+          std::sqrt(-2.0f * std::log(uni())) *
+          std::cos(6.283185307179586f * uni());
+      float act_v;
+      if (mode == 0) {  // PPO-C: Normal(tanh(mu), softplus(std)+1e-4)
+        const float mean = std::tanh(mu_raw);
+        const float sd =
+            std::log1p(std::exp(second)) + 1e-4f;  // softplus
+        act_v = mean + sd * n;
+        lp += -0.5f * n * n - std::log(sd) - kHalfLog2Pi;
+      } else {  // SAC-C: tanh-squashed, log_std clamped [-20, 2]
+        const float ls = std::min(std::max(second, -20.0f), 2.0f);
+        const float sd = std::exp(ls);
+        const float z = mu_raw + sd * n;
+        act_v = std::tanh(z);
+        lp += -0.5f * n * n - ls - kHalfLog2Pi -
+              std::log(1.0f - act_v * act_v + 1e-7f);
+      }
+      act_p[(long)m * A + a] = act_v;
+    }
+    lp_p[m] = lp;
+  }
+  rng.data_ptr<int64_t>()[0] = (int64_t)s;
+  return {action, logits, logp, h_out, c_out};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("act_batch_discrete", &act_batch_discrete,
         "batched CPU actor step: body+LSTM+logits+sample in one call");
+  m.def("act_batch_gaussian", &act_batch_gaussian,
+        "batched CPU actor step for Gaussian policies (PPO-C / SAC-C)");
 }

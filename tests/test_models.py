@@ -229,3 +229,56 @@ def test_worker_fast_act_engages_and_rolls():
     a, lg, lp, (h, c) = act(obs, (hx, cx))
     assert a.shape == (2, 1) and lg.shape == (2, 2) and lp.shape == (2, 1)
     assert h.shape == (2, p.hidden_size)
+
+
+def test_cpu_actor_gaussian_matches_eager():
+    """C++ Gaussian act (both sampling schemes) vs the eager actors: the
+    deterministic pieces (logits record, h, c) match; sampled actions and
+    log-probs are mutually consistent under the recorded distribution."""
+    pytest.importorskip("pdrl_amd.ops._cpu_actor")
+    import math
+
+    import torch
+
+    from pdrl_amd.networks import MlpLSTMActorContinuous, MlpLSTMContinuous
+    from pdrl_amd.ops import _cpu_actor
+
+    torch.manual_seed(5)
+    M, A = 5, 2
+
+    for mode, cls in ((0, MlpLSTMContinuous), (1, MlpLSTMActorContinuous)):
+        model = cls(3, A, 5, 64).eval()
+        core = model.core
+        obs = torch.randn(M, 3)
+        hx = torch.randn(M, 64) * 0.1
+        cx = torch.randn(M, 64) * 0.1
+        rng = torch.tensor([777], dtype=torch.int64)
+        a, lg, lp, h, c = _cpu_actor.act_batch_gaussian(
+            obs, hx, cx, core.body_w.detach(), core.body_b.detach(),
+            core.w_ih.detach(), core.w_hh.detach(), core.b_g.detach(),
+            core.heads_w.detach(), core.heads_b.detach(), A, mode, rng)
+
+        outs, he, ce = core.step(obs, hx, cx)
+        torch.testing.assert_close(h, he, rtol=1e-4, atol=1e-5)
+        torch.testing.assert_close(c, ce, rtol=1e-4, atol=1e-5)
+        second = outs["std"] if mode == 0 else outs["log_std"]
+        torch.testing.assert_close(
+            lg, torch.cat([outs["mu"], second], -1), rtol=1e-4, atol=1e-5)
+
+        # log-prob consistency with the recorded (mu, second) under the
+        # scheme's own density
+        mu, sec = lg[..., :A], lg[..., A:]
+        if mode == 0:
+            mean = torch.tanh(mu)
+            sd = torch.nn.functional.softplus(sec) + 1e-4
+            ref = (-0.5 * ((a - mean) / sd) ** 2 - sd.log()
+                   - 0.5 * math.log(2 * math.pi)).sum(-1, keepdim=True)
+        else:
+            ls = sec.clamp(-20.0, 2.0)
+            sd = ls.exp()
+            z = torch.atanh(a.clamp(-1 + 1e-6, 1 - 1e-6))
+            ref = (-0.5 * ((z - mu) / sd) ** 2 - ls
+                   - 0.5 * math.log(2 * math.pi)
+                   - torch.log(1 - a.pow(2) + 1e-7)).sum(-1, keepdim=True)
+            assert a.abs().max() < 1.0  # tanh-squashed
+        torch.testing.assert_close(lp, ref, rtol=1e-3, atol=1e-4)
