@@ -76,19 +76,15 @@ class BaseUpdater:
             return None
         from pdrl_amd.ops.fused_step import FusedOnPolicyStep
 
-        # multi-rank capture INCLUDES the RCCL all-reduce in the graph (the
-        # communicator is warmed up by the pre-capture side-stream runs). If
-        # capture fails on a rank it falls back to the stream-ordered fused
-        # path — correct either way, because graphed and non-graphed ranks
-        # issue the identical collective sequence. PDRL_GRAPH_RCCL=0 forces
-        # stream-ordered for all multi-rank steps.
+        # Graph policy: single-rank captures the whole step. Multi-rank
+        # (decided inside FusedOnPolicyStep.run) defaults to SPLIT-GRAPH —
+        # two captured graphs around a stream-ordered RCCL all-reduce —
+        # because full capture including the collective cannot be validated
+        # before the driver's one multi-GPU run; PDRL_GRAPH_RCCL=1 opts
+        # into full capture, PDRL_USE_GRAPH=0 forces stream-ordered.
         import os
 
         use_graph = bool(int(os.environ.get("PDRL_USE_GRAPH", "1")))
-        if self.grad_reducer is not None and not bool(
-            int(os.environ.get("PDRL_GRAPH_RCCL", "1"))
-        ):
-            use_graph = False
         return FusedOnPolicyStep(algo, core, self.params, optimizer,
                                  grad_reducer=self.grad_reducer,
                                  use_graph=use_graph, duals=duals)
