@@ -137,8 +137,7 @@ def main():
         # staging + training step + actor weight publish (D2H + encode + TCP
         # send on a bound PUB — the reference publishes after every update,
         # ppo/learning.py:108)
-        from pdrl_amd.agents.learner import (
-            AsyncWeightPublisher, BatchStager, WeightPublisher)
+        from pdrl_amd.agents.learner import BatchStager, WeightPublisher
         from pdrl_amd.transport import pub_bind
         from pdrl_amd.utils import Protocol, encode
 
@@ -147,13 +146,11 @@ def main():
         stager = BatchStager(device)
         pub = pub_bind("127.0.0.1", 35000 + 37 * rank) if rank == 0 else None
         actor = getattr(model, "actor", model)
-        # NOTE: measured both publishers here — the async (thread) variant
-        # LOST 13 µs/step to GIL contention (its 0.7 MB pickle blocks the
-        # Python-driven hot loop), so the bench publishes synchronously;
-        # AsyncWeightPublisher stays the learner-process default, where the
-        # loop has real idle gaps to absorb the thread.
+        # Synchronous publisher on purpose: the threaded AsyncWeightPublisher
+        # LOST 13 µs/step here to GIL contention (its 0.7 MB pickle blocks
+        # this Python-driven hot loop); it stays the learner-process
+        # default, where the loop has real idle gaps to absorb the thread.
         wpub = WeightPublisher(actor, device) if pub is not None else None
-        del AsyncWeightPublisher  # documented-above decision; unused here
 
         def step_fn():
             dev_batch = stager.stage(host_np)
