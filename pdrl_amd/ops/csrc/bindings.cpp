@@ -47,6 +47,21 @@ bool megastep_onpolicy_hip(
     double, double, double, double, double, double, double, double, double,
     double, double, double, double, double, double, bool, long);
 void barrier_bench_hip(at::Tensor&, long, long);
+void seq_lstm_fwd_loss_hip(
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
+    const c10::optional<at::Tensor>&, long, double, double, double, double,
+    double, double, double, double, double, double, double);
+void seq_lstm_bwd_fin_hip(
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, at::Tensor&, at::Tensor&,
+    at::Tensor&, at::Tensor&, at::Tensor&, long, double, double, double,
+    double);
 at::Tensor gae_hip(const at::Tensor&, double, double, const at::Tensor&);
 std::vector<at::Tensor> vtrace_hip(const at::Tensor&, const at::Tensor&,
                                    const at::Tensor&, const at::Tensor&,
@@ -190,6 +205,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "+RMSprop between grid barriers); false => shape not co-resident");
   m.def("barrier_bench", &barrier_bench_hip,
         "grid-barrier microbenchmark (N back-to-back barriers)");
+  m.def("seq_lstm_fwd_loss", &seq_lstm_fwd_loss_hip,
+        "forward + row-local IMPALA/PPO loss in ONE launch");
+  m.def("seq_lstm_bwd_fin", &seq_lstm_bwd_fin_hip,
+        "BPTT backward + loss-stat finalization in ONE launch");
   m.def("gae", &gae_hip, "GAE reverse scan");
   m.def("vtrace", &vtrace_hip, "fused V-trace scan",
         pybind11::arg("behav_lp"), pybind11::arg("target_lp"),

@@ -1,0 +1,175 @@
+// Row-local on-policy (IMPALA/PPO) loss for ONE batch row: categorical
+// stats, V-trace/GAE scan, analytic packed head grads, atomic loss-stat
+// partials. Shared by megastep.hip (whole-step kernel) and fwd_loss.hip
+// (loss fused into the forward launch): both exploit that these losses
+// have NO cross-row coupling except the monitoring means.
+#pragma once
+
+#include "common.h"
+
+constexpr int kAlgoImpala = 0;
+constexpr int kAlgoPpo = 1;
+constexpr int kMsThreads = 256;
+
+__device__ __forceinline__ float ms_huber(float d) {
+  const float a = fabsf(d);
+  return (a < 1.0f) ? 0.5f * d * d : a - 0.5f;
+}
+__device__ __forceinline__ float ms_huber_grad(float d) {
+  return fminf(fmaxf(d, -1.0f), 1.0f);
+}
+
+
+// Row-local IMPALA/PPO loss for batch row b: categorical stats, V-trace or
+// TD+GAE scan, analytic packed head grads, atomic loss-stat partials.
+// stats_acc: [pl, vl, es, ravg/rsum, rg]; mm: ordered-int {min,max} of the
+// PPO ratio (ratios are positive, so int-bit order == float order).
+__device__ void onpolicy_loss_row(
+    int algo, const float* __restrict__ outs, const float* __restrict__ act,
+    const float* __restrict__ behav, const float* __restrict__ rew,
+    const float* __restrict__ fir, float* __restrict__ gouts,
+    float* __restrict__ stats_acc, int* __restrict__ mm, int b, int B, int S,
+    int A, float gamma, float lmbda, float rho_bar, float rho_min,
+    float c_bar, float rew_scale, float cp, float cv, float ce,
+    float eps_clip, float creg, char* smem) {
+  const int D = A + 1;
+  const int T = S - 1;
+  const int BT = B * T;
+  const int tid = threadIdx.x;
+  float* s_logp = reinterpret_cast<float*>(smem);  // (S)
+  float* s_lse = s_logp + S;                       // (S)
+  float* s_ent = s_lse + S;                        // (S)
+  float* s_w = s_ent + S;                          // (T) rho (IMPALA)
+  float* s_adv = s_w + S;                          // (T)
+  float* s_ret = s_adv + S;                        // (T) vs (IMPALA) | td (PPO)
+  const long sb = (long)b * S;
+
+  if (tid < S) {
+    const float* z = outs + (sb + tid) * D;
+    float m = z[0];
+    for (int j = 1; j < A; ++j) m = fmaxf(m, z[j]);
+    float s = 0.f;
+    for (int j = 0; j < A; ++j) s += __expf(z[j] - m);
+    const float l = m + __logf(s);
+    float h = 0.f;
+    for (int j = 0; j < A; ++j) {
+      const float lp = z[j] - l;
+      h -= __expf(lp) * lp;
+    }
+    s_logp[tid] = z[(int)act[sb + tid]] - l;
+    s_lse[tid] = l;
+    s_ent[tid] = h;
+  }
+  __syncthreads();
+
+  if (tid == 0) {
+    // scans + this row's loss partials (T is tiny: serial on one lane)
+    float pl = 0.f, vl = 0.f, es = 0.f, rs = 0.f, rg = 0.f;
+    if (algo == kAlgoImpala) {
+      float acc = 0.f;
+      for (int t = T - 1; t >= 0; --t) {
+        const float ratio = __expf(s_logp[t] - behav[sb + t]);
+        const float rho = fminf(fmaxf(ratio, rho_min), rho_bar);
+        const float c = fminf(ratio, c_bar);
+        const float mask = 1.f - fir[sb + t + 1];
+        const float vt = outs[(sb + t) * D + A];
+        const float vn = outs[(sb + t + 1) * D + A];
+        const float delta =
+            rho * (rew[sb + t] * rew_scale + gamma * mask * vn - vt);
+        acc = fmaf(gamma * mask * c, acc, delta);
+        s_w[t] = rho;
+        s_ret[t] = vt + acc;
+      }
+      for (int t = 0; t < T; ++t) {
+        const float mask = 1.f - fir[sb + t + 1];
+        const float vnext = (t + 1 < T) ? s_ret[t + 1] : outs[(sb + T) * D + A];
+        s_adv[t] = s_w[t] * (rew[sb + t] * rew_scale + gamma * mask * vnext -
+                             outs[(sb + t) * D + A]);
+      }
+      for (int t = 0; t < T; ++t) {
+        pl -= s_logp[t] * s_adv[t];
+        vl += ms_huber(outs[(sb + t) * D + A] - s_ret[t]);
+        es += s_ent[t];
+        rs += s_w[t];
+        for (int j = 0; j < A; ++j) {
+          const float z = outs[(sb + t) * D + j];
+          rg = fmaf(z, z, rg);
+        }
+      }
+    } else {  // PPO: TD target + GAE, clipped surrogate
+      float run = 0.f;
+      for (int t = T - 1; t >= 0; --t) {
+        const float mask = 1.f - fir[sb + t + 1];
+        const float tdv = rew[sb + t] * rew_scale +
+                          gamma * mask * outs[(sb + t + 1) * D + A];
+        const float delta = tdv - outs[(sb + t) * D + A];
+        run = fmaf(gamma * lmbda * mask, run, delta);
+        s_ret[t] = tdv;
+        s_adv[t] = run;
+      }
+      float rmin = 1e30f, rmax = -1e30f;
+      for (int t = 0; t < T; ++t) {
+        const float r = __expf(s_logp[t] - behav[sb + t]);
+        const float a = s_adv[t];
+        const float s1 = r * a;
+        const float s2 =
+            fminf(fmaxf(r, 1.f - eps_clip), 1.f + eps_clip) * a;
+        pl -= fminf(s1, s2);
+        vl += ms_huber(outs[(sb + t) * D + A] - s_ret[t]);
+        es += s_ent[t];
+        rs += r;
+        rmin = fminf(rmin, r);
+        rmax = fmaxf(rmax, r);
+        for (int j = 0; j < A; ++j) {
+          const float z = outs[(sb + t) * D + j];
+          rg = fmaf(z, z, rg);
+        }
+      }
+      atomicMin(mm, __float_as_int(rmin));
+      atomicMax(mm + 1, __float_as_int(rmax));
+    }
+    atomicAdd(stats_acc + 0, pl);
+    atomicAdd(stats_acc + 1, vl);
+    atomicAdd(stats_acc + 2, es);
+    atomicAdd(stats_acc + 3, rs);
+    atomicAdd(stats_acc + 4, rg);
+  }
+  __syncthreads();
+
+  // analytic packed head grads, elementwise over this row
+  const float invN = 1.0f / BT;
+  const float dreg = 2.0f * creg * invN / A;
+  for (int idx = tid; idx < S * D; idx += (int)blockDim.x) {
+    const int t = idx / D, j = idx % D;
+    float* g = gouts + (sb + t) * D;
+    if (t >= T) {
+      g[j] = 0.f;
+      continue;
+    }
+    const float* z = outs + (sb + t) * D;
+    if (j == A) {
+      g[A] = cv * ms_huber_grad(z[A] - s_ret[t]) * invN;
+      continue;
+    }
+    float dlogp;
+    if (algo == kAlgoImpala) {
+      dlogp = -cp * s_adv[t] * invN;
+    } else {
+      const float r = __expf(s_logp[t] - behav[sb + t]);
+      const float a_v = s_adv[t];
+      const bool inside = (r > 1.f - eps_clip) && (r < 1.f + eps_clip);
+      const float s1 = r * a_v;
+      const float s2 = fminf(fmaxf(r, 1.f - eps_clip), 1.f + eps_clip) * a_v;
+      const float gr = (inside || s1 < s2) ? a_v * r : 0.f;
+      dlogp = -cp * gr * invN;
+    }
+    const float dH = -ce * invN;
+    const float H = s_ent[t];
+    const float lp = z[j] - s_lse[t];
+    const float pj = __expf(lp);
+    const int a = (int)act[sb + t];
+    g[j] = dlogp * ((j == a ? 1.f : 0.f) - pj) + dH * (-pj * (lp + H)) +
+           dreg * z[j];
+  }
+}
+

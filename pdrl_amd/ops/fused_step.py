@@ -364,6 +364,76 @@ class FusedOnPolicyStep(GraphableStep):
                     for i, name in enumerate(self.stat_names)}
         return super().run(batch)
 
+    # -- fwd+loss fusion: the on-policy loss is ROW-LOCAL, so it rides the
+    # forward launch (same block, no barrier) and the stats finalize at the
+    # head of the backward launch — removes the dedicated loss launch
+    # (~9 µs of the 52.8 µs step) with none of the megastep's barrier cost.
+    def _fwdloss_setup(self, batch):
+        c = self.core
+        x = batch["obs"]
+        B, S, _ = x.shape
+        dev = x.device
+        H = c.w_ih.size(0)
+        D = c.heads_w.size(1)
+
+        def mk(*shape):
+            return torch.empty(*shape, device=dev)
+
+        mm = torch.empty(2, dtype=torch.int32, device=dev)
+        mm[0] = torch.tensor(1e30).view(torch.int32)
+        mm[1] = torch.tensor(-1e30).view(torch.int32)
+        self._fl = {
+            "outs": mk(B, S, D), "hS": mk(B, H), "cS": mk(B, H),
+            "stash": mk(B, S, 7 * H), "gouts": mk(B, S, D),
+            "dgates": mk(B, S, 4 * H), "dxb": mk(B, S, H),
+            "stats_acc": torch.zeros(8, dtype=torch.float32, device=dev),
+            "mm": mm,
+        }
+        self._fl_shape = (B, S)
+
+    def _try_fwdloss(self, e, batch, x, act, behav, rew, fir, hx0, cx0,
+                     B, S, A, p, update) -> bool:
+        import os
+
+        if self.algo not in ("IMPALA", "PPO"):
+            return False
+        if self.core.w_ih.size(0) != 64:
+            return False  # kernels specialized for the H=64 model family
+        if not bool(int(os.environ.get("PDRL_FWDLOSS", "1"))):
+            return False
+        if getattr(self, "_fl_shape", None) != (B, S):
+            self._fwdloss_setup(batch)
+        ws = self._fl
+        c = self.core
+        norm = self._norm_buf()
+        algo_i = 0 if self.algo == "IMPALA" else 1
+        e.seq_lstm_fwd_loss(
+            x, hx0, cx0, c.body_w, c.body_b, c.w_ih, c.w_hh, c.b_g,
+            c.heads_w, c.heads_b, ws["outs"], ws["hS"], ws["cS"],
+            ws["stash"], act, behav, rew, fir, ws["gouts"], ws["stats_acc"],
+            ws["mm"], norm, algo_i, p.gamma, p.lmbda, 0.8, 0.1, 1.0,
+            p.reward_scale, p.policy_loss_coef, p.value_loss_coef,
+            p.entropy_coef, p.eps_clip,
+            float(getattr(p, "logit_reg", 0.0)),
+        )
+        e.seq_lstm_bwd_fin(
+            ws["gouts"], ws["stash"], x, cx0, c.body_w, c.w_ih, c.w_hh,
+            c.heads_w, ws["dgates"], ws["dxb"], self.stats_buf,
+            ws["stats_acc"], ws["mm"], algo_i, p.policy_loss_coef,
+            p.value_loss_coef, p.entropy_coef,
+            float(getattr(p, "logit_reg", 0.0)),
+        )
+        e.seq_lstm_wgrad_out(x, hx0, ws["stash"], ws["dgates"], ws["dxb"],
+                             ws["gouts"], *self._grad_views(), norm)
+        if not update:
+            return True
+        if self.grad_reducer is not None:
+            self.grad_reducer.all_reduce([self.optimizer.flat_grad])
+            self.optimizer.step()
+        else:
+            self.optimizer._update()
+        return True
+
     def _body(self, batch, update: bool = True):
         c, p, A = self.core, self.params, self.A
         e = ext()
@@ -382,6 +452,9 @@ class FusedOnPolicyStep(GraphableStep):
 
         if update and self._try_megastep(e, batch, x, act, behav, rew, fir,
                                          hx0, cx0, B, S, A, p):
+            return
+        if self._try_fwdloss(e, batch, x, act, behav, rew, fir, hx0, cx0,
+                             B, S, A, p, update):
             return
 
         mo, hS, cS, stash = e.seq_lstm_forward(

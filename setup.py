@@ -29,6 +29,7 @@ SRC = [
     "pdrl_amd/ops/csrc/scans.hip",
     "pdrl_amd/ops/csrc/multi_tensor.hip",
     "pdrl_amd/ops/csrc/megastep.hip",
+    "pdrl_amd/ops/csrc/fwd_loss.hip",
 ]
 
 setup(

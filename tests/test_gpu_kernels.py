@@ -586,9 +586,13 @@ def test_fused_impala_shape_sweep(B, S, A):
 
 
 def test_fused_impala_large_shape_fallback():
-    """Shapes beyond the mega-kernel LDS budget run the 4-kernel loss path;
-    gradients must still match eager autograd."""
+    """Large shapes, BOTH fused paths vs eager autograd: the fwd+loss
+    fused launch (row-local, no shape cap) and — with PDRL_FWDLOSS=0 —
+    the legacy route where the mega loss kernel exceeds its LDS budget
+    and falls back to the 4-kernel loss sequence."""
     _ops()
+    import os
+
     from pdrl_amd.agents.learner_module import ImpalaUpdater
     from pdrl_amd.networks import MlpLSTMSingle
     from pdrl_amd.utils import load_params
@@ -596,18 +600,25 @@ def test_fused_impala_large_shape_fallback():
 
     p = load_params()
     p.batch_size, p.seq_len, p.obs_dim, p.n_actions = 640, 5, 4, 2
-    torch.manual_seed(2)
-    model = MlpLSTMSingle(4, 2, p.seq_len, p.hidden_size)
-    upd = ImpalaUpdater(model, p, DEV)
-    batch = make_batch(p, device=DEV, seed=77)
-    upd.fused_step.compute_grads_only(batch)
-    fused = {n: q.grad.detach().clone() for n, q in model.named_parameters()}
-    upd.optimizer.zero_grad()
-    loss, _ = upd.compute_losses(batch)
-    loss.backward()
-    for n, q in model.named_parameters():
-        torch.testing.assert_close(fused[n], q.grad, rtol=2e-4, atol=2e-6,
-                                   msg=lambda m: f"{n}: {m}")
+    for fwdloss in ("1", "0"):
+        os.environ["PDRL_FWDLOSS"] = fwdloss
+        try:
+            torch.manual_seed(2)
+            model = MlpLSTMSingle(4, 2, p.seq_len, p.hidden_size)
+            upd = ImpalaUpdater(model, p, DEV)
+            batch = make_batch(p, device=DEV, seed=77)
+            upd.fused_step.compute_grads_only(batch)
+            fused = {n: q.grad.detach().clone()
+                     for n, q in model.named_parameters()}
+            upd.optimizer.zero_grad()
+            loss, _ = upd.compute_losses(batch)
+            loss.backward()
+            for n, q in model.named_parameters():
+                torch.testing.assert_close(
+                    fused[n], q.grad, rtol=2e-4, atol=2e-6,
+                    msg=lambda m: f"fwdloss={fwdloss} {n}: {m}")
+        finally:
+            os.environ.pop("PDRL_FWDLOSS", None)
 
 
 def test_sac_fused_step_parity():
