@@ -43,7 +43,7 @@ bool megastep_onpolicy_hip(
     at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
     at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
     at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
-    at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, long,
+    at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, long,
     double, double, double, double, double, double, double, double, double,
     double, double, double, double, double, double, bool, long);
 void barrier_bench_hip(at::Tensor&, long, long);
@@ -53,14 +53,14 @@ void seq_lstm_fwd_loss_hip(
     const at::Tensor&, const at::Tensor&, const at::Tensor&,
     const at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
     const at::Tensor&, const at::Tensor&, const at::Tensor&,
-    const at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
+    const at::Tensor&, at::Tensor&, at::Tensor&,
     const c10::optional<at::Tensor>&, long, double, double, double, double,
     double, double, double, double, double, double, double);
 void seq_lstm_bwd_fin_hip(
     const at::Tensor&, const at::Tensor&, const at::Tensor&,
     const at::Tensor&, const at::Tensor&, const at::Tensor&,
     const at::Tensor&, const at::Tensor&, at::Tensor&, at::Tensor&,
-    at::Tensor&, at::Tensor&, at::Tensor&, long, double, double, double,
+    at::Tensor&, at::Tensor&, long, double, double, double,
     double);
 at::Tensor gae_hip(const at::Tensor&, double, double, const at::Tensor&);
 std::vector<at::Tensor> vtrace_hip(const at::Tensor&, const at::Tensor&,

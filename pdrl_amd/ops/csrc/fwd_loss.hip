@@ -4,19 +4,18 @@
 // its row's forward it can compute that row's categorical stats,
 // V-trace/GAE scan and analytic head grads in the SAME launch — no grid
 // barrier, unlike the whole-step mega-kernel. The only cross-row products
-// are the monitoring sums, accumulated with atomics and FINALIZED at the
-// head of the backward launch (stream order guarantees all row atomics
-// landed), which also re-zeroes the accumulators for the next step.
-// This removes the dedicated single-block loss launch (~9 µs of the
-// 52.8 µs IMPALA step) without any of the megastep's barrier costs.
+// are the monitoring sums: each row stores its partials, and block 0 of
+// the backward launch reduces them (stream order guarantees every row's
+// stores landed). This removes the dedicated single-block loss launch
+// (~9 µs of the 52.8 µs IMPALA step) without the megastep's barriers.
 //
-// Accumulator protocol (race-free by stream ordering, no barriers):
+// Stats protocol (race-free by stream ordering, no barriers, no atomics):
 //   fwd_loss:  block 0 zeroes norm_sq (its next writer, the wgrad kernel,
-//              runs in a LATER launch); all row blocks atomicAdd stats_acc
-//              and atomicMin/Max mm (zeroed by the PREVIOUS step's
-//              bwd_fin; initial state zeroed at allocation).
-//   bwd_fin:   block 0 thread 0 reads the raw sums, writes the final
-//              stats vector, and re-zeroes stats_acc / mm.
+//              runs in a LATER launch); each row block PLAIN-stores its
+//              loss partials into its own stats_part row.
+//   bwd_fin:   block 0 reduces the (B, 8) partials in parallel and writes
+//              the final stats vector — nothing to re-zero (rows are
+//              fully overwritten every step).
 #include "common.h"
 #include "core_rows.h"
 #include "loss_row.h"
@@ -36,8 +35,8 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_fwd_loss_kernel(
     float* __restrict__ stash,
     const float* __restrict__ act, const float* __restrict__ behav,
     const float* __restrict__ rew, const float* __restrict__ fir,
-    float* __restrict__ gouts, float* __restrict__ stats_acc,
-    int* __restrict__ mm, float* __restrict__ norm_sq,
+    float* __restrict__ gouts, float* __restrict__ stats_part,
+    float* __restrict__ norm_sq,
     int algo, int B, int S, int F, int D, long h0s, float gamma, float lmbda,
     float rho_bar, float rho_min, float c_bar, float rew_scale, float cp,
     float cv, float ce, float eps_clip, float creg) {
@@ -50,7 +49,7 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_fwd_loss_kernel(
                       heads_b, outs, hS, cS, stash, b, S, F, D, h0s,
                       smem_raw);
   __syncthreads();  // smem_raw reused by the loss phase
-  onpolicy_loss_row(algo, outs, act, behav, rew, fir, gouts, stats_acc, mm,
+  onpolicy_loss_row(algo, outs, act, behav, rew, fir, gouts, stats_part,
                     b, B, S, D - 1, gamma, lmbda, rho_bar, rho_min, c_bar,
                     rew_scale, cp, cv, ce, eps_clip, creg, smem_raw);
 }
@@ -62,28 +61,68 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_fin_kernel(
     const float* __restrict__ body_w, const float* __restrict__ w_ih,
     const float* __restrict__ w_hh, const float* __restrict__ heads_w,
     float* __restrict__ dgates, float* __restrict__ dxb,
-    float* __restrict__ stats, float* __restrict__ stats_acc,
-    int* __restrict__ mm, int algo, int B, int S, int F, int D, long h0s,
+    float* __restrict__ stats, float* __restrict__ stats_part,
+    int algo, int B, int S, int F, int D, long h0s,
     float cp, float cv, float ce, float creg) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  if (blockIdx.x == 0 && threadIdx.x == 0) {
-    // all fwd_loss atomics are complete (stream order): finalize + re-zero
-    const int A = D - 1;
-    const float inv = 1.0f / (B * (S - 1));
-    const float p = stats_acc[0] * inv, v = stats_acc[1] * inv,
-                e = stats_acc[2] * inv;
-    stats[0] = cp * p + cv * v - ce * e + creg * stats_acc[4] * inv / A;
-    stats[1] = p;
-    stats[2] = v;
-    stats[3] = e;
-    stats[4] = stats_acc[3] * inv;
-    if (algo == kAlgoPpo) {
-      stats[5] = __int_as_float(mm[0]);
-      stats[6] = __int_as_float(mm[1]);
+  if (blockIdx.x == 0) {
+    // parallel reduce the (B, 8) per-row loss partials (all written by
+    // the fwd_loss launch — stream order) and finalize the stats vector
+    const int tid = threadIdx.x;
+    float pl = 0.f, vl = 0.f, es = 0.f, rs = 0.f, rg = 0.f;
+    float rmn = 1e30f, rmx = -1e30f;
+    for (int b = tid; b < B; b += (int)blockDim.x) {
+      const float* sp = stats_part + (long)b * 8;
+      pl += sp[0];
+      vl += sp[1];
+      es += sp[2];
+      rs += sp[3];
+      rg += sp[4];
+      if (algo == kAlgoPpo) {
+        rmn = fminf(rmn, sp[5]);
+        rmx = fmaxf(rmx, sp[6]);
+      }
     }
-    for (int i = 0; i < 5; ++i) stats_acc[i] = 0.f;
-    mm[0] = __float_as_int(1e30f);
-    mm[1] = __float_as_int(-1e30f);
+#pragma unroll
+    for (int off = kWave / 2; off > 0; off >>= 1) {
+      pl += __shfl_down(pl, off, kWave);
+      vl += __shfl_down(vl, off, kWave);
+      es += __shfl_down(es, off, kWave);
+      rs += __shfl_down(rs, off, kWave);
+      rg += __shfl_down(rg, off, kWave);
+      rmn = fminf(rmn, __shfl_down(rmn, off, kWave));
+      rmx = fmaxf(rmx, __shfl_down(rmx, off, kWave));
+    }
+    __shared__ float s74[7][8];
+    const int lane = tid & (kWave - 1), wave = tid >> 6;
+    if (lane == 0) {
+      s74[0][wave] = pl; s74[1][wave] = vl; s74[2][wave] = es;
+      s74[3][wave] = rs; s74[4][wave] = rg; s74[5][wave] = rmn;
+      s74[6][wave] = rmx;
+    }
+    __syncthreads();
+    if (tid == 0) {
+      const int nw = (int)blockDim.x / kWave;
+      float P = 0, V = 0, E = 0, R = 0, G2 = 0, MN = 1e30f, MX = -1e30f;
+      for (int w = 0; w < nw; ++w) {
+        P += s74[0][w]; V += s74[1][w]; E += s74[2][w];
+        R += s74[3][w]; G2 += s74[4][w];
+        MN = fminf(MN, s74[5][w]); MX = fmaxf(MX, s74[6][w]);
+      }
+      const int A = D - 1;
+      const float inv = 1.0f / (B * (S - 1));
+      stats[0] = cp * P * inv + cv * V * inv - ce * E * inv +
+                 creg * G2 * inv / A;
+      stats[1] = P * inv;
+      stats[2] = V * inv;
+      stats[3] = E * inv;
+      stats[4] = R * inv;
+      if (algo == kAlgoPpo) {
+        stats[5] = MN;
+        stats[6] = MX;
+      }
+    }
+    __syncthreads();  // s74 done before smem_raw phases start
   }
   seq_lstm_bwd_row<H>(gouts, nullptr, nullptr, stash, x, c0, body_w, w_ih,
                       w_hh, heads_w, nullptr, nullptr, nullptr, dgates, dxb,
@@ -99,7 +138,7 @@ void seq_lstm_fwd_loss_hip(
     const at::Tensor& heads_w, const at::Tensor& heads_b, at::Tensor& outs,
     at::Tensor& hS, at::Tensor& cS, at::Tensor& stash, const at::Tensor& act,
     const at::Tensor& behav, const at::Tensor& rew, const at::Tensor& fir,
-    at::Tensor& gouts, at::Tensor& stats_acc, at::Tensor& mm,
+    at::Tensor& gouts, at::Tensor& stats_part,
     const c10::optional<at::Tensor>& norm_sq, long algo, double gamma,
     double lmbda, double rho_bar, double rho_min, double c_bar,
     double rew_scale, double cp, double cv, double ce, double eps_clip,
@@ -120,8 +159,7 @@ void seq_lstm_fwd_loss_hip(
       outs.data_ptr<float>(), hS.data_ptr<float>(), cS.data_ptr<float>(),
       stash.data_ptr<float>(), act.data_ptr<float>(),
       behav.data_ptr<float>(), rew.data_ptr<float>(), fir.data_ptr<float>(),
-      gouts.data_ptr<float>(), stats_acc.data_ptr<float>(),
-      mm.data_ptr<int>(),
+      gouts.data_ptr<float>(), stats_part.data_ptr<float>(),
       norm_sq.has_value() ? norm_sq->data_ptr<float>() : nullptr, (int)algo,
       B, S, F, D, (long)h0.stride(0), (float)gamma, (float)lmbda,
       (float)rho_bar, (float)rho_min, (float)c_bar, (float)rew_scale,
@@ -133,8 +171,8 @@ void seq_lstm_bwd_fin_hip(
     const at::Tensor& gouts, const at::Tensor& stash, const at::Tensor& x,
     const at::Tensor& c0, const at::Tensor& body_w, const at::Tensor& w_ih,
     const at::Tensor& w_hh, const at::Tensor& heads_w, at::Tensor& dgates,
-    at::Tensor& dxb, at::Tensor& stats, at::Tensor& stats_acc,
-    at::Tensor& mm, long algo, double cp, double cv, double ce,
+    at::Tensor& dxb, at::Tensor& stats, at::Tensor& stats_part,
+    long algo, double cp, double cv, double ce,
     double creg) {
   CHECK_IN(x);
   const int B = x.size(0), S = x.size(1), F = x.size(2);
@@ -149,7 +187,7 @@ void seq_lstm_bwd_fin_hip(
       w_ih.data_ptr<float>(), w_hh.data_ptr<float>(),
       heads_w.data_ptr<float>(), dgates.data_ptr<float>(),
       dxb.data_ptr<float>(), stats.data_ptr<float>(),
-      stats_acc.data_ptr<float>(), mm.data_ptr<int>(), (int)algo, B, S, F, D,
+      stats_part.data_ptr<float>(), (int)algo, B, S, F, D,
       (long)c0.stride(0), (float)cp, (float)cv, (float)ce, (float)creg);
   HIP_CHECK_LAST();
 }

@@ -94,8 +94,8 @@ __global__ __launch_bounds__(kMsThreads) void megastep_kernel(
     float* __restrict__ outs, float* __restrict__ hS, float* __restrict__ cS,
     float* __restrict__ stash, float* __restrict__ gouts,
     float* __restrict__ dgates, float* __restrict__ dxb,
-    float* __restrict__ stats, float* __restrict__ stats_acc,
-    int* __restrict__ mm, unsigned* __restrict__ bar,
+    float* __restrict__ stats, float* __restrict__ stats_part,
+    unsigned* __restrict__ bar,
     // grad views (into flat_grad) + optimizer state
     float* __restrict__ dw_ih, float* __restrict__ dw_hh,
     float* __restrict__ dbody_w, float* __restrict__ dbody_b,
@@ -116,9 +116,6 @@ __global__ __launch_bounds__(kMsThreads) void megastep_kernel(
 
   if (b == 0 && threadIdx.x == 0) {
     *norm_sq = 0.f;
-    for (int i = 0; i < 5; ++i) stats_acc[i] = 0.f;
-    mm[0] = __float_as_int(1e30f);
-    mm[1] = __float_as_int(-1e30f);
   }
 
   // phase 1: forward, one block per batch row
@@ -135,7 +132,7 @@ __global__ __launch_bounds__(kMsThreads) void megastep_kernel(
 
   // phase 2: row-local loss + BPTT backward
   if (b < B) {
-    onpolicy_loss_row(algo, outs, act, behav, rew, fir, gouts, stats_acc, mm,
+    onpolicy_loss_row(algo, outs, act, behav, rew, fir, gouts, stats_part,
                       b, B, S, A, gamma, lmbda, rho_bar, rho_min, c_bar,
                       rew_scale, cp, cv, ce, eps_clip, creg, smem_raw);
     __syncthreads();
@@ -149,18 +146,27 @@ __global__ __launch_bounds__(kMsThreads) void megastep_kernel(
   // phase 3: weight grads (blocks re-mapped onto GEMM tiles + small grads);
   // the last block's lane 0 finalizes the loss stats concurrently
   if (b == nblocks - 1 && threadIdx.x == 0) {
+    // serial reduce of the (B,8) per-row partials (opt-in path: fine)
+    float pl = 0, vl = 0, es = 0, rs = 0, rg = 0;
+    float rmn = 1e30f, rmx = -1e30f;
+    for (int r = 0; r < B; ++r) {
+      const float* sp = stats_part + (long)r * 8;
+      pl += sp[0]; vl += sp[1]; es += sp[2]; rs += sp[3]; rg += sp[4];
+      if (algo == kAlgoPpo) {
+        rmn = fminf(rmn, sp[5]);
+        rmx = fmaxf(rmx, sp[6]);
+      }
+    }
     const float inv = 1.0f / (B * (S - 1));
-    const float p = stats_acc[0] * inv, v = stats_acc[1] * inv,
-                e = stats_acc[2] * inv;
-    stats[0] = cp * p + cv * v - ce * e + creg * stats_acc[4] * inv / A;
-    stats[1] = p;
-    stats[2] = v;
-    stats[3] = e;
-    stats[4] = stats_acc[3] * inv;
+    stats[0] = cp * pl * inv + cv * vl * inv - ce * es * inv +
+               creg * rg * inv / A;
+    stats[1] = pl * inv;
+    stats[2] = vl * inv;
+    stats[3] = es * inv;
+    stats[4] = rs * inv;
     if (algo == kAlgoPpo) {
-      stats[4] = stats_acc[3] * inv;
-      stats[5] = __int_as_float(mm[0]);
-      stats[6] = __int_as_float(mm[1]);
+      stats[5] = rmn;
+      stats[6] = rmx;
     }
   }
   {
@@ -232,7 +238,7 @@ bool megastep_onpolicy_hip(
     const at::Tensor& heads_w, const at::Tensor& heads_b, at::Tensor& outs,
     at::Tensor& hS, at::Tensor& cS, at::Tensor& stash, at::Tensor& gouts,
     at::Tensor& dgates, at::Tensor& dxb, at::Tensor& stats,
-    at::Tensor& stats_acc, at::Tensor& mm, at::Tensor& bar,
+    at::Tensor& stats_part, at::Tensor& bar,
     at::Tensor& dw_ih, at::Tensor& dw_hh, at::Tensor& dbody_w,
     at::Tensor& dbody_b, at::Tensor& db_g, at::Tensor& dheads_w,
     at::Tensor& dheads_b, at::Tensor& norm_sq, at::Tensor& flat_param,
@@ -282,8 +288,8 @@ bool megastep_onpolicy_hip(
       outs.data_ptr<float>(), hS.data_ptr<float>(), cS.data_ptr<float>(),
       stash.data_ptr<float>(), gouts.data_ptr<float>(),
       dgates.data_ptr<float>(), dxb.data_ptr<float>(),
-      stats.data_ptr<float>(), stats_acc.data_ptr<float>(),
-      mm.data_ptr<int>(), (unsigned*)bar.data_ptr<int>(),
+      stats.data_ptr<float>(), stats_part.data_ptr<float>(),
+      (unsigned*)bar.data_ptr<int>(),
       dw_ih.data_ptr<float>(), dw_hh.data_ptr<float>(),
       dbody_w.data_ptr<float>(), dbody_b.data_ptr<float>(),
       db_g.data_ptr<float>(), dheads_w.data_ptr<float>(),

@@ -247,8 +247,7 @@ class FusedOnPolicyStep(GraphableStep):
             "outs": mk(B, S, D), "hS": mk(B, H), "cS": mk(B, H),
             "stash": mk(B, S, 7 * H), "gouts": mk(B, S, D),
             "dgates": mk(B, S, 4 * H), "dxb": mk(B, S, H),
-            "stats_acc": torch.zeros(8, dtype=torch.float32, device=dev),
-            "mm": torch.zeros(2, dtype=torch.int32, device=dev),
+            "stats_part": mk(B, 8),
             "bar": torch.zeros(1024, dtype=torch.int32, device=dev),
         }
         self._mega_shape = (B, S)
@@ -286,8 +285,8 @@ class FusedOnPolicyStep(GraphableStep):
             x, hx0, cx0, act, behav, rew, fir,
             c.body_w, c.body_b, c.w_ih, c.w_hh, c.b_g, c.heads_w, c.heads_b,
             ws["outs"], ws["hS"], ws["cS"], ws["stash"], ws["gouts"],
-            ws["dgates"], ws["dxb"], self.stats_buf, ws["stats_acc"],
-            ws["mm"], ws["bar"],
+            ws["dgates"], ws["dxb"], self.stats_buf, ws["stats_part"],
+            ws["bar"],
             gv[0], gv[1], gv[2], gv[3], gv[4], gv[5], gv[6], opt.norm_sq,
             opt.flat_param, opt.flat_grad, opt.sq_avg,
             0 if self.algo == "IMPALA" else 1,
@@ -379,15 +378,11 @@ class FusedOnPolicyStep(GraphableStep):
         def mk(*shape):
             return torch.empty(*shape, device=dev)
 
-        mm = torch.empty(2, dtype=torch.int32, device=dev)
-        mm[0] = torch.tensor(1e30).view(torch.int32)
-        mm[1] = torch.tensor(-1e30).view(torch.int32)
         self._fl = {
             "outs": mk(B, S, D), "hS": mk(B, H), "cS": mk(B, H),
             "stash": mk(B, S, 7 * H), "gouts": mk(B, S, D),
             "dgates": mk(B, S, 4 * H), "dxb": mk(B, S, H),
-            "stats_acc": torch.zeros(8, dtype=torch.float32, device=dev),
-            "mm": mm,
+            "stats_part": mk(B, 8),  # per-row loss partials (plain stores)
         }
         self._fl_shape = (B, S)
 
@@ -410,8 +405,8 @@ class FusedOnPolicyStep(GraphableStep):
         e.seq_lstm_fwd_loss(
             x, hx0, cx0, c.body_w, c.body_b, c.w_ih, c.w_hh, c.b_g,
             c.heads_w, c.heads_b, ws["outs"], ws["hS"], ws["cS"],
-            ws["stash"], act, behav, rew, fir, ws["gouts"], ws["stats_acc"],
-            ws["mm"], norm, algo_i, p.gamma, p.lmbda, 0.8, 0.1, 1.0,
+            ws["stash"], act, behav, rew, fir, ws["gouts"], ws["stats_part"],
+            norm, algo_i, p.gamma, p.lmbda, 0.8, 0.1, 1.0,
             p.reward_scale, p.policy_loss_coef, p.value_loss_coef,
             p.entropy_coef, p.eps_clip,
             float(getattr(p, "logit_reg", 0.0)),
@@ -419,7 +414,7 @@ class FusedOnPolicyStep(GraphableStep):
         e.seq_lstm_bwd_fin(
             ws["gouts"], ws["stash"], x, cx0, c.body_w, c.w_ih, c.w_hh,
             c.heads_w, ws["dgates"], ws["dxb"], self.stats_buf,
-            ws["stats_acc"], ws["mm"], algo_i, p.policy_loss_coef,
+            ws["stats_part"], algo_i, p.policy_loss_coef,
             p.value_loss_coef, p.entropy_coef,
             float(getattr(p, "logit_reg", 0.0)),
         )
