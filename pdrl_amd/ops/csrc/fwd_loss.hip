@@ -49,9 +49,15 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_fwd_loss_kernel(
                       heads_b, outs, hS, cS, stash, b, S, F, D, h0s,
                       smem_raw);
   __syncthreads();  // smem_raw reused by the loss phase
-  onpolicy_loss_row(algo, outs, act, behav, rew, fir, gouts, stats_part,
-                    b, B, S, D - 1, gamma, lmbda, rho_bar, rho_min, c_bar,
-                    rew_scale, cp, cv, ce, eps_clip, creg, smem_raw);
+  if (algo == 2) {  // PPO-Continuous (Gaussian tanh-mean policy)
+    ppoc_loss_row(outs, act, behav, rew, fir, gouts, stats_part, b, B, S,
+                  (D - 1) / 2, gamma, lmbda, rew_scale, cp, cv, ce,
+                  eps_clip, creg, smem_raw);
+  } else {
+    onpolicy_loss_row(algo, outs, act, behav, rew, fir, gouts, stats_part,
+                      b, B, S, D - 1, gamma, lmbda, rho_bar, rho_min, c_bar,
+                      rew_scale, cp, cv, ce, eps_clip, creg, smem_raw);
+  }
 }
 
 template <int H>
@@ -78,7 +84,7 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_fin_kernel(
       es += sp[2];
       rs += sp[3];
       rg += sp[4];
-      if (algo == kAlgoPpo) {
+      if (algo != kAlgoImpala) {  // PPO and PPO-C track ratio min/max
         rmn = fminf(rmn, sp[5]);
         rmx = fmaxf(rmx, sp[6]);
       }
@@ -109,15 +115,15 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_fin_kernel(
         R += s74[3][w]; G2 += s74[4][w];
         MN = fminf(MN, s74[5][w]); MX = fmaxf(MX, s74[6][w]);
       }
-      const int A = D - 1;
+      const int nlog = D - 1;  // regularized heads: A, or 2A for PPO-C
       const float inv = 1.0f / (B * (S - 1));
       stats[0] = cp * P * inv + cv * V * inv - ce * E * inv +
-                 creg * G2 * inv / A;
+                 creg * G2 * inv / nlog;
       stats[1] = P * inv;
       stats[2] = V * inv;
       stats[3] = E * inv;
       stats[4] = R * inv;
-      if (algo == kAlgoPpo) {
+      if (algo != kAlgoImpala) {
         stats[5] = MN;
         stats[6] = MX;
       }

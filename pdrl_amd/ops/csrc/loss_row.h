@@ -177,3 +177,125 @@ __device__ void onpolicy_loss_row(
   }
 }
 
+
+// Row-local PPO-Continuous loss (Gaussian tanh-mean policy; reference math:
+// networks/models.py:103-118 + ppo/learning.py — identical numerics to
+// ppoc_loss.hip's mega kernel). Packed heads [mu | std | value], D = 2A+1.
+// Same per-row stats_part protocol as onpolicy_loss_row.
+__device__ inline void ppoc_loss_row(
+    const float* __restrict__ outs, const float* __restrict__ act,
+    const float* __restrict__ behav, const float* __restrict__ rew,
+    const float* __restrict__ fir, float* __restrict__ gouts,
+    float* __restrict__ stats_part, int b, int B, int S, int A, float gamma,
+    float lmbda, float rew_scale, float cp, float cv, float ce,
+    float eps_clip, float creg, char* smem) {
+  constexpr float kRowLogSqrt2Pi = 0.9189385332046727f;
+  constexpr float kRowEntConst = 1.4189385332046727f;  // 0.5*(1+log(2*pi))
+  const int D = 2 * A + 1;
+  const int T = S - 1;
+  const int BT = B * T;
+  const int tid = threadIdx.x;
+  float* s_logp = reinterpret_cast<float*>(smem);  // (S)
+  float* s_ent = s_logp + S;                       // (S)
+  float* s_adv = s_ent + S;                        // (T)
+  float* s_ret = s_adv + S;                        // (T)
+  const long sb = (long)b * S;
+
+  if (tid < S) {
+    const float* z = outs + (sb + tid) * D;
+    float lp = 0.f, h = 0.f;
+    for (int j = 0; j < A; ++j) {
+      const float mut = tanhf(z[j]);
+      const float sig = ((z[A + j] > 20.f) ? z[A + j]
+                                           : log1pf(__expf(z[A + j]))) + 1e-4f;
+      const float x = act[(sb + tid) * A + j];
+      const float d = (x - mut) / sig;
+      lp += -0.5f * d * d - __logf(sig) - kRowLogSqrt2Pi;
+      h += __logf(sig) + kRowEntConst;
+    }
+    s_logp[tid] = lp;
+    s_ent[tid] = h;
+  }
+  __syncthreads();
+
+  if (tid == 0) {
+    float run = 0.f;
+    for (int t = T - 1; t >= 0; --t) {
+      const float mask = 1.f - fir[sb + t + 1];
+      const float tdv = rew[sb + t] * rew_scale +
+                        gamma * mask * outs[(sb + t + 1) * D + 2 * A];
+      const float delta = tdv - outs[(sb + t) * D + 2 * A];
+      run = fmaf(gamma * lmbda * mask, run, delta);
+      s_ret[t] = tdv;
+      s_adv[t] = run;
+    }
+    float pl = 0.f, vl = 0.f, es = 0.f, rs = 0.f, rg = 0.f;
+    float rmin = 1e30f, rmax = -1e30f;
+    for (int t = 0; t < T; ++t) {
+      const float r = __expf(s_logp[t] - behav[sb + t]);
+      const float a = s_adv[t];
+      const float s1 = r * a;
+      const float s2 = fminf(fmaxf(r, 1.f - eps_clip), 1.f + eps_clip) * a;
+      pl -= fminf(s1, s2);
+      vl += ms_huber(outs[(sb + t) * D + 2 * A] - s_ret[t]);
+      es += s_ent[t];
+      rs += r;
+      rmin = fminf(rmin, r);
+      rmax = fmaxf(rmax, r);
+      for (int j = 0; j < 2 * A; ++j) {
+        const float z = outs[(sb + t) * D + j];
+        rg = fmaf(z, z, rg);
+      }
+    }
+    float* sp = stats_part + (long)b * 8;
+    sp[0] = pl;
+    sp[1] = vl;
+    sp[2] = es;
+    sp[3] = rs;
+    sp[4] = rg;
+    sp[5] = rmin;
+    sp[6] = rmax;
+  }
+  __syncthreads();
+
+  const float invN = 1.0f / BT;
+  const float dreg = 2.0f * creg * invN / (2 * A);
+  for (int idx = tid; idx < S * D; idx += (int)blockDim.x) {
+    const int t = idx / D, j = idx % D;
+    float* g = gouts + (sb + t) * D;
+    if (t >= T) {
+      g[j] = 0.f;
+      continue;
+    }
+    const float a_v = s_adv[t];
+    const float r = __expf(s_logp[t] - behav[sb + t]);
+    const bool inside = (r > 1.f - eps_clip) && (r < 1.f + eps_clip);
+    const float s1 = r * a_v;
+    const float s2 = fminf(fmaxf(r, 1.f - eps_clip), 1.f + eps_clip) * a_v;
+    const float gr = (inside || s1 < s2) ? a_v * r : 0.f;
+    const float dlogp = -cp * gr * invN;
+    const float dH = -ce * invN;
+    const float* z = outs + (sb + t) * D;
+    if (j == 2 * A) {
+      g[j] = cv * ms_huber_grad(z[2 * A] - s_ret[t]) * invN;
+    } else if (j < A) {
+      const float mut = tanhf(z[j]);
+      const float sig = ((z[A + j] > 20.f) ? z[A + j]
+                                           : log1pf(__expf(z[A + j]))) + 1e-4f;
+      const float x = act[(sb + t) * A + j];
+      const float diff = x - mut;
+      const float dl_dmut = diff / (sig * sig);
+      g[j] = dlogp * dl_dmut * (1.f - mut * mut) + dreg * z[j];
+    } else {
+      const int jj = j - A;
+      const float mut = tanhf(z[jj]);
+      const float sig = ((z[j] > 20.f) ? z[j] : log1pf(__expf(z[j]))) + 1e-4f;
+      const float x = act[(sb + t) * A + jj];
+      const float diff = x - mut;
+      const float dl_dsig = diff * diff / (sig * sig * sig) - 1.0f / sig;
+      const float dH_dsig = 1.0f / sig;
+      const float sgm = 1.0f / (1.0f + __expf(-z[j]));
+      g[j] = (dlogp * dl_dsig + dH * dH_dsig) * sgm + dreg * z[j];
+    }
+  }
+}

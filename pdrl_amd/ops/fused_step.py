@@ -128,12 +128,18 @@ class FusedOnPolicyStep(GraphableStep):
         return [gs[2], gs[3], gs[0], gs[1], gs[4], gs[5], gs[6]]
 
     def fits(self, batch) -> bool:
-        """Whether the single-launch loss kernel covers this shape (V-MPO
-        has no multi-kernel fallback; IMPALA/PPO do)."""
+        """Whether a fused loss path covers this shape (V-MPO's
+        single-launch kernel has an LDS cap and no multi-kernel fallback;
+        IMPALA/PPO/PPO-C have shape-unlimited paths)."""
+        import os
+
         B, S, _ = batch["obs"].shape
         if self.algo == "V-MPO":
             return (2 * B * S + 3 * B * (S - 1)) * 4 <= 56 * 1024
         if self.algo == "PPO-C":
+            if self.core.w_ih.size(0) == 64 and bool(
+                    int(os.environ.get("PDRL_FWDLOSS", "1"))):
+                return True  # row-local fwd+loss: no LDS shape limit
             return (2 * B * S + 2 * B * (S - 1)) * 4 <= 56 * 1024
         return True
 
@@ -390,7 +396,7 @@ class FusedOnPolicyStep(GraphableStep):
                      B, S, A, p, update) -> bool:
         import os
 
-        if self.algo not in ("IMPALA", "PPO"):
+        if self.algo not in ("IMPALA", "PPO", "PPO-C"):
             return False
         if self.core.w_ih.size(0) != 64:
             return False  # kernels specialized for the H=64 model family
@@ -401,7 +407,7 @@ class FusedOnPolicyStep(GraphableStep):
         ws = self._fl
         c = self.core
         norm = self._norm_buf()
-        algo_i = 0 if self.algo == "IMPALA" else 1
+        algo_i = {"IMPALA": 0, "PPO": 1, "PPO-C": 2}[self.algo]
         e.seq_lstm_fwd_loss(
             x, hx0, cx0, c.body_w, c.body_b, c.w_ih, c.w_hh, c.b_g,
             c.heads_w, c.heads_b, ws["outs"], ws["hS"], ws["cS"],
