@@ -55,13 +55,24 @@ void seq_lstm_fwd_loss_hip(
     const at::Tensor&, const at::Tensor&, const at::Tensor&,
     const at::Tensor&, at::Tensor&, at::Tensor&,
     const c10::optional<at::Tensor>&, long, double, double, double, double,
-    double, double, double, double, double, double, double);
+    double, double, double, double, double, double, double,
+    const c10::optional<at::Tensor>&, const c10::optional<at::Tensor>&,
+    const c10::optional<at::Tensor>&, const c10::optional<at::Tensor>&);
 void seq_lstm_bwd_fin_hip(
     const at::Tensor&, const at::Tensor&, const at::Tensor&,
     const at::Tensor&, const at::Tensor&, const at::Tensor&,
     const at::Tensor&, const at::Tensor&, at::Tensor&, at::Tensor&,
     at::Tensor&, at::Tensor&, long, double, double, double,
-    double);
+    double, const c10::optional<at::Tensor>&,
+    const c10::optional<at::Tensor>&, const c10::optional<at::Tensor>&,
+    const c10::optional<at::Tensor>&, const c10::optional<at::Tensor>&,
+    const c10::optional<at::Tensor>&, const c10::optional<at::Tensor>&);
+bool vmpo_mid_hip(const at::Tensor&, const at::Tensor&, const at::Tensor&,
+                  const at::Tensor&, const at::Tensor&, const at::Tensor&,
+                  const at::Tensor&, const at::Tensor&, at::Tensor&,
+                  at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
+                  const c10::optional<at::Tensor>&, at::Tensor&, long,
+                  double, double, double, double, double, double);
 at::Tensor gae_hip(const at::Tensor&, double, double, const at::Tensor&);
 std::vector<at::Tensor> vtrace_hip(const at::Tensor&, const at::Tensor&,
                                    const at::Tensor&, const at::Tensor&,
@@ -206,9 +217,34 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("barrier_bench", &barrier_bench_hip,
         "grid-barrier microbenchmark (N back-to-back barriers)");
   m.def("seq_lstm_fwd_loss", &seq_lstm_fwd_loss_hip,
-        "forward + row-local IMPALA/PPO loss in ONE launch");
+        "forward + row-local loss (IMPALA/PPO/PPO-C, V-MPO pre) in ONE "
+        "launch",
+        py::arg("x"), py::arg("h0"), py::arg("c0"), py::arg("body_w"),
+        py::arg("body_b"), py::arg("w_ih"), py::arg("w_hh"), py::arg("b_g"),
+        py::arg("heads_w"), py::arg("heads_b"), py::arg("outs"),
+        py::arg("hS"), py::arg("cS"), py::arg("stash"), py::arg("act"),
+        py::arg("behav"), py::arg("rew"), py::arg("fir"), py::arg("gouts"),
+        py::arg("stats_part"), py::arg("norm_sq"), py::arg("algo"),
+        py::arg("gamma"), py::arg("lmbda"), py::arg("rho_bar"),
+        py::arg("rho_min"), py::arg("c_bar"), py::arg("rew_scale"),
+        py::arg("cp"), py::arg("cv"), py::arg("ce"), py::arg("eps_clip"),
+        py::arg("creg"), py::arg("vm_lse") = c10::nullopt,
+        py::arg("vm_logp") = c10::nullopt, py::arg("vm_adv") = c10::nullopt,
+        py::arg("vm_td") = c10::nullopt);
   m.def("seq_lstm_bwd_fin", &seq_lstm_bwd_fin_hip,
-        "BPTT backward + loss-stat finalization in ONE launch");
+        "BPTT backward + stat finalization (or V-MPO grad emission)",
+        py::arg("gouts"), py::arg("stash"), py::arg("x"), py::arg("c0"),
+        py::arg("body_w"), py::arg("w_ih"), py::arg("w_hh"),
+        py::arg("heads_w"), py::arg("dgates"), py::arg("dxb"),
+        py::arg("stats"), py::arg("stats_part"), py::arg("algo"),
+        py::arg("cp"), py::arg("cv"), py::arg("ce"), py::arg("creg"),
+        py::arg("act") = c10::nullopt, py::arg("behav") = c10::nullopt,
+        py::arg("vm_lse") = c10::nullopt, py::arg("vm_psi") = c10::nullopt,
+        py::arg("vm_td") = c10::nullopt,
+        py::arg("vm_scalars") = c10::nullopt,
+        py::arg("vm_outs") = c10::nullopt);
+  m.def("vmpo_mid", &vmpo_mid_hip,
+        "V-MPO cross-row middle kernel: selection + psi + duals + stats");
   m.def("gae", &gae_hip, "GAE reverse scan");
   m.def("vtrace", &vtrace_hip, "fused V-trace scan",
         pybind11::arg("behav_lp"), pybind11::arg("target_lp"),

@@ -37,6 +37,8 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_fwd_loss_kernel(
     const float* __restrict__ rew, const float* __restrict__ fir,
     float* __restrict__ gouts, float* __restrict__ stats_part,
     float* __restrict__ norm_sq,
+    float* __restrict__ vm_lse, float* __restrict__ vm_logp,
+    float* __restrict__ vm_adv, float* __restrict__ vm_td,
     int algo, int B, int S, int F, int D, long h0s, float gamma, float lmbda,
     float rho_bar, float rho_min, float c_bar, float rew_scale, float cp,
     float cv, float ce, float eps_clip, float creg) {
@@ -49,7 +51,10 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_fwd_loss_kernel(
                       heads_b, outs, hS, cS, stash, b, S, F, D, h0s,
                       smem_raw);
   __syncthreads();  // smem_raw reused by the loss phase
-  if (algo == 2) {  // PPO-Continuous (Gaussian tanh-mean policy)
+  if (algo == 3) {  // V-MPO pre-phases: log-softmax + GAE to global scratch
+    vmpo_pre_row(outs, act, rew, fir, vm_lse, vm_logp, vm_adv, vm_td, b, S,
+                 D - 1, gamma, lmbda, rew_scale);
+  } else if (algo == 2) {  // PPO-Continuous (Gaussian tanh-mean policy)
     ppoc_loss_row(outs, act, behav, rew, fir, gouts, stats_part, b, B, S,
                   (D - 1) / 2, gamma, lmbda, rew_scale, cp, cv, ce,
                   eps_clip, creg, smem_raw);
@@ -68,10 +73,22 @@ __global__ __launch_bounds__(4 * H) void seq_lstm_bwd_fin_kernel(
     const float* __restrict__ w_hh, const float* __restrict__ heads_w,
     float* __restrict__ dgates, float* __restrict__ dxb,
     float* __restrict__ stats, float* __restrict__ stats_part,
+    const float* __restrict__ act, const float* __restrict__ behav,
+    const float* __restrict__ vm_lse, const float* __restrict__ vm_psi,
+    const float* __restrict__ vm_td, const float* __restrict__ vm_scalars,
     int algo, int B, int S, int F, int D, long h0s,
-    float cp, float cv, float ce, float creg) {
+    float cp, float cv, float ce, float creg,
+    const float* __restrict__ vm_outs) {
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  if (blockIdx.x == 0) {
+  if (algo == 3) {
+    // V-MPO: emit this row's packed head grads from the middle kernel's
+    // psi/scalars (stats already finalized there), then backward
+    vmpo_grad_row(vm_outs, act, behav, vm_lse, vm_psi, vm_td, vm_scalars,
+                  // gouts is written then consumed by bwd_row below
+                  const_cast<float*>(gouts), blockIdx.x, B, S, D - 1, cp,
+                  cv, creg);
+    __syncthreads();
+  } else if (blockIdx.x == 0) {
     // parallel reduce the (B, 8) per-row loss partials (all written by
     // the fwd_loss launch — stream order) and finalize the stats vector
     const int tid = threadIdx.x;
@@ -148,7 +165,10 @@ void seq_lstm_fwd_loss_hip(
     const c10::optional<at::Tensor>& norm_sq, long algo, double gamma,
     double lmbda, double rho_bar, double rho_min, double c_bar,
     double rew_scale, double cp, double cv, double ce, double eps_clip,
-    double creg) {
+    double creg, const c10::optional<at::Tensor>& vm_lse,
+    const c10::optional<at::Tensor>& vm_logp,
+    const c10::optional<at::Tensor>& vm_adv,
+    const c10::optional<at::Tensor>& vm_td) {
   CHECK_IN(x);
   const int B = x.size(0), S = x.size(1), F = x.size(2);
   const int H = h0.size(1), D = heads_w.size(1);
@@ -166,7 +186,11 @@ void seq_lstm_fwd_loss_hip(
       stash.data_ptr<float>(), act.data_ptr<float>(),
       behav.data_ptr<float>(), rew.data_ptr<float>(), fir.data_ptr<float>(),
       gouts.data_ptr<float>(), stats_part.data_ptr<float>(),
-      norm_sq.has_value() ? norm_sq->data_ptr<float>() : nullptr, (int)algo,
+      norm_sq.has_value() ? norm_sq->data_ptr<float>() : nullptr,
+      vm_lse.has_value() ? vm_lse->data_ptr<float>() : nullptr,
+      vm_logp.has_value() ? vm_logp->data_ptr<float>() : nullptr,
+      vm_adv.has_value() ? vm_adv->data_ptr<float>() : nullptr,
+      vm_td.has_value() ? vm_td->data_ptr<float>() : nullptr, (int)algo,
       B, S, F, D, (long)h0.stride(0), (float)gamma, (float)lmbda,
       (float)rho_bar, (float)rho_min, (float)c_bar, (float)rew_scale,
       (float)cp, (float)cv, (float)ce, (float)eps_clip, (float)creg);
@@ -179,7 +203,13 @@ void seq_lstm_bwd_fin_hip(
     const at::Tensor& w_hh, const at::Tensor& heads_w, at::Tensor& dgates,
     at::Tensor& dxb, at::Tensor& stats, at::Tensor& stats_part,
     long algo, double cp, double cv, double ce,
-    double creg) {
+    double creg, const c10::optional<at::Tensor>& act,
+    const c10::optional<at::Tensor>& behav,
+    const c10::optional<at::Tensor>& vm_lse,
+    const c10::optional<at::Tensor>& vm_psi,
+    const c10::optional<at::Tensor>& vm_td,
+    const c10::optional<at::Tensor>& vm_scalars,
+    const c10::optional<at::Tensor>& vm_outs) {
   CHECK_IN(x);
   const int B = x.size(0), S = x.size(1), F = x.size(2);
   const int H = c0.size(1), D = heads_w.size(1);
@@ -193,7 +223,15 @@ void seq_lstm_bwd_fin_hip(
       w_ih.data_ptr<float>(), w_hh.data_ptr<float>(),
       heads_w.data_ptr<float>(), dgates.data_ptr<float>(),
       dxb.data_ptr<float>(), stats.data_ptr<float>(),
-      stats_part.data_ptr<float>(), (int)algo, B, S, F, D,
-      (long)c0.stride(0), (float)cp, (float)cv, (float)ce, (float)creg);
+      stats_part.data_ptr<float>(),
+      act.has_value() ? act->data_ptr<float>() : nullptr,
+      behav.has_value() ? behav->data_ptr<float>() : nullptr,
+      vm_lse.has_value() ? vm_lse->data_ptr<float>() : nullptr,
+      vm_psi.has_value() ? vm_psi->data_ptr<float>() : nullptr,
+      vm_td.has_value() ? vm_td->data_ptr<float>() : nullptr,
+      vm_scalars.has_value() ? vm_scalars->data_ptr<float>() : nullptr,
+      (int)algo, B, S, F, D,
+      (long)c0.stride(0), (float)cp, (float)cv, (float)ce, (float)creg,
+      vm_outs.has_value() ? vm_outs->data_ptr<float>() : nullptr);
   HIP_CHECK_LAST();
 }

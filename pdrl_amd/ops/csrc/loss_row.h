@@ -299,3 +299,91 @@ __device__ inline void ppoc_loss_row(
     }
   }
 }
+
+// ---- V-MPO row-local phases (reference math: v_mpo/learning.py:49-124,
+// identical numerics to vmpo_loss.hip's mega kernel, which keeps only the
+// cross-row phases: top-half selection, psi softmax, dual losses, stats).
+
+// phases A+B for one row: log-softmax stats and the GAE scan, written to
+// GLOBAL scratch consumed by the single-block middle kernel.
+__device__ inline void vmpo_pre_row(
+    const float* __restrict__ outs, const float* __restrict__ act,
+    const float* __restrict__ rew, const float* __restrict__ fir,
+    float* __restrict__ lse_g, float* __restrict__ logp_g,
+    float* __restrict__ adv_g, float* __restrict__ td_g, int b, int S,
+    int A, float gamma, float lmbda, float rew_scale) {
+  const int D = A + 1;
+  const int T = S - 1;
+  const int tid = threadIdx.x;
+  const long sb = (long)b * S, tb = (long)b * T;
+
+  if (tid < S) {
+    const float* z = outs + (sb + tid) * D;
+    float mx = z[0];
+    for (int j = 1; j < A; ++j) mx = fmaxf(mx, z[j]);
+    float sum = 0.f;
+    for (int j = 0; j < A; ++j) sum += __expf(z[j] - mx);
+    const float l = mx + __logf(sum);
+    lse_g[sb + tid] = l;
+    logp_g[sb + tid] = z[(int)act[sb + tid]] - l;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    float run = 0.f;
+    for (int t = T - 1; t >= 0; --t) {
+      const float mask = 1.f - fir[sb + t + 1];
+      const float tdv = rew[sb + t] * rew_scale +
+                        gamma * mask * outs[(sb + t + 1) * D + A];
+      const float delta = tdv - outs[(sb + t) * D + A];
+      run = fmaf(gamma * lmbda * mask, run, delta);
+      td_g[tb + t] = tdv;
+      adv_g[tb + t] = run;
+    }
+  }
+}
+
+// phase E for one row: analytic packed head grads from the middle kernel's
+// psi weights and scalars ([7] = alpha).
+__device__ inline void vmpo_grad_row(
+    const float* __restrict__ outs, const float* __restrict__ act,
+    const float* __restrict__ behav,  // (N,A) behaviour LOGITS
+    const float* __restrict__ lse_g, const float* __restrict__ psi_g,
+    const float* __restrict__ td_g, const float* __restrict__ scalars_g,
+    float* __restrict__ gouts, int b, int B, int S, int A, float cp,
+    float cv, float creg) {
+  const int D = A + 1;
+  const int T = S - 1;
+  const int BT = B * T;
+  const int tid = threadIdx.x;
+  const long sb = (long)b * S, tb = (long)b * T;
+  const float alpha_v = scalars_g[7];
+  const float invBT = 1.0f / BT;
+  const float dreg = 2.0f * creg * invBT / A;
+
+  for (int idx = tid; idx < S * D; idx += (int)blockDim.x) {
+    const int t = idx / D, j = idx % D;
+    float* g = gouts + (sb + t) * D;
+    if (t >= T) {
+      g[j] = 0.f;
+      continue;
+    }
+    const float* zq = outs + (sb + t) * D;
+    if (j == A) {
+      const float dv = fminf(fmaxf(zq[A] - td_g[tb + t], -1.0f), 1.0f);
+      g[A] = cv * dv * invBT;
+      continue;
+    }
+    const float dlogp = -cp * psi_g[tb + t];  // zero for unselected
+    const float* zb = behav + (sb + t) * A;
+    float mb = zb[0];
+    for (int k = 1; k < A; ++k) mb = fmaxf(mb, zb[k]);
+    float sb_ = 0.f;
+    for (int k = 0; k < A; ++k) sb_ += __expf(zb[k] - mb);
+    const float lb = mb + __logf(sb_);
+    const int a = (int)act[sb + t];
+    const float q = __expf(zq[j] - lse_g[sb + t]);
+    const float pb = __expf(zb[j] - lb);
+    g[j] = dlogp * ((j == a ? 1.f : 0.f) - q) + alpha_v * (q - pb) * invBT +
+           dreg * zq[j];
+  }
+}

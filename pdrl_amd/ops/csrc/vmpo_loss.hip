@@ -449,3 +449,263 @@ bool vmpo_loss_mega_hip(const at::Tensor& mo, const at::Tensor& act,
   HIP_CHECK_LAST();
   return true;
 }
+
+// ---- split-phase V-MPO: the row-local phases (log-softmax+GAE, grads)
+// ride the fwd/bwd launches (loss_row.h); this SLIM single-block kernel
+// keeps only the genuinely cross-row work — top-half selection, psi
+// softmax, dual losses, stats — on global scratch. Smaller code also
+// matters per se: the fat mega kernel was instruction-fetch sensitive.
+namespace {
+
+__global__ __launch_bounds__(kThreads) void vmpo_mid_kernel(
+    const float* __restrict__ mo,      // (N,D)
+    const float* __restrict__ behav,   // (N,A) behaviour logits
+    const float* __restrict__ logp_g,  // (N)
+    const float* __restrict__ lse_g,   // (N)
+    const float* __restrict__ adv_g,   // (BT)
+    const float* __restrict__ td_g,    // (BT)
+    const float* __restrict__ log_eta_p, const float* __restrict__ log_alpha_p,
+    float* __restrict__ psi_g,         // (BT) out
+    float* __restrict__ scalars_g,     // (8) out: [7] = alpha
+    float* __restrict__ g_eta, float* __restrict__ g_alpha,
+    float* __restrict__ stats, float* __restrict__ norm_sq,
+    unsigned* __restrict__ rng_state, int B, int S, int A, float cp,
+    float cv, float creg, float eps_eta, float alpha_below,
+    float alpha_upper) {
+  const int D = A + 1;
+  const int T = S - 1;
+  const int BT = B * T;
+  const int tid = threadIdx.x;
+
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  float* s_adv = reinterpret_cast<float*>(smem_raw);  // (BT)
+  float* s_psi = s_adv + BT;                          // (BT)
+  __shared__ float s4[kWaves];
+  __shared__ float s4b[3][kWaves];
+  __shared__ float s_scalars[8];
+
+  if (tid == 0) {
+    unsigned st = *rng_state * 1664525u + 1013904223u;
+    *rng_state = st;
+    const float u = (st >> 8) * (1.0f / 16777216.0f);
+    s_scalars[0] = alpha_below + (alpha_upper - alpha_below) * u;
+    s_scalars[6] = __expf(*log_eta_p);
+    s_scalars[7] = __expf(*log_alpha_p);
+  }
+  for (int i = tid; i < BT; i += kThreads) s_adv[i] = adv_g[i];
+  __syncthreads();
+
+  const int K = BT / 2 > 0 ? BT / 2 : 1;
+  unsigned thresh_u;
+  int n_gt;
+  {
+    __shared__ float s_hist[kWaves][256];
+    __shared__ unsigned s_sel[4];
+    unsigned prefix = 0;
+    int above = 0;
+    const int mywave = tid >> 6;
+    for (int shift = 24; shift >= 0; shift -= 8) {
+      const unsigned pmask = (shift == 24) ? 0u : (0xFFFFFFFFu << (shift + 8));
+#pragma unroll
+      for (int w = 0; w < kWaves; ++w) s_hist[w][tid] = 0.f;
+      __syncthreads();
+      for (int i = tid; i < BT; i += kThreads) {
+        const unsigned u = f2u(s_adv[i]);
+        if ((u & pmask) == prefix && u != 0u) {
+          atomicAdd(&s_hist[mywave][(u >> shift) & 255u], 1.0f);
+        }
+      }
+      __syncthreads();
+      float mine = 0.f;
+#pragma unroll
+      for (int w = 0; w < kWaves; ++w) mine += s_hist[w][tid];
+      float suf = mine;
+#pragma unroll
+      for (int off = 1; off < kWave; off <<= 1) {
+        const float up = __shfl_down(suf, off, kWave);
+        if ((tid & (kWave - 1)) + off < kWave) suf += up;
+      }
+      if ((tid & (kWave - 1)) == 0) s4[tid >> 6] = suf;
+      __syncthreads();
+      float higher = 0.f;
+#pragma unroll
+      for (int w = 0; w < kWaves; ++w)
+        if (w > (tid >> 6)) higher += s4[w];
+      const int suf_ge = (int)(suf + higher);
+      const int suf_gt = suf_ge - (int)mine;
+      if (above + suf_gt < K && K <= above + suf_ge) {
+        s_sel[0] = prefix | ((unsigned)tid << shift);
+        s_sel[1] = (unsigned)(above + suf_gt);
+      }
+      __syncthreads();
+      prefix = s_sel[0];
+      above = (int)s_sel[1];
+      __syncthreads();
+    }
+    thresh_u = prefix;
+    n_gt = above;
+  }
+
+  {
+    const int need_eq = K - n_gt;
+    for (int i = tid; i < BT; i += kThreads) {
+      s_psi[i] = (f2u(s_adv[i]) > thresh_u) ? 1.f : 0.f;
+    }
+    __syncthreads();
+    if (need_eq > 0) {
+      const int chunk = (BT + kThreads - 1) / kThreads;
+      const int lo = tid * chunk;
+      const int hi = min(lo + chunk, BT);
+      int cnt = 0;
+      for (int i = lo; i < hi; ++i) cnt += (f2u(s_adv[i]) == thresh_u);
+      float inc = (float)cnt;
+#pragma unroll
+      for (int off = 1; off < kWave; off <<= 1) {
+        const float dn = __shfl_up(inc, off, kWave);
+        if ((tid & (kWave - 1)) >= off) inc += dn;
+      }
+      if ((tid & (kWave - 1)) == kWave - 1) s4[tid >> 6] = inc;
+      __syncthreads();
+      float base = 0.f;
+      for (int w = 0; w < (tid >> 6); ++w) base += s4[w];
+      int rank = (int)(base + inc) - cnt;
+      for (int i = lo; i < hi && rank < need_eq; ++i) {
+        if (f2u(s_adv[i]) == thresh_u) {
+          s_psi[i] = 1.f;
+          ++rank;
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  {
+    float mx = -1e30f;
+    for (int i = tid; i < BT; i += kThreads)
+      if (s_psi[i] > 0.f) mx = fmaxf(mx, s_adv[i]);
+    const float m = block_max(mx, s4);
+    const float eta = s_scalars[6];
+    float z = 0.f, wa = 0.f;
+    for (int i = tid; i < BT; i += kThreads) {
+      if (s_psi[i] > 0.f) {
+        const float e = __expf((s_adv[i] - m) / eta);
+        z += e;
+        wa = fmaf(e, s_adv[i], wa);
+      }
+    }
+    z = wred_sum(z);
+    wa = wred_sum(wa);
+    __syncthreads();
+    if ((tid & (kWave - 1)) == 0) {
+      s4b[0][tid >> 6] = z;
+      s4b[1][tid >> 6] = wa;
+    }
+    __syncthreads();
+    float Z = 0.f, WA = 0.f;
+#pragma unroll
+    for (int w = 0; w < kWaves; ++w) {
+      Z += s4b[0][w];
+      WA += s4b[1][w];
+    }
+    if (tid == 0) { s_scalars[2] = m; s_scalars[3] = Z; s_scalars[4] = WA; }
+    for (int i = tid; i < BT; i += kThreads) {
+      const float p = (s_psi[i] > 0.f)
+                          ? __expf((s_adv[i] - m) / eta) / Z : 0.f;
+      s_psi[i] = p;
+      psi_g[i] = p;
+    }
+  }
+  __syncthreads();
+
+  {
+    float pl = 0.f, vl = 0.f, kl = 0.f, rg = 0.f;
+    for (int i = tid; i < BT; i += kThreads) {
+      const int b = i / T, t = i % T;
+      const long si = (long)b * S + t;
+      pl -= s_psi[i] * logp_g[si];
+      vl += huber_v(mo[si * D + A] - td_g[i]);
+      const float* zb = behav + si * A;
+      const float* zq = mo + si * D;
+      float mb = zb[0];
+      for (int j = 1; j < A; ++j) mb = fmaxf(mb, zb[j]);
+      float sb_ = 0.f;
+      for (int j = 0; j < A; ++j) sb_ += __expf(zb[j] - mb);
+      const float lb = mb + __logf(sb_);
+      for (int j = 0; j < A; ++j) {
+        const float lpb = zb[j] - lb;
+        const float lpq = zq[j] - lse_g[si];
+        kl += __expf(lpb) * (lpb - lpq);
+        rg = fmaf(zq[j], zq[j], rg);
+      }
+    }
+    pl = wred_sum(pl); vl = wred_sum(vl); kl = wred_sum(kl); rg = wred_sum(rg);
+    __syncthreads();
+    if ((tid & (kWave - 1)) == 0) {
+      const int w = tid >> 6;
+      s4[w] = pl; s4b[0][w] = vl; s4b[1][w] = kl; s4b[2][w] = rg;
+    }
+    __syncthreads();
+    if (tid == 0) {
+      float red0 = 0.f, red1 = 0.f, red2 = 0.f, red3 = 0.f;
+#pragma unroll
+      for (int w = 0; w < kWaves; ++w) {
+        red0 += s4[w]; red1 += s4b[0][w]; red2 += s4b[1][w]; red3 += s4b[2][w];
+      }
+      const float eta_v = s_scalars[6], alpha_v = s_scalars[7];
+      const float m = s_scalars[2], Z = s_scalars[3], wa = s_scalars[4];
+      const float eps_alpha = s_scalars[0];
+      const float kl_mean = red2 / BT;
+      const float pl_v = red0;
+      const float vl_v = red1 / BT;
+      const float lse_sel = m / eta_v + __logf(Z);
+      const float eta_loss = eta_v * eps_eta + eta_v * (lse_sel - __logf((float)K));
+      const float reg_v = creg * red3 / (BT * A);
+      stats[0] = cp * pl_v + cv * vl_v + eta_loss + (alpha_v * eps_alpha) + reg_v;
+      stats[1] = pl_v;
+      stats[2] = vl_v;
+      stats[3] = eta_v;
+      stats[4] = alpha_v;
+      stats[5] = kl_mean;
+      const float deta = eps_eta + (lse_sel - __logf((float)K)) - (wa / Z) / eta_v;
+      g_eta[0] = deta * eta_v;
+      g_alpha[0] = alpha_v * (eps_alpha - kl_mean);
+      scalars_g[7] = alpha_v;
+      if (norm_sq != nullptr) {
+        atomicAdd(norm_sq, g_eta[0] * g_eta[0] + g_alpha[0] * g_alpha[0]);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+bool vmpo_mid_hip(const at::Tensor& mo, const at::Tensor& behav,
+                  const at::Tensor& logp_g, const at::Tensor& lse_g,
+                  const at::Tensor& adv_g, const at::Tensor& td_g,
+                  const at::Tensor& log_eta, const at::Tensor& log_alpha,
+                  at::Tensor& psi_g, at::Tensor& scalars_g, at::Tensor& g_eta,
+                  at::Tensor& g_alpha, at::Tensor& stats,
+                  const c10::optional<at::Tensor>& norm_sq,
+                  at::Tensor& rng_state, long A, double cp, double cv,
+                  double creg, double eps_eta, double alpha_below,
+                  double alpha_upper) {
+  const int B = mo.size(0), S = mo.size(1);
+  const int BT = B * (S - 1);
+  const long lds = 2L * BT * sizeof(float);
+  if (lds > 56 * 1024) return false;
+  hipLaunchKernelGGL(
+      vmpo_mid_kernel, dim3(1), dim3(256), lds, current_stream(),
+      mo.data_ptr<float>(), behav.data_ptr<float>(),
+      logp_g.data_ptr<float>(), lse_g.data_ptr<float>(),
+      adv_g.data_ptr<float>(), td_g.data_ptr<float>(),
+      log_eta.data_ptr<float>(), log_alpha.data_ptr<float>(),
+      psi_g.data_ptr<float>(), scalars_g.data_ptr<float>(),
+      g_eta.data_ptr<float>(), g_alpha.data_ptr<float>(),
+      stats.data_ptr<float>(),
+      norm_sq.has_value() ? norm_sq->data_ptr<float>() : nullptr,
+      (unsigned*)rng_state.data_ptr<int>(), B, S, (int)A, (float)cp,
+      (float)cv, (float)creg, (float)eps_eta, (float)alpha_below,
+      (float)alpha_upper);
+  HIP_CHECK_LAST();
+  return true;
+}

@@ -135,6 +135,10 @@ class FusedOnPolicyStep(GraphableStep):
 
         B, S, _ = batch["obs"].shape
         if self.algo == "V-MPO":
+            if self.core.w_ih.size(0) == 64 and bool(
+                    int(os.environ.get("PDRL_FWDLOSS", "1"))):
+                # split path: only the middle kernel's s_adv+s_psi in LDS
+                return 2 * B * (S - 1) * 4 <= 56 * 1024
             return (2 * B * S + 3 * B * (S - 1)) * 4 <= 56 * 1024
         if self.algo == "PPO-C":
             if self.core.w_ih.size(0) == 64 and bool(
@@ -390,40 +394,74 @@ class FusedOnPolicyStep(GraphableStep):
             "dgates": mk(B, S, 4 * H), "dxb": mk(B, S, H),
             "stats_part": mk(B, 8),  # per-row loss partials (plain stores)
         }
+        if self.algo == "V-MPO":
+            BT = B * (S - 1)
+            self._fl.update({
+                "vm_lse": mk(B * S), "vm_logp": mk(B * S),
+                "vm_adv": mk(BT), "vm_td": mk(BT), "vm_psi": mk(BT),
+                "vm_scalars": mk(8),
+            })
         self._fl_shape = (B, S)
 
     def _try_fwdloss(self, e, batch, x, act, behav, rew, fir, hx0, cx0,
                      B, S, A, p, update) -> bool:
         import os
 
-        if self.algo not in ("IMPALA", "PPO", "PPO-C"):
+        if self.algo not in ("IMPALA", "PPO", "PPO-C", "V-MPO"):
             return False
         if self.core.w_ih.size(0) != 64:
             return False  # kernels specialized for the H=64 model family
         if not bool(int(os.environ.get("PDRL_FWDLOSS", "1"))):
             return False
+        if self.algo == "V-MPO" and 2 * B * (S - 1) * 4 > 56 * 1024:
+            return False  # middle kernel's LDS cap (s_adv + s_psi)
         if getattr(self, "_fl_shape", None) != (B, S):
             self._fwdloss_setup(batch)
         ws = self._fl
         c = self.core
         norm = self._norm_buf()
-        algo_i = {"IMPALA": 0, "PPO": 1, "PPO-C": 2}[self.algo]
+        creg = float(getattr(p, "logit_reg", 0.0))
+        algo_i = {"IMPALA": 0, "PPO": 1, "PPO-C": 2, "V-MPO": 3}[self.algo]
+        vm = {}
+        if self.algo == "V-MPO":
+            vm = {"vm_lse": ws["vm_lse"], "vm_logp": ws["vm_logp"],
+                  "vm_adv": ws["vm_adv"], "vm_td": ws["vm_td"]}
         e.seq_lstm_fwd_loss(
             x, hx0, cx0, c.body_w, c.body_b, c.w_ih, c.w_hh, c.b_g,
             c.heads_w, c.heads_b, ws["outs"], ws["hS"], ws["cS"],
             ws["stash"], act, behav, rew, fir, ws["gouts"], ws["stats_part"],
             norm, algo_i, p.gamma, p.lmbda, 0.8, 0.1, 1.0,
             p.reward_scale, p.policy_loss_coef, p.value_loss_coef,
-            p.entropy_coef, p.eps_clip,
-            float(getattr(p, "logit_reg", 0.0)),
+            p.entropy_coef, p.eps_clip, creg, **vm,
         )
-        e.seq_lstm_bwd_fin(
-            ws["gouts"], ws["stash"], x, cx0, c.body_w, c.w_ih, c.w_hh,
-            c.heads_w, ws["dgates"], ws["dxb"], self.stats_buf,
-            ws["stats_part"], algo_i, p.policy_loss_coef,
-            p.value_loss_coef, p.entropy_coef,
-            float(getattr(p, "logit_reg", 0.0)),
-        )
+        if self.algo == "V-MPO":
+            log_eta, log_alpha = self.duals
+            ok = e.vmpo_mid(
+                ws["outs"], self._behav_logits, ws["vm_logp"], ws["vm_lse"],
+                ws["vm_adv"], ws["vm_td"], log_eta.data.view(1),
+                log_alpha.data.view(1), ws["vm_psi"], ws["vm_scalars"],
+                log_eta.grad.view(1), log_alpha.grad.view(1), self.stats_buf,
+                norm, self.rng_state, A, p.policy_loss_coef,
+                p.value_loss_coef, creg, p.coef_eta, p.coef_alpha_below,
+                p.coef_alpha_upper,
+            )
+            assert ok  # gated above by the python-side LDS check
+            e.seq_lstm_bwd_fin(
+                ws["gouts"], ws["stash"], x, cx0, c.body_w, c.w_ih, c.w_hh,
+                c.heads_w, ws["dgates"], ws["dxb"], self.stats_buf,
+                ws["stats_part"], algo_i, p.policy_loss_coef,
+                p.value_loss_coef, p.entropy_coef, creg,
+                act=act, behav=self._behav_logits, vm_lse=ws["vm_lse"],
+                vm_psi=ws["vm_psi"], vm_td=ws["vm_td"],
+                vm_scalars=ws["vm_scalars"], vm_outs=ws["outs"],
+            )
+        else:
+            e.seq_lstm_bwd_fin(
+                ws["gouts"], ws["stash"], x, cx0, c.body_w, c.w_ih, c.w_hh,
+                c.heads_w, ws["dgates"], ws["dxb"], self.stats_buf,
+                ws["stats_part"], algo_i, p.policy_loss_coef,
+                p.value_loss_coef, p.entropy_coef, creg,
+            )
         e.seq_lstm_wgrad_out(x, hx0, ws["stash"], ws["dgates"], ws["dxb"],
                              ws["gouts"], *self._grad_views(), norm)
         if not update:
