@@ -463,45 +463,60 @@ def test_fused_grad_parity_vs_autograd(algo):
                                    msg=lambda m: f"{n}: {m}")
 
 
-def test_vmpo_fused_grad_parity_vs_autograd():
-    """Fused single-launch V-MPO loss (top-k + duals + analytic bwd) vs GPU
-    eager autograd, with the sampled dual coefficient pinned."""
+@pytest.mark.parametrize("bsz,fwdloss", [
+    (16, "1"), (16, "0"),   # split-phase path vs mega-kernel path
+    (1024, "1"),            # split path beyond the mega kernel's LDS cap
+])
+def test_vmpo_fused_grad_parity_vs_autograd(bsz, fwdloss):
+    """Fused V-MPO loss (top-half selection + duals + analytic bwd) vs GPU
+    eager autograd, with the sampled dual coefficient pinned. Covers both
+    the split-phase path (row-local pre/grad fused into the fwd/bwd
+    launches, PDRL_FWDLOSS=1) and the legacy single mega kernel."""
     _ops()
+    import os
+
     from pdrl_amd.agents.learner_module import VMPOUpdater
     from pdrl_amd.networks import MlpLSTMSingle
     from pdrl_amd.utils import load_params
     from tests.conftest import make_batch
 
     p = load_params()
-    p.batch_size, p.seq_len, p.obs_dim, p.n_actions = 16, 5, 4, 2
+    p.batch_size, p.seq_len, p.obs_dim, p.n_actions = bsz, 5, 4, 2
     p.coef_alpha_below = p.coef_alpha_upper = 0.0075  # pin the sampled dual
-    torch.manual_seed(5)
-    model = MlpLSTMSingle(4, 2, p.seq_len, p.hidden_size)
-    upd = VMPOUpdater(model, p, DEV)
-    assert upd.fused_step is not None
-    batch = make_batch(p, seed=31, device=DEV)
-    assert upd.fused_step.fits(batch)
+    os.environ["PDRL_FWDLOSS"] = fwdloss
+    try:
+        torch.manual_seed(5)
+        model = MlpLSTMSingle(4, 2, p.seq_len, p.hidden_size)
+        upd = VMPOUpdater(model, p, DEV)
+        assert upd.fused_step is not None
+        batch = make_batch(p, seed=31, device=DEV)
+        assert upd.fused_step.fits(batch)
 
-    upd.fused_step.compute_grads_only(batch)
-    fused = {n: q.grad.detach().clone() for n, q in model.named_parameters()}
-    fused_eta = upd.log_eta.grad.detach().clone()
-    fused_alpha = upd.log_alpha.grad.detach().clone()
+        upd.fused_step.compute_grads_only(batch)
+        fused = {n: q.grad.detach().clone()
+                 for n, q in model.named_parameters()}
+        fused_eta = upd.log_eta.grad.detach().clone()
+        fused_alpha = upd.log_alpha.grad.detach().clone()
 
-    upd.optimizer.zero_grad()
-    loss, stats_e = upd.compute_losses(batch)
-    loss.backward()
-    for n, q in model.named_parameters():
-        torch.testing.assert_close(fused[n], q.grad, rtol=2e-4, atol=2e-6,
-                                   msg=lambda m: f"{n}: {m}")
-    torch.testing.assert_close(fused_eta.squeeze(), upd.log_eta.grad.squeeze(),
-                               rtol=1e-3, atol=1e-5)
-    torch.testing.assert_close(fused_alpha.squeeze(),
-                               upd.log_alpha.grad.squeeze(),
-                               rtol=1e-3, atol=1e-5)
-    # loss value parity
-    upd.fused_step._body(batch, update=False)
-    torch.testing.assert_close(upd.fused_step.stats_buf[0], loss.detach(),
-                               rtol=1e-3, atol=1e-4)
+        upd.optimizer.zero_grad()
+        loss, stats_e = upd.compute_losses(batch)
+        loss.backward()
+        for n, q in model.named_parameters():
+            torch.testing.assert_close(fused[n], q.grad, rtol=2e-4,
+                                       atol=2e-6,
+                                       msg=lambda m: f"{n}: {m}")
+        torch.testing.assert_close(fused_eta.squeeze(),
+                                   upd.log_eta.grad.squeeze(),
+                                   rtol=1e-3, atol=1e-5)
+        torch.testing.assert_close(fused_alpha.squeeze(),
+                                   upd.log_alpha.grad.squeeze(),
+                                   rtol=1e-3, atol=1e-5)
+        # loss value parity
+        upd.fused_step._body(batch, update=False)
+        torch.testing.assert_close(upd.fused_step.stats_buf[0],
+                                   loss.detach(), rtol=1e-3, atol=1e-4)
+    finally:
+        os.environ.pop("PDRL_FWDLOSS", None)
 
 
 def test_ppo_continuous_fused_grad_parity():
