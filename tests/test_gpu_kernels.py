@@ -483,6 +483,9 @@ def test_vmpo_fused_grad_parity_vs_autograd(bsz, fwdloss):
     p = load_params()
     p.batch_size, p.seq_len, p.obs_dim, p.n_actions = bsz, 5, 4, 2
     p.coef_alpha_below = p.coef_alpha_upper = 0.0075  # pin the sampled dual
+    # at BT=4096 the psi softmax sums 2048 exp terms: fp32 accumulation
+    # order alone moves head grads ~5e-4 rel (selection + loss identical)
+    rtol, atol = (1e-3, 2e-5) if bsz >= 512 else (2e-4, 2e-6)
     os.environ["PDRL_FWDLOSS"] = fwdloss
     try:
         torch.manual_seed(5)
@@ -502,8 +505,8 @@ def test_vmpo_fused_grad_parity_vs_autograd(bsz, fwdloss):
         loss, stats_e = upd.compute_losses(batch)
         loss.backward()
         for n, q in model.named_parameters():
-            torch.testing.assert_close(fused[n], q.grad, rtol=2e-4,
-                                       atol=2e-6,
+            torch.testing.assert_close(fused[n], q.grad, rtol=rtol,
+                                       atol=atol,
                                        msg=lambda m: f"{n}: {m}")
         torch.testing.assert_close(fused_eta.squeeze(),
                                    upd.log_eta.grad.squeeze(),
