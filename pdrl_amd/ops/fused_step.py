@@ -445,7 +445,10 @@ class FusedOnPolicyStep(GraphableStep):
                 p.value_loss_coef, creg, p.coef_eta, p.coef_alpha_below,
                 p.coef_alpha_upper,
             )
-            assert ok  # gated above by the python-side LDS check
+            if not ok:  # unreachable: _try_fwdloss gates on the same cap
+                raise RuntimeError(
+                    "vmpo_mid refused a shape the python-side LDS gate "
+                    "accepted — the 2*BT*4 <= 56K checks are out of sync")
             e.seq_lstm_bwd_fin(
                 ws["gouts"], ws["stash"], x, cx0, c.body_w, c.w_ih, c.w_hh,
                 c.heads_w, ws["dgates"], ws["dxb"], self.stats_buf,
