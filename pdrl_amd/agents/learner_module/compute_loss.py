@@ -109,7 +109,11 @@ def soft_update(net: torch.nn.Module, target_net: torch.nn.Module, tau: float = 
 
         key = (id(net), id(target_net))
         tab = _soft_update_tables.get(key)
-        if tab is None or tab[3] != [p.data.data_ptr() for p in params]:
+        # revalidate BOTH sides: param storages move when an optimizer (or
+        # the fused SAC step's Polyak flat buffer) re-homes them
+        ptrs = [p.data.data_ptr() for p in params] + \
+               [t.data.data_ptr() for t in tparams]
+        if tab is None or tab[3] != ptrs:
             dev = params[0].device
             sp = torch.tensor([p.data.data_ptr() for p in params],
                               dtype=torch.int64).to(dev)
@@ -118,7 +122,7 @@ def soft_update(net: torch.nn.Module, target_net: torch.nn.Module, tau: float = 
             ne = torch.tensor([p.numel() for p in params],
                               dtype=torch.int64).to(dev)
             mx = max(p.numel() for p in params)
-            tab = (sp, dp, ne, [p.data.data_ptr() for p in params], mx)
+            tab = (sp, dp, ne, ptrs, mx)
             _soft_update_tables[key] = tab
         ops.ext().soft_update_cached(tab[0], tab[1], tab[2], len(params),
                                      tab[4], float(tau))

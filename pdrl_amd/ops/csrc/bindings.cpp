@@ -160,7 +160,35 @@ void rmsprop_step_hip(at::Tensor&, const at::Tensor&, at::Tensor&,
                       const at::Tensor&, double, double, double, double);
 void adam_step_hip(at::Tensor&, const at::Tensor&, at::Tensor&, at::Tensor&,
                    at::Tensor&, const at::Tensor&, double, double, double,
-                   double, double, bool);
+                   double, double, bool, const c10::optional<at::Tensor>&,
+                   const c10::optional<at::Tensor>&, double,
+                   const c10::optional<at::Tensor>&, double);
+void sac_actor_bwd_hip(const at::Tensor&, const at::Tensor&,
+                       const at::Tensor&, const at::Tensor&, at::Tensor&,
+                       at::Tensor&, const c10::optional<at::Tensor>&,
+                       const at::Tensor&, const at::Tensor&,
+                       const at::Tensor&, const at::Tensor&,
+                       const at::Tensor&, const at::Tensor&,
+                       const at::Tensor&, at::Tensor&, at::Tensor&);
+void sac_actor_wgrad_hip(const at::Tensor&, const at::Tensor&,
+                         const at::Tensor&, const at::Tensor&,
+                         const at::Tensor&, const at::Tensor&, at::Tensor&,
+                         at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
+                         at::Tensor&, at::Tensor&,
+                         const c10::optional<at::Tensor>&, const at::Tensor&,
+                         const at::Tensor&, at::Tensor&, at::Tensor&,
+                         const c10::optional<at::Tensor>&,
+                         const c10::optional<at::Tensor>&, double, double,
+                         double);
+void sac_fwd2_critic_loss_hip(
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, at::Tensor&, at::Tensor&,
+    at::Tensor&, const c10::optional<at::Tensor>&, double, double);
 void adam_prep_hip(at::Tensor&, double, double);
 void adam_multi_hip(const at::Tensor&, const at::Tensor&, const at::Tensor&,
                     long, long, double, double, double);
@@ -284,6 +312,23 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("beta1") = 0.9, py::arg("beta2") = 0.999);
   m.def("sac_critic_loss", &sac_critic_loss_hip,
         "SAC discrete soft-Q target + twin critic loss grads");
+  m.def("sac_actor_bwd", &sac_actor_bwd_hip,
+        "SAC discrete actor loss (row-local) fused into the actor BPTT "
+        "launch; per-row {ubar, entropy} partials to stats_part");
+  m.def("sac_actor_wgrad", &sac_actor_wgrad_hip,
+        "actor MFMA wgrad + actor-loss cross-row reduce (g_alpha, stats, "
+        "alpha_norm, shared Adam clock prep) in one extra block",
+        py::arg("x"), py::arg("h0"), py::arg("stash"), py::arg("dgates"),
+        py::arg("dxb"), py::arg("gouts"), py::arg("dw_ih"), py::arg("dw_hh"),
+        py::arg("dbody_w"), py::arg("dbody_b"), py::arg("db_g"),
+        py::arg("dheads_w"), py::arg("dheads_b"), py::arg("norm_sq"),
+        py::arg("stats_part"), py::arg("log_alpha"), py::arg("g_alpha"),
+        py::arg("stats4"), py::arg("alpha_norm") = c10::nullopt,
+        py::arg("clock") = c10::nullopt, py::arg("target_entropy") = 0.0,
+        py::arg("beta1") = 0.9, py::arg("beta2") = 0.999);
+  m.def("sac_fwd2_critic_loss", &sac_fwd2_critic_loss_hip,
+        "post-update actor forward + SAC critic loss in one launch "
+        "(fwd+loss pattern; vl partials reduced by the critic Adam)");
   m.def("sacc_sample", &sacc_sample_hip,
         "reparameterized tanh-Gaussian sample + log-prob (graph-safe RNG)");
   m.def("sacc_actor_grad", &sacc_actor_grad_hip,
@@ -301,11 +346,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "accumulator");
   m.def("l2norm_sq", &l2norm_sq_hip, "squared L2 norm into a device scalar");
   m.def("rmsprop_step", &rmsprop_step_hip, "fused clip+RMSprop on flat buffers");
-  m.def("adam_step", &adam_step_hip, "fused clip+Adam on flat buffers",
+  m.def("adam_step", &adam_step_hip,
+        "fused clip+Adam on flat buffers (optional per-row stats reduce in "
+        "block 0 and inline Polyak target tracking)",
         py::arg("p"), py::arg("g"), py::arg("m"), py::arg("v"),
         py::arg("state3"), py::arg("norm_sq"), py::arg("lr"), py::arg("beta1"),
         py::arg("beta2"), py::arg("eps"), py::arg("max_norm"),
-        py::arg("do_prep") = true);
+        py::arg("do_prep") = true, py::arg("stats_part") = c10::nullopt,
+        py::arg("stats_out") = c10::nullopt, py::arg("part_scale") = 1.0,
+        py::arg("polyak") = c10::nullopt, py::arg("tau") = 0.0);
   m.def("adam_prep", &adam_prep_hip,
         "advance the shared device Adam step clock");
   m.def("adam_multi", &adam_multi_hip,

@@ -71,12 +71,35 @@ __global__ void adam_prep_kernel(float* __restrict__ state3, float beta1,
   state3[2] = 1.0f - __powf(beta2, t);
 }
 
+// Optional extras fold two more SAC launches into this one:
+//   stats_part/stats_out — block 0 sums the (n_part) per-row loss partials
+//     written by an earlier kernel and stores sum*part_scale (the scalar
+//     value-loss stat; nothing downstream reads it in-step);
+//   polyak/tau — Polyak target tracking applied INLINE on the freshly
+//     updated parameter (θ' ← θ' + τ(θ_new − θ')), replacing the separate
+//     soft_update launch (the target flat buffer is laid out to match p).
 __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
                             float* __restrict__ m, float* __restrict__ v,
                             const float* __restrict__ state3,
                             const float* __restrict__ norm_sq, long n,
                             float lr, float beta1, float beta2, float eps,
-                            float max_norm) {
+                            float max_norm,
+                            const float* __restrict__ stats_part,
+                            float* __restrict__ stats_out,
+                            int n_part, float part_scale,
+                            float* __restrict__ polyak, float tau) {
+  if (stats_part != nullptr && blockIdx.x == 0) {
+    __shared__ float red[256];
+    float acc = 0.0f;
+    for (int i = threadIdx.x; i < n_part; i += 256) acc += stats_part[i];
+    red[threadIdx.x] = acc;
+    __syncthreads();
+    for (int off = 128; off > 0; off >>= 1) {
+      if ((int)threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) stats_out[0] = red[0] * part_scale;
+  }
   const float scale = clip_scale(norm_sq, max_norm);
   const float bc1 = state3[1], bc2 = state3[2];
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
@@ -86,7 +109,9 @@ __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
     const float vi = beta2 * v[i] + (1.0f - beta2) * gi * gi;
     m[i] = mi;
     v[i] = vi;
-    p[i] -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+    const float pn = p[i] - lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+    p[i] = pn;
+    if (polyak != nullptr) polyak[i] = fmaf(tau, pn - polyak[i], polyak[i]);
   }
 }
 
@@ -172,18 +197,32 @@ void rmsprop_step_hip(at::Tensor& p, const at::Tensor& g, at::Tensor& sq_avg,
 void adam_step_hip(at::Tensor& p, const at::Tensor& g, at::Tensor& m,
                    at::Tensor& v, at::Tensor& state3,
                    const at::Tensor& norm_sq, double lr, double beta1,
-                   double beta2, double eps, double max_norm, bool do_prep) {
+                   double beta2, double eps, double max_norm, bool do_prep,
+                   const c10::optional<at::Tensor>& stats_part,
+                   const c10::optional<at::Tensor>& stats_out,
+                   double part_scale,
+                   const c10::optional<at::Tensor>& polyak, double tau) {
   CHECK_IN(p); CHECK_IN(g); CHECK_IN(m); CHECK_IN(v);
   if (do_prep) {
     hipLaunchKernelGGL(adam_prep_kernel, dim3(1), dim3(1), 0, current_stream(),
                        state3.data_ptr<float>(), (float)beta1, (float)beta2);
+  }
+  if (polyak.has_value()) {
+    TORCH_CHECK(polyak->numel() == p.numel(),
+                "polyak target numel mismatches the flat param buffer");
   }
   hipLaunchKernelGGL(adam_kernel, dim3(grid_for(p.numel())), dim3(kThreads), 0,
                      current_stream(), p.data_ptr<float>(),
                      g.data_ptr<float>(), m.data_ptr<float>(),
                      v.data_ptr<float>(), state3.data_ptr<float>(),
                      norm_sq.data_ptr<float>(), p.numel(), (float)lr,
-                     (float)beta1, (float)beta2, (float)eps, (float)max_norm);
+                     (float)beta1, (float)beta2, (float)eps, (float)max_norm,
+                     stats_part.has_value() ? stats_part->data_ptr<float>() : nullptr,
+                     stats_out.has_value() ? stats_out->data_ptr<float>() : nullptr,
+                     stats_part.has_value() ? (int)stats_part->numel() : 0,
+                     (float)part_scale,
+                     polyak.has_value() ? polyak->data_ptr<float>() : nullptr,
+                     (float)tau);
   HIP_CHECK_LAST();
 }
 

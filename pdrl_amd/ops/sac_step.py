@@ -46,6 +46,15 @@ class FusedSacStep(GraphableStep):
         self.stats_buf = torch.zeros(8, dtype=torch.float32, device=dev)
         self.stat_names = _SAC_STATS
         self.use_graph = use_graph and self.grad_reducer is None
+        import os
+        # 8-launch restructured DAG (H=64, single-rank): actor loss rides
+        # the actor BPTT, its reduce + Adam clock ride the actor wgrad,
+        # the post-update forward rides the critic loss, and the vl reduce
+        # + Polyak ride the critic Adam. PDRL_SAC8=0 falls back to the
+        # 10-launch DAG (kept for multi-rank and other widths).
+        self._fast8 = (self.grad_reducer is None
+                       and self.cores["actor"].w_ih.size(0) == 64
+                       and bool(int(os.environ.get("PDRL_SAC8", "1"))))
         # actor+alpha Adam updates happen back-to-back at the same DAG
         # point → one multi-group launch (shared device step clock)
         self.adam_aa = None
@@ -80,6 +89,25 @@ class FusedSacStep(GraphableStep):
         def mk(*shape):
             return torch.empty(*shape, device=dev)
 
+        # Re-home the target critic's params into ONE flat fp32 buffer laid
+        # out exactly like critic_optimizer.flat_param (same param order —
+        # target is a deepcopy), so the Polyak update can ride the critic
+        # Adam kernel. Must precede the pointer-table builds below (p.data
+        # storage moves). Params stay views, so checkpoint/load still work.
+        if not hasattr(self, "_t_flat"):
+            cps = u.critic_optimizer.space.params
+            tps = list(u.target_critic.parameters())
+            assert len(cps) == len(tps)
+            tf = torch.empty(sum(p.numel() for p in tps), device=dev)
+            off = 0
+            for cp, tp in zip(cps, tps):
+                assert cp.shape == tp.shape, "critic/target param order skew"
+                n = tp.numel()
+                tf[off:off + n].copy_(tp.data.reshape(-1))
+                tp.data = tf[off:off + n].view_as(tp.data)
+                off += n
+            self._t_flat = tf
+
         buf = {}
         for name in ("actor", "q1", "q2", "t1", "t2"):
             buf[name] = {"outs": mk(B, S, D), "hS": mk(B, H), "cS": mk(B, H),
@@ -88,6 +116,20 @@ class FusedSacStep(GraphableStep):
             buf[name]["dgates"] = mk(B, S, 4 * H)
             buf[name]["dxb"] = mk(B, S, H)
             buf[name]["gq"] = mk(B, S, D)
+        if self._fast8:
+            a = buf["actor"]
+            a["gA"] = mk(B, S, D)
+            a["dgates"] = mk(B, S, 4 * H)
+            a["dxb"] = mk(B, S, H)
+            buf["mo2"] = mk(B, S, D)
+            self._sp_a = mk(B, 2)   # actor-loss {ubar, entropy} partials
+            self._sp_c = mk(B)      # critic huber partials
+            ag = self.cores["actor"]
+            gs = [ag.body_w.grad, ag.body_b.grad, ag.w_ih.grad,
+                  ag.w_hh.grad, ag.b_g.grad, ag.heads_w.grad,
+                  ag.heads_b.grad]
+            assert all(g is not None for g in gs)
+            self._actor_grads = gs
         self.buf = buf
 
         def t64(rows):
@@ -172,6 +214,87 @@ class FusedSacStep(GraphableStep):
             optimizer._update()
 
     def _body(self, batch):
+        if self._fast8:
+            return self._body8(batch)
+        return self._body_legacy(batch)
+
+    def _body8(self, batch):
+        """8-launch restructured DAG (see __init__); math identical to the
+        legacy sequence (GPU parity test vs the eager updater)."""
+        u, p = self.u, self.params
+        e = ext()
+        x = batch["obs"]
+        B, S, _ = x.shape
+        hx0 = batch["hx"][:, 0]
+        cx0 = batch["cx"][:, 0]
+        act = batch["act"].reshape(-1)
+        rew = batch["rew"].reshape(B, S)
+        fir = batch["is_fir"].reshape(B, S)
+        log_alpha = u.log_alpha.data.view(1)
+        if getattr(self, "_mshape", None) != (B, S):
+            self._multi_setup(batch)
+        buf, D = self.buf, self._D
+        ac = self.cores["actor"]
+        ab = buf["actor"]
+
+        # 1. actor + twin critics + twin TARGET critics — ONE launch
+        e.seq_lstm_forward_multi(x, hx0, cx0, self.fwd1_cores,
+                                 self.fwd1_outs, 5, D)
+        moA1 = ab["outs"]
+        mq1, mq2 = buf["q1"]["outs"], buf["q2"]["outs"]
+
+        # 2. actor loss (row-local analytic dlogits) + actor BPTT — ONE
+        #    launch; per-row loss partials to _sp_a; zeroes the actor norm
+        e.sac_actor_bwd(moA1, mq1, mq2, log_alpha, ab["gA"], self._sp_a,
+                        u.actor_optimizer.norm_sq, ab["stash"], x, cx0,
+                        ac.body_w, ac.w_ih, ac.w_hh, ac.heads_w,
+                        ab["dgates"], ab["dxb"])
+        # 3. actor MFMA wgrad + loss reduce (g_alpha, stats[0..3]) + shared
+        #    Adam clock prep — ONE launch (one extra block on the wgrad grid)
+        gs = self._actor_grads
+        clk = u.actor_optimizer if self.adam_aa is not None else None
+        e.sac_actor_wgrad(
+            x, hx0, ab["stash"], ab["dgates"], ab["dxb"], ab["gA"],
+            gs[2], gs[3], gs[0], gs[1], gs[4], gs[5], gs[6],
+            u.actor_optimizer.norm_sq, self._sp_a, log_alpha,
+            u.log_alpha.grad.view(1), self.stats_buf[:4], alpha_norm=None,
+            clock=clk.state3 if clk is not None else None,
+            target_entropy=u.target_entropy,
+            beta1=clk.beta1 if clk is not None else 0.9,
+            beta2=clk.beta2 if clk is not None else 0.999,
+        )
+        # 4. actor + alpha Adam — ONE multi-group launch
+        if self.adam_aa is not None:
+            self.adam_aa.update(tick=False)  # clock prepped by the wgrad
+        else:
+            u.actor_optimizer._update()
+            u.alpha_optimizer._update()  # no clip: norm unused
+
+        # 5. post-update actor forward + critic loss — ONE launch (the
+        #    actor stash/hS/cS scratch is dead after step 3 and is reused;
+        #    zeroes the critic norm; vl partials to _sp_c)
+        gq1, gq2 = buf["q1"]["gq"], buf["q2"]["gq"]
+        e.sac_fwd2_critic_loss(
+            x, hx0, cx0, ac.body_w, ac.body_b, ac.w_ih, ac.w_hh, ac.b_g,
+            ac.heads_w, ac.heads_b, buf["mo2"], ab["hS"], ab["cS"],
+            ab["stash"], mq1, mq2, buf["t1"]["outs"], buf["t2"]["outs"],
+            act, rew, fir, log_alpha, gq1, gq2, self._sp_c,
+            u.critic_optimizer.norm_sq, p.gamma, p.reward_scale,
+        )
+        # 6-7. twin-critic BPTT + MFMA wgrads — one launch each
+        e.seq_lstm_backward_multi(x, cx0, self.bwd_in, self.bwd_out, 2, D)
+        e.seq_lstm_wgrad_multi(x, hx0, self.wg_tab, 2, D)
+        # 8. critic Adam + vl-stat reduce + Polyak target — ONE launch
+        co = u.critic_optimizer
+        e.adam_step(
+            co.space.flat_param, co.space.flat_grad, co.exp_avg,
+            co.exp_avg_sq, co.state3, co.norm_sq, co.lr, co.beta1, co.beta2,
+            co.eps, co.max_norm, do_prep=False, stats_part=self._sp_c,
+            stats_out=self.stats_buf[4:5], part_scale=1.0 / (B * (S - 1)),
+            polyak=self._t_flat, tau=u.TAU,
+        )
+
+    def _body_legacy(self, batch):
         from pdrl_amd.agents.learner_module.compute_loss import soft_update
 
         u, p = self.u, self.params
