@@ -639,10 +639,12 @@ def test_fused_impala_large_shape_fallback():
             os.environ.pop("PDRL_FWDLOSS", None)
 
 
-def test_sac_fused_step_parity():
+@pytest.mark.parametrize("fast8", ["1", "0"])
+def test_sac_fused_step_parity(fast8):
     """Fused SAC-discrete DAG vs the eager updater: identical parameter
     trajectories from identical init over 3 updates (gradients are analytic,
-    no sampling in the discrete SAC step)."""
+    no sampling in the discrete SAC step). Covers the 8-launch restructured
+    DAG (PDRL_SAC8=1) and the legacy 10-launch sequence."""
     _ops()
     import os
 
@@ -651,6 +653,7 @@ def test_sac_fused_step_parity():
     from pdrl_amd.utils import load_params
     from tests.conftest import make_batch
 
+    os.environ["PDRL_SAC8"] = fast8
     p = load_params()
     p.batch_size, p.seq_len, p.obs_dim, p.n_actions = 16, 5, 4, 2
     p.lr = 1e-4
@@ -677,9 +680,13 @@ def test_sac_fused_step_parity():
     for k in ("loss-actor", "loss-value", "loss-alpha", "alpha", "entropy"):
         assert abs(float(sf[k]) - float(se[k])) < 5e-2, (k, float(sf[k]), float(se[k]))
     # target critics moved in both
-    for tf, te in zip(upd_f.target_critic.parameters(),
-                      upd_e.target_critic.parameters()):
-        torch.testing.assert_close(tf.detach(), te.detach(), rtol=5e-3, atol=5e-4)
+    try:
+        for tf, te in zip(upd_f.target_critic.parameters(),
+                          upd_e.target_critic.parameters()):
+            torch.testing.assert_close(tf.detach(), te.detach(), rtol=5e-3,
+                                       atol=5e-4)
+    finally:
+        os.environ.pop("PDRL_SAC8", None)
 
 
 # --------------------------------------------------------------------------- #
