@@ -189,6 +189,37 @@ void sac_fwd2_critic_loss_hip(
     const at::Tensor&, const at::Tensor&, const at::Tensor&,
     const at::Tensor&, const at::Tensor&, at::Tensor&, at::Tensor&,
     at::Tensor&, const c10::optional<at::Tensor>&, double, double);
+void sacc_fwd_sample_hip(
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
+    at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&,
+    const c10::optional<at::Tensor>&, const c10::optional<at::Tensor>&,
+    const c10::optional<at::Tensor>&);
+void sacc_minmask_bwd_hip(
+    const at::Tensor&, const at::Tensor&, at::Tensor&, at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&, at::Tensor&,
+    at::Tensor&, at::Tensor&, at::Tensor&, at::Tensor&, long);
+void sacc_actor_bwd_hip(
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, at::Tensor&, at::Tensor&,
+    const c10::optional<at::Tensor>&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, at::Tensor&, at::Tensor&);
+void sacc_critic_bwd_hip(
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, at::Tensor&, at::Tensor&,
+    at::Tensor&, const c10::optional<at::Tensor>&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&, at::Tensor&,
+    at::Tensor&, at::Tensor&, at::Tensor&, double, double);
 void adam_prep_hip(at::Tensor&, double, double);
 void adam_multi_hip(const at::Tensor&, const at::Tensor&, const at::Tensor&,
                     long, long, double, double, double);
@@ -341,6 +372,25 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("beta2") = 0.999);
   m.def("sacc_critic_loss", &sacc_critic_loss_hip,
         "SAC-continuous soft-Q target + twin critic loss grads");
+  m.def("sacc_fwd_sample", &sacc_fwd_sample_hip,
+        "actor forward + reparameterized tanh-Gaussian sample in one "
+        "launch (optional dQ/da zeroing and behaviour-action staging)",
+        py::arg("x"), py::arg("h0"), py::arg("c0"), py::arg("body_w"),
+        py::arg("body_b"), py::arg("w_ih"), py::arg("w_hh"), py::arg("b_g"),
+        py::arg("heads_w"), py::arg("heads_b"), py::arg("moA"),
+        py::arg("hS"), py::arg("cS"), py::arg("stash"), py::arg("rng"),
+        py::arg("eps"), py::arg("act"), py::arg("logpi"),
+        py::arg("dact_zero") = c10::nullopt,
+        py::arg("act_src") = c10::nullopt, py::arg("actb") = c10::nullopt);
+  m.def("sacc_minmask_bwd", &sacc_minmask_bwd_hip,
+        "min-critic selection + twin-critic input-grad backward in one "
+        "launch (dminQ/da accumulated into the shared buffer)");
+  m.def("sacc_actor_bwd", &sacc_actor_bwd_hip,
+        "SAC-continuous analytic actor grad + actor BPTT in one launch; "
+        "per-row partials reduced by sac_actor_wgrad");
+  m.def("sacc_critic_bwd", &sacc_critic_bwd_hip,
+        "SAC-continuous soft-Q critic loss + twin BPTT in one launch; "
+        "huber partials reduced by the critic Adam");
   m.def("sacc_min_mask", &sacc_min_mask_hip,
         "min-critic selection masks for dE[-minQ]/dq + zero the dQ/da "
         "accumulator");

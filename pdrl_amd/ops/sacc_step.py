@@ -53,6 +53,16 @@ class FusedSacContinuousStep(GraphableStep):
         self.stat_names = _SACC_STATS
         self.rng = torch.randint(1, 1 << 30, (1,), dtype=torch.int32, device=dev)
         self.use_graph = use_graph and self.grad_reducer is None
+        import os
+        # 11-launch restructured DAG (H=64, single-rank) mirroring the
+        # SAC-discrete restructure: sampling rides the actor forwards,
+        # min-mask rides the input-grad backward, actor grad rides the
+        # actor BPTT (reduce + clock on the wgrad extra block), critic
+        # loss rides the critic BPTT, vl-reduce + Polyak ride the critic
+        # Adam. PDRL_SAC8=0 falls back to the 18-launch DAG.
+        self._fast11 = (self.grad_reducer is None
+                        and self.actor_core.w_ih.size(0) == 64
+                        and bool(int(os.environ.get("PDRL_SAC8", "1"))))
         # actor+alpha Adam updates batch into one launch (shared clock)
         self.adam_aa = None
         if self.grad_reducer is None and getattr(
@@ -87,6 +97,24 @@ class FusedSacContinuousStep(GraphableStep):
         def mk(*shape):
             return torch.empty(*shape, device=dev)
 
+        # Re-home the target critic's params into ONE flat buffer laid out
+        # like critic_optimizer.flat_param (see sac_step.py) so the Polyak
+        # update can ride the critic Adam kernel. Must precede the pointer
+        # tables below (p.data storage moves).
+        if not hasattr(self, "_t_flat"):
+            cps = u.critic_optimizer.space.params
+            tps = list(u.target_critic.parameters())
+            assert len(cps) == len(tps)
+            tf = torch.empty(sum(p.numel() for p in tps), device=dev)
+            off = 0
+            for cp, tp in zip(cps, tps):
+                assert cp.shape == tp.shape, "critic/target param order skew"
+                n = tp.numel()
+                tf[off:off + n].copy_(tp.data.reshape(-1))
+                tp.data = tf[off:off + n].view_as(tp.data)
+                off += n
+            self._t_flat = tf
+
         buf = {
             "a1": mk(B, S, A), "eps1": mk(B, S, A), "logpi1": mk(B, S, 1),
             "a2": mk(B, S, A), "eps2": mk(B, S, A), "logpi2": mk(B, S, 1),
@@ -94,6 +122,23 @@ class FusedSacContinuousStep(GraphableStep):
             "dact": mk(B, S, A),           # accumulated dminQ/da
             "gq1p": mk(B, S, 1), "gq2p": mk(B, S, 1),
         }
+        if self._fast11:
+            buf["moA"] = mk(B, S, 2 * A)
+            buf["moA2"] = mk(B, S, 2 * A)
+            buf["dmoA"] = mk(B, S, 2 * A)
+            buf["stA"] = mk(B, S, 7 * H)
+            buf["hA"] = mk(B, H)
+            buf["cA"] = mk(B, H)
+            buf["adg"] = mk(B, S, 4 * H)
+            buf["adx"] = mk(B, S, H)
+            self._sp_a = mk(B, 2)   # actor partials {l_sum, -lp_sum}
+            self._sp_c = mk(B, 2)   # per-(row, critic) huber partials
+            ag = ac
+            gs = [ag.body_w.grad, ag.body_b.grad, ag.w_ih.grad,
+                  ag.w_hh.grad, ag.b_g.grad, ag.heads_w.grad,
+                  ag.heads_b.grad]
+            assert all(g is not None for g in gs)
+            self._actor_grads = gs
         # per-network activation slots (critics D=1)
         for name in ("q1s", "q2s", "t1", "t2", "q1b", "q2b"):
             buf[name] = {"outs": mk(B, S, 1), "hS": mk(B, H), "cS": mk(B, H),
@@ -206,6 +251,114 @@ class FusedSacContinuousStep(GraphableStep):
 
     # ------------------------------------------------------------------ #
     def _body(self, batch):
+        if self._fast11:
+            return self._body11(batch)
+        return self._body_legacy(batch)
+
+    def _body11(self, batch):
+        """11-launch restructured DAG (see __init__); math identical to the
+        legacy sequence (GPU parity tests vs eager)."""
+        u, p = self.u, self.params
+        e = ext()
+        x = batch["obs"]
+        B, S, _ = x.shape
+        if self._mshape != (B, S):
+            self._multi_setup(batch)
+        buf = self.buf
+        A = self._A
+        half = self._half
+        hx0 = batch["hx"][:, 0]
+        cx0 = batch["cx"][:, 0]
+        rew = batch["rew"].reshape(B, S)
+        fir = batch["is_fir"].reshape(B, S)
+        log_alpha = u.log_alpha.data.view(1)
+        ac = self.actor_core
+        q = self.q
+
+        # 1. actor fwd + reparameterized sample + zero the dQ/da buffer
+        e.sacc_fwd_sample(x, hx0, cx0, ac.body_w, ac.body_b, ac.w_ih,
+                          ac.w_hh, ac.b_g, ac.heads_w, ac.heads_b,
+                          buf["moA"], buf["hA"], buf["cA"], buf["stA"],
+                          self.rng, buf["eps1"], buf["a1"], buf["logpi1"],
+                          dact_zero=buf["dact"])
+        # 2. twin critics on the fresh sample — ONE dual-body launch
+        e.seq_lstm_forward_multi(x, hx0, cx0, self.fwdS_cores, self.fwdS_outs,
+                                 2, 1, F2=A, half=half)
+        qp1 = buf["q1s"]["outs"]
+        qp2 = buf["q2s"]["outs"]
+        # 3. min-critic selection + input-grad backward — ONE launch,
+        #    dminQ/da atomically accumulated into dact
+        e.sacc_minmask_bwd(
+            qp1.reshape(-1), qp2.reshape(-1), buf["gq1p"], buf["gq2p"],
+            buf["q1s"]["stash"], buf["q2s"]["stash"], x, cx0,
+            q["q1"].w_ih, q["q1"].w_hh, q["q1"].heads_w, q["q1"].body2_w,
+            q["q2"].w_ih, q["q2"].w_hh, q["q2"].heads_w, q["q2"].body2_w,
+            buf["q1s"]["dgates"], buf["q1s"]["dxb"],
+            buf["q2s"]["dgates"], buf["q2s"]["dxb"], buf["dact"], half)
+        # 4. analytic actor grad (row-local) + actor BPTT — ONE launch
+        e.sacc_actor_bwd(
+            buf["moA"], buf["eps1"], buf["a1"], buf["dact"],
+            qp1.reshape(-1), qp2.reshape(-1), log_alpha, buf["dmoA"],
+            self._sp_a, u.actor_optimizer.norm_sq, buf["stA"], x, cx0,
+            ac.w_ih, ac.w_hh, ac.heads_w, buf["adg"], buf["adx"])
+        # 5. actor wgrad + loss reduce + shared Adam clock prep (the
+        #    continuous partials are stored in the discrete form, so the
+        #    discrete reduce block applies verbatim)
+        gs = self._actor_grads
+        clk = u.actor_optimizer if self.adam_aa is not None else None
+        e.sac_actor_wgrad(
+            x, hx0, buf["stA"], buf["adg"], buf["adx"], buf["dmoA"],
+            gs[2], gs[3], gs[0], gs[1], gs[4], gs[5], gs[6],
+            u.actor_optimizer.norm_sq, self._sp_a, log_alpha,
+            u.log_alpha.grad.view(1), self.stats_buf[:4], alpha_norm=None,
+            clock=clk.state3 if clk is not None else None,
+            target_entropy=u.target_entropy,
+            beta1=clk.beta1 if clk is not None else 0.9,
+            beta2=clk.beta2 if clk is not None else 0.999,
+        )
+        # 6. actor + alpha Adam — ONE multi-group launch
+        if self.adam_aa is not None:
+            self.adam_aa.update(tick=False)
+        else:
+            u.actor_optimizer._update()
+            u.alpha_optimizer._update()
+        # 7. post-update actor fwd + sample + behaviour-action staging
+        e.sacc_fwd_sample(x, hx0, cx0, ac.body_w, ac.body_b, ac.w_ih,
+                          ac.w_hh, ac.b_g, ac.heads_w, ac.heads_b,
+                          buf["moA2"], buf["hA"], buf["cA"], buf["stA"],
+                          self.rng, buf["eps2"], buf["a2"], buf["logpi2"],
+                          act_src=batch["act"].reshape(B, S, A).contiguous(),
+                          actb=buf["actb"])
+        # 8. target critics on a2 + behaviour critics — ONE 4-network launch
+        e.seq_lstm_forward_multi(x, hx0, cx0, self.fwdT_cores, self.fwdT_outs,
+                                 4, 1, F2=A, half=half)
+        # 9. critic loss (row-local) + twin-critic BPTT — ONE launch
+        e.sacc_critic_bwd(
+            buf["q1b"]["outs"].reshape(-1), buf["q2b"]["outs"].reshape(-1),
+            buf["t1"]["outs"].reshape(-1), buf["t2"]["outs"].reshape(-1),
+            buf["logpi2"].reshape(-1), rew, fir, log_alpha,
+            buf["q1b"]["gq"].reshape(-1), buf["q2b"]["gq"].reshape(-1),
+            self._sp_c, u.critic_optimizer.norm_sq,
+            buf["q1b"]["stash"], buf["q2b"]["stash"], x, cx0,
+            q["q1"].w_ih, q["q1"].w_hh, q["q1"].heads_w,
+            q["q2"].w_ih, q["q2"].w_hh, q["q2"].heads_w,
+            buf["q1b"]["dgates"], buf["q1b"]["dxb"],
+            buf["q2b"]["dgates"], buf["q2b"]["dxb"],
+            p.gamma, p.reward_scale)
+        # 10. twin-critic MFMA wgrads (encoder grads ride the same launch)
+        e.seq_lstm_wgrad_multi(x, hx0, self.wgB_tab, 2, 1, x2=buf["actb"],
+                               F2=A, half=half)
+        # 11. critic Adam + vl-stat reduce + Polyak target — ONE launch
+        co = u.critic_optimizer
+        e.adam_step(
+            co.space.flat_param, co.space.flat_grad, co.exp_avg,
+            co.exp_avg_sq, co.state3, co.norm_sq, co.lr, co.beta1, co.beta2,
+            co.eps, co.max_norm, do_prep=False, stats_part=self._sp_c,
+            stats_out=self.stats_buf[4:5], part_scale=1.0 / (B * (S - 1)),
+            polyak=self._t_flat, tau=u.TAU,
+        )
+
+    def _body_legacy(self, batch):
         from pdrl_amd.agents.learner_module.compute_loss import soft_update
 
         u, p = self.u, self.params
