@@ -116,7 +116,10 @@ class SeqLSTMCore(nn.Module):
         if x.is_cuda:
             from pdrl_amd import ops
 
-            if ops.available():
+            # the MFMA wgrad kernels stage a (B*S)-entry row-pointer table
+            # in LDS (64 KB) — beyond that, degrade to the eager torch path
+            # instead of raising mid-training (profiles/batch_scaling_r02.md)
+            if ops.available() and x.shape[0] * x.shape[1] <= 8192:
                 return self._forward_fused(x, hx, cx, x2)
         return self._forward_eager(x, hx, cx, x2)
 

@@ -134,6 +134,8 @@ class FusedOnPolicyStep(GraphableStep):
         import os
 
         B, S, _ = batch["obs"].shape
+        if B * S > 8192:
+            return False  # wgrad row-pointer table exceeds LDS
         if self.algo == "V-MPO":
             if self.core.w_ih.size(0) == 64 and bool(
                     int(os.environ.get("PDRL_FWDLOSS", "1"))):
