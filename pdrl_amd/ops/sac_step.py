@@ -66,7 +66,9 @@ class FusedSacStep(GraphableStep):
                 [updater.actor_optimizer, updater.alpha_optimizer])
 
     def fits(self, batch) -> bool:
-        return True  # the loss kernels grid-stride; no LDS shape limit
+        B, S, _ = batch["obs"].shape
+        # wgrad kernels stage a (B*S)-entry row-pointer table in LDS
+        return B * S <= 8192
 
     # ------------------------------------------------------------------ #
     def _multi_setup(self, batch):

@@ -74,7 +74,9 @@ class FusedSacContinuousStep(GraphableStep):
         self._mshape = None
 
     def fits(self, batch) -> bool:
-        return True  # loss kernels grid-stride; no LDS shape limit
+        B, S, _ = batch["obs"].shape
+        # wgrad kernels stage a (B*S)-entry row-pointer table in LDS
+        return B * S <= 8192
 
     # ------------------------------------------------------------------ #
     def _multi_setup(self, batch):
