@@ -15,25 +15,41 @@ from pathlib import Path
 
 
 class SummaryWriter:
-    def __init__(self, log_dir: str):
+    """Writes BOTH a JSONL stream (easy to tail/parse offline) and a native
+    TensorBoard event file (utils/tb_events.py — hand-encoded Event protos,
+    no tensorboardX needed), so ``tensorboard --logdir results/`` renders
+    the curves directly, matching the reference's tooling surface."""
+
+    def __init__(self, log_dir: str, tensorboard: bool = True):
         self.log_dir = Path(log_dir)
         self.log_dir.mkdir(parents=True, exist_ok=True)
         self._path = self.log_dir / "scalars.jsonl"
         self._f = open(self._path, "a", buffering=1)
+        self._tb = None
+        if tensorboard:
+            from .tb_events import EventFileWriter
+
+            self._tb = EventFileWriter(str(self.log_dir))
 
     def add_scalar(self, tag: str, value, step: int):
         rec = {"tag": tag, "value": float(value), "step": int(step), "wall": time.time()}
         self._f.write(json.dumps(rec) + "\n")
+        if self._tb is not None:
+            self._tb.add_scalar(tag, value, step)
 
     def flush(self):
         self._f.flush()
         os.fsync(self._f.fileno())
+        if self._tb is not None:
+            self._tb.flush()
 
     def close(self):
         try:
             self._f.close()
         except Exception:
             pass
+        if self._tb is not None:
+            self._tb.close()
 
     def __del__(self):
         self.close()
