@@ -96,6 +96,10 @@ def main():
     ap.add_argument("--warmup", type=int, default=50)
     ap.add_argument("--batch-size", type=int, default=128)
     ap.add_argument("--seq-len", type=int, default=5)
+    ap.add_argument("--hidden", type=int, default=None,
+                    help="hidden size (default: params.hidden_size=64; "
+                         "64 runs the fully fused loss DAGs, other widths "
+                         "use the fused cores + general loss path)")
     ap.add_argument("--mode", default="resident", choices=["resident", "staged"],
                     help="resident: batch lives on-device (kernel-DAG ceiling); "
                          "staged: ring drain + pinned H2D + step + weight "
@@ -120,6 +124,8 @@ def main():
     params.algo = args.algo
     params.batch_size = args.batch_size
     params.seq_len = args.seq_len
+    if args.hidden is not None:
+        params.hidden_size = args.hidden
     obs_dim, n_act, continuous = env_shape(args.algo)
     params.obs_dim, params.n_actions = obs_dim, n_act
 

@@ -860,3 +860,28 @@ def test_sacc_fused_step_runs_and_learns():
     t_moved = sum(int(not torch.allclose(a, b.detach()))
                   for a, b in zip(t_before, upd.target_critic.parameters()))
     assert t_moved > 0
+
+
+@pytest.mark.parametrize("hidden", [32, 128])
+def test_updater_step_nonflagship_hidden(hidden):
+    """H≠64 widths: the fused SeqLSTM core kernels (templated 32/64/128)
+    drive autograd while the losses take the general path — one update
+    must run and produce finite grads/params."""
+    _ops()
+    from pdrl_amd.agents.learner_module import switch_module
+    from pdrl_amd.utils import load_params
+    from tests.conftest import make_batch
+
+    torch.manual_seed(3)
+    p = load_params()
+    p.algo = "IMPALA"
+    p.batch_size, p.seq_len, p.obs_dim, p.n_actions = 16, 5, 4, 2
+    p.hidden_size = hidden
+    upd_cls, model_cls = switch_module(p.algo)
+    model = model_cls(p.obs_dim, p.n_actions, p.seq_len, hidden)
+    upd = upd_cls(model, p, DEV)
+    batch = make_batch(p, device=DEV)
+    for _ in range(2):
+        stats = upd.step(batch)
+    assert all(torch.isfinite(q).all() for q in model.parameters())
+    assert all(abs(float(v)) < 1e6 for v in stats.values())
