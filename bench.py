@@ -158,13 +158,23 @@ def main():
         # default, where the loop has real idle gaps to absorb the thread.
         wpub = WeightPublisher(actor, device) if pub is not None else None
 
+        # Software-pipelined iteration: the pinned-host fill and the
+        # encode+TCP send overlap the GPU step. Every update's weights are
+        # still broadcast (same wire traffic as the reference's
+        # publish-after-every-update); each payload is SENT during the
+        # following iteration, once its async D2H has completed.
+        state = {"pending": False}
+
         def step_fn():
-            dev_batch = stager.stage(host_np)
+            dev_batch = stager.stage(host_np)   # fill overlaps prev GPU tail
             updater.step(dev_batch)
             if wpub is not None:
-                header, payload = encode(Protocol.Model, wpub.payload(),
-                                         compress=False)
-                pub.send(header, payload)
+                if state["pending"]:
+                    header, payload = encode(Protocol.Model, wpub.finish(),
+                                             compress=False)
+                    pub.send(header, payload)   # overlaps THIS step on GPU
+                wpub.begin()                    # D2H queued after the step
+                state["pending"] = True
     else:
         batch = make_synthetic_batch(params, device, seed=100 + rank)
 

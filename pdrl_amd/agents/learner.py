@@ -114,9 +114,21 @@ class WeightPublisher:
                                    pin_memory=True)
         self._host = self._pinned.numpy()
 
-    def payload(self) -> dict:
-        from pdrl_amd.buffers.wire import pack_weights
+    def begin(self):
+        """Queue the fused gather + async D2H of the current weights and
+        record an event — no host sync. Pair with finish(); finish() MUST
+        be called before the next begin() (single snapshot buffer)."""
+        if not self.use_cuda:
+            return
+        torch._foreach_copy_(self._views, self.params)
+        self._pinned.copy_(self._dev, non_blocking=True)
+        if not hasattr(self, "_ev"):
+            self._ev = torch.cuda.Event()
+        self._ev.record()
 
+    def finish(self) -> dict:
+        """Wait for the begin() D2H and return the packed payload. encode()
+        pickles (copies) the buffer before the next begin() reuses it."""
         if not self.use_cuda:
             return {
                 "wschema": self.schema,
@@ -124,11 +136,12 @@ class WeightPublisher:
                     [p.detach().cpu().numpy().reshape(-1) for p in self.params]
                 ).astype(np.float32, copy=False),
             }
-        torch._foreach_copy_(self._views, self.params)
-        self._pinned.copy_(self._dev, non_blocking=True)
-        torch.cuda.current_stream(self.device).synchronize()
-        # encode() pickles (copies) the buffer before the next publish reuses it
+        self._ev.synchronize()
         return {"wschema": self.schema, "wbuf": self._host}
+
+    def payload(self) -> dict:
+        self.begin()
+        return self.finish()
 
 
 class AsyncWeightPublisher:
