@@ -110,10 +110,11 @@ class Worker:
         heads = list(core.head_names or [])
         rng = torch.tensor([((seed or 0) * 0x9E3779B97F4A7C15 + 0x2545F491) &
                             ((1 << 63) - 1) | 1], dtype=torch.int64)
-        wargs = lambda: (core.body_w.detach(), core.body_b.detach(),  # noqa: E731
-                         core.w_ih.detach(), core.w_hh.detach(),
-                         core.b_g.detach(), core.heads_w.detach(),
-                         core.heads_b.detach())
+        # captured ONCE: in-place load_state_dict keeps the same storages,
+        # so hot-reloads still apply; the C++ kernel only reads data_ptr
+        # (per-call .detach() cost ~12% of the tick in profiling)
+        wargs = (core.body_w, core.body_b, core.w_ih, core.w_hh,
+                 core.b_g, core.heads_w, core.heads_b)
 
         if not self._continuous and heads and heads[0] == "logits":
             A = core.head_dims["logits"]
@@ -122,7 +123,7 @@ class Worker:
                 hx, cx = hxs
                 a, lg, lp, h, c = _cpu_actor.act_batch_discrete(
                     obs.contiguous(), hx.contiguous(), cx.contiguous(),
-                    *wargs(), A, rng)
+                    *wargs, A, rng)
                 return a, lg, lp, (h, c)
 
             return fast
@@ -134,7 +135,7 @@ class Worker:
                 hx, cx = hxs
                 a, lg, lp, h, c = _cpu_actor.act_batch_gaussian(
                     obs.contiguous(), hx.contiguous(), cx.contiguous(),
-                    *wargs(), A, mode, rng)
+                    *wargs, A, mode, rng)
                 return a, lg, lp, (h, c)
 
             return fastc
@@ -249,6 +250,37 @@ class Worker:
             if max_episodes is not None and episodes >= max_episodes:
                 break
 
+    def _make_batch_env(self):
+        """Vectorized native-env physics: ONE C++ call steps all M envs
+        (ops/csrc/cpu_actor.cpp replicates envs/{cartpole,mountain_car}.py
+        dynamics; parity-tested in tests/test_envs.py). Resets stay with
+        the per-env objects (their own RNG); the caller re-syncs the state
+        row after each reset. Returns None for mixed/foreign env types
+        (gymnasium adapter etc.) — the per-env python loop handles those."""
+        try:
+            from pdrl_amd.ops import _cpu_actor
+        except ImportError:
+            return None
+        import numpy as np
+
+        raws = [getattr(e, "env", None) for e in self.envs]
+        names = {type(r).__name__ for r in raws}
+        if names == {"CartPoleEnv"}:
+            fn, dim = _cpu_actor.cartpole_step_batch, 4
+        elif names == {"MountainCarContinuousEnv"}:
+            fn, dim = _cpu_actor.mcc_step_batch, 2
+        else:
+            return None
+        M = len(raws)
+        state = torch.empty(M, dim, dtype=torch.float64)
+        for i, r in enumerate(raws):
+            if getattr(r, "_state", None) is None:
+                return None  # env not reset yet
+            state[i] = torch.from_numpy(np.asarray(r._state))
+        return {"fn": fn, "state": state,
+                "steps": torch.zeros(M, dtype=torch.int64),
+                "max_steps": int(raws[0].MAX_EPISODE_STEPS), "raws": raws}
+
     def _collect_vec(self, max_episodes: int | None = None):
         """Vectorized rollout: M envs, ONE batched model.act per tick. Each
         env keeps its own episode uuid / recurrent-state row / reward
@@ -264,6 +296,7 @@ class Worker:
         p = self.params
         H, M = p.hidden_size, self.num_envs
         obs = torch.cat([e.reset() for e in self.envs], dim=0)  # (M, F)
+        benv = self._make_batch_env()  # after reset: states are live
         hx = torch.zeros(M, H)
         cx = torch.zeros(M, H)
         epi_rew = [0.0] * M
@@ -290,19 +323,41 @@ class Worker:
                 chunk_ids: list = []
             n0 = len(chunk_ids)
             sl = slice(n0, n0 + M)
+            # packing + env stepping both consume numpy views — convert the
+            # act outputs ONCE per tick (torch reshape/float per tick cost
+            # ~20% of the loop in profiling)
+            act_np = action.detach().numpy()
+            if self._continuous:
+                acts = [act_np[i].astype(np.float32, copy=False)
+                        for i in range(M)]
+            else:
+                acts = [int(a) for a in act_np.reshape(-1)]
             chunk_buf[sl, offs[0]:offs[1]] = obs.numpy()
             chunk_buf[sl, offs[1]:offs[2]] = \
-                action.reshape(M, -1).float().numpy()
+                act_np.reshape(M, -1).astype(np.float32, copy=False)
             chunk_buf[sl, offs[3]:offs[4]] = logits.numpy()
-            chunk_buf[sl, offs[4]:offs[5]] = log_prob.reshape(M, -1).numpy()
+            chunk_buf[sl, offs[4]:offs[5]] = log_prob.numpy().reshape(M, -1)
             chunk_buf[sl, offs[5]] = is_fir
             chunk_buf[sl, offs[7]:offs[8]] = hx.numpy()
             chunk_buf[sl, offs[8]:offs[9]] = cx.numpy()
 
             any_done = False
             next_rows = []
+            if benv is not None:
+                act_f32 = torch.from_numpy(
+                    act_np.reshape(M, -1)[:, 0].astype(np.float32))
+                b_obs, b_rew, b_done = benv["fn"](
+                    benv["state"], act_f32, benv["steps"], benv["max_steps"])
+                b_obs_np = b_obs.numpy()
+                b_rew_np = b_rew.numpy()
+                b_done_np = b_done.numpy()
             for i, env in enumerate(self.envs):
-                next_obs, rew, done, _ = env.step(action[i])
+                if benv is not None:
+                    rew = float(b_rew_np[i])
+                    done = bool(b_done_np[i])
+                    next_obs = None  # row read from b_obs_np below
+                else:
+                    next_obs, rew, done, _ = env.step(acts[i])
                 epi_rew[i] += rew
                 chunk_buf[n0 + i, offs[2]] = rew
                 chunk_buf[n0 + i, offs[6]] = float(done)
@@ -314,6 +369,11 @@ class Worker:
                     self.pub_stat(epi_rew[i])
                     episodes += 1
                     next_obs = env.reset()
+                    if benv is not None:
+                        benv["state"][i] = torch.from_numpy(
+                            np.asarray(benv["raws"][i]._state))
+                        benv["steps"][i] = 0
+                        b_obs_np[i] = next_obs.numpy()[0]
                     next_hx[i] = 0.0
                     next_cx[i] = 0.0
                     if self._ou_state is not None:
@@ -324,7 +384,8 @@ class Worker:
                     epi_steps[i] = 0
                 else:
                     is_fir[i] = 0.0
-                next_rows.append(next_obs)
+                if benv is None:
+                    next_rows.append(next_obs)
             n = len(chunk_ids)
             if n >= self.batch_steps or any_done:
                 header, payload = encode(
@@ -335,7 +396,8 @@ class Worker:
                 )
                 self.pub.send(header, payload)
                 chunk_ids = []
-            obs = torch.cat(next_rows, dim=0)
+            obs = (torch.from_numpy(b_obs_np) if benv is not None
+                   else torch.cat(next_rows, dim=0))
             hx, cx = next_hx, next_cx
             if self.heartbeat is not None:
                 self.heartbeat.value = time.time()

@@ -266,9 +266,113 @@ std::vector<at::Tensor> act_batch_gaussian(
   return {action, logits, logp, h_out, c_out};
 }
 
+
+// ------------------------------------------------------------------ //
+// Batched native env physics (vectorized worker fast path): one call
+// steps all M envs, writing obs/rew/done and updating state/steps
+// in-place. Dynamics replicate pdrl_amd/envs/{cartpole,mountain_car}.py
+// exactly (same float64 state, same formulas; parity-tested vs the
+// python envs in tests/test_envs.py). Resets stay in Python (per-env
+// RNG ownership).
+
+// CartPole-v1 (envs/cartpole.py:54-84): state (M,4) f64, act (M) f32
+// (0/1), steps (M) i64; done = terminated || steps >= max_steps.
+std::vector<at::Tensor> cartpole_step_batch(at::Tensor& state,
+                                            const at::Tensor& act,
+                                            at::Tensor& steps, long max_steps) {
+  const long M = state.size(0);
+  auto obs = at::empty({M, 4}, act.options());
+  auto rew = at::empty({M}, act.options());
+  auto done = at::empty({M}, act.options());
+  double* st = state.data_ptr<double>();
+  const float* ac = act.data_ptr<float>();
+  int64_t* sp = steps.data_ptr<int64_t>();
+  float* ob = obs.data_ptr<float>();
+  float* rw = rew.data_ptr<float>();
+  float* dn = done.data_ptr<float>();
+  constexpr double kGrav = 9.8, kMassPole = 0.1, kTotalMass = 1.1;
+  constexpr double kLen = 0.5, kPoleMassLen = 0.05, kForceMag = 10.0;
+  constexpr double kTau = 0.02, kXThr = 2.4;
+  const double kThetaThr = 12.0 * 2.0 * M_PI / 360.0;
+  for (long m = 0; m < M; ++m) {
+    double x = st[m * 4], x_dot = st[m * 4 + 1];
+    double th = st[m * 4 + 2], th_dot = st[m * 4 + 3];
+    const double force = (ac[m] >= 0.5f) ? kForceMag : -kForceMag;
+    const double cos_t = std::cos(th), sin_t = std::sin(th);
+    const double temp =
+        (force + kPoleMassLen * th_dot * th_dot * sin_t) / kTotalMass;
+    const double th_acc =
+        (kGrav * sin_t - cos_t * temp) /
+        (kLen * (4.0 / 3.0 - kMassPole * cos_t * cos_t / kTotalMass));
+    const double x_acc = temp - kPoleMassLen * th_acc * cos_t / kTotalMass;
+    x += kTau * x_dot;
+    x_dot += kTau * x_acc;
+    th += kTau * th_dot;
+    th_dot += kTau * th_acc;
+    st[m * 4] = x;
+    st[m * 4 + 1] = x_dot;
+    st[m * 4 + 2] = th;
+    st[m * 4 + 3] = th_dot;
+    sp[m] += 1;
+    const bool term = (x < -kXThr) || (x > kXThr) || (th < -kThetaThr) ||
+                      (th > kThetaThr);
+    ob[m * 4] = (float)x;
+    ob[m * 4 + 1] = (float)x_dot;
+    ob[m * 4 + 2] = (float)th;
+    ob[m * 4 + 3] = (float)th_dot;
+    rw[m] = 1.0f;
+    dn[m] = (term || sp[m] >= max_steps) ? 1.0f : 0.0f;
+  }
+  return {obs, rew, done};
+}
+
+// MountainCarContinuous-v0 (envs/mountain_car.py:44-64): state (M,2)
+// f64, act (M) f32 in [-1,1].
+std::vector<at::Tensor> mcc_step_batch(at::Tensor& state,
+                                       const at::Tensor& act,
+                                       at::Tensor& steps, long max_steps) {
+  const long M = state.size(0);
+  auto obs = at::empty({M, 2}, act.options());
+  auto rew = at::empty({M}, act.options());
+  auto done = at::empty({M}, act.options());
+  double* st = state.data_ptr<double>();
+  const float* ac = act.data_ptr<float>();
+  int64_t* sp = steps.data_ptr<int64_t>();
+  float* ob = obs.data_ptr<float>();
+  float* rw = rew.data_ptr<float>();
+  float* dn = done.data_ptr<float>();
+  constexpr double kMinPos = -1.2, kMaxPos = 0.6, kMaxSpeed = 0.07;
+  constexpr double kGoalPos = 0.45, kGoalVel = 0.0, kPower = 0.0015;
+  for (long m = 0; m < M; ++m) {
+    double pos = st[m * 2], vel = st[m * 2 + 1];
+    const double force =
+        std::min(std::max((double)ac[m], -1.0), 1.0);
+    vel += force * kPower - 0.0025 * std::cos(3.0 * pos);
+    vel = std::min(std::max(vel, -kMaxSpeed), kMaxSpeed);
+    pos += vel;
+    pos = std::min(std::max(pos, kMinPos), kMaxPos);
+    if (pos <= kMinPos && vel < 0.0) vel = 0.0;
+    st[m * 2] = pos;
+    st[m * 2 + 1] = vel;
+    sp[m] += 1;
+    const bool term = (pos >= kGoalPos) && (vel >= kGoalVel);
+    ob[m * 2] = (float)pos;
+    ob[m * 2 + 1] = (float)vel;
+    double r = -0.1 * force * force;
+    if (term) r += 100.0;
+    rw[m] = (float)r;
+    dn[m] = (term || sp[m] >= max_steps) ? 1.0f : 0.0f;
+  }
+  return {obs, rew, done};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("act_batch_discrete", &act_batch_discrete,
         "batched CPU actor step: body+LSTM+logits+sample in one call");
   m.def("act_batch_gaussian", &act_batch_gaussian,
         "batched CPU actor step for Gaussian policies (PPO-C / SAC-C)");
+  m.def("cartpole_step_batch", &cartpole_step_batch,
+        "vectorized CartPole-v1 physics (native env fast path)");
+  m.def("mcc_step_batch", &mcc_step_batch,
+        "vectorized MountainCarContinuous-v0 physics");
 }

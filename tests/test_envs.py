@@ -265,3 +265,65 @@ def test_vectorized_worker_collects_complete_trajectories():
         assert np.isfinite(tr["obs"].numpy()).all()
     w.close()
     mgr_sub.close()
+
+
+def test_cpp_batched_cartpole_physics_parity():
+    """cartpole_step_batch (cpu_actor.cpp) vs the python env: identical
+    obs/rew/done trajectories from identical states over random actions."""
+    pytest.importorskip("pdrl_amd.ops._cpu_actor")
+    import numpy as np
+    import torch
+    from pdrl_amd.ops import _cpu_actor
+    from pdrl_amd.envs.cartpole import CartPoleEnv
+
+    rng = np.random.default_rng(0)
+    M = 8
+    envs = [CartPoleEnv(seed=i) for i in range(M)]
+    for e in envs:
+        e.reset()
+    state = torch.tensor(np.stack([e._state for e in envs]))
+    steps = torch.zeros(M, dtype=torch.int64)
+    for t in range(300):
+        acts = rng.integers(0, 2, size=M)
+        b_obs, b_rew, b_done = _cpu_actor.cartpole_step_batch(
+            state, torch.tensor(acts, dtype=torch.float32), steps,
+            CartPoleEnv.MAX_EPISODE_STEPS)
+        for i, e in enumerate(envs):
+            obs, rew, term, trunc, _ = e.step(int(acts[i]))
+            np.testing.assert_allclose(b_obs[i].numpy(), obs, rtol=1e-6)
+            assert float(b_rew[i]) == rew
+            assert bool(b_done[i]) == bool(term or trunc)
+            if term or trunc:
+                e.reset()
+                state[i] = torch.tensor(e._state)
+                steps[i] = 0
+
+
+def test_cpp_batched_mcc_physics_parity():
+    pytest.importorskip("pdrl_amd.ops._cpu_actor")
+    import numpy as np
+    import torch
+    from pdrl_amd.ops import _cpu_actor
+    from pdrl_amd.envs.mountain_car import MountainCarContinuousEnv as MCC
+
+    rng = np.random.default_rng(1)
+    M = 4
+    envs = [MCC(seed=i) for i in range(M)]
+    for e in envs:
+        e.reset()
+    state = torch.tensor(np.stack([e._state for e in envs]))
+    steps = torch.zeros(M, dtype=torch.int64)
+    for t in range(400):
+        acts = rng.uniform(-1, 1, size=M).astype(np.float32)
+        b_obs, b_rew, b_done = _cpu_actor.mcc_step_batch(
+            state, torch.tensor(acts), steps, MCC.MAX_EPISODE_STEPS)
+        for i, e in enumerate(envs):
+            obs, rew, term, trunc, _ = e.step(acts[i:i + 1])
+            np.testing.assert_allclose(b_obs[i].numpy(), obs, rtol=1e-6,
+                                       atol=1e-7)
+            np.testing.assert_allclose(float(b_rew[i]), rew, rtol=1e-5)
+            assert bool(b_done[i]) == bool(term or trunc)
+            if term or trunc:
+                e.reset()
+                state[i] = torch.tensor(e._state)
+                steps[i] = 0
