@@ -201,8 +201,16 @@ class Worker:
 
     # ------------------------------------------------------------------ #
     def collect(self, max_episodes: int | None = None):
-        """Roll out episodes forever (or for max_episodes, for tests)."""
-        if self.num_envs > 1:
+        """Roll out episodes forever (or for max_episodes, for tests).
+
+        All env counts route through the vectorized loop (packed-chunk
+        records + C++ batched physics — at M=1 it is simply faster than
+        the per-step dict path below, which is kept for reference/debug
+        via PDRL_SCALAR_WORKER=1)."""
+        import os
+
+        if self.num_envs > 1 or not bool(
+                int(os.environ.get("PDRL_SCALAR_WORKER", "0"))):
             return self._collect_vec(max_episodes)
         H = self.params.hidden_size
         episodes = 0
@@ -403,6 +411,8 @@ class Worker:
                 self.heartbeat.value = time.time()
             if self.step_sleep > 0:
                 time.sleep(self.step_sleep)
+            else:
+                time.sleep(0)  # GIL handoff for in-process (thread) fleets
             if max_episodes is not None and episodes >= max_episodes:
                 break
 

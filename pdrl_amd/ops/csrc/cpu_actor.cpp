@@ -14,6 +14,7 @@
 //
 // Vector-friendly loop order: inner loops stream rows of the transposed
 // weights (unit stride) so -O3 autovectorizes them; no torch ops inside.
+#include <pybind11/pybind11.h>
 #include <torch/extension.h>
 
 #include <cmath>
@@ -56,6 +57,9 @@ std::vector<at::Tensor> act_batch_discrete(
   auto logp = at::empty({M, 1}, opt);
   auto h_out = at::empty({M, H}, opt);
   auto c_out = at::empty({M, H}, opt);
+  // release the GIL for the compute: worker threads (tests / in-process
+  // fleets) must not starve the python threads sharing the interpreter
+  pybind11::gil_scoped_release nogil;
 
   const float* ob = obs.data_ptr<float>();
   const float* hp = hx.data_ptr<float>();
@@ -172,6 +176,9 @@ std::vector<at::Tensor> act_batch_gaussian(
   auto logp = at::empty({M, 1}, opt);
   auto h_out = at::empty({M, H}, opt);
   auto c_out = at::empty({M, H}, opt);
+  // release the GIL for the compute: worker threads (tests / in-process
+  // fleets) must not starve the python threads sharing the interpreter
+  pybind11::gil_scoped_release nogil;
 
   const float* ob = obs.data_ptr<float>();
   const float* hp = hx.data_ptr<float>();
@@ -284,6 +291,7 @@ std::vector<at::Tensor> cartpole_step_batch(at::Tensor& state,
   auto obs = at::empty({M, 4}, act.options());
   auto rew = at::empty({M}, act.options());
   auto done = at::empty({M}, act.options());
+  pybind11::gil_scoped_release nogil;
   double* st = state.data_ptr<double>();
   const float* ac = act.data_ptr<float>();
   int64_t* sp = steps.data_ptr<int64_t>();
@@ -335,6 +343,7 @@ std::vector<at::Tensor> mcc_step_batch(at::Tensor& state,
   auto obs = at::empty({M, 2}, act.options());
   auto rew = at::empty({M}, act.options());
   auto done = at::empty({M}, act.options());
+  pybind11::gil_scoped_release nogil;
   double* st = state.data_ptr<double>();
   const float* ac = act.data_ptr<float>();
   int64_t* sp = steps.data_ptr<int64_t>();
