@@ -51,6 +51,10 @@ def pipeline_params(params):
     params.obs_dim, params.n_actions, params.continuous = 4, 2, False
     params.batch_size = 4
     params.seq_len = 5
+    # everything runs as THREADS here (prod uses processes): give the
+    # worker loops a real sleep so the manager/storage/learner threads
+    # are scheduled — the optimized tick otherwise outpaces GIL handoffs
+    params.worker_step_sleep = 0.002
     return params
 
 
