@@ -198,3 +198,27 @@ def test_learner_resume_from_checkpoint(pipeline_params, tmp_path):
                       lrn.updater.trainable_modules()["model"].parameters()):
         torch.testing.assert_close(p1.detach(), p2.detach())
     lrn.close()
+
+
+def test_scalar_worker_path_still_works(pipeline_params, monkeypatch):
+    """The per-step dict worker loop (PDRL_SCALAR_WORKER=1 debug path)
+    keeps producing decodable rollout messages."""
+    import torch
+
+    from pdrl_amd.agents import Worker
+    from pdrl_amd.networks import MlpLSTMSingle
+    from pdrl_amd.transport import Endpoint
+
+    monkeypatch.setenv("PDRL_SCALAR_WORKER", "1")
+    p = pipeline_params
+    model = MlpLSTMSingle(p.obs_dim, p.n_actions, p.seq_len, p.hidden_size)
+    sub = Endpoint(bind=("127.0.0.1", 0))
+    w = Worker(model, 0, "127.0.0.1", sub.bound_port, "127.0.0.1", 1, p,
+               seed=0)
+    w.collect(max_episodes=2)
+    n = 0
+    while sub.recv(timeout=0.3) is not None:
+        n += 1
+    w.close()
+    sub.close()
+    assert n >= 2  # at least per-episode flushes arrived
