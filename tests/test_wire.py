@@ -94,3 +94,25 @@ def test_packed_weights_roundtrip():
     dst.actor.load_state_dict(unpack_weights(obj))
     for a, b in zip(src.actor.parameters(), dst.actor.parameters()):
         torch.testing.assert_close(a, b)
+
+
+def test_weight_publisher_begin_finish_cpu_roundtrip():
+    """WeightPublisher.begin()/finish() (the staged-mode pipelined split)
+    on CPU: payload unpacks back into the exact state_dict."""
+    import torch
+
+    from pdrl_amd.agents.learner import WeightPublisher
+    from pdrl_amd.buffers.wire import is_packed_weights, unpack_weights
+    from pdrl_amd.networks import MlpLSTMSingle
+
+    torch.manual_seed(0)
+    m = MlpLSTMSingle(4, 2, 5, 64)
+    pub = WeightPublisher(m.actor, "cpu")
+    pub.begin()  # no-op on CPU, must not raise
+    payload = pub.finish()
+    assert is_packed_weights(payload)
+    sd = unpack_weights(payload)
+    ref = m.actor.state_dict()
+    assert set(sd) == set(ref)
+    for k in ref:
+        torch.testing.assert_close(sd[k], ref[k])
