@@ -58,3 +58,19 @@ def test_summary_writer_emits_both(tmp_path):
     assert len(files) == 1
     got = [parse_scalar(r) for r in read_records(files[0])[1:]]
     assert got == [("a", 1.5, 0), ("a", 2.5, 1)]
+
+
+def test_varint_large_step_roundtrip(tmp_path):
+    """Steps beyond 2^32 must survive the varint encoding."""
+    from pdrl_amd.utils.tb_events import (EventFileWriter, parse_scalar,
+                                          read_records)
+    import glob
+
+    w = EventFileWriter(str(tmp_path))
+    big = (1 << 40) + 12345
+    w.add_scalar("x", -1.0, big)
+    w.close()
+    f = glob.glob(str(tmp_path / "events.out.tfevents.*"))[0]
+    tag, v, s = parse_scalar(read_records(f)[1])
+    assert (tag, s) == ("x", big)
+    assert abs(v + 1.0) < 1e-6
