@@ -885,3 +885,26 @@ def test_updater_step_nonflagship_hidden(hidden):
         stats = upd.step(batch)
     assert all(torch.isfinite(q).all() for q in model.parameters())
     assert all(abs(float(v)) < 1e6 for v in stats.values())
+
+
+def test_weight_publisher_begin_finish_gpu_roundtrip():
+    """WeightPublisher.begin()/finish() (the staged-bench pipelined split):
+    payload matches the weights captured at begin() even if they change
+    before finish()."""
+    _ops()
+    from pdrl_amd.agents.learner import WeightPublisher
+    from pdrl_amd.buffers.wire import unpack_weights
+    from pdrl_amd.networks import MlpLSTMSingle
+
+    torch.manual_seed(2)
+    m = MlpLSTMSingle(4, 2, 5, 64).to(DEV)
+    pub = WeightPublisher(m.actor, DEV)
+    ref = {k: v.detach().clone() for k, v in m.actor.state_dict().items()}
+    pub.begin()
+    # mutate AFTER begin: the D2H snapshot must hold the begin-time values
+    with torch.no_grad():
+        for q in m.actor.parameters():
+            q.add_(1.0)
+    sd = unpack_weights(pub.finish())
+    for k in ref:
+        torch.testing.assert_close(sd[k], ref[k].cpu())
