@@ -1,22 +1,42 @@
-import sys, time, multiprocessing as mp
+"""Actor-fleet ingest ceiling: N worker PROCESSES (each M vectorized envs,
+C++ act + C++ batched env physics) collecting flat-out into local sinks.
+
+    python scripts/fleet_bench.py [N] [--envs M] [--seconds S] [--algo A]
+
+Measured end of round 2 on the MI355X box: 16 workers x 4 envs =
+1.35 M env-steps/s aggregate (84 K/worker); 32 workers saturate the
+box's cores at ~1.21 M. (Reference actor plane: <= 600 steps/s
+cluster-wide — its workers sleep 50 ms per step by design.)
+"""
+from __future__ import annotations
+
+import argparse
+import multiprocessing as mp
+import sys
+import time
 from pathlib import Path
+
 sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 
-def run_worker(idx, seconds, counter):
-    import torch
-    torch.set_num_threads(1)
+def run_worker(idx, seconds, envs, algo, env_name, counter):
     import threading
+
+    import torch
+
+    torch.set_num_threads(1)
+    import main as main_mod
     from pdrl_amd.agents import Worker
     from pdrl_amd.transport import Endpoint
     from pdrl_amd.utils import load_params
-    import main as main_mod
-    p = load_params(); p.algo = "IMPALA"; p.env = "CartPole-v1"
-    p.num_envs_per_worker = 4
+
+    p = load_params()
+    p.algo, p.env = algo, env_name
+    p.num_envs_per_worker = envs
     main_mod.probe_env_spaces(p)
     model = main_mod.build_model(p)
-    sub = Endpoint(bind=("127.0.0.1", 0))
-    w = Worker(model, idx, "127.0.0.1", sub.bound_port, "127.0.0.1", 1, p,
+    sink = Endpoint(bind=("127.0.0.1", 0))
+    w = Worker(model, idx, "127.0.0.1", sink.bound_port, "127.0.0.1", 1, p,
                seed=idx)
     stop = threading.Event()
     w.stop_event = stop
@@ -26,17 +46,33 @@ def run_worker(idx, seconds, counter):
     counter.value = w._total_steps
 
 
-if __name__ == "__main__":
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("workers", type=int, nargs="?", default=16)
+    ap.add_argument("--envs", type=int, default=4)
+    ap.add_argument("--seconds", type=float, default=8.0)
+    ap.add_argument("--algo", default="IMPALA")
+    ap.add_argument("--env", default="CartPole-v1")
+    args = ap.parse_args()
+
     mp.set_start_method("spawn", force=True)
-    N = int(sys.argv[1]) if len(sys.argv) > 1 else 16
-    secs = 8
-    cs = [mp.Value("l", 0) for _ in range(N)]
-    ps = [mp.Process(target=run_worker, args=(i, secs, cs[i]))
-          for i in range(N)]
-    for p_ in ps:
+    counters = [mp.Value("l", 0) for _ in range(args.workers)]
+    procs = [
+        mp.Process(target=run_worker,
+                   args=(i, args.seconds, args.envs, args.algo, args.env,
+                         counters[i]))
+        for i in range(args.workers)
+    ]
+    for p_ in procs:
         p_.start()
-    for p_ in ps:
-        p_.join(timeout=90)
-    tot = sum(c.value for c in cs)
-    print(f"{N} workers x M=4: {tot} steps in {secs}s -> "
-          f"{tot/secs:.0f} steps/s aggregate ({tot/secs/N:.0f}/worker)")
+    for p_ in procs:
+        p_.join(timeout=max(90.0, args.seconds * 4))
+    total = sum(c.value for c in counters)
+    rate = total / args.seconds
+    print(f"{args.workers} workers x M={args.envs}: {total} steps in "
+          f"{args.seconds:.0f}s -> {rate:.0f} steps/s aggregate "
+          f"({rate / args.workers:.0f}/worker)")
+
+
+if __name__ == "__main__":
+    main()
