@@ -44,3 +44,12 @@ def test_bench_multirank_self_exec_reports_true_world():
     assert rec["n_gpus"] == 2
     assert rec["config"]["parallelism"] == "dp2"
     assert rec["config"]["global_batch"] == 256
+
+
+def test_bench_staged_mode_cpu():
+    """--mode staged (pinned fill + step + pipelined weight publish) runs
+    on CPU and reports the same schema with mode labeled."""
+    rec = _run_bench("--mode", "staged", "--steps", "3", "--warmup", "1")
+    assert rec["config"]["mode"] == "staged"
+    assert rec["metric"] == "learner_env_steps_per_sec"
+    assert rec["value"] > 0
