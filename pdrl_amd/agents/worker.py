@@ -58,6 +58,13 @@ class Worker:
         self.pub = pub_connect(manager_ip, manager_port)
         self.sub = sub_connect(learner_ip, learner_port + 1)
         self.step_sleep = float(getattr(params, "worker_step_sleep", 0.0))
+        # optional ingest throttle (env-steps/s per worker, 0 = unthrottled):
+        # the C++ actor plane is fast enough that FULL-rate fleets can push
+        # on-policy algorithms into policy-lag instability (see
+        # profiles/training_runs/README.md) — this caps the rate directly
+        # instead of forcing fleet resizing
+        self.max_rate = float(getattr(params, "worker_max_steps_per_sec",
+                                      0.0))
         # steps per wire message: one decode per chunk instead of per step
         # (a per-step wire protocol capped the single storage process at
         # ~10K steps/s; seq_len-sized chunks amortize pickle+zlib 5×)
@@ -314,6 +321,7 @@ class Worker:
         episodes = 0
 
         chunk_buf = None  # built lazily once the record widths are known
+        t_start = time.perf_counter()
         while not self._stopped():
             self.poll_model()
             action, logits, log_prob, (next_hx, next_cx) = self._act(obs, (hx, cx))
@@ -411,6 +419,12 @@ class Worker:
                 self.heartbeat.value = time.time()
             if self.step_sleep > 0:
                 time.sleep(self.step_sleep)
+            elif self.max_rate > 0:
+                # token-bucket pacing toward max_rate env-steps/s
+                target = self._total_steps / self.max_rate
+                elapsed = time.perf_counter() - t_start
+                if target > elapsed:
+                    time.sleep(min(target - elapsed, 0.05))
             else:
                 time.sleep(0)  # GIL handoff for in-process (thread) fleets
             if max_episodes is not None and episodes >= max_episodes:

@@ -327,3 +327,37 @@ def test_cpp_batched_mcc_physics_parity():
                 e.reset()
                 state[i] = torch.tensor(e._state)
                 steps[i] = 0
+
+
+def test_worker_rate_throttle():
+    """worker_max_steps_per_sec caps the collection rate (token bucket)."""
+    import time
+
+    import torch
+
+    import main as main_mod
+    from pdrl_amd.agents import Worker
+    from pdrl_amd.transport import Endpoint
+    from pdrl_amd.utils import load_params
+
+    p = load_params()
+    p.algo, p.env = "IMPALA", "CartPole-v1"
+    p.worker_max_steps_per_sec = 400.0
+    main_mod.probe_env_spaces(p)
+    model = main_mod.build_model(p)
+    sink = Endpoint(bind=("127.0.0.1", 0))
+    w = Worker(model, 0, "127.0.0.1", sink.bound_port, "127.0.0.1", 1, p,
+               seed=0)
+    import threading
+    stop = threading.Event()
+    w.stop_event = stop
+    threading.Thread(target=lambda: (time.sleep(1.0), stop.set()),
+                     daemon=True).start()
+    t0 = time.perf_counter()
+    w.collect()
+    dt = time.perf_counter() - t0
+    rate = w._total_steps / dt
+    assert rate <= 700.0, rate  # capped (generous margin for timing noise)
+    assert rate >= 100.0, rate  # but still collecting
+    w.close()
+    sink.close()
