@@ -1,9 +1,21 @@
 """FusedSacStep: the SAC (discrete) training iteration as a fixed HIP kernel
 DAG — the discrete half of K11, hipGraph-capturable.
 
-Follows the EAGER reference ordering exactly
-(agents/learner_module/sac/learning.py), in 11 launches (launch latency IS
-the step time at this model size — profiles/algo_breakdown_r02a.md):
+Default path (_body8, PDRL_SAC8=1): EIGHT launches — launch latency IS the
+step time at this size (profiles/algo_breakdown_r02c.md, 103 µs/step):
+  1. 5-network forward (actor + twin critics + twin targets)
+  2. actor loss (row-local dlogits) + actor BPTT          [sac_actor_bwd]
+  3. actor MFMA wgrad + loss reduce + shared Adam clock   [sac_actor_wgrad]
+  4. actor + alpha Adam (one multi-group launch)
+  5. post-update actor forward + critic loss       [sac_fwd2_critic_loss]
+  6. twin-critic BPTT (multi)   7. twin-critic MFMA wgrads (multi)
+  8. critic Adam + value-loss reduce + Polyak target (adam_step extras;
+     the target critic's params live in a flat buffer aligned with
+     critic_optimizer.flat_param)
+
+Legacy 10-launch sequence (_body_legacy, also the multi-rank path),
+following the EAGER reference ordering exactly
+(agents/learner_module/sac/learning.py):
   1.  actor + twin critics + twin TARGET critics — ONE 5-network launch
       (target params are constant until step 9, so their forward commutes
       with the actor update)

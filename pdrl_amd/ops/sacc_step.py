@@ -1,10 +1,18 @@
 """FusedSacContinuousStep: the SAC-Continuous training iteration as a fixed
 HIP kernel DAG — the continuous half of K11,
-hipGraph-capturable, ~18 launches per step (the round-1 version ran ~95:
-eager encoder GEMMs/cat/clamp/pow around the fused kernels —
-profiles/algo_breakdown_r02a.md).
+hipGraph-capturable.
 
-Follows the EAGER reference ordering exactly
+Default path (_body11, PDRL_SAC8=1): ELEVEN launches
+(profiles/algo_breakdown_r02c.md, 137 µs/step) — sampling rides the actor
+forwards (+ dQ/da zeroing / behaviour-action staging), the min-critic
+masks ride the twin input-grad backward, the analytic actor grad rides
+the actor BPTT (cross-row reduce + Adam clock on the wgrad extra block,
+reusing the discrete reduce verbatim), the soft-Q critic loss rides the
+twin-critic BPTT, and the value-loss reduce + Polyak ride the critic
+Adam. (Round-1 eager version ran ~95 launches; the first fused DAG 18.)
+
+Legacy 18-launch sequence (_body_legacy, also the multi-rank path),
+following the EAGER reference ordering exactly
 (agents/learner_module/sac_continuous/learning.py, reference
 sac_continuous/learning.py:13-151):
   1.  actor fwd → reparameterized tanh-Gaussian sample (sacc_sample:
