@@ -1,7 +1,14 @@
 // Fused V-MPO loss for CDNA4 (gfx950) — K10 of SURVEY.md §2.4.
 //
-// One single-block launch computes the ENTIRE V-MPO loss and its analytic
-// backward (reference math: agents/learner_module/v_mpo/learning.py:49-124):
+// TWO paths share the math here (reference:
+// agents/learner_module/v_mpo/learning.py:49-124):
+//  * split-phase (default, PDRL_FWDLOSS=1): the row-local phases A/B ride
+//    the forward launch (vmpo_pre_row in loss_row.h via fwd_loss.hip) and
+//    phase E rides the backward launch (vmpo_grad_row); only the cross-row
+//    work runs here, in the slim single-block vmpo_mid_kernel (radix-256
+//    top-half selection + psi softmax + duals; LDS 2·BT·4 B ⇒ BT ≤ 7168).
+//  * vmpo_loss_mega_kernel (fallback): ONE single-block launch computes
+//    the ENTIRE loss and its analytic backward:
 //   A. log-softmax stats per (b,t)
 //   B. GAE advantages + TD targets (per-row scan)
 //   C. top-half advantage selection: exact k-th-largest via monotonic

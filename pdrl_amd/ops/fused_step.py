@@ -1,23 +1,30 @@
-"""FusedOnPolicyStep: the entire IMPALA/PPO training iteration as a fixed
-~10-kernel HIP DAG, bypassing autograd, with optional hipGraph capture.
+"""FusedOnPolicyStep: the entire IMPALA/PPO/PPO-C/V-MPO training iteration
+as a fixed HIP kernel DAG, bypassing autograd, with optional hipGraph
+capture.
 
-Per step (kernel sequence, one stream):
-  1. seq_lstm_forward      — body+LSTM+heads, packed out (B,S,D)
-  2. cat_stats             — log pi(a), entropy, lse
-  3. vtrace | ppo_td_gae   — return/advantage scan (reads value col of 1.)
-  4. *_loss_reduce         — device stats vector (loss parts + monitors)
-  5. *_loss_bwd            — analytic dlogits/dvalue into packed gouts
-  6. seq_lstm_backward_core— BPTT per batch row → dgates/dxb
-  7. seq_lstm_wgrad_out    — MFMA weight-grad GEMMs written DIRECTLY into the
-                             flat grad buffer's parameter views (no zero_grad
-                             needed: every grad is fully overwritten)
-  8. [RCCL all-reduce of the flat grad bucket]      (world > 1)
-  9. l2norm_sq + rmsprop   — fused clip + optimizer update
+Default path for the H=64 family (_try_fwdloss, PDRL_FWDLOSS=1) — FOUR
+launches (47 µs/step, profiles/algo_breakdown_r02c.md):
+  1. seq_lstm_fwd_loss  — forward + the row-local loss (V-trace/GAE scans,
+                          analytic head grads, per-row stat partials); for
+                          V-MPO, the row-local log-softmax + GAE phases
+  2. [vmpo_mid]         — V-MPO only: slim single-block cross-row kernel
+                          (radix-256 top-half selection, psi softmax,
+                          eta/alpha duals)
+  3. seq_lstm_bwd_fin   — stat-partial reduce (block 0) + BPTT per row;
+                          for V-MPO, the analytic grad emission per row
+  4. seq_lstm_wgrad_out — MFMA weight-grad GEMMs written DIRECTLY into the
+                          flat grad buffer's parameter views
+  5. l2norm_sq + rmsprop/adam — fused clip + update
+     [world > 1: RCCL all-reduce of the flat bucket before the update]
+
+Legacy sequence (PDRL_FWDLOSS=0 or H≠64): separate forward, mega loss
+kernel (or 4-kernel loss split beyond its LDS cap), backward, wgrad.
 
 No host syncs anywhere; stats are read back only at the log interval.
-With hipGraph capture (`use_graph`), steps 1-9 replay as one graph launch —
-the launch-overhead answer to the reference's ~hundreds of eager dispatches
-per iteration.
+With hipGraph capture (`use_graph`), the whole step replays as one graph
+launch — single-rank; multi-rank splits into two graphs around the
+stream-ordered collective (see run()). The launch-overhead answer to the
+reference's ~hundreds of eager dispatches per iteration.
 """
 from __future__ import annotations
 
