@@ -58,8 +58,7 @@ class LearnerStorage:
                         steps = unpack_steps(data)
                     else:  # plain step dict(s) — compatibility path
                         steps = data if isinstance(data, list) else [data]
-                    for step in steps:
-                        await self.assembler.push(step)
+                    await self.assembler.push_many(steps)
                     self.n_ingested += len(steps)
                 elif protocol is Protocol.Stat:
                     if self.shared_stat is not None:

@@ -21,6 +21,7 @@ from __future__ import annotations
 import asyncio
 import time
 
+import numpy as np
 import torch
 
 from .trajectory import Trajectory
@@ -40,13 +41,27 @@ REQUIRED_KEYS = {
 
 
 def stack_trajectory(steps: list[dict]) -> dict[str, torch.Tensor]:
-    """[{field: tensor(feat)} x seq] → {field: tensor(seq, feat)}."""
+    """[{field: array(feat)} x seq] → {field: tensor(seq, feat)}.
+
+    numpy-first: packed-chunk fields arrive as float32 numpy views, so one
+    np.stack + zero-copy from_numpy per field replaces per-(step, field)
+    torch.as_tensor/reshape calls — stacking was the measured ingest-shard
+    ceiling (~75% of decode+assemble time) once the actor plane went
+    native. Falls back to the generic torch path for exotic inputs."""
     out = {}
     for key in steps[0]:
         if key == "id":
             continue
-        vals = [torch.as_tensor(s[key], dtype=torch.float32).reshape(-1) for s in steps]
-        out[key] = torch.stack(vals, dim=0)
+        v0 = steps[0][key]
+        if isinstance(v0, np.ndarray) and v0.dtype == np.float32:
+            out[key] = torch.from_numpy(np.stack([s[key] for s in steps]))
+        elif isinstance(v0, float):
+            out[key] = torch.from_numpy(
+                np.array([[s[key]] for s in steps], dtype=np.float32))
+        else:
+            vals = [torch.as_tensor(s[key], dtype=torch.float32).reshape(-1)
+                    for s in steps]
+            out[key] = torch.stack(vals, dim=0)
     return out
 
 
@@ -60,9 +75,19 @@ class RolloutAssembler:
         self._next_evict = 0.0  # staleness sweep is amortized (time-based)
 
     async def push(self, step: dict):
-        missing = REQUIRED_KEYS - set(step)
+        await self.push_many((step,))
+
+    async def push_many(self, steps):
+        """Batch push: one coroutine per CHUNK instead of per step (the
+        per-step coroutine dispatch measurably capped the storage shard's
+        ingest rate once the actor plane went native). Field presence is
+        asserted on the first step only — packed chunks (wire.unpack_steps)
+        build every row with the same keys."""
+        if not steps:
+            return
+        missing = REQUIRED_KEYS - set(steps[0] if isinstance(steps, (list, tuple))
+                                      else next(iter(steps)))
         assert not missing, f"rollout step missing fields: {missing}"
-        eid = step["id"]
 
         # staleness eviction of partial trajectories (swept at most every
         # stale_s/4 — a per-push scan of the active dict measurably taxed
@@ -73,27 +98,32 @@ class RolloutAssembler:
             for k in [k for k, tr in self.active.items() if tr.age > self.stale_s]:
                 del self.active[k]
 
-        if eid in self.active:
-            traj = self.active[eid]
-        elif self.parked_done:
-            # splice a fresh episode onto the smallest parked finished one
-            smallest = min(self.parked_done, key=lambda k: len(self.parked_done[k]))
-            traj = self.parked_done.pop(smallest)
-            step = dict(step)
-            step["is_fir"] = 1.0
-            self.active[eid] = traj
-        else:
-            traj = Trajectory(self.seq_len)
-            self.active[eid] = traj
+        active = self.active
+        parked = self.parked_done
+        seq_len = self.seq_len
+        for step in steps:
+            eid = step["id"]
+            traj = active.get(eid)
+            if traj is None:
+                if parked:
+                    # splice a fresh episode onto the smallest parked
+                    # finished one (reference semantics)
+                    smallest = min(parked, key=lambda k: len(parked[k]))
+                    traj = parked.pop(smallest)
+                    step = dict(step)
+                    step["is_fir"] = 1.0
+                else:
+                    traj = Trajectory(seq_len)
+                active[eid] = traj
 
-        traj.append(step)
+            traj.append(step)
 
-        if len(traj) == self.seq_len:
-            del self.active[eid]
-            await self.out_queue.put(stack_trajectory(traj.steps))
-        elif step.get("done", False):
-            self.active.pop(eid, None)
-            self.parked_done[eid] = traj
+            if len(traj) == seq_len:
+                del active[eid]
+                await self.out_queue.put(stack_trajectory(traj.steps))
+            elif step.get("done", False):
+                active.pop(eid, None)
+                parked[eid] = traj
 
     async def pop(self) -> dict[str, torch.Tensor]:
         return await self.out_queue.get()
