@@ -71,14 +71,22 @@ class LearnerStorage:
                 self.heartbeat.value = time.time()
 
     async def store_task(self):
-        """Assembler → shared ring."""
+        """Assembler → shared ring. Drains in bursts: one await per WAKE,
+        then get_nowait until empty (per-trajectory awaits measurably
+        capped the shard once decode/stacking got fast)."""
+        q = self.assembler.out_queue
         while not self._stopped():
             try:
-                traj = await asyncio.wait_for(self.assembler.pop(), timeout=0.5)
+                traj = await asyncio.wait_for(q.get(), timeout=0.5)
             except asyncio.TimeoutError:
                 continue
-            self.ring.put(traj)
-            self.n_stored += 1
+            while True:
+                self.ring.put(traj)
+                self.n_stored += 1
+                try:
+                    traj = q.get_nowait()
+                except asyncio.QueueEmpty:
+                    break
 
     async def chain(self):
         await asyncio.gather(self.ingest_task(), self.store_task())

@@ -48,11 +48,17 @@ def stack_trajectory(steps: list[dict]) -> dict[str, torch.Tensor]:
     torch.as_tensor/reshape calls — stacking was the measured ingest-shard
     ceiling (~75% of decode+assemble time) once the actor plane went
     native. Falls back to the generic torch path for exotic inputs."""
+    s0 = steps[0]
+    if "_row" in s0 and all("_row" in s for s in steps):
+        # packed-chunk fast path: ONE stack of the full rows, fields are
+        # zero-copy column views of the stacked (seq, W) matrix
+        mat = torch.from_numpy(np.stack([s["_row"] for s in steps]))
+        return {k: mat[:, lo:hi] for k, (lo, hi) in s0["_offs"].items()}
     out = {}
-    for key in steps[0]:
-        if key == "id":
+    for key in s0:
+        if key in ("id", "_row", "_offs"):
             continue
-        v0 = steps[0][key]
+        v0 = s0[key]
         if isinstance(v0, np.ndarray) and v0.dtype == np.float32:
             out[key] = torch.from_numpy(np.stack([s[key] for s in steps]))
         elif isinstance(v0, float):
@@ -112,6 +118,11 @@ class RolloutAssembler:
                     traj = parked.pop(smallest)
                     step = dict(step)
                     step["is_fir"] = 1.0
+                    if "_row" in step:  # the stacked row must agree
+                        row = step["_row"].copy()
+                        lo, _hi = step["_offs"]["is_fir"]
+                        row[lo] = 1.0
+                        step["_row"] = row
                 else:
                     traj = Trajectory(seq_len)
                 active[eid] = traj

@@ -38,10 +38,15 @@ def pack_steps(steps: list[dict]) -> dict:
 
 def unpack_steps(obj: dict) -> list[dict]:
     """Inverse of pack_steps. Field values are VIEWS into the chunk matrix
-    (zero copies; the assembler stacks them into owned trajectory tensors)."""
+    (zero copies; the assembler stacks them into owned trajectory tensors).
+    Each step also carries its FULL row (``_row``) and the shared field
+    span map (``_offs``) so stack_trajectory can stack a whole trajectory
+    with ONE np.stack and slice fields zero-copy."""
     pk = obj["pk"]
     widths = obj["widths"]
     offs = np.cumsum([0] + list(widths))
+    span = {k: (int(offs[j]), int(offs[j + 1]))
+            for j, k in enumerate(FIELD_ORDER)}
     out = []
     for i, eid in enumerate(obj["ids"]):
         row = pk[i]
@@ -54,6 +59,8 @@ def unpack_steps(obj: dict) -> list[dict]:
         step["done"] = float(step["done"][0])
         step["is_fir"] = float(step["is_fir"][0])
         step["id"] = eid
+        step["_row"] = row
+        step["_offs"] = span
         out.append(step)
     return out
 
