@@ -188,10 +188,10 @@ def test_worker_ou_warmup_exploration():
             steps = unpack_steps(data) if is_packed(data) else data
             for step in steps:
                 acts.append(float(step["act"][0]))
-                assert set(step) == {
+                assert set(step) >= {
                     "obs", "act", "rew", "logits", "log_prob", "is_fir",
                     "done", "hx", "cx", "id",
-                }
+                }  # (+ _row/_offs fast-stack carriers on packed chunks)
                 assert np.isfinite(step["log_prob"]).all()
     assert len(acts) >= 50
     a = np.array(acts)
