@@ -162,3 +162,23 @@ def test_checkpoint_persists_loose_duals(params, tmp_path):
     vmpo2.load(tmp_path / "V-MPO_1.pt")
     assert float(vmpo2.log_eta) == pytest.approx(0.777)
     assert float(vmpo2.log_alpha) == pytest.approx(-0.333)
+
+
+@pytest.mark.parametrize("B,S", [(3, 3), (7, 9), (1, 5)])
+@pytest.mark.parametrize("algo", ["PPO", "IMPALA", "V-MPO", "SAC"])
+def test_updaters_odd_shapes(algo, B, S, params):
+    """Non-default batch/sequence shapes step without shape assumptions
+    leaking (the GPU fused paths gate on shape; the CPU reference path
+    must be shape-agnostic)."""
+    from pdrl_amd.agents.learner_module import switch_module
+
+    torch.manual_seed(1)
+    params.algo = algo
+    params.batch_size, params.seq_len = B, S
+    params.obs_dim, params.n_actions = 4, 2
+    upd_cls, model_cls = switch_module(algo)
+    model = model_cls(4, 2, S, params.hidden_size)
+    upd = upd_cls(model, params, "cpu")
+    batch = make_batch(params)
+    stats = upd.step(batch)
+    assert all(np.isfinite(float(v)) for v in stats.values()), (algo, B, S)
