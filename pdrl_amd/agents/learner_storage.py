@@ -54,8 +54,8 @@ class LearnerStorage:
             for msg in msgs:
                 protocol, data = decode(*msg)
                 if protocol is Protocol.Rollout:
-                    if is_packed(data):  # packed chunk: one matrix, n×field views
-                        steps = unpack_steps(data)
+                    if is_packed(data):  # packed chunk: one matrix per chunk
+                        steps = unpack_steps(data, lean=True)
                     else:  # plain step dict(s) — compatibility path
                         steps = data if isinstance(data, list) else [data]
                     await self.assembler.push_many(steps)

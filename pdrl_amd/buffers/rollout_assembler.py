@@ -13,8 +13,9 @@ including its two quirky-but-load-bearing semantics:
   the splice point so the learner resets recurrent state there
   (reference: rollout_assembler.py:61-73).
 
-Completed trajectories are stacked into ``{key: tensor(seq, feat)}`` and
-pushed to an asyncio queue.
+Completed trajectories are stacked into ``{key: float32 array(seq, feat)}``
+and pushed to an asyncio queue (numpy — the ring copies into shm; learner
+tensors are built at batch-staging time).
 """
 from __future__ import annotations
 
@@ -40,19 +41,18 @@ REQUIRED_KEYS = {
 }
 
 
-def stack_trajectory(steps: list[dict]) -> dict[str, torch.Tensor]:
-    """[{field: array(feat)} x seq] → {field: tensor(seq, feat)}.
+def stack_trajectory(steps: list[dict]) -> dict[str, np.ndarray]:
+    """[{field: array(feat)} x seq] → {field: float32 array(seq, feat)}.
 
-    numpy-first: packed-chunk fields arrive as float32 numpy views, so one
-    np.stack + zero-copy from_numpy per field replaces per-(step, field)
-    torch.as_tensor/reshape calls — stacking was the measured ingest-shard
-    ceiling (~75% of decode+assemble time) once the actor plane went
-    native. Falls back to the generic torch path for exotic inputs."""
+    numpy end-to-end: the only production consumer is the shared ring
+    (which copies numpy into shared memory), so no torch tensors are
+    created here — per-(step, field) torch.as_tensor/reshape and even the
+    per-field from_numpy wrappers were the measured ingest-shard ceiling
+    once the actor plane went native. Packed chunks stack their FULL rows
+    once; fields are zero-copy column views of the (seq, W) matrix."""
     s0 = steps[0]
     if "_row" in s0 and all("_row" in s for s in steps):
-        # packed-chunk fast path: ONE stack of the full rows, fields are
-        # zero-copy column views of the stacked (seq, W) matrix
-        mat = torch.from_numpy(np.stack([s["_row"] for s in steps]))
+        mat = np.stack([s["_row"] for s in steps])
         return {k: mat[:, lo:hi] for k, (lo, hi) in s0["_offs"].items()}
     out = {}
     for key in s0:
@@ -60,14 +60,13 @@ def stack_trajectory(steps: list[dict]) -> dict[str, torch.Tensor]:
             continue
         v0 = s0[key]
         if isinstance(v0, np.ndarray) and v0.dtype == np.float32:
-            out[key] = torch.from_numpy(np.stack([s[key] for s in steps]))
+            out[key] = np.stack([s[key] for s in steps])
         elif isinstance(v0, float):
-            out[key] = torch.from_numpy(
-                np.array([[s[key]] for s in steps], dtype=np.float32))
+            out[key] = np.array([[s[key]] for s in steps], dtype=np.float32)
         else:
-            vals = [torch.as_tensor(s[key], dtype=torch.float32).reshape(-1)
-                    for s in steps]
-            out[key] = torch.stack(vals, dim=0)
+            out[key] = np.stack(
+                [np.asarray(s[key], dtype=np.float32).reshape(-1)
+                 for s in steps])
     return out
 
 
@@ -91,9 +90,10 @@ class RolloutAssembler:
         build every row with the same keys."""
         if not steps:
             return
-        missing = REQUIRED_KEYS - set(steps[0] if isinstance(steps, (list, tuple))
-                                      else next(iter(steps)))
-        assert not missing, f"rollout step missing fields: {missing}"
+        first = steps[0] if isinstance(steps, (list, tuple)) else next(iter(steps))
+        if "_row" not in first:  # packed rows carry the schema by construction
+            missing = REQUIRED_KEYS - set(first)
+            assert not missing, f"rollout step missing fields: {missing}"
 
         # staleness eviction of partial trajectories (swept at most every
         # stale_s/4 — a per-push scan of the active dict measurably taxed
@@ -136,7 +136,7 @@ class RolloutAssembler:
                 active.pop(eid, None)
                 parked[eid] = traj
 
-    async def pop(self) -> dict[str, torch.Tensor]:
+    async def pop(self) -> dict[str, np.ndarray]:
         return await self.out_queue.get()
 
     def qsize(self) -> int:

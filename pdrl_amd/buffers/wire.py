@@ -36,7 +36,7 @@ def pack_steps(steps: list[dict]) -> dict:
     return {"ids": [s["id"] for s in steps], "widths": widths, "pk": pk}
 
 
-def unpack_steps(obj: dict) -> list[dict]:
+def unpack_steps(obj: dict, lean: bool = False) -> list[dict]:
     """Inverse of pack_steps. Field values are VIEWS into the chunk matrix
     (zero copies; the assembler stacks them into owned trajectory tensors).
     Each step also carries its FULL row (``_row``) and the shared field
@@ -48,6 +48,16 @@ def unpack_steps(obj: dict) -> list[dict]:
     span = {k: (int(offs[j]), int(offs[j + 1]))
             for j, k in enumerate(FIELD_ORDER)}
     out = []
+    if lean:
+        # assembler fast path: it only routes on id/done (+ is_fir via the
+        # row on splice) and stacks whole rows — skip the 9 per-step field
+        # views (measured ~4 µs/step of pure dict building)
+        done_lo = span["done"][0]
+        for i, eid in enumerate(obj["ids"]):
+            row = pk[i]
+            out.append({"id": eid, "done": float(row[done_lo]),
+                        "_row": row, "_offs": span})
+        return out
     for i, eid in enumerate(obj["ids"]):
         row = pk[i]
         step = {

@@ -82,6 +82,7 @@ def _manager(mport, lport, seconds, relayed, shards=1, drops=None):
 def _storage(ring, lport, seconds, ingested, stored):
     sys.path.insert(0, str(REPO))
     import asyncio
+    import os
 
     import torch
 
@@ -107,7 +108,18 @@ def _storage(ring, lport, seconds, ingested, stored):
         for task in tasks:
             task.cancel()
 
-    asyncio.run(run())
+    if os.environ.get("PDRL_PROF_STORAGE"):
+        import cProfile
+        import pstats
+
+        pr = cProfile.Profile()
+        pr.enable()
+        asyncio.run(run())
+        pr.disable()
+        with open("/tmp/pdrl_storage_prof.txt", "w") as fh:
+            pstats.Stats(pr, stream=fh).sort_stats("tottime").print_stats(16)
+    else:
+        asyncio.run(run())
 
 
 def main():

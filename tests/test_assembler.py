@@ -39,7 +39,8 @@ def test_full_trajectory_assembly_and_stacking():
         assert traj["obs"].shape == (5, 4)
         assert traj["rew"].shape == (5, 1)
         # steps in order
-        torch.testing.assert_close(traj["rew"].squeeze(-1), torch.arange(5.0))
+        torch.testing.assert_close(torch.as_tensor(traj["rew"]).squeeze(-1),
+                                   torch.arange(5.0))
         assert asm.qsize() == 0
 
     run(go())

@@ -262,7 +262,7 @@ def test_vectorized_worker_collects_complete_trajectories():
     assert trajs, "assembler completed no trajectories from vec worker"
     for tr in trajs:
         assert tr["obs"].shape == (p.seq_len, 4)
-        assert np.isfinite(tr["obs"].numpy()).all()
+        assert np.isfinite(np.asarray(tr["obs"])).all()
     w.close()
     mgr_sub.close()
 
