@@ -19,6 +19,8 @@ import numpy as np
 FIELD_ORDER = ("obs", "act", "rew", "logits", "log_prob", "is_fir", "done",
                "hx", "cx")
 
+_SPAN_CACHE: dict[tuple, dict] = {}  # widths-tuple → field span map
+
 
 def pack_steps(steps: list[dict]) -> dict:
     """[{field: array-like} × n] → {"ids", "widths", "pk"} (one float32 mat)."""
@@ -44,9 +46,13 @@ def unpack_steps(obj: dict, lean: bool = False) -> list[dict]:
     with ONE np.stack and slice fields zero-copy."""
     pk = obj["pk"]
     widths = obj["widths"]
-    offs = np.cumsum([0] + list(widths))
-    span = {k: (int(offs[j]), int(offs[j + 1]))
-            for j, k in enumerate(FIELD_ORDER)}
+    key = tuple(widths)
+    span = _SPAN_CACHE.get(key)
+    if span is None:  # identical for every chunk of a run — build once
+        offs_l = np.cumsum([0] + list(widths))
+        span = {k: (int(offs_l[j]), int(offs_l[j + 1]))
+                for j, k in enumerate(FIELD_ORDER)}
+        _SPAN_CACHE[key] = span
     out = []
     if lean:
         # assembler fast path: it only routes on id/done (+ is_fir via the
@@ -58,6 +64,7 @@ def unpack_steps(obj: dict, lean: bool = False) -> list[dict]:
             out.append({"id": eid, "done": float(row[done_lo]),
                         "_row": row, "_offs": span})
         return out
+    offs = np.cumsum([0] + list(widths))
     for i, eid in enumerate(obj["ids"]):
         row = pk[i]
         step = {
