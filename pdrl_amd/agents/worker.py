@@ -126,25 +126,33 @@ class Worker:
         if not self._continuous and heads and heads[0] == "logits":
             A = core.head_dims["logits"]
 
-            def fast(obs, hxs):
+            def fast(obs, hxs, outs=None):
                 hx, cx = hxs
                 a, lg, lp, h, c = _cpu_actor.act_batch_discrete(
                     obs.contiguous(), hx.contiguous(), cx.contiguous(),
-                    *wargs, A, rng)
+                    *wargs, A, rng, outs=outs)
                 return a, lg, lp, (h, c)
 
+            fast.out_shapes = lambda M, H: [
+                (torch.int64, (M, 1)), (torch.float32, (M, A)),
+                (torch.float32, (M, 1)), (torch.float32, (M, H)),
+                (torch.float32, (M, H))]
             return fast
         if self._continuous and heads[:2] in (["mu", "std"], ["mu", "log_std"]):
             A = core.head_dims["mu"]
             mode = 0 if heads[1] == "std" else 1  # PPO-C | SAC-C sampling
 
-            def fastc(obs, hxs):
+            def fastc(obs, hxs, outs=None):
                 hx, cx = hxs
                 a, lg, lp, h, c = _cpu_actor.act_batch_gaussian(
                     obs.contiguous(), hx.contiguous(), cx.contiguous(),
-                    *wargs, A, mode, rng)
+                    *wargs, A, mode, rng, outs=outs)
                 return a, lg, lp, (h, c)
 
+            fastc.out_shapes = lambda M, H: [
+                (torch.float32, (M, A)), (torch.float32, (M, 2 * A)),
+                (torch.float32, (M, 1)), (torch.float32, (M, H)),
+                (torch.float32, (M, H))]
             return fastc
         return None
 
@@ -292,9 +300,15 @@ class Worker:
             if getattr(r, "_state", None) is None:
                 return None  # env not reset yet
             state[i] = torch.from_numpy(np.asarray(r._state))
+        act_f32 = torch.empty(M, dtype=torch.float32)
+        outs = [torch.empty(M, dim, dtype=torch.float32),
+                torch.empty(M, dtype=torch.float32),
+                torch.empty(M, dtype=torch.float32)]
         return {"fn": fn, "state": state,
                 "steps": torch.zeros(M, dtype=torch.int64),
-                "max_steps": int(raws[0].MAX_EPISODE_STEPS), "raws": raws}
+                "max_steps": int(raws[0].MAX_EPISODE_STEPS), "raws": raws,
+                "act_f32": act_f32, "act_np": act_f32.numpy(),
+                "outs": outs, "onp": [t.numpy() for t in outs]}
 
     def _collect_vec(self, max_episodes: int | None = None):
         """Vectorized rollout: M envs, ONE batched model.act per tick. Each
@@ -314,6 +328,20 @@ class Worker:
         benv = self._make_batch_env()  # after reset: states are live
         hx = torch.zeros(M, H)
         cx = torch.zeros(M, H)
+        # ping-ponged persistent act output buffers + cached numpy views:
+        # kills the 5 allocations and ~7 tensor.numpy() calls per tick
+        # (the C++ act writes the new state while reading the old, so two
+        # slots alternate; packing reads the INPUT state's cached views)
+        slots = None
+        if hasattr(self._act, "out_shapes"):
+            def mk_slot():
+                ts = [torch.empty(shape, dtype=dt)
+                      for dt, shape in self._act.out_shapes(M, H)]
+                return {"t": ts, "np": [t.numpy() for t in ts]}
+
+            slots = (mk_slot(), mk_slot())
+            cur = 0
+            hx_np, cx_np = hx.numpy(), cx.numpy()
         epi_rew = [0.0] * M
         epi_id = [uuid.uuid4().hex for _ in range(M)]
         is_fir = [1.0] * M
@@ -321,10 +349,17 @@ class Worker:
         episodes = 0
 
         chunk_buf = None  # built lazily once the record widths are known
+        obs_np = None  # numpy alias of obs when the C++ env path is active
         t_start = time.perf_counter()
         while not self._stopped():
             self.poll_model()
-            action, logits, log_prob, (next_hx, next_cx) = self._act(obs, (hx, cx))
+            if slots is not None:
+                slot = slots[cur]
+                action, logits, log_prob, (next_hx, next_cx) = self._act(
+                    obs, (hx, cx), outs=slot["t"])
+            else:
+                action, logits, log_prob, (next_hx, next_cx) = \
+                    self._act(obs, (hx, cx))
             if self._continuous and self._total_steps < self.explore_warmup_steps:
                 action, log_prob = self._ou_explore(action, logits)
             self._total_steps += M
@@ -342,31 +377,44 @@ class Worker:
             # packing + env stepping both consume numpy views — convert the
             # act outputs ONCE per tick (torch reshape/float per tick cost
             # ~20% of the loop in profiling)
-            act_np = action.detach().numpy()
-            if self._continuous:
+            if slots is not None and action is slot["t"][0]:
+                act_np, logits_np, logp_np = (slot["np"][0], slot["np"][1],
+                                              slot["np"][2])
+            else:  # OU override or eager act: fresh tensors
+                act_np = action.detach().numpy()
+                logits_np = logits.detach().numpy()
+                logp_np = log_prob.detach().numpy()
+            if benv is not None:
+                acts = None  # C++ batch stepper consumes act_np directly
+            elif self._continuous:
                 acts = [act_np[i].astype(np.float32, copy=False)
                         for i in range(M)]
             else:
                 acts = [int(a) for a in act_np.reshape(-1)]
-            chunk_buf[sl, offs[0]:offs[1]] = obs.numpy()
+            chunk_buf[sl, offs[0]:offs[1]] = obs_np if obs_np is not None \
+                else obs.numpy()
             chunk_buf[sl, offs[1]:offs[2]] = \
                 act_np.reshape(M, -1).astype(np.float32, copy=False)
-            chunk_buf[sl, offs[3]:offs[4]] = logits.numpy()
-            chunk_buf[sl, offs[4]:offs[5]] = log_prob.numpy().reshape(M, -1)
+            chunk_buf[sl, offs[3]:offs[4]] = logits_np
+            chunk_buf[sl, offs[4]:offs[5]] = logp_np.reshape(M, -1)
             chunk_buf[sl, offs[5]] = is_fir
-            chunk_buf[sl, offs[7]:offs[8]] = hx.numpy()
-            chunk_buf[sl, offs[8]:offs[9]] = cx.numpy()
+            chunk_buf[sl, offs[7]:offs[8]] = hx_np if slots is not None \
+                else hx.numpy()
+            chunk_buf[sl, offs[8]:offs[9]] = cx_np if slots is not None \
+                else cx.numpy()
 
             any_done = False
             next_rows = []
             if benv is not None:
-                act_f32 = torch.from_numpy(
-                    act_np.reshape(M, -1)[:, 0].astype(np.float32))
-                b_obs, b_rew, b_done = benv["fn"](
-                    benv["state"], act_f32, benv["steps"], benv["max_steps"])
-                b_obs_np = b_obs.numpy()
-                b_rew_np = b_rew.numpy()
-                b_done_np = b_done.numpy()
+                np.copyto(benv["act_np"], act_np.reshape(M, -1)[:, 0],
+                          casting="unsafe")
+                benv["fn"](benv["state"], benv["act_f32"], benv["steps"],
+                           benv["max_steps"], outs=benv["outs"])
+                b_obs_np, b_rew_np, b_done_np = benv["onp"]
+                # vectorized column writes (tensor/scalar indexing per env
+                # measurably taxed the tick)
+                chunk_buf[sl, offs[2]] = b_rew_np
+                chunk_buf[sl, offs[6]] = b_done_np
             for i, env in enumerate(self.envs):
                 if benv is not None:
                     rew = float(b_rew_np[i])
@@ -374,9 +422,9 @@ class Worker:
                     next_obs = None  # row read from b_obs_np below
                 else:
                     next_obs, rew, done, _ = env.step(acts[i])
+                    chunk_buf[n0 + i, offs[2]] = rew
+                    chunk_buf[n0 + i, offs[6]] = float(done)
                 epi_rew[i] += rew
-                chunk_buf[n0 + i, offs[2]] = rew
-                chunk_buf[n0 + i, offs[6]] = float(done)
                 chunk_ids.append(epi_id[i])
                 any_done = any_done or done
                 epi_steps[i] += 1
@@ -412,9 +460,13 @@ class Worker:
                 )
                 self.pub.send(header, payload)
                 chunk_ids = []
-            obs = (torch.from_numpy(b_obs_np) if benv is not None
+            obs = (benv["outs"][0] if benv is not None
                    else torch.cat(next_rows, dim=0))
+            obs_np = b_obs_np if benv is not None else None
             hx, cx = next_hx, next_cx
+            if slots is not None:
+                hx_np, cx_np = slot["np"][3], slot["np"][4]
+                cur = 1 - cur
             if self.heartbeat is not None:
                 self.heartbeat.value = time.time()
             if self.step_sleep > 0:

@@ -37,12 +37,18 @@ inline uint64_t xorshift64(uint64_t& s) {
 // obs (M,F), hx/cx (M,H) fp32; weights in the SeqLSTMCore transposed layout;
 // logits live in head columns [0, A). rng: int64[1] state (advanced here).
 // Returns (action (M,1) int64, logits (M,A), log_prob (M,1), h (M,H), c (M,H)).
+// optional ``outs`` (5 preallocated tensors: action i64 (M,1),
+// logits (M,A), logp (M,1), h_out (M,H), c_out (M,H)) skip the per-call
+// allocations — the vectorized worker reuses ping-ponged buffers and
+// cached numpy views. h_out/c_out MUST NOT alias hx/cx (the kernel reads
+// state while writing the new one).
 std::vector<at::Tensor> act_batch_discrete(
     const at::Tensor& obs, const at::Tensor& hx, const at::Tensor& cx,
     const at::Tensor& body_w, const at::Tensor& body_b,
     const at::Tensor& w_ih, const at::Tensor& w_hh, const at::Tensor& b_g,
     const at::Tensor& heads_w, const at::Tensor& heads_b, int64_t A,
-    at::Tensor& rng) {
+    at::Tensor& rng,
+    const c10::optional<std::vector<at::Tensor>>& outs = c10::nullopt) {
   TORCH_CHECK(obs.device().is_cpu() && obs.dtype() == at::kFloat);
   TORCH_CHECK(obs.is_contiguous() && hx.is_contiguous() && cx.is_contiguous());
   const int M = obs.size(0), F = obs.size(1), H = hx.size(1);
@@ -52,11 +58,16 @@ std::vector<at::Tensor> act_batch_discrete(
   TORCH_CHECK(A <= D);
 
   auto opt = obs.options();
-  auto action = at::empty({M, 1}, opt.dtype(at::kLong));
-  auto logits = at::empty({M, (long)A}, opt);
-  auto logp = at::empty({M, 1}, opt);
-  auto h_out = at::empty({M, H}, opt);
-  auto c_out = at::empty({M, H}, opt);
+  const bool reuse = outs.has_value();
+  TORCH_CHECK(!reuse || outs->size() == 5, "outs must hold 5 tensors");
+  auto action = reuse ? (*outs)[0] : at::empty({M, 1}, opt.dtype(at::kLong));
+  auto logits = reuse ? (*outs)[1] : at::empty({M, (long)A}, opt);
+  auto logp = reuse ? (*outs)[2] : at::empty({M, 1}, opt);
+  auto h_out = reuse ? (*outs)[3] : at::empty({M, H}, opt);
+  auto c_out = reuse ? (*outs)[4] : at::empty({M, H}, opt);
+  TORCH_CHECK(!reuse || (h_out.data_ptr() != hx.data_ptr() &&
+                         c_out.data_ptr() != cx.data_ptr()),
+              "out state buffers must not alias the input state");
   // release the GIL for the compute: worker threads (tests / in-process
   // fleets) must not starve the python threads sharing the interpreter
   pybind11::gil_scoped_release nogil;
@@ -162,7 +173,8 @@ std::vector<at::Tensor> act_batch_gaussian(
     const at::Tensor& body_w, const at::Tensor& body_b,
     const at::Tensor& w_ih, const at::Tensor& w_hh, const at::Tensor& b_g,
     const at::Tensor& heads_w, const at::Tensor& heads_b, int64_t A,
-    int64_t mode, at::Tensor& rng) {
+    int64_t mode, at::Tensor& rng,
+    const c10::optional<std::vector<at::Tensor>>& outs = c10::nullopt) {
   TORCH_CHECK(obs.device().is_cpu() && obs.dtype() == at::kFloat);
   TORCH_CHECK(obs.is_contiguous() && hx.is_contiguous() && cx.is_contiguous());
   const int M = obs.size(0), F = obs.size(1), H = hx.size(1);
@@ -171,11 +183,16 @@ std::vector<at::Tensor> act_batch_gaussian(
   constexpr float kHalfLog2Pi = 0.91893853320467274f;
 
   auto opt = obs.options();
-  auto action = at::empty({M, (long)A}, opt);
-  auto logits = at::empty({M, (long)(2 * A)}, opt);
-  auto logp = at::empty({M, 1}, opt);
-  auto h_out = at::empty({M, H}, opt);
-  auto c_out = at::empty({M, H}, opt);
+  const bool reuse = outs.has_value();
+  TORCH_CHECK(!reuse || outs->size() == 5, "outs must hold 5 tensors");
+  auto action = reuse ? (*outs)[0] : at::empty({M, (long)A}, opt);
+  auto logits = reuse ? (*outs)[1] : at::empty({M, (long)(2 * A)}, opt);
+  auto logp = reuse ? (*outs)[2] : at::empty({M, 1}, opt);
+  auto h_out = reuse ? (*outs)[3] : at::empty({M, H}, opt);
+  auto c_out = reuse ? (*outs)[4] : at::empty({M, H}, opt);
+  TORCH_CHECK(!reuse || (h_out.data_ptr() != hx.data_ptr() &&
+                         c_out.data_ptr() != cx.data_ptr()),
+              "out state buffers must not alias the input state");
   // release the GIL for the compute: worker threads (tests / in-process
   // fleets) must not starve the python threads sharing the interpreter
   pybind11::gil_scoped_release nogil;
@@ -286,11 +303,14 @@ std::vector<at::Tensor> act_batch_gaussian(
 // (0/1), steps (M) i64; done = terminated || steps >= max_steps.
 std::vector<at::Tensor> cartpole_step_batch(at::Tensor& state,
                                             const at::Tensor& act,
-                                            at::Tensor& steps, long max_steps) {
+                                            at::Tensor& steps, long max_steps,
+                                            const c10::optional<std::vector<at::Tensor>>& outs = c10::nullopt) {
   const long M = state.size(0);
-  auto obs = at::empty({M, 4}, act.options());
-  auto rew = at::empty({M}, act.options());
-  auto done = at::empty({M}, act.options());
+  const bool reuse = outs.has_value();
+  TORCH_CHECK(!reuse || outs->size() == 3, "outs must hold 3 tensors");
+  auto obs = reuse ? (*outs)[0] : at::empty({M, 4}, act.options());
+  auto rew = reuse ? (*outs)[1] : at::empty({M}, act.options());
+  auto done = reuse ? (*outs)[2] : at::empty({M}, act.options());
   pybind11::gil_scoped_release nogil;
   double* st = state.data_ptr<double>();
   const float* ac = act.data_ptr<float>();
@@ -338,11 +358,14 @@ std::vector<at::Tensor> cartpole_step_batch(at::Tensor& state,
 // f64, act (M) f32 in [-1,1].
 std::vector<at::Tensor> mcc_step_batch(at::Tensor& state,
                                        const at::Tensor& act,
-                                       at::Tensor& steps, long max_steps) {
+                                       at::Tensor& steps, long max_steps,
+                                       const c10::optional<std::vector<at::Tensor>>& outs = c10::nullopt) {
   const long M = state.size(0);
-  auto obs = at::empty({M, 2}, act.options());
-  auto rew = at::empty({M}, act.options());
-  auto done = at::empty({M}, act.options());
+  const bool reuse = outs.has_value();
+  TORCH_CHECK(!reuse || outs->size() == 3, "outs must hold 3 tensors");
+  auto obs = reuse ? (*outs)[0] : at::empty({M, 2}, act.options());
+  auto rew = reuse ? (*outs)[1] : at::empty({M}, act.options());
+  auto done = reuse ? (*outs)[2] : at::empty({M}, act.options());
   pybind11::gil_scoped_release nogil;
   double* st = state.data_ptr<double>();
   const float* ac = act.data_ptr<float>();
@@ -377,11 +400,27 @@ std::vector<at::Tensor> mcc_step_batch(at::Tensor& state,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("act_batch_discrete", &act_batch_discrete,
-        "batched CPU actor step: body+LSTM+logits+sample in one call");
+        "batched CPU actor step: body+LSTM+logits+sample in one call",
+        pybind11::arg("obs"), pybind11::arg("hx"), pybind11::arg("cx"),
+        pybind11::arg("body_w"), pybind11::arg("body_b"),
+        pybind11::arg("w_ih"), pybind11::arg("w_hh"), pybind11::arg("b_g"),
+        pybind11::arg("heads_w"), pybind11::arg("heads_b"),
+        pybind11::arg("A"), pybind11::arg("rng"),
+        pybind11::arg("outs") = pybind11::none());
   m.def("act_batch_gaussian", &act_batch_gaussian,
-        "batched CPU actor step for Gaussian policies (PPO-C / SAC-C)");
+        "batched CPU actor step for Gaussian policies (PPO-C / SAC-C)",
+        pybind11::arg("obs"), pybind11::arg("hx"), pybind11::arg("cx"),
+        pybind11::arg("body_w"), pybind11::arg("body_b"),
+        pybind11::arg("w_ih"), pybind11::arg("w_hh"), pybind11::arg("b_g"),
+        pybind11::arg("heads_w"), pybind11::arg("heads_b"),
+        pybind11::arg("A"), pybind11::arg("mode"), pybind11::arg("rng"),
+        pybind11::arg("outs") = pybind11::none());
   m.def("cartpole_step_batch", &cartpole_step_batch,
-        "vectorized CartPole-v1 physics (native env fast path)");
+        "vectorized CartPole-v1 physics (native env fast path)",
+        pybind11::arg("state"), pybind11::arg("act"), pybind11::arg("steps"),
+        pybind11::arg("max_steps"), pybind11::arg("outs") = pybind11::none());
   m.def("mcc_step_batch", &mcc_step_batch,
-        "vectorized MountainCarContinuous-v0 physics");
+        "vectorized MountainCarContinuous-v0 physics",
+        pybind11::arg("state"), pybind11::arg("act"), pybind11::arg("steps"),
+        pybind11::arg("max_steps"), pybind11::arg("outs") = pybind11::none());
 }
