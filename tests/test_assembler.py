@@ -125,3 +125,41 @@ def test_splice_prefers_smallest_parked():
         assert len(asm.active) == 1
 
     run(go())
+
+
+def test_packed_chunk_splice_rewrites_is_fir_in_row():
+    """Lean packed steps (wire.unpack_steps(lean=True)) through the splice:
+    the parked-done trajectory absorbs the new episode's first step with
+    is_fir forced to 1.0 INSIDE the stacked row (the fast stacking path
+    reads rows, not the per-field dict)."""
+    import numpy as np
+
+    from pdrl_amd.buffers.wire import FIELD_ORDER, pack_steps, unpack_steps
+
+    def mk(eid, t, done=0.0, is_fir=0.0):
+        return {"obs": np.full(4, t, dtype=np.float32),
+                "act": np.zeros(1, dtype=np.float32), "rew": float(t),
+                "logits": np.zeros(2, dtype=np.float32),
+                "log_prob": np.zeros(1, dtype=np.float32),
+                "is_fir": is_fir, "done": done,
+                "hx": np.zeros(64, dtype=np.float32),
+                "cx": np.zeros(64, dtype=np.float32), "id": eid}
+
+    async def go():
+        asm = RolloutAssembler(seq_len=4)
+        # episode A ends after 2 steps (done) → parked
+        chunk_a = pack_steps([mk("A", 0, is_fir=1.0), mk("A", 1, done=1.0)])
+        await asm.push_many(unpack_steps(chunk_a, lean=True))
+        assert asm.qsize() == 0
+        # episode B starts: first step splices onto parked A with is_fir=1
+        chunk_b = pack_steps([mk("B", 10), mk("B", 11)])
+        await asm.push_many(unpack_steps(chunk_b, lean=True))
+        traj = await asm.pop()
+        rew = np.asarray(traj["rew"]).reshape(-1)
+        np.testing.assert_allclose(rew, [0.0, 1.0, 10.0, 11.0])
+        fir = np.asarray(traj["is_fir"]).reshape(-1)
+        np.testing.assert_allclose(fir, [1.0, 0.0, 1.0, 0.0])  # splice mark
+        done = np.asarray(traj["done"]).reshape(-1)
+        np.testing.assert_allclose(done, [0.0, 1.0, 0.0, 0.0])
+
+    run(go())
